@@ -1,0 +1,854 @@
+/*
+ * core.cpp — engine implementation. See core.hpp for the layer map and
+ * reference anchors.
+ */
+#include "core.hpp"
+
+#include <math.h>
+#include <stdio.h>
+#include <string.h>
+
+#include <algorithm>
+#include <random>
+
+#include "json.hpp"
+
+namespace vgamma {
+
+/* ------------------------------------------------------------ KillRegistry */
+KillRegistry &KillRegistry::inst() {
+  static KillRegistry k;
+  return k;
+}
+void KillRegistry::set(const std::string &rid, int pid) {
+  std::lock_guard<std::mutex> g(mu_);
+  map_[{rid, pid}] = 1;
+}
+void KillRegistry::del(const std::string &rid, int pid) {
+  std::lock_guard<std::mutex> g(mu_);
+  map_.erase({rid, pid});
+}
+bool KillRegistry::killed(const std::string &rid, int pid) {
+  if (rid.empty()) return false;
+  std::lock_guard<std::mutex> g(mu_);
+  return map_.count({rid, pid}) > 0;
+}
+
+/* --------------------------------------------------------------- DeviceBuf */
+int DeviceBuf::reserve(size_t bytes) {
+  if (bytes <= bytes_) return 0;
+  void *np = nullptr;
+  if (hipMalloc(&np, bytes) != hipSuccess) return -1;
+  if (p_) hipFree(p_);
+  p_ = np;
+  bytes_ = bytes;
+  return 0;
+}
+void DeviceBuf::free() {
+  if (p_) hipFree(p_);
+  p_ = nullptr;
+  bytes_ = 0;
+}
+
+/* ---------------------------------------------------------------- RawStore */
+int RawStore::init(int d) {
+  d_ = d;
+  return 0;
+}
+
+int RawStore::ensure_capacity(int64_t n_new, hipStream_t s) {
+  const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
+  size_t need_segs = (size_t)((n_new + seg_n - 1) >> SEG_SHIFT);
+  bool grew = false;
+  while (dev_segs_.size() < need_segs) {
+    void *p = nullptr;
+    if (hipMalloc(&p, (size_t)seg_n * d_ * 4) != hipSuccess) return -1;
+    dev_segs_.push_back(p);
+    host_segs_.emplace_back();
+    host_segs_.back().resize((size_t)seg_n * d_);
+    grew = true;
+  }
+  if (grew) {
+    if (seg_table_.reserve(dev_segs_.size() * sizeof(float *))) return -1;
+    if (hipMemcpy(seg_table_.get(), dev_segs_.data(),
+                  dev_segs_.size() * sizeof(float *),
+                  hipMemcpyHostToDevice) != hipSuccess)
+      return -1;
+  }
+  if (n_new > norms_cap_) {
+    int64_t cap = std::max<int64_t>(n_new * 2, seg_n);
+    DeviceBuf nb;
+    if (nb.reserve((size_t)cap * 4)) return -1;
+    if (norms_cap_ > 0)
+      hipMemcpy(nb.get(), norms_.get(), (size_t)n_ * 4,
+                hipMemcpyDeviceToDevice);
+    /* steal */
+    norms_.free();
+    if (norms_.reserve(0)) return -1;
+    /* move nb into norms_: DeviceBuf lacks move; emulate */
+    static_assert(sizeof(DeviceBuf) == sizeof(void *) + sizeof(size_t),
+                  "layout");
+    memcpy((void *)&norms_, (void *)&nb, sizeof(DeviceBuf));
+    memset((void *)&nb, 0, sizeof(DeviceBuf));
+    norms_cap_ = cap;
+  }
+  return 0;
+}
+
+int RawStore::add(const float *x, int64_t cnt, hipStream_t s) {
+  if (cnt <= 0) return 0;
+  if (ensure_capacity(n_ + cnt, s)) return -1;
+  const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
+  int64_t done = 0;
+  while (done < cnt) {
+    int64_t vid = n_ + done;
+    int64_t seg = vid >> SEG_SHIFT;
+    int64_t off = vid & (seg_n - 1);
+    int64_t room = seg_n - off;
+    int64_t take = std::min(room, cnt - done);
+    memcpy(host_segs_[seg].data() + (size_t)off * d_, x + (size_t)done * d_,
+           (size_t)take * d_ * 4);
+    if (hipMemcpyAsync((char *)dev_segs_[seg] + (size_t)off * d_ * 4,
+                       x + (size_t)done * d_, (size_t)take * d_ * 4,
+                       hipMemcpyHostToDevice, s) != hipSuccess)
+      return -1;
+    if (gk::row_norms(s, (const float *)dev_segs_[seg] + (size_t)off * d_,
+                      take, d_, norms_.as<float>() + vid) != hipSuccess)
+      return -1;
+    done += take;
+  }
+  if (hipStreamSynchronize(s) != hipSuccess) return -1;
+  n_ += cnt;
+  return 0;
+}
+
+const float *RawStore::host_row(int64_t vid) const {
+  const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
+  return host_segs_[vid >> SEG_SHIFT].data() + (size_t)(vid & (seg_n - 1)) * d_;
+}
+
+void RawStore::host_copy(int64_t start, int64_t cnt, float *out) const {
+  for (int64_t i = 0; i < cnt;) {
+    int64_t vid = start + i;
+    const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
+    int64_t off = vid & (seg_n - 1);
+    int64_t take = std::min(seg_n - off, cnt - i);
+    memcpy(out + (size_t)i * d_, host_row(vid), (size_t)take * d_ * 4);
+    i += take;
+  }
+}
+
+const float *RawStore::dev_run(int64_t vid, int64_t *run_len) const {
+  const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
+  int64_t seg = vid >> SEG_SHIFT, off = vid & (seg_n - 1);
+  int64_t end = std::min(n_, (seg + 1) << SEG_SHIFT);
+  *run_len = end - vid;
+  return (const float *)dev_segs_[seg] + (size_t)off * d_;
+}
+
+int RawStore::dump(FILE *f) const {
+  fwrite(&d_, 4, 1, f);
+  fwrite(&n_, 8, 1, f);
+  const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
+  for (int64_t s = 0; s * seg_n < n_; s++) {
+    int64_t take = std::min(seg_n, n_ - s * seg_n);
+    fwrite(host_segs_[s].data(), 4, (size_t)take * d_, f);
+  }
+  return 0;
+}
+
+int RawStore::load(FILE *f, hipStream_t s) {
+  int d = 0;
+  int64_t n = 0;
+  if (fread(&d, 4, 1, f) != 1 || fread(&n, 8, 1, f) != 1) return -1;
+  d_ = d;
+  std::vector<float> buf((size_t)std::min<int64_t>(n, 65536) * d_);
+  for (int64_t done = 0; done < n;) {
+    int64_t take = std::min<int64_t>(65536, n - done);
+    if (fread(buf.data(), 4, (size_t)take * d_, f) != (size_t)take * d_)
+      return -1;
+    if (add(buf.data(), take, s)) return -1;
+    done += take;
+  }
+  return 0;
+}
+
+/* ------------------------------------------------------------------ Bitmap */
+int Bitmap::ensure(int64_t nbits, hipStream_t s) {
+  if (nbits <= bits_) return 0;
+  int64_t words = (std::max<int64_t>(nbits, 1 << 20) + 31) / 32;
+  words = std::max<int64_t>(words * 2, (int64_t)host_.size());
+  size_t old_words = host_.size();
+  host_.resize(words, 0);
+  DeviceBuf nd;
+  if (nd.reserve((size_t)words * 4)) return -1;
+  hipMemset(nd.get(), 0, (size_t)words * 4);
+  if (old_words)
+    hipMemcpy(nd.get(), dev_.get(), old_words * 4, hipMemcpyDeviceToDevice);
+  dev_.free();
+  memcpy((void *)&dev_, (void *)&nd, sizeof(DeviceBuf));
+  memset((void *)&nd, 0, sizeof(DeviceBuf));
+  bits_ = words * 32;
+  return 0;
+}
+int Bitmap::set(int64_t vid, hipStream_t s) {
+  if (ensure(vid + 1, s)) return -1;
+  host_[vid >> 5] |= 1u << (vid & 31);
+  /* mirror the single word to device */
+  if (hipMemcpy((uint32_t *)dev_.get() + (vid >> 5), &host_[vid >> 5], 4,
+                hipMemcpyHostToDevice) != hipSuccess)
+    return -1;
+  return 0;
+}
+bool Bitmap::test(int64_t vid) const {
+  if (vid < 0 || vid >= bits_) return false;
+  return (host_[vid >> 5] >> (vid & 31)) & 1u;
+}
+int64_t Bitmap::popcount() const {
+  int64_t c = 0;
+  for (uint32_t w : host_) c += __builtin_popcount(w);
+  return c;
+}
+int Bitmap::dump(FILE *f) const {
+  int64_t words = (int64_t)host_.size();
+  fwrite(&words, 8, 1, f);
+  fwrite(host_.data(), 4, host_.size(), f);
+  return 0;
+}
+int Bitmap::load(FILE *f, hipStream_t s) {
+  int64_t words = 0;
+  if (fread(&words, 8, 1, f) != 1) return -1;
+  std::vector<uint32_t> tmp(words);
+  if (words && fread(tmp.data(), 4, words, f) != (size_t)words) return -1;
+  if (ensure(words * 32, s)) return -1;
+  std::copy(tmp.begin(), tmp.end(), host_.begin());
+  if (words)
+    hipMemcpy(dev_.get(), host_.data(), (size_t)words * 4,
+              hipMemcpyHostToDevice);
+  return 0;
+}
+
+/* ------------------------------------------------------------- IndexParams */
+int IndexParams::parse(const std::string &json, std::string *err) {
+  if (json.empty()) return 0;
+  gjson::Value v;
+  if (!gjson::parse(json, v)) {
+    if (err) *err = "parse index params error: " + json;
+    return -1;
+  }
+  int x;
+  if (v.get_int("ncentroids", x) && x > 0) ncentroids = x;
+  if (v.get_int("nsubvector", x) && x > 0) nsubvector = x;
+  if (v.get_int("nbits_per_idx", x) && x > 0) nbits = x;
+  if (v.get_int("nprobe", x) && x > 0) nprobe = x;
+  if (v.get_int("bucket_init_size", x) && x > 0) bucket_init_size = x;
+  if (v.get_int("bucket_max_size", x) && x > 0) bucket_max_size = x;
+  if (v.get_int("training_threshold", x) && x > 0) training_threshold = x;
+  std::string mt;
+  if (v.get_str("metric_type", mt)) {
+    if (!strcasecmp(mt.c_str(), "L2")) metric_ip = false;
+    else if (!strcasecmp(mt.c_str(), "InnerProduct")) metric_ip = true;
+    else {
+      if (err) *err = "invalid metric_type = " + mt;
+      return -1;
+    }
+  }
+  if (nbits != 8) {
+    if (err) *err = "only nbits_per_idx=8 supported";
+    return -1;
+  }
+  if (v.has("hnsw")) {
+    if (err) *err = "hnsw coarse quantizer not supported (out of scope)";
+    return -1;
+  }
+  if (v.has("opq")) {
+    if (err) *err = "opq not supported yet (SURVEY 8f-4)";
+    return -1;
+  }
+  return 0;
+}
+
+/* ---------------------------------------------------------------- IVFIndex */
+int IVFIndex::init(int d, const IndexParams &p) {
+  params_ = p;
+  d_ = d;
+  nlist_ = p.ncentroids;
+  if (params_.kind == IndexKind::IVFPQ) {
+    M_ = p.nsubvector > 0 ? p.nsubvector : d / 2; /* ivfpq.cc:122-124 */
+    if (M_ <= 0 || d % M_ != 0 || M_ % 4 != 0) return -1;
+    dsub_ = d / M_;
+    code_size_ = M_;
+  } else {
+    M_ = 0;
+    dsub_ = 0;
+    code_size_ = d * 4;
+  }
+  buckets_.resize(nlist_);
+  return 0;
+}
+
+int IVFIndex::kmeans_gpu(const float *x_host, int64_t n, int ncl, int niter,
+                         bool spherical, std::vector<float> &cent_out,
+                         hipStream_t s) {
+  /* Lloyd, faiss-Clustering style (ivfpq.cc:188-191): random-sample init,
+   * largest-cluster split for empties, optional spherical renorm. */
+  DeviceBuf xd, xnorm, cd, cnorm, dots, asg;
+  GAMMA_CHECK(xd.reserve((size_t)n * d_ * 4) ? hipErrorOutOfMemory
+                                             : hipSuccess);
+  GAMMA_CHECK(hipMemcpy(xd.get(), x_host, (size_t)n * d_ * 4,
+                        hipMemcpyHostToDevice));
+  GAMMA_CHECK(xnorm.reserve((size_t)n * 4) ? hipErrorOutOfMemory
+                                           : hipSuccess);
+  GAMMA_CHECK(gk::row_norms(s, xd.as<float>(), n, d_, xnorm.as<float>()));
+  const int64_t chunk = 16384;
+  GAMMA_CHECK(dots.reserve((size_t)std::min(n, chunk) * ncl * 4)
+                  ? hipErrorOutOfMemory
+                  : hipSuccess);
+  GAMMA_CHECK(asg.reserve((size_t)n * 4) ? hipErrorOutOfMemory : hipSuccess);
+  GAMMA_CHECK(cd.reserve((size_t)ncl * d_ * 4) ? hipErrorOutOfMemory
+                                               : hipSuccess);
+  GAMMA_CHECK(cnorm.reserve((size_t)ncl * 4) ? hipErrorOutOfMemory
+                                             : hipSuccess);
+
+  cent_out.resize((size_t)ncl * d_);
+  std::mt19937_64 rng(42);
+  std::vector<int64_t> perm(n);
+  for (int64_t i = 0; i < n; i++) perm[i] = i;
+  for (int64_t i = 0; i < std::min<int64_t>(ncl, n); i++) {
+    std::swap(perm[i], perm[i + (int64_t)(rng() % (uint64_t)(n - i))]);
+  }
+  for (int c = 0; c < ncl; c++) {
+    int64_t src = perm[c % n];
+    memcpy(cent_out.data() + (size_t)c * d_, x_host + (size_t)src * d_,
+           (size_t)d_ * 4);
+  }
+  auto renorm = [&](std::vector<float> &cent) {
+    if (!spherical) return;
+    for (int c = 0; c < ncl; c++) {
+      double nn = 0;
+      float *row = cent.data() + (size_t)c * d_;
+      for (int j = 0; j < d_; j++) nn += (double)row[j] * row[j];
+      float inv = (float)(1.0 / std::max(sqrt(nn), 1e-20));
+      for (int j = 0; j < d_; j++) row[j] *= inv;
+    }
+  };
+  renorm(cent_out);
+
+  std::vector<int32_t> assign_h(n);
+  std::vector<double> sums((size_t)ncl * d_);
+  std::vector<int64_t> counts(ncl);
+
+  for (int it = 0; it < niter; it++) {
+    GAMMA_CHECK(hipMemcpy(cd.get(), cent_out.data(), (size_t)ncl * d_ * 4,
+                          hipMemcpyHostToDevice));
+    GAMMA_CHECK(gk::row_norms(s, cd.as<float>(), ncl, d_,
+                              cnorm.as<float>()));
+    for (int64_t r0 = 0; r0 < n; r0 += chunk) {
+      int64_t rn = std::min(chunk, n - r0);
+      GAMMA_CHECK(gk::dots_mfma(s, xd.as<float>() + (size_t)r0 * d_, (int)rn,
+                                cd.as<float>(), ncl, d_, dots.as<float>()));
+      GAMMA_CHECK(gk::argmin_rows(s, rn, ncl, dots.as<float>(),
+                                  xnorm.as<float>() + r0, cnorm.as<float>(),
+                                  !spherical, asg.as<int32_t>() + r0));
+    }
+    GAMMA_CHECK(hipMemcpy(assign_h.data(), asg.get(), (size_t)n * 4,
+                          hipMemcpyDeviceToHost));
+    std::fill(sums.begin(), sums.end(), 0.0);
+    std::fill(counts.begin(), counts.end(), 0);
+    for (int64_t i = 0; i < n; i++) {
+      int c = assign_h[i];
+      const float *row = x_host + (size_t)i * d_;
+      double *srow = sums.data() + (size_t)c * d_;
+      for (int j = 0; j < d_; j++) srow[j] += row[j];
+      counts[c]++;
+    }
+    for (int c = 0; c < ncl; c++) {
+      if (counts[c] == 0) { /* split the largest cluster */
+        int big = (int)(std::max_element(counts.begin(), counts.end()) -
+                        counts.begin());
+        if (counts[big] < 2) continue;
+        for (int j = 0; j < d_; j++) {
+          double mean = sums[(size_t)big * d_ + j] / counts[big];
+          double eps = 1e-5 * (1.0 + fabs(mean));
+          sums[(size_t)c * d_ + j] = (mean + eps) * (counts[big] / 2);
+          sums[(size_t)big * d_ + j] =
+              (mean - eps) * (counts[big] - counts[big] / 2);
+        }
+        counts[c] = counts[big] / 2;
+        counts[big] -= counts[c];
+      }
+    }
+    for (int c = 0; c < ncl; c++) {
+      float *row = cent_out.data() + (size_t)c * d_;
+      for (int j = 0; j < d_; j++)
+        row[j] = (float)(sums[(size_t)c * d_ + j] /
+                         std::max<int64_t>(counts[c], 1));
+    }
+    renorm(cent_out);
+  }
+  return 0;
+}
+
+int IVFIndex::train(const float *xt, int64_t n, hipStream_t s,
+                    std::string *err) {
+  std::vector<float> cent;
+  if (kmeans_gpu(xt, n, nlist_, 10, params_.metric_ip, cent, s)) {
+    if (err) *err = "coarse k-means failed";
+    return -1;
+  }
+  if (centroids_.reserve((size_t)nlist_ * d_ * 4)) return -1;
+  if (hipMemcpy(centroids_.get(), cent.data(), (size_t)nlist_ * d_ * 4,
+                hipMemcpyHostToDevice) != hipSuccess)
+    return -1;
+  if (cent_norms_.reserve((size_t)nlist_ * 4)) return -1;
+  if (gk::row_norms(s, centroids_.as<float>(), nlist_, d_,
+                    cent_norms_.as<float>()) != hipSuccess)
+    return -1;
+
+  if (params_.kind == IndexKind::IVFPQ) {
+    /* residuals of the training set under the final centroids */
+    DeviceBuf xd, dots, asg, resid;
+    if (xd.reserve((size_t)n * d_ * 4)) return -1;
+    hipMemcpy(xd.get(), xt, (size_t)n * d_ * 4, hipMemcpyHostToDevice);
+    DeviceBuf xnorm;
+    xnorm.reserve((size_t)n * 4);
+    gk::row_norms(s, xd.as<float>(), n, d_, xnorm.as<float>());
+    const int64_t chunk = 16384;
+    dots.reserve((size_t)std::min(n, chunk) * nlist_ * 4);
+    asg.reserve((size_t)n * 4);
+    for (int64_t r0 = 0; r0 < n; r0 += chunk) {
+      int64_t rn = std::min(chunk, n - r0);
+      if (gk::dots_mfma(s, xd.as<float>() + (size_t)r0 * d_, (int)rn,
+                        centroids_.as<float>(), nlist_, d_,
+                        dots.as<float>()) != hipSuccess)
+        return -1;
+      if (gk::argmin_rows(s, rn, nlist_, dots.as<float>(),
+                          xnorm.as<float>() + r0, cent_norms_.as<float>(),
+                          !params_.metric_ip,
+                          asg.as<int32_t>() + r0) != hipSuccess)
+        return -1;
+    }
+    resid.reserve((size_t)n * d_ * 4);
+    if (gk::residuals(s, n, d_, xd.as<float>(), centroids_.as<float>(),
+                      asg.as<int32_t>(), resid.as<float>()) != hipSuccess)
+      return -1;
+    std::vector<float> resid_h((size_t)n * d_);
+    hipMemcpy(resid_h.data(), resid.get(), (size_t)n * d_ * 4,
+              hipMemcpyDeviceToHost);
+
+    /* per-subspace k-means, ksub=256, niter=25 (faiss PQ default) */
+    std::vector<float> books((size_t)M_ * ksub_ * dsub_);
+    std::vector<float> sub((size_t)n * dsub_);
+    for (int m = 0; m < M_; m++) {
+      for (int64_t i = 0; i < n; i++)
+        memcpy(sub.data() + (size_t)i * dsub_,
+               resid_h.data() + (size_t)i * d_ + (size_t)m * dsub_,
+               (size_t)dsub_ * 4);
+      std::vector<float> cb;
+      if (pq_subspace_kmeans_(sub.data(), n, cb, s, m)) {
+        if (err) *err = "pq train failed";
+        return -1;
+      }
+      memcpy(books.data() + (size_t)m * ksub_ * dsub_, cb.data(),
+             (size_t)ksub_ * dsub_ * 4);
+    }
+    if (codebooks_.reserve((size_t)M_ * ksub_ * dsub_ * 4)) return -1;
+    hipMemcpy(codebooks_.get(), books.data(), (size_t)M_ * ksub_ * dsub_ * 4,
+              hipMemcpyHostToDevice);
+  }
+  trained_ = true;
+  return 0;
+}
+
+/* per-subspace k-means via the pq_encode kernel as the assign step */
+int IVFIndex::pq_subspace_kmeans_(const float *sub_host, int64_t n,
+                                  std::vector<float> &cb, hipStream_t s,
+                                  int seed_off) {
+  const int niter = 25;
+  DeviceBuf xd, cbd, codes;
+  if (xd.reserve((size_t)n * dsub_ * 4)) return -1;
+  hipMemcpy(xd.get(), sub_host, (size_t)n * dsub_ * 4,
+            hipMemcpyHostToDevice);
+  if (cbd.reserve((size_t)ksub_ * dsub_ * 4)) return -1;
+  if (codes.reserve((size_t)n)) return -1;
+
+  cb.resize((size_t)ksub_ * dsub_);
+  std::mt19937_64 rng(42 + seed_off);
+  std::vector<int64_t> perm(n);
+  for (int64_t i = 0; i < n; i++) perm[i] = i;
+  for (int64_t i = 0; i < std::min<int64_t>(ksub_, n); i++)
+    std::swap(perm[i], perm[i + (int64_t)(rng() % (uint64_t)(n - i))]);
+  for (int c = 0; c < ksub_; c++)
+    memcpy(cb.data() + (size_t)c * dsub_,
+           sub_host + (size_t)perm[c % n] * dsub_, (size_t)dsub_ * 4);
+
+  std::vector<uint8_t> codes_h(n);
+  std::vector<double> sums((size_t)ksub_ * dsub_);
+  std::vector<int64_t> counts(ksub_);
+  for (int it = 0; it < niter; it++) {
+    hipMemcpy(cbd.get(), cb.data(), (size_t)ksub_ * dsub_ * 4,
+              hipMemcpyHostToDevice);
+    if (gk::pq_encode(s, n, dsub_, 1, ksub_, xd.as<float>(),
+                      cbd.as<float>(), codes.as<uint8_t>()) != hipSuccess)
+      return -1;
+    hipMemcpy(codes_h.data(), codes.get(), (size_t)n, hipMemcpyDeviceToHost);
+    std::fill(sums.begin(), sums.end(), 0.0);
+    std::fill(counts.begin(), counts.end(), 0);
+    for (int64_t i = 0; i < n; i++) {
+      int c = codes_h[i];
+      const float *row = sub_host + (size_t)i * dsub_;
+      for (int j = 0; j < dsub_; j++) sums[(size_t)c * dsub_ + j] += row[j];
+      counts[c]++;
+    }
+    for (int c = 0; c < ksub_; c++) {
+      if (counts[c] == 0) {
+        int big = (int)(std::max_element(counts.begin(), counts.end()) -
+                        counts.begin());
+        if (counts[big] < 2) continue;
+        for (int j = 0; j < dsub_; j++) {
+          double mean = sums[(size_t)big * dsub_ + j] / counts[big];
+          double eps = 1e-5 * (1.0 + fabs(mean));
+          sums[(size_t)c * dsub_ + j] = (mean + eps) * (counts[big] / 2);
+          sums[(size_t)big * dsub_ + j] =
+              (mean - eps) * (counts[big] - counts[big] / 2);
+        }
+        counts[c] = counts[big] / 2;
+        counts[big] -= counts[c];
+      }
+    }
+    for (int c = 0; c < ksub_; c++)
+      for (int j = 0; j < dsub_; j++)
+        cb[(size_t)c * dsub_ + j] =
+            (float)(sums[(size_t)c * dsub_ + j] /
+                    std::max<int64_t>(counts[c], 1));
+  }
+  return 0;
+}
+
+int IVFIndex::update_dev_buckets(hipStream_t s) {
+  if (!dev_buckets_dirty_) return 0;
+  std::vector<GammaBucketDev> h(nlist_);
+  for (int i = 0; i < nlist_; i++) {
+    h[i].ids = buckets_[i].ids ? buckets_[i].ids->as<int64_t>() : nullptr;
+    h[i].data = buckets_[i].data ? buckets_[i].data->get() : nullptr;
+    h[i].size = buckets_[i].size;
+  }
+  if (dev_buckets_.reserve(nlist_ * sizeof(GammaBucketDev))) return -1;
+  if (hipMemcpy(dev_buckets_.get(), h.data(),
+                nlist_ * sizeof(GammaBucketDev),
+                hipMemcpyHostToDevice) != hipSuccess)
+    return -1;
+  dev_buckets_dirty_ = false;
+  return 0;
+}
+
+int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
+                  hipStream_t s) {
+  if (!trained_ || n <= 0) return trained_ ? 0 : -1;
+  const int64_t chunk = 65536;
+  const size_t entry =
+      params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;
+
+  DeviceBuf xd, xnorm, dots, asg, resid, codes;
+  std::vector<int32_t> asg_h(std::min(n, chunk));
+  std::vector<uint8_t> codes_h;
+
+  for (int64_t c0 = 0; c0 < n; c0 += chunk) {
+    int64_t cn = std::min(chunk, n - c0);
+    if (xd.reserve((size_t)cn * d_ * 4)) return -1;
+    hipMemcpy(xd.get(), x_host + (size_t)c0 * d_, (size_t)cn * d_ * 4,
+              hipMemcpyHostToDevice);
+    if (xnorm.reserve((size_t)cn * 4)) return -1;
+    gk::row_norms(s, xd.as<float>(), cn, d_, xnorm.as<float>());
+    const int64_t sub = 16384;
+    if (dots.reserve((size_t)std::min(cn, sub) * nlist_ * 4)) return -1;
+    if (asg.reserve((size_t)cn * 4)) return -1;
+    for (int64_t r0 = 0; r0 < cn; r0 += sub) {
+      int64_t rn = std::min(sub, cn - r0);
+      if (gk::dots_mfma(s, xd.as<float>() + (size_t)r0 * d_, (int)rn,
+                        centroids_.as<float>(), nlist_, d_,
+                        dots.as<float>()) != hipSuccess)
+        return -1;
+      if (gk::argmin_rows(s, rn, nlist_, dots.as<float>(),
+                          xnorm.as<float>() + r0, cent_norms_.as<float>(),
+                          !params_.metric_ip,
+                          asg.as<int32_t>() + r0) != hipSuccess)
+        return -1;
+    }
+    hipMemcpy(asg_h.data(), asg.get(), (size_t)cn * 4,
+              hipMemcpyDeviceToHost);
+
+    const uint8_t *payload_h = nullptr;
+    if (params_.kind == IndexKind::IVFPQ) {
+      if (resid.reserve((size_t)cn * d_ * 4)) return -1;
+      if (gk::residuals(s, cn, d_, xd.as<float>(), centroids_.as<float>(),
+                        asg.as<int32_t>(), resid.as<float>()) != hipSuccess)
+        return -1;
+      if (codes.reserve((size_t)cn * code_size_)) return -1;
+      if (gk::pq_encode(s, cn, d_, M_, ksub_, resid.as<float>(),
+                        codebooks_.as<float>(),
+                        codes.as<uint8_t>()) != hipSuccess)
+        return -1;
+      codes_h.resize((size_t)cn * code_size_);
+      hipMemcpy(codes_h.data(), codes.get(), (size_t)cn * code_size_,
+                hipMemcpyDeviceToHost);
+      payload_h = codes_h.data();
+    } else {
+      payload_h = (const uint8_t *)(x_host + (size_t)c0 * d_);
+    }
+
+    /* group by bucket (AddKeys analog, realtime_mem_data.cc) */
+    std::map<int32_t, std::pair<std::vector<int64_t>, std::vector<uint8_t>>>
+        groups;
+    for (int64_t i = 0; i < cn; i++) {
+      int32_t b = asg_h[i];
+      if (b < 0 || b >= nlist_) b = (int32_t)(vids[c0 + i] % nlist_);
+      auto &g = groups[b];
+      g.first.push_back(vids[c0 + i]);
+      size_t off = g.second.size();
+      g.second.resize(off + entry);
+      memcpy(g.second.data() + off, payload_h + (size_t)i * entry, entry);
+    }
+    for (auto &kv : groups) {
+      Bucket &bk = buckets_[kv.first];
+      int64_t add_n = (int64_t)kv.second.first.size();
+      if (bk.size + add_n > bk.cap) {
+        long long ncap =
+            std::max<long long>({(long long)params_.bucket_init_size,
+                                 bk.cap * 2, bk.size + add_n});
+        ncap = std::min<long long>(
+            std::max<long long>(ncap, bk.size + add_n),
+            std::max<long long>((long long)params_.bucket_max_size,
+                                bk.size + add_n));
+        auto nids = std::make_unique<DeviceBuf>();
+        auto ndata = std::make_unique<DeviceBuf>();
+        if (nids->reserve((size_t)ncap * 8)) return -1;
+        if (ndata->reserve((size_t)ncap * entry)) return -1;
+        if (bk.size > 0) {
+          hipMemcpy(nids->get(), bk.ids->get(), (size_t)bk.size * 8,
+                    hipMemcpyDeviceToDevice);
+          hipMemcpy(ndata->get(), bk.data->get(), (size_t)bk.size * entry,
+                    hipMemcpyDeviceToDevice);
+        }
+        bk.ids = std::move(nids);
+        bk.data = std::move(ndata);
+        bk.cap = ncap;
+        dev_buckets_dirty_ = true;
+      }
+      hipMemcpy(bk.ids->as<int64_t>() + bk.size, kv.second.first.data(),
+                (size_t)add_n * 8, hipMemcpyHostToDevice);
+      hipMemcpy((uint8_t *)bk.data->get() + (size_t)bk.size * entry,
+                kv.second.second.data(), (size_t)add_n * entry,
+                hipMemcpyHostToDevice);
+      for (int64_t i = 0; i < add_n; i++)
+        vid_loc_[kv.second.first[i]] = {kv.first, bk.size + i};
+      bk.size += add_n;
+      dev_buckets_dirty_ = true;
+    }
+  }
+  ntotal_ += n;
+  return update_dev_buckets(s);
+}
+
+int IVFIndex::del(int64_t vid, hipStream_t s) {
+  auto it = vid_loc_.find(vid);
+  if (it == vid_loc_.end()) return 0;
+  Bucket &bk = buckets_[it->second.first];
+  int64_t marked = vid | (int64_t)((uint64_t)1 << 63); /* kDelIdxMask */
+  if (hipMemcpy(bk.ids->as<int64_t>() + it->second.second, &marked, 8,
+                hipMemcpyHostToDevice) != hipSuccess)
+    return -1;
+  return 0;
+}
+
+int IVFIndex::coarse_assign(const float *q_dev, int nq, int nprobe, bool ip,
+                            const float *q_norms_dev, hipStream_t s,
+                            int64_t *probes_dev, float *probe_dists_dev) {
+  const int64_t sub = 16384;
+  if (scratch_dots_.reserve((size_t)std::min<int64_t>(nq, sub) * nlist_ * 4))
+    return -1;
+  if (scratch_keys_.reserve((size_t)nq * nprobe * 8)) return -1;
+  for (int64_t r0 = 0; r0 < nq; r0 += sub) {
+    int64_t rn = std::min<int64_t>(sub, nq - r0);
+    GAMMA_CHECK(gk::dots_mfma(s, q_dev + (size_t)r0 * d_, (int)rn,
+                              centroids_.as<float>(), nlist_, d_,
+                              scratch_dots_.as<float>()));
+    GAMMA_CHECK(gk::select_from_dots(
+        s, (int)rn, nlist_, 0, nlist_, scratch_dots_.as<float>(),
+        q_norms_dev + r0, cent_norms_.as<float>(), !ip, ip, nullptr, nprobe,
+        scratch_keys_.as<uint64_t>() + (size_t)r0 * nprobe, false));
+  }
+  GAMMA_CHECK(gk::unpack_keys(s, (int64_t)nq * nprobe,
+                              scratch_keys_.as<uint64_t>(), ip,
+                              probe_dists_dev, probes_dev));
+  return 0;
+}
+
+int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
+                     const uint32_t *bitmap_dev, bool metric_ip,
+                     hipStream_t s, uint64_t *out_keys_dev,
+                     const float *q_norms_dev, double *t_assign_ms,
+                     double *t_scan_ms) {
+  if (!trained_) return -1;
+  nprobe = std::min(nprobe, nlist_);
+  if (scratch_probes_.reserve((size_t)nq * nprobe * 8)) return -1;
+  if (scratch_pdists_.reserve((size_t)nq * nprobe * 4)) return -1;
+  if (update_dev_buckets(s)) return -1;
+
+  hipEvent_t e0, e1, e2;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  hipEventCreate(&e2);
+  hipEventRecord(e0, s);
+  GAMMA_CHECK(coarse_assign(q_dev, nq, nprobe, metric_ip, q_norms_dev, s,
+                            scratch_probes_.as<int64_t>(),
+                            scratch_pdists_.as<float>()));
+  hipEventRecord(e1, s);
+  if (params_.kind == IndexKind::IVFPQ) {
+    GAMMA_CHECK(gk::ivfpq_scan(s, nq, d_, M_, nprobe, k2, q_dev,
+                               centroids_.as<float>(),
+                               codebooks_.as<float>(),
+                               dev_buckets_.as<GammaBucketDev>(), nlist_,
+                               scratch_probes_.as<int64_t>(), bitmap_dev,
+                               metric_ip, out_keys_dev));
+  } else {
+    GAMMA_CHECK(gk::ivfflat_scan(s, nq, d_, nprobe, k2, q_dev,
+                                 dev_buckets_.as<GammaBucketDev>(), nlist_,
+                                 scratch_probes_.as<int64_t>(), bitmap_dev,
+                                 metric_ip, out_keys_dev));
+  }
+  hipEventRecord(e2, s);
+  GAMMA_CHECK(hipEventSynchronize(e2));
+  float ms = 0;
+  hipEventElapsedTime(&ms, e0, e1);
+  if (t_assign_ms) *t_assign_ms = ms;
+  hipEventElapsedTime(&ms, e1, e2);
+  if (t_scan_ms) *t_scan_ms = ms;
+  hipEventDestroy(e0);
+  hipEventDestroy(e1);
+  hipEventDestroy(e2);
+  return 0;
+}
+
+int64_t IVFIndex::list_size(int64_t ln) const {
+  if (ln < 0 || ln >= nlist_) return -1;
+  return buckets_[ln].size;
+}
+
+int IVFIndex::copy_list_to_host(int64_t ln, int64_t *ids, uint8_t *codes,
+                                hipStream_t s) const {
+  if (ln < 0 || ln >= nlist_) return -1;
+  const Bucket &bk = buckets_[ln];
+  const size_t entry =
+      params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;
+  if (bk.size == 0) return 0;
+  if (ids)
+    GAMMA_CHECK(hipMemcpy(ids, bk.ids->get(), (size_t)bk.size * 8,
+                          hipMemcpyDeviceToHost));
+  if (codes)
+    GAMMA_CHECK(hipMemcpy(codes, bk.data->get(), (size_t)bk.size * entry,
+                          hipMemcpyDeviceToHost));
+  return 0;
+}
+
+int IVFIndex::copy_model_to_host(float *centroids, float *codebooks,
+                                 hipStream_t s) const {
+  if (!trained_) return -1;
+  if (centroids)
+    GAMMA_CHECK(hipMemcpy(centroids, centroids_.get(),
+                          (size_t)nlist_ * d_ * 4, hipMemcpyDeviceToHost));
+  if (codebooks && params_.kind == IndexKind::IVFPQ)
+    GAMMA_CHECK(hipMemcpy(codebooks, codebooks_.get(),
+                          (size_t)M_ * ksub_ * dsub_ * 4,
+                          hipMemcpyDeviceToHost));
+  return 0;
+}
+
+int IVFIndex::dump(FILE *f, hipStream_t s) const {
+  int tr = trained_ ? 1 : 0;
+  fwrite(&tr, 4, 1, f);
+  if (!trained_) return 0;
+  std::vector<float> cent((size_t)nlist_ * d_);
+  hipMemcpy(cent.data(), centroids_.get(), cent.size() * 4,
+            hipMemcpyDeviceToHost);
+  fwrite(cent.data(), 4, cent.size(), f);
+  if (params_.kind == IndexKind::IVFPQ) {
+    std::vector<float> books((size_t)M_ * ksub_ * dsub_);
+    hipMemcpy(books.data(), codebooks_.get(), books.size() * 4,
+              hipMemcpyDeviceToHost);
+    fwrite(books.data(), 4, books.size(), f);
+  }
+  const size_t entry =
+      params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;
+  fwrite(&ntotal_, 8, 1, f);
+  for (int i = 0; i < nlist_; i++) {
+    long long sz = buckets_[i].size;
+    fwrite(&sz, 8, 1, f);
+    if (sz > 0) {
+      std::vector<int64_t> ids(sz);
+      std::vector<uint8_t> data((size_t)sz * entry);
+      hipMemcpy(ids.data(), buckets_[i].ids->get(), (size_t)sz * 8,
+                hipMemcpyDeviceToHost);
+      hipMemcpy(data.data(), buckets_[i].data->get(), (size_t)sz * entry,
+                hipMemcpyDeviceToHost);
+      fwrite(ids.data(), 8, sz, f);
+      fwrite(data.data(), 1, data.size(), f);
+    }
+  }
+  return 0;
+}
+
+int IVFIndex::load(FILE *f, hipStream_t s) {
+  int tr = 0;
+  if (fread(&tr, 4, 1, f) != 1) return -1;
+  if (!tr) return 0;
+  std::vector<float> cent((size_t)nlist_ * d_);
+  if (fread(cent.data(), 4, cent.size(), f) != cent.size()) return -1;
+  if (centroids_.reserve(cent.size() * 4)) return -1;
+  hipMemcpy(centroids_.get(), cent.data(), cent.size() * 4,
+            hipMemcpyHostToDevice);
+  if (cent_norms_.reserve((size_t)nlist_ * 4)) return -1;
+  gk::row_norms(s, centroids_.as<float>(), nlist_, d_,
+                cent_norms_.as<float>());
+  if (params_.kind == IndexKind::IVFPQ) {
+    std::vector<float> books((size_t)M_ * ksub_ * dsub_);
+    if (fread(books.data(), 4, books.size(), f) != books.size()) return -1;
+    if (codebooks_.reserve(books.size() * 4)) return -1;
+    hipMemcpy(codebooks_.get(), books.data(), books.size() * 4,
+              hipMemcpyHostToDevice);
+  }
+  const size_t entry =
+      params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;
+  if (fread(&ntotal_, 8, 1, f) != 1) return -1;
+  for (int i = 0; i < nlist_; i++) {
+    long long sz = 0;
+    if (fread(&sz, 8, 1, f) != 1) return -1;
+    Bucket &bk = buckets_[i];
+    bk.size = 0;
+    bk.cap = 0;
+    if (sz > 0) {
+      std::vector<int64_t> ids(sz);
+      std::vector<uint8_t> data((size_t)sz * entry);
+      if (fread(ids.data(), 8, sz, f) != (size_t)sz) return -1;
+      if (fread(data.data(), 1, data.size(), f) != data.size()) return -1;
+      bk.ids = std::make_unique<DeviceBuf>();
+      bk.data = std::make_unique<DeviceBuf>();
+      if (bk.ids->reserve((size_t)sz * 8)) return -1;
+      if (bk.data->reserve(data.size())) return -1;
+      hipMemcpy(bk.ids->get(), ids.data(), (size_t)sz * 8,
+                hipMemcpyHostToDevice);
+      hipMemcpy(bk.data->get(), data.data(), data.size(),
+                hipMemcpyHostToDevice);
+      bk.size = bk.cap = sz;
+      for (long long j = 0; j < sz; j++) {
+        int64_t vid = ids[j];
+        if (!((uint64_t)vid >> 63)) vid_loc_[vid] = {i, j};
+      }
+    }
+  }
+  dev_buckets_dirty_ = true;
+  trained_ = true;
+  return update_dev_buckets(s);
+}
+
+}  // namespace vgamma

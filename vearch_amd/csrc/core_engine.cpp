@@ -1,0 +1,486 @@
+/*
+ * core_engine.cpp — Engine: the thin reimplementation of
+ * search/engine.cc + vector/vector_manager.cc above the index models.
+ */
+#include <math.h>
+#include <sys/stat.h>
+
+#include <algorithm>
+
+#include "core.hpp"
+#include "json.hpp"
+
+namespace vgamma {
+
+static const std::string kEmpty;
+
+int Engine::init(const std::string &config_json, std::string *err) {
+  /* engine config JSON: path, log_dir, space_name (gamma_api.cc:36-70) */
+  gjson::Value v;
+  if (!config_json.empty() && !gjson::parse(config_json, v)) {
+    if (err) *err = "bad engine config json";
+    return -1;
+  }
+  v.get_str("path", path_);
+  v.get_str("log_dir", log_dir_);
+  v.get_str("space_name", space_name_);
+  if (path_.empty()) path_ = ".";
+  int dev_count = 0;
+  if (hipGetDeviceCount(&dev_count) != hipSuccess || dev_count == 0) {
+    if (err) *err = "no HIP device: the MI355X engine requires a GPU";
+    return -1;
+  }
+  GAMMA_CHECK(hipStreamCreate(&stream_));
+  return 0;
+}
+
+bool Engine::metric_ip_default() const { return params_.metric_ip; }
+
+int Engine::create_table(const std::string &name,
+                         const std::vector<FieldMeta> &scalar_fields,
+                         const std::string &vec_name, int dimension,
+                         const std::string &index_type,
+                         const std::string &index_params_json,
+                         int training_threshold, std::string *err) {
+  if (table_created_) {
+    if (err) *err = "table already created";
+    return -1;
+  }
+  if (dimension <= 0 || dimension % 4 != 0) {
+    if (err) *err = "dimension must be a positive multiple of 4";
+    return -1;
+  }
+  space_name_ = space_name_.empty() ? name : space_name_;
+  fields_ = scalar_fields;
+  for (auto &f : fields_) field_vals_[f.name] = {};
+  vec_name_ = vec_name;
+  dim_ = dimension;
+  index_type_ = index_type.empty() ? "IVFPQ" : index_type;
+
+  if (params_.parse(index_params_json, err)) return -1;
+  if (index_type_ == "IVFPQ") params_.kind = IndexKind::IVFPQ;
+  else if (index_type_ == "IVFFLAT") params_.kind = IndexKind::IVFFLAT;
+  else if (index_type_ == "FLAT") params_.kind = IndexKind::FLAT;
+  else {
+    if (err)
+      *err = "index type " + index_type_ +
+             " not in this build's scope (FLAT/IVFFLAT/IVFPQ)";
+    return -1;
+  }
+  if (raw_.init(dim_)) return -1;
+
+  if (params_.kind != IndexKind::FLAT) {
+    index_ = std::make_unique<IVFIndex>();
+    if (index_->init(dim_, params_)) {
+      if (err) *err = "index init failed (check nsubvector divides d, %4)";
+      return -1;
+    }
+  }
+  /* training_threshold default: max(nlist*39, 256*39) vectors
+   * (ivfpq.cc:137-144 with default_points_per_centroid=39) */
+  if (training_threshold > 0) training_threshold_ = training_threshold;
+  else if (params_.training_threshold > 0)
+    training_threshold_ = params_.training_threshold;
+  else
+    training_threshold_ = std::max(params_.ncentroids * 39, 256 * 39);
+  table_created_ = true;
+  return 0;
+}
+
+int Engine::add_doc(
+    const std::string &p_key,
+    const std::vector<std::pair<std::string, std::string>> &fields,
+    const float *vec, int vec_len) {
+  if (!table_created_ || vec_len != dim_) return -1;
+  std::unique_lock<std::shared_mutex> g(rw_);
+  auto it = pkey2docid_.find(p_key);
+  if (it != pkey2docid_.end()) {
+    /* update = delete old + add new (engine.cc AddOrUpdate semantics) */
+    int64_t old = it->second;
+    bitmap_.set(old, stream_);
+    if (index_) index_->del(old, stream_);
+  }
+  int64_t docid = max_docid_++;
+  pkey2docid_[p_key] = docid;
+  docid2pkey_.push_back(p_key);
+  bitmap_.ensure(max_docid_, stream_);
+  for (auto &f : fields_) {
+    auto &col = field_vals_[f.name];
+    col.resize(docid + 1);
+    for (auto &kv : fields)
+      if (kv.first == f.name) col[docid] = kv.second;
+  }
+  if (raw_.add(vec, 1, stream_)) return -1;
+  if (index_ && index_->trained()) {
+    if (index_->add(vec, &docid, 1, stream_)) return -1;
+    indexed_count_++;
+  }
+  return 0;
+}
+
+int Engine::bulk_add(int64_t n, const float *vecs) {
+  if (!table_created_ || n <= 0) return -1;
+  std::unique_lock<std::shared_mutex> g(rw_);
+  int64_t base = max_docid_;
+  if ((uint64_t)(base + n) > 0xffffffffull) return -1; /* u32 vid space */
+  docid2pkey_.reserve(base + n);
+  char buf[24];
+  for (int64_t i = 0; i < n; i++) {
+    int64_t docid = base + i;
+    snprintf(buf, sizeof buf, "%lld", (long long)docid);
+    pkey2docid_.emplace(buf, docid);
+    docid2pkey_.push_back(buf);
+  }
+  max_docid_ += n;
+  bitmap_.ensure(max_docid_, stream_);
+  if (raw_.add(vecs, n, stream_)) return -1;
+  if (index_ && index_->trained()) {
+    std::vector<int64_t> vids(n);
+    for (int64_t i = 0; i < n; i++) vids[i] = base + i;
+    if (index_->add(vecs, vids.data(), n, stream_)) return -1;
+    indexed_count_ += n;
+  }
+  return 0;
+}
+
+int Engine::delete_doc(const std::string &p_key) {
+  std::unique_lock<std::shared_mutex> g(rw_);
+  auto it = pkey2docid_.find(p_key);
+  if (it == pkey2docid_.end()) return -1;
+  bitmap_.set(it->second, stream_);
+  if (index_) index_->del(it->second, stream_);
+  pkey2docid_.erase(it);
+  return 0;
+}
+
+int Engine::build_index(std::string *err) {
+  if (!table_created_) {
+    if (err) *err = "no table";
+    return -1;
+  }
+  if (params_.kind == IndexKind::FLAT) return 0;
+  std::unique_lock<std::shared_mutex> g(rw_);
+  if (index_->trained()) return 0;
+  /* training size clamp (ivfpq.cc:296-329): [nlist*39, nlist*256] */
+  int64_t nlist = params_.ncentroids;
+  int64_t num = training_threshold_;
+  if (num < nlist) num = nlist * 39;
+  else if (num > nlist * 256) num = nlist * 256;
+  else if (num < nlist * 39) { /* warning case, keep as is */
+  }
+  if (num > raw_.size()) {
+    if (err)
+      *err = "vector count " + std::to_string(raw_.size()) +
+             " less than training threshold " + std::to_string(num);
+    return -1;
+  }
+  std::vector<float> xt((size_t)num * dim_);
+  raw_.host_copy(0, num, xt.data());
+  if (index_->train(xt.data(), num, stream_, err)) return -1;
+  /* index every existing vector (Indexing/AddRTVecsToIndex loop,
+   * engine.cc:1084-1127) */
+  const int64_t chunk = 262144;
+  std::vector<float> buf;
+  std::vector<int64_t> vids;
+  for (int64_t c0 = 0; c0 < raw_.size(); c0 += chunk) {
+    int64_t cn = std::min(chunk, raw_.size() - c0);
+    buf.resize((size_t)cn * dim_);
+    raw_.host_copy(c0, cn, buf.data());
+    vids.resize(cn);
+    for (int64_t i = 0; i < cn; i++) vids[i] = c0 + i;
+    if (index_->add(buf.data(), vids.data(), cn, stream_)) return -1;
+  }
+  indexed_count_ = raw_.size();
+  return 0;
+}
+
+int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
+                             const float *q_norms_dev, bool ip,
+                             hipStream_t s, uint64_t *out_keys_dev) {
+  const int64_t n = raw_.size();
+  if (nq < 512 || n < 200000) {
+    GAMMA_CHECK(gk::flat_stream_scan(s, nq, n, dim_, k2, q_dev,
+                                     raw_.dev_seg_table(),
+                                     RawStore::SEG_SHIFT, bitmap_.dev(), ip,
+                                     out_keys_dev));
+    return 0;
+  }
+  /* chunked MFMA GEMM + seeded select (nq large): per segment-run chunks */
+  const int64_t chunk = 65536;
+  if (flat_dots_.reserve((size_t)nq * chunk * 4)) return -1;
+  bool seeded = false;
+  for (int64_t v0 = 0; v0 < n;) {
+    int64_t run = 0;
+    const float *seg = raw_.dev_run(v0, &run);
+    int64_t take = std::min(run, chunk);
+    GAMMA_CHECK(gk::dots_mfma(s, q_dev, nq, seg, take, dim_,
+                              flat_dots_.as<float>()));
+    GAMMA_CHECK(gk::select_from_dots(s, nq, take, v0, take,
+                                     flat_dots_.as<float>(), q_norms_dev,
+                                     raw_.dev_norms(), !ip, ip,
+                                     bitmap_.dev(), k2, out_keys_dev,
+                                     seeded));
+    seeded = true;
+    v0 += take;
+  }
+  return 0;
+}
+
+int Engine::search(int nq, const float *xq, int k, int nprobe,
+                   int recall_num, int metric, bool brute_force,
+                   const std::string &request_id, float *out_dists,
+                   int64_t *out_ids, bool l2_sqrt) {
+  if (!table_created_ || nq <= 0 || k <= 0) return -1;
+  std::shared_lock<std::shared_mutex> g(rw_);
+  const int pid = 0; /* partition id is carried by the kill key */
+  bool ip = metric == 0 ? params_.metric_ip : (metric == 2);
+  hipStream_t s = stream_;
+
+  struct Timer {
+    hipEvent_t ev[6];
+    hipStream_t s;
+    explicit Timer(hipStream_t s_) : s(s_) {
+      for (auto &e : ev) hipEventCreate(&e);
+    }
+    ~Timer() {
+      for (auto &e : ev) hipEventDestroy(e);
+    }
+    void rec(int i) { hipEventRecord(ev[i], s); }
+    double ms(int a, int b) {
+      float m = 0;
+      hipEventElapsedTime(&m, ev[a], ev[b]);
+      return m;
+    }
+  } tm(s);
+
+  if (KillRegistry::inst().killed(request_id, pid)) return -2;
+
+  /* k2: scan-phase candidates; recall_num>k -> rerank leg
+   * (search_preassigned, ivfpq.cc:765-776) */
+  int k2 = std::max(k, recall_num);
+  bool rerank = recall_num > 0;
+  if (k2 > 1536) return -1;
+
+  tm.rec(0);
+  if (q_dev_.reserve((size_t)nq * dim_ * 4)) return -1;
+  GAMMA_CHECK(hipMemcpyAsync(q_dev_.get(), xq, (size_t)nq * dim_ * 4,
+                             hipMemcpyHostToDevice, s));
+  if (q_norms_dev_.reserve((size_t)nq * 4)) return -1;
+  GAMMA_CHECK(gk::row_norms(s, q_dev_.as<float>(), nq, dim_,
+                            q_norms_dev_.as<float>()));
+  if (keys_dev_.reserve((size_t)nq * k2 * 8)) return -1;
+  if (out_d_dev_.reserve((size_t)nq * k * 4)) return -1;
+  if (out_i_dev_.reserve((size_t)nq * k * 8)) return -1;
+  tm.rec(1);
+
+  bool use_flat = brute_force || params_.kind == IndexKind::FLAT ||
+                  !index_ || !index_->trained();
+  double t_assign = 0, t_scan = 0;
+  bool need_canonical_rerank = true; /* FLAT & IVFFLAT canonicalize */
+  if (use_flat) {
+    /* FLAT margin so GEMM-order rounding cannot evict a true top-k
+     * member before the canonical re-rank (DESIGN.md numerics note) */
+    int kf = std::min<int64_t>((int64_t)k2 + 64, 1536);
+    kf = (int)std::min<int64_t>(kf, std::max<int64_t>(raw_.size(), 1));
+    if (keys_dev_.reserve((size_t)nq * kf * 8)) return -1;
+    if (flat_search_keys(q_dev_.as<float>(), nq, kf,
+                         q_norms_dev_.as<float>(), ip, s,
+                         keys_dev_.as<uint64_t>()))
+      return -1;
+    k2 = kf;
+  } else {
+    if (index_->search(q_dev_.as<float>(), nq, k2, nprobe, bitmap_.dev(),
+                       ip, s, keys_dev_.as<uint64_t>(),
+                       q_norms_dev_.as<float>(), &t_assign, &t_scan))
+      return -1;
+    /* ADC distances already match the oracle bit-for-bit; canonical
+     * re-rank only when the caller asked for the exact rerank leg */
+    need_canonical_rerank =
+        rerank || params_.kind == IndexKind::IVFFLAT;
+  }
+  tm.rec(2);
+  if (KillRegistry::inst().killed(request_id, pid)) return -2;
+
+  if (need_canonical_rerank) {
+    GAMMA_CHECK(gk::rerank(s, nq, k2, dim_, q_dev_.as<float>(),
+                           raw_.dev_seg_table(), RawStore::SEG_SHIFT, ip,
+                           keys_dev_.as<uint64_t>(),
+                           keys_dev_.as<uint64_t>()));
+  }
+  GAMMA_CHECK(gk::sort_rows(s, nq, k2, k, keys_dev_.as<uint64_t>(), ip,
+                            out_d_dev_.as<float>(),
+                            out_i_dev_.as<int64_t>()));
+  tm.rec(3);
+  GAMMA_CHECK(hipMemcpyAsync(out_dists, out_d_dev_.get(),
+                             (size_t)nq * k * 4, hipMemcpyDeviceToHost, s));
+  GAMMA_CHECK(hipMemcpyAsync(out_ids, out_i_dev_.get(), (size_t)nq * k * 8,
+                             hipMemcpyDeviceToHost, s));
+  tm.rec(4);
+  GAMMA_CHECK(hipStreamSynchronize(s));
+
+  if (l2_sqrt && !ip) {
+    for (int64_t i = 0; i < (int64_t)nq * k; i++)
+      if (out_ids[i] >= 0) out_dists[i] = sqrtf(out_dists[i]);
+  }
+  last_timing[0] = tm.ms(0, 1) * 1000.0;
+  last_timing[1] = (use_flat ? 0.0 : t_assign) * 1000.0;
+  last_timing[2] = (use_flat ? tm.ms(1, 2) : t_scan) * 1000.0;
+  last_timing[3] = tm.ms(2, 3) * 1000.0;
+  last_timing[4] = tm.ms(3, 4) * 1000.0;
+  last_timing[5] = tm.ms(0, 4) * 1000.0;
+  return 0;
+}
+
+int64_t Engine::docid_of(const std::string &p_key) const {
+  auto it = pkey2docid_.find(p_key);
+  return it == pkey2docid_.end() ? -1 : it->second;
+}
+
+const std::string &Engine::pkey_of(int64_t docid) const {
+  if (docid < 0 || docid >= (int64_t)docid2pkey_.size()) return kEmpty;
+  return docid2pkey_[docid];
+}
+
+const std::string *Engine::field_value(int64_t docid,
+                                       const std::string &f) const {
+  auto it = field_vals_.find(f);
+  if (it == field_vals_.end()) return nullptr;
+  if (docid < 0 || docid >= (int64_t)it->second.size()) return nullptr;
+  return &it->second[docid];
+}
+
+std::string Engine::status_json() const {
+  char buf[512];
+  snprintf(buf, sizeof buf,
+           "{\"doc_count\": %lld, \"index_status\": %d, \"min_indexed_num\": "
+           "%lld, \"max_docid\": %lld, \"table_name\": \"%s\"}",
+           (long long)(max_docid_ - bitmap_.popcount()),
+           (index_ && index_->trained()) ? 2 : 0,
+           (long long)indexed_count_, (long long)(max_docid_ - 1),
+           gjson::escape(space_name_).c_str());
+  return buf;
+}
+
+static const uint32_t kDumpMagic = 0x47414D41; /* "GAMA" */
+
+int Engine::dump(std::string *err) {
+  std::unique_lock<std::shared_mutex> g(rw_);
+  mkdir(path_.c_str(), 0755);
+  std::string fn = path_ + "/gamma.dump";
+  FILE *f = fopen((fn + ".tmp").c_str(), "wb");
+  if (!f) {
+    if (err) *err = "cannot open dump file " + fn;
+    return -1;
+  }
+  fwrite(&kDumpMagic, 4, 1, f);
+  int ver = 1;
+  fwrite(&ver, 4, 1, f);
+  auto wstr = [&](const std::string &s_) {
+    int64_t n = (int64_t)s_.size();
+    fwrite(&n, 8, 1, f);
+    fwrite(s_.data(), 1, n, f);
+  };
+  wstr(space_name_);
+  wstr(index_type_);
+  wstr(vec_name_);
+  fwrite(&dim_, 4, 1, f);
+  fwrite(&training_threshold_, 4, 1, f);
+  fwrite(&max_docid_, 8, 1, f);
+  int nfields = (int)fields_.size();
+  fwrite(&nfields, 4, 1, f);
+  for (auto &fm : fields_) {
+    wstr(fm.name);
+    fwrite(&fm.data_type, 4, 1, f);
+  }
+  for (int64_t i = 0; i < max_docid_; i++) wstr(docid2pkey_[i]);
+  for (auto &fm : fields_) {
+    auto &col = field_vals_.at(fm.name);
+    for (int64_t i = 0; i < max_docid_; i++)
+      wstr(i < (int64_t)col.size() ? col[i] : kEmpty);
+  }
+  raw_.dump(f);
+  bitmap_.dump(f);
+  int has_index = index_ ? 1 : 0;
+  fwrite(&has_index, 4, 1, f);
+  if (index_) index_->dump(f, stream_);
+  fclose(f);
+  rename((fn + ".tmp").c_str(), fn.c_str());
+  return 0;
+}
+
+int Engine::load(std::string *err) {
+  std::unique_lock<std::shared_mutex> g(rw_);
+  std::string fn = path_ + "/gamma.dump";
+  FILE *f = fopen(fn.c_str(), "rb");
+  if (!f) {
+    /* nothing dumped yet: reference Load returns ok on empty dir */
+    return 0;
+  }
+  uint32_t magic = 0;
+  int ver = 0;
+  if (fread(&magic, 4, 1, f) != 1 || magic != kDumpMagic ||
+      fread(&ver, 4, 1, f) != 1) {
+    fclose(f);
+    if (err) *err = "bad dump file";
+    return -1;
+  }
+  auto rstr = [&](std::string &s_) {
+    int64_t n = 0;
+    if (fread(&n, 8, 1, f) != 1 || n < 0 || n > (int64_t)1 << 30) return -1;
+    s_.resize(n);
+    if (n && fread(&s_[0], 1, n, f) != (size_t)n) return -1;
+    return 0;
+  };
+  std::string idx_type, vecn;
+  if (rstr(space_name_) || rstr(idx_type) || rstr(vecn)) {
+    fclose(f);
+    return -1;
+  }
+  int dim = 0, tt = 0;
+  int64_t maxdoc = 0;
+  fread(&dim, 4, 1, f);
+  fread(&tt, 4, 1, f);
+  fread(&maxdoc, 8, 1, f);
+  int nfields = 0;
+  fread(&nfields, 4, 1, f);
+  std::vector<FieldMeta> fms(nfields);
+  for (auto &fm : fms) {
+    if (rstr(fm.name)) { fclose(f); return -1; }
+    fread(&fm.data_type, 4, 1, f);
+  }
+  if (!table_created_) {
+    /* table should have been created before Load (reference flow);
+     * recreate from the dump if not */
+    std::string e2;
+    if (create_table(space_name_, fms, vecn, dim, idx_type, "", tt, &e2)) {
+      fclose(f);
+      if (err) *err = e2;
+      return -1;
+    }
+  }
+  max_docid_ = maxdoc;
+  docid2pkey_.resize(maxdoc);
+  for (int64_t i = 0; i < maxdoc; i++) {
+    if (rstr(docid2pkey_[i])) { fclose(f); return -1; }
+    pkey2docid_[docid2pkey_[i]] = i;
+  }
+  for (auto &fm : fms) {
+    auto &col = field_vals_[fm.name];
+    col.resize(maxdoc);
+    for (int64_t i = 0; i < maxdoc; i++)
+      if (rstr(col[i])) { fclose(f); return -1; }
+  }
+  if (raw_.load(f, stream_)) { fclose(f); return -1; }
+  if (bitmap_.load(f, stream_)) { fclose(f); return -1; }
+  bitmap_.ensure(max_docid_, stream_);
+  int has_index = 0;
+  fread(&has_index, 4, 1, f);
+  if (has_index && index_) {
+    if (index_->load(f, stream_)) { fclose(f); return -1; }
+    indexed_count_ = index_->ntotal();
+  }
+  fclose(f);
+  return 0;
+}
+
+}  // namespace vgamma

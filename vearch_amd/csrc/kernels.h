@@ -1,0 +1,92 @@
+/* kernels.h — host API of the gfx950 HIP kernels (kernels.hip). */
+#pragma once
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+/* One IVF bucket as the device sees it (RT list storage, SURVEY §8 a7;
+ * semantics of realtime_mem_data.cc:57-68: append-only SoA, ids with
+ * bit-63 delete mask). For IVFPQ `data` is uint8 codes (code_size per
+ * entry); for IVFFLAT it is fp32 vectors (d floats per entry). */
+struct GammaBucketDev {
+  const int64_t *ids;
+  const void *data;
+  long long size;
+};
+
+namespace gk {
+
+/* out[i*n+j] = dot(Q_i, B_j); f32 MFMA 16x16x4. */
+hipError_t dots_mfma(hipStream_t s, const float *Q, int nq, const float *B,
+                     int64_t n, int d, float *out);
+
+/* norms[i] = sum_j v[i][j]^2, canonical sequential fmaf. */
+hipError_t row_norms(hipStream_t s, const float *v, int64_t n, int d,
+                     float *norms);
+
+/* Exact top-k2 select per row of a dots matrix.
+ * dist = l2 ? (qn[q] + bn[col_base+c] - 2*dot) : dot; id = col_base + c.
+ * bitmap: optional delete bitmap over global ids (u32 words).
+ * state_keys (nq x k2 u64) is seeded from itself when `seeded`. */
+hipError_t select_from_dots(hipStream_t s, int nq, int64_t ncols,
+                            int64_t col_base, int64_t ld, const float *dots,
+                            const float *qnorms, const float *bnorms,
+                            bool l2, bool ip_order, const uint32_t *bitmap,
+                            int k2, uint64_t *state_keys, bool seeded);
+
+/* argmin over each row (ties -> lowest col). out int32[nrows]. */
+hipError_t argmin_rows(hipStream_t s, int64_t nrows, int ncols,
+                       const float *dots, const float *qnorms,
+                       const float *bnorms, bool l2, int32_t *out);
+
+/* IVFPQ fused search: one workgroup per query; per probed list builds the
+ * residual ADC table in LDS (gamma_index_ivfpq.h:243-249) and scans the
+ * list (h:923-953). out_keys: nq x k2. */
+hipError_t ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
+                      int k2, const float *queries, const float *centroids,
+                      const float *codebooks, const GammaBucketDev *buckets,
+                      int nlist, const int64_t *probes,
+                      const uint32_t *bitmap, bool ip, uint64_t *out_keys);
+
+/* IVFFLAT fused search (gamma_index_ivfflat.h:36-91). */
+hipError_t ivfflat_scan(hipStream_t s, int nq, int d, int nprobe, int k2,
+                        const float *queries, const GammaBucketDev *buckets,
+                        int nlist, const int64_t *probes,
+                        const uint32_t *bitmap, bool ip, uint64_t *out_keys);
+
+/* Canonical-order exact re-rank (ivfpq.cc:675-726 + parity canonicalizer):
+ * for each key in keys_in (nq x ncand), gather the raw vector through the
+ * segment table and recompute the distance with the sequential fmaf loop.
+ * keys_out may alias keys_in. */
+hipError_t rerank(hipStream_t s, int nq, int ncand, int d,
+                  const float *queries, const float *const *segs,
+                  int seg_shift, bool ip, const uint64_t *keys_in,
+                  uint64_t *keys_out);
+
+/* Per-row sort of ncand keys (<= 2048), emit top-k (dists, ids). */
+hipError_t sort_rows(hipStream_t s, int nq, int ncand, int k,
+                     const uint64_t *keys, bool ip, float *out_dists,
+                     int64_t *out_ids);
+
+/* unpack keys -> (dists, ids) without sorting (keys already sorted). */
+hipError_t unpack_keys(hipStream_t s, int64_t n, const uint64_t *keys,
+                       bool ip, float *out_dists, int64_t *out_ids);
+
+/* residuals[i] = x[i] - centroids[assign[i]] (elementwise exact). */
+hipError_t residuals(hipStream_t s, int64_t n, int d, const float *x,
+                     const float *centroids, const int32_t *assign,
+                     float *out);
+
+/* PQ encode: codes[i][m] = argmin_j ||x_sub - codebook[m][j]||^2
+ * (pq.compute_codes, ivfpq.cc:494; ties -> lowest j). */
+hipError_t pq_encode(hipStream_t s, int64_t n, int d, int M, int ksub,
+                     const float *x, const float *codebooks, uint8_t *codes);
+
+/* FLAT brute-force over the raw-vector segment table (no materialized
+ * dist matrix): per query workgroup streams all n vectors. Used for small
+ * nq; large nq goes through dots_mfma chunks + select_from_dots. */
+hipError_t flat_stream_scan(hipStream_t s, int nq, int64_t n, int d, int k2,
+                            const float *queries, const float *const *segs,
+                            int seg_shift, const uint32_t *bitmap, bool ip,
+                            uint64_t *out_keys);
+
+}  // namespace gk

@@ -1,0 +1,652 @@
+/*
+ * kernels.hip — hand-written gfx950 (CDNA4) kernels for the Gamma hot
+ * path. No CUDA shims, no hipify output: wave64, MFMA, LDS-staged tables.
+ *
+ * Parity contract (DESIGN.md "Parity model"): every accumulation whose
+ * value is returned to the caller uses the same sequential fmaf/add order
+ * as oracle/ref_scan.c. Approximate orders (MFMA GEMM, shuffle-tree
+ * reductions) are only used for candidate *selection* and are always
+ * followed by the canonical re-rank.
+ */
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#include "kernels.h"
+#include "select.hpp"
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define WG 256  /* 4 waves of 64 */
+
+/* ------------------------------------------------------------------ norms */
+__global__ void k_row_norms(const float *__restrict__ v, int64_t n, int d,
+                            float *__restrict__ norms) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const float *row = v + i * d;
+  float acc = 0.0f;
+  for (int j = 0; j < d; j++) acc = fmaf(row[j], row[j], acc);
+  norms[i] = acc;
+}
+
+hipError_t gk::row_norms(hipStream_t s, const float *v, int64_t n, int d,
+                         float *norms) {
+  if (n == 0) return hipSuccess;
+  int64_t blocks = (n + WG - 1) / WG;
+  k_row_norms<<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(v, n, d, norms);
+  return hipGetLastError();
+}
+
+/* ------------------------------------------------------- MFMA dot GEMM
+ * D[i][j] = dot(Q_i, B_j) on v_mfma_f32_16x16x4_f32 (exact f32 chain,
+ * 155 TF ceiling — guide §3). One workgroup = 4 waves stacked over 64
+ * query rows x 16 base columns; k-loop over d in steps of 4.
+ * Lane maps (guide §3): A[i=l&15][k=l>>4], B[k=l>>4][j=l&15];
+ * D: col=l&15, row=(l>>4)*4+reg. */
+__global__ void __launch_bounds__(WG)
+k_dots_mfma(const float *__restrict__ Q, int nq,
+            const float *__restrict__ B, int64_t n, int d,
+            float *__restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t col0 = (int64_t)blockIdx.x * 16;
+  const int row0 = blockIdx.y * 64 + wave * 16;
+
+  const int a_row = row0 + (lane & 15);     /* query row this lane loads */
+  const int64_t b_row = col0 + (lane & 15); /* base row this lane loads */
+  const int kk0 = lane >> 4;                /* k sub-index 0..3 */
+  const bool a_ok = a_row < nq;
+  const bool b_ok = b_row < n;
+  const float *qa = Q + (int64_t)(a_ok ? a_row : 0) * d;
+  const float *bb = B + (b_ok ? b_row : 0) * d;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int kk = kk0; kk < d; kk += 4) {
+    float a = a_ok ? qa[kk] : 0.0f;
+    float b = b_ok ? bb[kk] : 0.0f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  }
+  const int64_t col = col0 + (lane & 15);
+  if (col >= n) return;
+#pragma unroll
+  for (int r = 0; r < 4; r++) {
+    int qi = row0 + (lane >> 4) * 4 + r;
+    if (qi < nq) out[(int64_t)qi * n + col] = acc[r];
+  }
+}
+
+hipError_t gk::dots_mfma(hipStream_t s, const float *Q, int nq,
+                         const float *B, int64_t n, int d, float *out) {
+  if (nq == 0 || n == 0) return hipSuccess;
+  dim3 grid((uint32_t)((n + 15) / 16), (uint32_t)((nq + 63) / 64));
+  k_dots_mfma<<<grid, dim3(WG), 0, s>>>(Q, nq, B, n, d, out);
+  return hipGetLastError();
+}
+
+/* -------------------------------------------------- select from dist rows */
+template <bool IP>
+__global__ void __launch_bounds__(WG)
+k_select_from_dots(int nq, int64_t ncols, int64_t col_base, int64_t ld,
+                   const float *__restrict__ dots,
+                   const float *__restrict__ qnorms,
+                   const float *__restrict__ bnorms, int l2,
+                   const uint32_t *__restrict__ bitmap, int k2,
+                   uint64_t *__restrict__ state_keys, int seeded) {
+  extern __shared__ char smem[];
+  uint64_t *sortbuf = (uint64_t *)smem;
+  uint64_t *res = sortbuf + GAMMA_SORT_CAP;
+  int *state = (int *)(res + k2);
+
+  const int q = blockIdx.x;
+  if (q >= nq) return;
+  const float *row = dots + (int64_t)q * ld;
+  const float qn = l2 ? qnorms[q] : 0.0f;
+
+  GammaSelector sel;
+  sel.init(sortbuf, res, state, k2);
+  if (seeded) sel.seed(state_keys + (int64_t)q * k2, k2);
+
+  const int64_t chunk = (int64_t)blockDim.x * GAMMA_SEL_CHUNK;
+  for (int64_t c0 = 0; c0 < ncols; c0 += chunk) {
+    int64_t cend = min(c0 + chunk, ncols);
+    for (int64_t c = c0 + threadIdx.x; c < cend; c += blockDim.x) {
+      int64_t id = col_base + c;
+      if (gamma_bitmap_test(bitmap, (uint64_t)id)) continue;
+      float dot = row[c];
+      float dist = l2 ? (qn + bnorms[id] - 2.0f * dot) : dot;
+      sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
+    }
+    sel.maybe_flush();
+  }
+  sel.finish();
+  for (int i = threadIdx.x; i < k2; i += blockDim.x)
+    state_keys[(int64_t)q * k2 + i] = res[i];
+}
+
+hipError_t gk::select_from_dots(hipStream_t s, int nq, int64_t ncols,
+                                int64_t col_base, int64_t ld,
+                                const float *dots, const float *qnorms,
+                                const float *bnorms, bool l2, bool ip_order,
+                                const uint32_t *bitmap, int k2,
+                                uint64_t *state_keys, bool seeded) {
+  size_t smem = (GAMMA_SORT_CAP + k2) * 8 + 4 * sizeof(int);
+  if (ip_order)
+    k_select_from_dots<true><<<dim3(nq), dim3(WG), smem, s>>>(
+        nq, ncols, col_base, ld, dots, qnorms, bnorms, l2 ? 1 : 0, bitmap,
+        k2, state_keys, seeded ? 1 : 0);
+  else
+    k_select_from_dots<false><<<dim3(nq), dim3(WG), smem, s>>>(
+        nq, ncols, col_base, ld, dots, qnorms, bnorms, l2 ? 1 : 0, bitmap,
+        k2, state_keys, seeded ? 1 : 0);
+  return hipGetLastError();
+}
+
+/* --------------------------------------------------------------- argmin */
+__global__ void k_argmin_rows(int64_t nrows, int ncols,
+                              const float *__restrict__ dots,
+                              const float *__restrict__ qnorms,
+                              const float *__restrict__ bnorms, int l2,
+                              int32_t *__restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= nrows) return;
+  const float *row = dots + i * ncols;
+  float qn = l2 ? qnorms[i] : 0.0f;
+  float best = INFINITY;
+  int bestj = 0;
+  for (int j = 0; j < ncols; j++) {
+    float v = l2 ? (qn + bnorms[j] - 2.0f * row[j]) : -row[j];
+    if (v < best) { best = v; bestj = j; }
+  }
+  out[i] = bestj;
+}
+
+hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
+                           const float *dots, const float *qnorms,
+                           const float *bnorms, bool l2, int32_t *out) {
+  if (nrows == 0) return hipSuccess;
+  int64_t blocks = (nrows + WG - 1) / WG;
+  k_argmin_rows<<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(
+      nrows, ncols, dots, qnorms, bnorms, l2 ? 1 : 0, out);
+  return hipGetLastError();
+}
+
+/* ------------------------------------------------------- IVFPQ fused scan
+ * One workgroup per query (grid fills 256 CUs at nq >= ~512; at the
+ * north-star nq=10k it is ~40 rounds of blocks per CU).
+ * L2 by-residual (use_precomputed_table=0, ivfpq.cc:196): per probed list
+ * build T[m][j] = ||r_m - cw_mj||^2 in LDS, dis0 = 0.
+ * IP: one query-level table (h:164-167), dis0 = dot(q, c_list) (h:223-236,
+ * canonical order on one thread).
+ * Scan (h:923-953): skip bit-63-deleted ids and bitmap-deleted docs,
+ * dis = dis0 + sum_m T[m][code_m] in ascending m (plain adds — matches
+ * oracle_adc_scan_list bit-for-bit). */
+template <bool IP>
+__global__ void __launch_bounds__(WG)
+k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
+             const float *__restrict__ queries,
+             const float *__restrict__ centroids,
+             const float *__restrict__ codebooks,
+             const GammaBucketDev *__restrict__ buckets, int nlist,
+             const int64_t *__restrict__ probes,
+             const uint32_t *__restrict__ bitmap,
+             uint64_t *__restrict__ out_keys) {
+  extern __shared__ char smem[];
+  const int ksub = 256;
+  const int dsub = d / M;
+  float *lut = (float *)smem;                       /* M*ksub */
+  uint64_t *sortbuf = (uint64_t *)(smem + ((size_t)M * ksub * 4 + 7) / 8 * 8);
+  uint64_t *res = sortbuf + GAMMA_SORT_CAP;
+  float *qs = (float *)(res + k2);                  /* d */
+  float *rs = qs + d;                               /* d residual */
+  float *dis0s = rs + d;                            /* 1 */
+  int *state = (int *)(dis0s + 1) + 1;              /* int[2], 8B aligned-ish */
+
+  const int q = blockIdx.x;
+  if (q >= nq) return;
+  const float *qg = queries + (int64_t)q * d;
+  for (int i = threadIdx.x; i < d; i += blockDim.x) qs[i] = qg[i];
+
+  GammaSelector sel;
+  sel.init(sortbuf, res, state, k2);  /* has the barrier qs needs */
+
+  if (IP) {
+    /* query-level table T[m][j] = dot(q_m, cw_mj), canonical per entry */
+    for (int e = threadIdx.x; e < M * ksub; e += blockDim.x) {
+      int m = e >> 8, j = e & 255;
+      const float *cw = codebooks + ((size_t)m * ksub + j) * dsub;
+      const float *qm = qs + m * dsub;
+      float acc = 0.0f;
+      for (int t = 0; t < dsub; t++) acc = fmaf(qm[t], cw[t], acc);
+      lut[e] = acc;
+    }
+    __syncthreads();
+  }
+
+  for (int p = 0; p < nprobe; p++) {
+    int64_t ln = probes[(int64_t)q * nprobe + p];
+    if (ln < 0 || ln >= nlist) continue;
+    GammaBucketDev bk = buckets[ln];
+    if (bk.size <= 0) continue;
+    const float *cent = centroids + (size_t)ln * d;
+
+    if (IP) {
+      if (threadIdx.x == 0) {  /* dis0 = dot(q, c), canonical order */
+        float acc = 0.0f;
+        for (int t = 0; t < d; t++) acc = fmaf(qs[t], cent[t], acc);
+        dis0s[0] = acc;
+      }
+      __syncthreads();
+    } else {
+      /* residual + per-list table */
+      for (int i = threadIdx.x; i < d; i += blockDim.x)
+        rs[i] = qs[i] - cent[i];
+      __syncthreads();
+      for (int e = threadIdx.x; e < M * ksub; e += blockDim.x) {
+        int m = e >> 8, j = e & 255;
+        const float *cw = codebooks + ((size_t)m * ksub + j) * dsub;
+        const float *rm = rs + m * dsub;
+        float acc = 0.0f;
+        for (int t = 0; t < dsub; t++) {
+          float diff = rm[t] - cw[t];
+          acc = fmaf(diff, diff, acc);
+        }
+        lut[e] = acc;
+      }
+      __syncthreads();
+    }
+    const float dis0 = IP ? dis0s[0] : 0.0f;
+
+    const int64_t *ids = bk.ids;
+    const uint8_t *codes = (const uint8_t *)bk.data;
+    const int mwords = M >> 2;
+    for (long long j0 = 0; j0 < bk.size; j0 += blockDim.x) {
+      long long j = j0 + threadIdx.x;
+      if (j < bk.size) {
+        int64_t id = ids[j];
+        if (!((uint64_t)id >> 63) &&
+            !gamma_bitmap_test(bitmap, (uint64_t)id)) {
+          const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
+          float dis = dis0;
+          const float *tab = lut;
+          for (int mw = 0; mw < mwords; mw++) {
+            uint32_t w = cw[mw];
+            dis += tab[w & 255u];         tab += ksub;
+            dis += tab[(w >> 8) & 255u];  tab += ksub;
+            dis += tab[(w >> 16) & 255u]; tab += ksub;
+            dis += tab[w >> 24];          tab += ksub;
+          }
+          sel.push(gamma_make_key<IP>(dis, (uint32_t)id));
+        }
+      }
+      sel.maybe_flush();
+    }
+    __syncthreads();  /* lut rebuilt next list: scan readers done */
+  }
+  sel.finish();
+  for (int i = threadIdx.x; i < k2; i += blockDim.x)
+    out_keys[(int64_t)q * k2 + i] = res[i];
+}
+
+hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
+                          int k2, const float *queries,
+                          const float *centroids, const float *codebooks,
+                          const GammaBucketDev *buckets, int nlist,
+                          const int64_t *probes, const uint32_t *bitmap,
+                          bool ip, uint64_t *out_keys) {
+  size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
+                (GAMMA_SORT_CAP + k2) * 8 + (2 * d + 1) * 4 + 4 * sizeof(int);
+  if (smem > 160 * 1024) return hipErrorInvalidValue;
+  if (ip)
+    k_ivfpq_scan<true><<<dim3(nq), dim3(WG), smem, s>>>(
+        nq, d, M, nprobe, k2, queries, centroids, codebooks, buckets, nlist,
+        probes, bitmap, out_keys);
+  else
+    k_ivfpq_scan<false><<<dim3(nq), dim3(WG), smem, s>>>(
+        nq, d, M, nprobe, k2, queries, centroids, codebooks, buckets, nlist,
+        probes, bitmap, out_keys);
+  return hipGetLastError();
+}
+
+/* ------------------------------------------------------ IVFFLAT fused scan
+ * Lists store raw fp32 vectors (gamma_index_ivfflat.h:63-91). One wave
+ * handles one list vector per pass: 64 lanes x ceil(d/64) elements each,
+ * coalesced 512 B reads at d=128, shuffle-tree reduction (selection only;
+ * final distances are canonicalized by the re-rank kernel). */
+template <bool IP>
+__global__ void __launch_bounds__(WG)
+k_ivfflat_scan(int nq, int d, int nprobe, int k2,
+               const float *__restrict__ queries,
+               const GammaBucketDev *__restrict__ buckets, int nlist,
+               const int64_t *__restrict__ probes,
+               const uint32_t *__restrict__ bitmap,
+               uint64_t *__restrict__ out_keys) {
+  extern __shared__ char smem[];
+  uint64_t *sortbuf = (uint64_t *)smem;
+  uint64_t *res = sortbuf + GAMMA_SORT_CAP;
+  float *qs = (float *)(res + k2);
+  int *state = (int *)(qs + d);
+
+  const int q = blockIdx.x;
+  if (q >= nq) return;
+  const float *qg = queries + (int64_t)q * d;
+  for (int i = threadIdx.x; i < d; i += blockDim.x) qs[i] = qg[i];
+
+  GammaSelector sel;
+  sel.init(sortbuf, res, state, k2);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int nwave = blockDim.x >> 6;
+  const int epl = (d + 63) >> 6; /* elements per lane */
+
+  for (int p = 0; p < nprobe; p++) {
+    int64_t ln = probes[(int64_t)q * nprobe + p];
+    if (ln < 0 || ln >= nlist) continue;
+    GammaBucketDev bk = buckets[ln];
+    const int64_t *ids = bk.ids;
+    const float *vecs = (const float *)bk.data;
+
+    long long iters = (bk.size + nwave - 1) / nwave;
+    for (long long it = 0; it < iters; it++) {
+      long long j = it * nwave + wave;
+      bool live = false;
+      uint64_t id = 0;
+      float dist = 0.0f;
+      if (j < bk.size) {
+        id = (uint64_t)ids[j];
+        live = !(id >> 63) && !gamma_bitmap_test(bitmap, id);
+        if (live) {
+          const float *v = vecs + (size_t)j * d;
+          float acc = 0.0f;
+          for (int e = 0; e < epl; e++) {
+            int t = lane + (e << 6);
+            if (t < d) {
+              if (IP) acc = fmaf(qs[t], v[t], acc);
+              else {
+                float diff = qs[t] - v[t];
+                acc = fmaf(diff, diff, acc);
+              }
+            }
+          }
+          for (int off = 32; off > 0; off >>= 1)
+            acc += __shfl_xor(acc, off, 64);
+          dist = acc;
+        }
+      }
+      /* rotate the pushing lane so buffers fill evenly */
+      if (live && lane == (int)(j & 63))
+        sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
+      if ((it & 63) == 63) sel.maybe_flush();
+    }
+    sel.maybe_flush();
+  }
+  sel.finish();
+  for (int i = threadIdx.x; i < k2; i += blockDim.x)
+    out_keys[(int64_t)q * k2 + i] = res[i];
+}
+
+hipError_t gk::ivfflat_scan(hipStream_t s, int nq, int d, int nprobe, int k2,
+                            const float *queries,
+                            const GammaBucketDev *buckets, int nlist,
+                            const int64_t *probes, const uint32_t *bitmap,
+                            bool ip, uint64_t *out_keys) {
+  size_t smem = (GAMMA_SORT_CAP + k2) * 8 + d * 4 + 4 * sizeof(int);
+  if (smem > 160 * 1024) return hipErrorInvalidValue;
+  if (ip)
+    k_ivfflat_scan<true><<<dim3(nq), dim3(WG), smem, s>>>(
+        nq, d, nprobe, k2, queries, buckets, nlist, probes, bitmap,
+        out_keys);
+  else
+    k_ivfflat_scan<false><<<dim3(nq), dim3(WG), smem, s>>>(
+        nq, d, nprobe, k2, queries, buckets, nlist, probes, bitmap,
+        out_keys);
+  return hipGetLastError();
+}
+
+/* ------------------------------------------------------ FLAT stream scan
+ * Per-query workgroup streaming every segment (small-nq path; HBM-bound).
+ * Distances here are selection-only (canonical re-rank follows). */
+template <bool IP>
+__global__ void __launch_bounds__(WG)
+k_flat_stream(int nq, int64_t n, int d, int k2,
+              const float *__restrict__ queries,
+              const float *const *__restrict__ segs, int seg_shift,
+              const uint32_t *__restrict__ bitmap,
+              uint64_t *__restrict__ out_keys) {
+  extern __shared__ char smem[];
+  uint64_t *sortbuf = (uint64_t *)smem;
+  uint64_t *res = sortbuf + GAMMA_SORT_CAP;
+  float *qs = (float *)(res + k2);
+  int *state = (int *)(qs + d);
+
+  const int q = blockIdx.x;
+  if (q >= nq) return;
+  const float *qg = queries + (int64_t)q * d;
+  for (int i = threadIdx.x; i < d; i += blockDim.x) qs[i] = qg[i];
+
+  GammaSelector sel;
+  sel.init(sortbuf, res, state, k2);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int nwave = blockDim.x >> 6;
+  const int epl = (d + 63) >> 6;
+  const int64_t seg_mask = ((int64_t)1 << seg_shift) - 1;
+
+  long long iters = (n + nwave - 1) / nwave;
+  for (long long it = 0; it < iters; it++) {
+    int64_t j = it * nwave + wave;
+    bool live = false;
+    float dist = 0.0f;
+    if (j < n) {
+      live = !gamma_bitmap_test(bitmap, (uint64_t)j);
+      if (live) {
+        const float *v = segs[j >> seg_shift] + (size_t)(j & seg_mask) * d;
+        float acc = 0.0f;
+        for (int e = 0; e < epl; e++) {
+          int t = lane + (e << 6);
+          if (t < d) {
+            if (IP) acc = fmaf(qs[t], v[t], acc);
+            else {
+              float diff = qs[t] - v[t];
+              acc = fmaf(diff, diff, acc);
+            }
+          }
+        }
+        for (int off = 32; off > 0; off >>= 1)
+          acc += __shfl_xor(acc, off, 64);
+        dist = acc;
+      }
+    }
+    if (live && lane == (int)(j & 63))
+      sel.push(gamma_make_key<IP>(dist, (uint32_t)j));
+    if ((it & 63) == 63) sel.maybe_flush();
+  }
+  sel.finish();
+  for (int i = threadIdx.x; i < k2; i += blockDim.x)
+    out_keys[(int64_t)q * k2 + i] = res[i];
+}
+
+hipError_t gk::flat_stream_scan(hipStream_t s, int nq, int64_t n, int d,
+                                int k2, const float *queries,
+                                const float *const *segs, int seg_shift,
+                                const uint32_t *bitmap, bool ip,
+                                uint64_t *out_keys) {
+  size_t smem = (GAMMA_SORT_CAP + k2) * 8 + d * 4 + 4 * sizeof(int);
+  if (smem > 160 * 1024) return hipErrorInvalidValue;
+  if (ip)
+    k_flat_stream<true><<<dim3(nq), dim3(WG), smem, s>>>(
+        nq, n, d, k2, queries, segs, seg_shift, bitmap, out_keys);
+  else
+    k_flat_stream<false><<<dim3(nq), dim3(WG), smem, s>>>(
+        nq, n, d, k2, queries, segs, seg_shift, bitmap, out_keys);
+  return hipGetLastError();
+}
+
+/* ------------------------------------------------------------- re-rank
+ * Canonical-order exact distances (matches oracle_l2sqr/oracle_ip).
+ * Thread per (query, candidate); consecutive threads share the query so
+ * its elements broadcast from L1. */
+template <bool IP>
+__global__ void k_rerank(int nq, int ncand, int d,
+                         const float *__restrict__ queries,
+                         const float *const *__restrict__ segs,
+                         int seg_shift, const uint64_t *__restrict__ keys_in,
+                         uint64_t *__restrict__ keys_out) {
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (int64_t)nq * ncand) return;
+  int q = (int)(idx / ncand);
+  uint64_t key = keys_in[idx];
+  if (key == GAMMA_KEY_EMPTY) { keys_out[idx] = GAMMA_KEY_EMPTY; return; }
+  uint32_t id = (uint32_t)(key & 0xffffffffu);
+  const int64_t seg_mask = ((int64_t)1 << seg_shift) - 1;
+  const float *v = segs[id >> seg_shift] + (size_t)(id & seg_mask) * d;
+  const float *qv = queries + (int64_t)q * d;
+  float acc = 0.0f;
+  for (int t = 0; t < d; t++) {
+    if (IP) acc = fmaf(qv[t], v[t], acc);
+    else {
+      float diff = qv[t] - v[t];
+      acc = fmaf(diff, diff, acc);
+    }
+  }
+  keys_out[idx] = gamma_make_key<IP>(acc, id);
+}
+
+hipError_t gk::rerank(hipStream_t s, int nq, int ncand, int d,
+                      const float *queries, const float *const *segs,
+                      int seg_shift, bool ip, const uint64_t *keys_in,
+                      uint64_t *keys_out) {
+  int64_t total = (int64_t)nq * ncand;
+  if (total == 0) return hipSuccess;
+  int64_t blocks = (total + WG - 1) / WG;
+  if (ip)
+    k_rerank<true><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(
+        nq, ncand, d, queries, segs, seg_shift, keys_in, keys_out);
+  else
+    k_rerank<false><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(
+        nq, ncand, d, queries, segs, seg_shift, keys_in, keys_out);
+  return hipGetLastError();
+}
+
+/* ------------------------------------------------------------ row sort */
+template <bool IP>
+__global__ void k_sort_rows(int nq, int ncand, int k,
+                            const uint64_t *__restrict__ keys,
+                            float *__restrict__ out_dists,
+                            int64_t *__restrict__ out_ids) {
+  extern __shared__ char smem[];
+  uint64_t *buf = (uint64_t *)smem;
+  const int q = blockIdx.x;
+  if (q >= nq) return;
+  int n = 1;
+  while (n < ncand) n <<= 1;
+  for (int i = threadIdx.x; i < n; i += blockDim.x)
+    buf[i] = (i < ncand) ? keys[(int64_t)q * ncand + i] : GAMMA_KEY_EMPTY;
+  gamma_bitonic_sort(buf, n);
+  for (int i = threadIdx.x; i < k; i += blockDim.x) {
+    uint64_t key = buf[i];
+    out_dists[(int64_t)q * k + i] =
+        (key == GAMMA_KEY_EMPTY) ? -1.0f : gamma_key_dist<IP>(key);
+    out_ids[(int64_t)q * k + i] = gamma_key_id(key);
+  }
+}
+
+hipError_t gk::sort_rows(hipStream_t s, int nq, int ncand, int k,
+                         const uint64_t *keys, bool ip, float *out_dists,
+                         int64_t *out_ids) {
+  int n = 1;
+  while (n < ncand) n <<= 1;
+  size_t smem = (size_t)n * 8;
+  if (smem > 160 * 1024) return hipErrorInvalidValue;
+  if (ip)
+    k_sort_rows<true><<<dim3(nq), dim3(WG), smem, s>>>(nq, ncand, k, keys,
+                                                       out_dists, out_ids);
+  else
+    k_sort_rows<false><<<dim3(nq), dim3(WG), smem, s>>>(nq, ncand, k, keys,
+                                                        out_dists, out_ids);
+  return hipGetLastError();
+}
+
+/* ----------------------------------------------------------- unpack keys */
+template <bool IP>
+__global__ void k_unpack(int64_t n, const uint64_t *__restrict__ keys,
+                         float *__restrict__ out_dists,
+                         int64_t *__restrict__ out_ids) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint64_t key = keys[i];
+  out_dists[i] = (key == GAMMA_KEY_EMPTY) ? -1.0f : gamma_key_dist<IP>(key);
+  out_ids[i] = gamma_key_id(key);
+}
+
+hipError_t gk::unpack_keys(hipStream_t s, int64_t n, const uint64_t *keys,
+                           bool ip, float *out_dists, int64_t *out_ids) {
+  if (n == 0) return hipSuccess;
+  int64_t blocks = (n + WG - 1) / WG;
+  if (ip)
+    k_unpack<true><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(
+        n, keys, out_dists, out_ids);
+  else
+    k_unpack<false><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(
+        n, keys, out_dists, out_ids);
+  return hipGetLastError();
+}
+
+/* ----------------------------------------------------------- residuals */
+__global__ void k_residuals(int64_t n, int d, const float *__restrict__ x,
+                            const float *__restrict__ centroids,
+                            const int32_t *__restrict__ assign,
+                            float *__restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n * d) return;
+  int64_t row = i / d;
+  int col = (int)(i - row * d);
+  out[i] = x[i] - centroids[(size_t)assign[row] * d + col];
+}
+
+hipError_t gk::residuals(hipStream_t s, int64_t n, int d, const float *x,
+                         const float *centroids, const int32_t *assign,
+                         float *out) {
+  if (n == 0) return hipSuccess;
+  int64_t blocks = (n * d + WG - 1) / WG;
+  k_residuals<<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(n, d, x, centroids,
+                                                          assign, out);
+  return hipGetLastError();
+}
+
+/* ----------------------------------------------------------- PQ encode */
+__global__ void k_pq_encode(int64_t n, int d, int M, int ksub,
+                            const float *__restrict__ x,
+                            const float *__restrict__ codebooks,
+                            uint8_t *__restrict__ codes) {
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= n * M) return;
+  int64_t i = idx / M;
+  int m = (int)(idx - i * M);
+  int dsub = d / M;
+  const float *xm = x + i * d + m * dsub;
+  const float *cb = codebooks + (size_t)m * ksub * dsub;
+  float best = INFINITY;
+  int bestj = 0;
+  for (int j = 0; j < ksub; j++) {
+    const float *cw = cb + (size_t)j * dsub;
+    float acc = 0.0f;
+    for (int t = 0; t < dsub; t++) {
+      float diff = xm[t] - cw[t];
+      acc = fmaf(diff, diff, acc);
+    }
+    if (acc < best) { best = acc; bestj = j; }
+  }
+  codes[idx] = (uint8_t)bestj;
+}
+
+hipError_t gk::pq_encode(hipStream_t s, int64_t n, int d, int M, int ksub,
+                         const float *x, const float *codebooks,
+                         uint8_t *codes) {
+  if (n == 0) return hipSuccess;
+  int64_t blocks = (n * M + WG - 1) / WG;
+  k_pq_encode<<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(n, d, M, ksub, x,
+                                                          codebooks, codes);
+  return hipGetLastError();
+}

@@ -1,0 +1,165 @@
+/*
+ * select.hpp — device-side exact top-k selection on u64 (dist,id) keys.
+ *
+ * Key layout: upper 32 bits = monotone u32 image of the fp32 distance
+ * (sign-flip trick; inverted for inner-product so ascending key ==
+ * descending similarity), lower 32 bits = docid. One integer compare gives
+ * the exact (dist, id) total order — ties break by id deterministically
+ * (SURVEY §8c pin (i)); replaces the reference's faiss CMin/CMax heaps
+ * (gamma_index_ivfpq.cc:606-631, gamma_index_flat.cc:80-83).
+ *
+ * GammaSelector: each thread keeps a register buffer of candidates better
+ * than the current block-wide threshold; the buffers are periodically
+ * dumped to LDS and bitonic-merged with the running top-k, which
+ * re-tightens the threshold. Exact: a candidate is dropped only when it is
+ * >= the k-th best key seen so far.
+ *
+ * Protocol: between two maybe_flush() calls, each thread may push() at
+ * most GAMMA_SEL_CHUNK candidates. All threads of the block must reach
+ * maybe_flush()/finish() together (they contain barriers).
+ */
+#pragma once
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define GAMMA_KEY_EMPTY 0xffffffffffffffffull
+
+__device__ __forceinline__ uint32_t gamma_f32_key(float f) {
+  uint32_t u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+
+__device__ __forceinline__ float gamma_key_f32(uint32_t dk) {
+  uint32_t u = (dk & 0x80000000u) ? (dk & 0x7fffffffu) : ~dk;
+  return __uint_as_float(u);
+}
+
+template <bool IP>
+__device__ __forceinline__ uint64_t gamma_make_key(float dist, uint32_t id) {
+  uint32_t dk = gamma_f32_key(dist);
+  if (IP) dk = ~dk;
+  return ((uint64_t)dk << 32) | (uint64_t)id;
+}
+
+template <bool IP>
+__device__ __forceinline__ float gamma_key_dist(uint64_t key) {
+  uint32_t dk = (uint32_t)(key >> 32);
+  if (IP) dk = ~dk;
+  return gamma_key_f32(dk);
+}
+
+__device__ __forceinline__ int64_t gamma_key_id(uint64_t key) {
+  return (key == GAMMA_KEY_EMPTY) ? (int64_t)-1
+                                  : (int64_t)(key & 0xffffffffull);
+}
+
+/* In-LDS bitonic sort (ascending) of buf[0..n), n a power of two.
+ * Call with all threads of the block; barriers inside. */
+__device__ inline void gamma_bitonic_sort(uint64_t *buf, int n) {
+  const int tid = threadIdx.x, nt = blockDim.x;
+  for (int len = 2; len <= n; len <<= 1) {
+    for (int inc = len >> 1; inc > 0; inc >>= 1) {
+      __syncthreads();
+      for (int i = tid; i < n; i += nt) {
+        int j = i ^ inc;
+        if (j > i) {
+          bool up = (i & len) == 0;
+          uint64_t a = buf[i], b = buf[j];
+          if ((a > b) == up) { buf[i] = b; buf[j] = a; }
+        }
+      }
+    }
+  }
+  __syncthreads();
+}
+
+#define GAMMA_SORT_CAP 2048   /* LDS sort scratch, u64 x 2048 = 16 KB */
+#define GAMMA_SEL_BUF 16      /* per-thread candidate registers */
+#define GAMMA_SEL_CHUNK 8     /* max pushes per thread between flush checks */
+
+struct GammaSelector {
+  uint64_t *sortbuf;  /* uint64_t[GAMMA_SORT_CAP] in LDS */
+  uint64_t *res;      /* uint64_t[k] in LDS, sorted ascending after flush */
+  int *state;         /* int[2] in LDS: [0]=dump counter, [1]=flags */
+  int k;              /* 1 <= k <= GAMMA_SORT_CAP/2 */
+  uint64_t thresh;
+  uint64_t buf[GAMMA_SEL_BUF];
+  int cnt;
+
+  __device__ void init(uint64_t *sortbuf_, uint64_t *res_, int *state_,
+                       int k_) {
+    sortbuf = sortbuf_; res = res_; state = state_; k = k_;
+    thresh = GAMMA_KEY_EMPTY;
+    cnt = 0;
+    for (int i = threadIdx.x; i < k; i += blockDim.x) res[i] = GAMMA_KEY_EMPTY;
+    if (threadIdx.x == 0) { state[0] = 0; state[1] = 0; }
+    __syncthreads();
+  }
+
+  /* seed the running result with previously selected keys (unsorted ok) */
+  __device__ void seed(const uint64_t *keys, int n) {
+    for (int i = threadIdx.x; i < k; i += blockDim.x)
+      res[i] = (i < n) ? keys[i] : GAMMA_KEY_EMPTY;
+    flush_();
+  }
+
+  __device__ __forceinline__ void push(uint64_t key) {
+    if (key < thresh) {
+      buf[cnt++] = key;
+      if (cnt > GAMMA_SEL_BUF - GAMMA_SEL_CHUNK) atomicOr(&state[1], 1);
+    }
+  }
+
+  /* all threads must arrive; flushes when any thread is near-full */
+  __device__ __forceinline__ void maybe_flush() {
+    __syncthreads();
+    if (state[1]) flush_();
+  }
+
+  __device__ void finish() {
+    __syncthreads();
+    flush_();
+  }
+
+ private:
+  /* Drain every thread's buffer into sortbuf (in rounds if needed),
+   * merge with res, keep top-k, refresh threshold. */
+  __device__ void flush_() {
+    const int nt = blockDim.x, tid = threadIdx.x;
+    const int cap = GAMMA_SORT_CAP - k;
+    for (;;) {
+      if (tid == 0) { state[0] = 0; state[1] = 0; }
+      __syncthreads();
+      int base = (cnt > 0) ? atomicAdd(&state[0], cnt) : 0;
+      int fit = 0;
+      if (cnt > 0 && base < cap) fit = min(cnt, cap - base);
+      if (cnt > fit) atomicOr(&state[1], 1); /* leftovers -> another round */
+      __syncthreads();
+      int total = min(state[0], cap);
+      int more = state[1];
+      __syncthreads();               /* everyone read state before reuse */
+      /* pad candidate region, then append res at [total, total+k) */
+      int n = 1;
+      while (n < total + k) n <<= 1;
+      for (int i = tid; i < n; i += nt) sortbuf[i] = GAMMA_KEY_EMPTY;
+      __syncthreads();
+      for (int i = 0; i < fit; i++) sortbuf[base + i] = buf[i];
+      for (int i = fit; i < cnt; i++) buf[i - fit] = buf[i];
+      cnt -= fit;
+      for (int i = tid; i < k; i += nt) sortbuf[total + i] = res[i];
+      gamma_bitonic_sort(sortbuf, n);
+      for (int i = tid; i < k; i += nt) res[i] = sortbuf[i];
+      __syncthreads();
+      thresh = res[k - 1];
+      if (!more) break;
+    }
+    if (threadIdx.x == 0) { state[0] = 0; state[1] = 0; }
+    __syncthreads();
+  }
+};
+
+/* delete-bitmap test (1 = deleted), u32 words */
+__device__ __forceinline__ bool gamma_bitmap_test(const uint32_t *bm,
+                                                  uint64_t id) {
+  return bm && ((bm[id >> 5] >> (id & 31)) & 1u);
+}

@@ -42,6 +42,13 @@ int GammaRawSearch(void *engine, int nq, const float *xq, int k, int nprobe,
                    int rerank, int metric, float *out_dists,
                    int64_t *out_ids);
 
+/* Upload nq*d queries once into HBM; GammaRawSearchCached then times the
+ * hot path with inputs already device-resident (bench contract). */
+int GammaCacheQueries(void *engine, int nq, const float *xq);
+int GammaRawSearchCached(void *engine, int nq, int k, int nprobe,
+                         int rerank, int metric, float *out_dists,
+                         int64_t *out_ids);
+
 /* Debug: run only the coarse-assign stage (quantizer->search equivalent,
  * reference ivfpq.cc:595): top-nprobe centroids per query. */
 int GammaDebugCoarseAssign(void *engine, int nq, const float *xq, int nprobe,

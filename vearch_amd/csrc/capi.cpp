@@ -1,0 +1,566 @@
+/*
+ * capi.cpp — the drop-in C ABI (include/gamma_api.h) plus the bench/debug
+ * extensions (include/gamma_bench.h). Mirrors the dispatch of the
+ * reference's c_api/gamma_api.cc onto the MI355X engine.
+ */
+#include <math.h>
+#include <stdlib.h>
+#include <string.h>
+
+#include <string>
+#include <vector>
+
+#include "../../include/gamma_api.h"
+#include "../../include/gamma_bench.h"
+#include "core.hpp"
+#include "fbs.hpp"
+#include "json.hpp"
+#include "pbcodec.hpp"
+
+using vgamma::Engine;
+using vgamma::FieldMeta;
+using vgamma::KillRegistry;
+
+static char *dup_malloc(const std::string &s) {
+  char *p = (char *)malloc(s.size() + 1);
+  memcpy(p, s.data(), s.size());
+  p[s.size()] = 0;
+  return p;
+}
+
+static CStatus ok_status() { return CStatus{0, nullptr}; }
+static CStatus err_status(int code, const std::string &msg) {
+  /* msg malloc'd (the Go side frees with C.free — gamma.go:186-188;
+   * the reference's new[] at gamma_api.cc:150 was a latent mismatch) */
+  return CStatus{code, dup_malloc(msg)};
+}
+
+static int g_memory_limit_mb = 0;
+
+extern "C" {
+
+void *Init(const char *config_str, int len) {
+  auto *e = new Engine();
+  std::string err;
+  if (e->init(std::string(config_str ? config_str : "", len > 0 ? len : 0),
+              &err)) {
+    fprintf(stderr, "[gamma] Init failed: %s\n", err.c_str());
+    delete e;
+    return nullptr;
+  }
+  return e;
+}
+
+int Close(void *engine) {
+  if (!engine) return 1;
+  delete static_cast<Engine *>(engine);
+  return 0;
+}
+
+struct CStatus CreateTable(void *engine, const char *table_str, int len) {
+  if (!engine) return err_status(1, "null engine");
+  gfb::TableSchema ts;
+  if (!ts.parse(table_str, (size_t)len))
+    return err_status(1, "bad table flatbuffer");
+  if (ts.vectors.empty())
+    return err_status(1, "table has no vector field");
+  if (ts.vectors.size() > 1)
+    return err_status(1, "multi-vector tables not supported this round");
+  std::vector<FieldMeta> fields;
+  for (auto &f : ts.fields) fields.push_back({f.name, f.data_type});
+  std::string index_type = ts.index_type;
+  std::string index_params = ts.index_params;
+  if (index_type.empty() && !ts.indexes.empty()) {
+    index_type = ts.indexes[0].type;
+    index_params = ts.indexes[0].params;
+  }
+  /* training_threshold comes from the index-params JSON (table.cc:122) */
+  int tt = 0;
+  {
+    gjson::Value v;
+    if (!index_params.empty() && gjson::parse(index_params, v))
+      v.get_int("training_threshold", tt);
+  }
+  std::string err;
+  if (static_cast<Engine *>(engine)->create_table(
+          ts.name, fields, ts.vectors[0].name, ts.vectors[0].dimension,
+          index_type, index_params, tt, &err))
+    return err_status(1, err);
+  return ok_status();
+}
+
+int AddOrUpdateDoc(void *engine, const char *doc_str, int len) {
+  if (!engine) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  gfb::Doc doc;
+  if (!doc.parse(doc_str, (size_t)len)) return -1;
+  std::string p_key;
+  const float *vec = nullptr;
+  int vec_len = 0;
+  std::vector<std::pair<std::string, std::string>> fields;
+  for (auto &f : doc.fields) {
+    if (f.name == "_id") {
+      p_key = f.value;
+    } else if (f.data_type == gfb::VECTOR) {
+      vec = (const float *)f.value.data();
+      vec_len = (int)(f.value.size() / 4);
+    } else {
+      fields.emplace_back(f.name, f.value);
+    }
+  }
+  if (p_key.empty() || !vec) return -1;
+  return e->add_doc(p_key, fields, vec, vec_len);
+}
+
+int DeleteDoc(void *engine, const char *docid, int docid_len) {
+  if (!engine) return -1;
+  return static_cast<Engine *>(engine)->delete_doc(
+      std::string(docid, docid_len));
+}
+
+void GetEngineStatus(void *engine, char **status, int *len) {
+  std::string s = engine ? static_cast<Engine *>(engine)->status_json()
+                         : std::string("{}");
+  *status = dup_malloc(s);
+  *len = (int)s.size();
+}
+
+void GetMemoryInfo(void *engine, char **memory_info, int *len) {
+  char buf[256];
+  auto *e = static_cast<Engine *>(engine);
+  long long nd = e ? e->num_docs() : 0;
+  long long dim = e ? e->dimension() : 0;
+  snprintf(buf, sizeof buf,
+           "{\"table_mem_bytes\": 0, \"vector_mem_bytes\": %lld, "
+           "\"index_mem_bytes\": 0, \"bitmap_mem_bytes\": %lld}",
+           nd * dim * 4, (nd + 7) / 8);
+  std::string s(buf);
+  *memory_info = dup_malloc(s);
+  *len = (int)s.size();
+}
+
+static int serialize_doc(Engine *e, int64_t docid, char **doc_str,
+                         int *len) {
+  gfb::Doc doc;
+  doc.fields.push_back({"_id", e->pkey_of(docid), gfb::STRING});
+  for (auto &fm : e->scalar_fields()) {
+    const std::string *v = e->field_value(docid, fm.name);
+    doc.fields.push_back({fm.name, v ? *v : "", fm.data_type});
+  }
+  const float *vec = e->raw().host_row(docid);
+  doc.fields.push_back(
+      {e->vec_field_name(),
+       std::string((const char *)vec, (size_t)e->dimension() * 4),
+       gfb::VECTOR});
+  std::string out = doc.serialize();
+  *doc_str = dup_malloc(out);
+  *len = (int)out.size();
+  return 0;
+}
+
+int GetDocByID(void *engine, const char *docid, int docid_len,
+               char **doc_str, int *len) {
+  if (!engine) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  int64_t id = e->docid_of(std::string(docid, docid_len));
+  if (id < 0 || e->bitmap().test(id)) return -1;
+  return serialize_doc(e, id, doc_str, len);
+}
+
+int GetDocByDocID(void *engine, int docid, char next, char **doc_str,
+                  int *len) {
+  if (!engine) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  int64_t id = docid;
+  if (next) { /* next undeleted doc after docid (gamma_api.h:86) */
+    id++;
+    while (id < e->num_docs() && e->bitmap().test(id)) id++;
+  }
+  if (id < 0 || id >= e->num_docs() || e->bitmap().test(id)) return -1;
+  return serialize_doc(e, id, doc_str, len);
+}
+
+int BuildIndex(void *engine) {
+  if (!engine) return -1;
+  std::string err;
+  int rc = static_cast<Engine *>(engine)->build_index(&err);
+  if (rc) fprintf(stderr, "[gamma] BuildIndex: %s\n", err.c_str());
+  return rc;
+}
+
+int RebuildIndex(void *engine, int drop_before_rebuild, int limit_cpu,
+                 int describe) {
+  (void)drop_before_rebuild;
+  (void)limit_cpu;
+  (void)describe;
+  return BuildIndex(engine);
+}
+
+int Dump(void *engine) {
+  if (!engine) return -1;
+  std::string err;
+  int rc = static_cast<Engine *>(engine)->dump(&err);
+  if (rc) fprintf(stderr, "[gamma] Dump: %s\n", err.c_str());
+  return rc;
+}
+
+int Load(void *engine) {
+  if (!engine) return -1;
+  std::string err;
+  int rc = static_cast<Engine *>(engine)->load(&err);
+  if (rc) fprintf(stderr, "[gamma] Load: %s\n", err.c_str());
+  return rc;
+}
+
+struct CStatus Search(void *engine, const char *request_str, int req_len,
+                      char **response_str, int *res_len) {
+  if (!engine) return err_status(1, "null engine");
+  auto *e = static_cast<Engine *>(engine);
+  gpb::SearchRequest req;
+  if (!req.parse(request_str, req_len))
+    return err_status(1, "parse search request failed");
+  if (req.vec_fields.empty())
+    return err_status(1, "no vector query (scalar-only search via Query)");
+  if (req.vec_fields.size() > 1)
+    return err_status(1, "multi-vector ranking not supported this round");
+  if (req.n_filters > 0)
+    return err_status(1,
+                      "scalar filtered search not supported this round "
+                      "(SURVEY 8f-2)");
+  if (req.topn <= 0) return err_status(1, "limit[topN] is zero");
+
+  const gpb::VectorQuery &vq = req.vec_fields[0];
+  int d = e->dimension();
+  int nq = (int)(vq.value.size() / ((size_t)d * 4));
+  if (nq <= 0) return err_status(1, "Search n shouldn't less than 0!");
+
+  /* retrieval params JSON (ivfpq.cc:233-294): nprobe, recall_num,
+   * metric_type, parallel_on_queries (N/A on GPU) */
+  int nprobe = 0, recall_num = 0, metric = 0;
+  if (!req.index_params.empty()) {
+    gjson::Value v;
+    if (gjson::parse(req.index_params, v)) {
+      v.get_int("nprobe", nprobe);
+      v.get_int("recall_num", recall_num);
+      std::string mt;
+      if (v.get_str("metric_type", mt))
+        metric = strcasecmp(mt.c_str(), "L2") == 0 ? 1 : 2;
+    }
+  }
+  int k = req.topn + req.offset;
+  std::vector<float> dists((size_t)nq * k);
+  std::vector<int64_t> ids((size_t)nq * k);
+  int rc = e->search(nq, (const float *)vq.value.data(), k, nprobe,
+                     recall_num, metric, req.brute == 1, req.request_id,
+                     dists.data(), ids.data(), req.l2_sqrt);
+  if (rc == -2) return err_status(-2, "request killed");
+  if (rc != 0) return err_status(1, "search failed");
+
+  /* score-range filter (SearchCondition::IsSimilarScoreValid,
+   * gamma_common_data.h:94) applied on the final top-k */
+  bool has_range = vq.has_min || vq.has_max;
+
+  std::vector<gpb::SearchResult> results(nq);
+  int64_t total = e->num_docs() - e->bitmap().popcount();
+  bool want_vec = req.is_vector_value;
+  for (int i = 0; i < nq; i++) {
+    gpb::SearchResult &res = results[i];
+    res.total = (int)total;
+    for (int j = req.offset; j < k; j++) {
+      int64_t id = ids[(size_t)i * k + j];
+      if (id < 0) continue;
+      double score = dists[(size_t)i * k + j];
+      if (has_range && (score < vq.min_score || score > vq.max_score))
+        continue;
+      gpb::ResultItem item;
+      item.score = score;
+      item.fields.push_back({"_id", e->pkey_of(id)});
+      for (auto &fname : req.fields) {
+        if (fname == "_id") continue;
+        if (fname == e->vec_field_name()) continue;
+        const std::string *v = e->field_value(id, fname);
+        if (v) item.fields.push_back({fname, *v});
+      }
+      if (want_vec) {
+        const float *vec = e->raw().host_row(id);
+        item.fields.push_back(
+            {e->vec_field_name(),
+             std::string((const char *)vec, (size_t)d * 4)});
+      }
+      res.max_score = std::max(res.max_score, item.score);
+      res.items.push_back(std::move(item));
+    }
+  }
+  std::string out = gpb::encode_search_response(results);
+  *response_str = (char *)malloc(out.size());
+  memcpy(*response_str, out.data(), out.size());
+  *res_len = (int)out.size();
+  return ok_status();
+}
+
+struct CStatus Query(void *engine, const char *request_str, int req_len,
+                     char **response_str, int *res_len) {
+  if (!engine) return err_status(1, "null engine");
+  auto *e = static_cast<Engine *>(engine);
+  gpb::QueryRequest req;
+  if (!req.parse(request_str, req_len))
+    return err_status(1, "parse query request failed");
+  if (req.n_filters > 0)
+    return err_status(1, "filtered query not supported this round");
+  std::vector<gpb::SearchResult> results(1);
+  gpb::SearchResult &res = results[0];
+  res.total = (int)req.document_ids.size();
+  for (auto &pk : req.document_ids) {
+    int64_t id = e->docid_of(pk);
+    if (id < 0 || e->bitmap().test(id)) continue;
+    gpb::ResultItem item;
+    item.score = 0;
+    item.fields.push_back({"_id", e->pkey_of(id)});
+    for (auto &fname : req.fields) {
+      if (fname == "_id" || fname == e->vec_field_name()) continue;
+      const std::string *v = e->field_value(id, fname);
+      if (v) item.fields.push_back({fname, *v});
+    }
+    if (req.is_vector_value) {
+      const float *vec = e->raw().host_row(id);
+      item.fields.push_back(
+          {e->vec_field_name(),
+           std::string((const char *)vec, (size_t)e->dimension() * 4)});
+    }
+    res.items.push_back(std::move(item));
+  }
+  std::string out = gpb::encode_search_response(results);
+  *response_str = (char *)malloc(out.size());
+  memcpy(*response_str, out.data(), out.size());
+  *res_len = (int)out.size();
+  return ok_status();
+}
+
+int SetConfig(void *engine, const char *config_str, int len) {
+  (void)engine;
+  (void)config_str;
+  (void)len;
+  return 0; /* cache sizes are N/A for the GPU engine */
+}
+
+int GetConfig(void *engine, char **config_str, int *len) {
+  std::string s = "{\"path\": \"\", \"cache_sizes\": []}";
+  *config_str = dup_malloc(s);
+  *len = (int)s.size();
+  return 0;
+}
+
+struct CStatus Backup(void *engine, int command) {
+  (void)engine;
+  (void)command;
+  return err_status(1, "backup not supported this round");
+}
+
+struct CStatus AddFieldIndexWithParams(
+    void *engine, const char *index_name, int index_name_len,
+    const char *const *field_names, const int *field_name_lens,
+    int field_name_count, const char *index_type, int index_type_len,
+    const char *index_params, int index_params_len) {
+  (void)engine;
+  (void)index_name;
+  (void)index_name_len;
+  (void)field_names;
+  (void)field_name_lens;
+  (void)field_name_count;
+  (void)index_type;
+  (void)index_type_len;
+  (void)index_params;
+  (void)index_params_len;
+  return err_status(1,
+                    "dynamic field indexes (scalar/composite) are out of "
+                    "scope this round");
+}
+
+struct CStatus RemoveFieldIndex(void *engine, const char *index_name,
+                                int index_name_len) {
+  (void)engine;
+  (void)index_name;
+  (void)index_name_len;
+  return err_status(1, "dynamic field indexes are out of scope this round");
+}
+
+void SetMemoryLimitConfig(int memory_limit) {
+  g_memory_limit_mb = memory_limit;
+}
+
+void SetKillStatus(const char *request_id, int partition_id, int reason) {
+  (void)reason;
+  KillRegistry::inst().set(request_id ? request_id : "", partition_id);
+}
+
+void DeleteKillStatus(const char *request_id, int partition_id) {
+  KillRegistry::inst().del(request_id ? request_id : "", partition_id);
+}
+
+/* ----------------------------------------------------- bench extensions */
+
+int GammaBulkAdd(void *engine, const char *field, int field_len, int n,
+                 const float *vecs) {
+  (void)field;
+  (void)field_len;
+  if (!engine) return -1;
+  return static_cast<Engine *>(engine)->bulk_add(n, vecs);
+}
+
+int GammaRawSearch(void *engine, int nq, const float *xq, int k, int nprobe,
+                   int rerank, int metric, float *out_dists,
+                   int64_t *out_ids) {
+  if (!engine) return -1;
+  return static_cast<Engine *>(engine)->search(
+      nq, xq, k, nprobe, rerank, metric, false, "", out_dists, out_ids);
+}
+
+int GammaCacheQueries(void *engine, int nq, const float *xq) {
+  if (!engine) return -1;
+  return static_cast<Engine *>(engine)->cache_queries(nq, xq);
+}
+
+int GammaRawSearchCached(void *engine, int nq, int k, int nprobe,
+                         int rerank, int metric, float *out_dists,
+                         int64_t *out_ids) {
+  if (!engine) return -1;
+  return static_cast<Engine *>(engine)->search(
+      nq, nullptr, k, nprobe, rerank, metric, false, "", out_dists,
+      out_ids);
+}
+
+int GammaDebugCoarseAssign(void *engine, int nq, const float *xq,
+                           int nprobe, int64_t *out_lists,
+                           float *out_dists) {
+  if (!engine) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  auto *ix = e->index();
+  if (!ix || !ix->trained()) return -1;
+  hipStream_t s = e->stream();
+  vgamma::DeviceBuf qd, qn, probes, pdists;
+  int d = e->dimension();
+  if (qd.reserve((size_t)nq * d * 4)) return -1;
+  hipMemcpy(qd.get(), xq, (size_t)nq * d * 4, hipMemcpyHostToDevice);
+  if (qn.reserve((size_t)nq * 4)) return -1;
+  gk::row_norms(s, qd.as<float>(), nq, d, qn.as<float>());
+  if (probes.reserve((size_t)nq * nprobe * 8)) return -1;
+  if (pdists.reserve((size_t)nq * nprobe * 4)) return -1;
+  if (ix->coarse_assign(qd.as<float>(), nq, nprobe,
+                        e->metric_ip_default(), qn.as<float>(), s,
+                        probes.as<int64_t>(), pdists.as<float>()))
+    return -1;
+  hipStreamSynchronize(s);
+  hipMemcpy(out_lists, probes.get(), (size_t)nq * nprobe * 8,
+            hipMemcpyDeviceToHost);
+  hipMemcpy(out_dists, pdists.get(), (size_t)nq * nprobe * 4,
+            hipMemcpyDeviceToHost);
+  return 0;
+}
+
+int GammaDebugGetModel(void *engine, float *centroids, float *codebooks) {
+  if (!engine) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  if (!e->index()) return -1;
+  return e->index()->copy_model_to_host(centroids, codebooks, e->stream());
+}
+
+int64_t GammaDebugGetList(void *engine, int64_t list_no, int64_t *ids,
+                          uint8_t *codes) {
+  if (!engine) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  if (!e->index()) return -1;
+  int64_t sz = e->index()->list_size(list_no);
+  if (sz < 0) return -1;
+  if (ids || codes)
+    if (e->index()->copy_list_to_host(list_no, ids, codes, e->stream()))
+      return -1;
+  return sz;
+}
+
+int64_t GammaDebugNumDocs(void *engine) {
+  if (!engine) return -1;
+  return static_cast<Engine *>(engine)->num_docs();
+}
+
+int GammaLastSearchTiming(void *engine, double *us6) {
+  if (!engine) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  for (int i = 0; i < 6; i++) us6[i] = e->last_timing[i];
+  return 0;
+}
+
+/* ------------------------------------------- codec self-tests (CPU-only)
+ * Used by tests/test_capi_cpu.py to exercise the hand-written protobuf /
+ * FlatBuffers codecs without a GPU. */
+
+int GammaTestParseSearchRequest(const char *buf, int len, char **json_out,
+                                int *json_len) {
+  gpb::SearchRequest req;
+  if (!req.parse(buf, len)) return -1;
+  std::string s = "{";
+  s += "\"request_id\": \"" + gjson::escape(req.request_id) + "\",";
+  s += "\"partition_id\": " + std::to_string(req.partition_id) + ",";
+  s += "\"req_num\": " + std::to_string(req.req_num) + ",";
+  s += "\"topn\": " + std::to_string(req.topn) + ",";
+  s += "\"brute\": " + std::to_string(req.brute) + ",";
+  s += "\"n_vec\": " + std::to_string(req.vec_fields.size()) + ",";
+  if (!req.vec_fields.empty()) {
+    s += "\"vec_name\": \"" + gjson::escape(req.vec_fields[0].name) + "\",";
+    s += "\"vec_bytes\": " +
+         std::to_string(req.vec_fields[0].value.size()) + ",";
+    s += "\"min_score\": " + std::to_string(req.vec_fields[0].min_score) +
+         ",";
+    s += "\"max_score\": " + std::to_string(req.vec_fields[0].max_score) +
+         ",";
+  }
+  s += "\"index_params\": \"" + gjson::escape(req.index_params) + "\",";
+  s += "\"l2_sqrt\": " + std::to_string(req.l2_sqrt ? 1 : 0) + ",";
+  s += "\"n_fields\": " + std::to_string(req.fields.size()) + ",";
+  s += "\"n_filters\": " + std::to_string(req.n_filters) + "}";
+  *json_out = dup_malloc(s);
+  *json_len = (int)s.size();
+  return 0;
+}
+
+int GammaTestParseTable(const char *buf, int len, char **json_out,
+                        int *json_len) {
+  gfb::TableSchema ts;
+  if (!ts.parse(buf, (size_t)len)) return -1;
+  std::string s = "{";
+  s += "\"name\": \"" + gjson::escape(ts.name) + "\",";
+  s += "\"index_type\": \"" + gjson::escape(ts.index_type) + "\",";
+  s += "\"index_params\": \"" + gjson::escape(ts.index_params) + "\",";
+  s += "\"n_fields\": " + std::to_string(ts.fields.size()) + ",";
+  s += "\"n_vectors\": " + std::to_string(ts.vectors.size()) + ",";
+  if (!ts.vectors.empty()) {
+    s += "\"vec_name\": \"" + gjson::escape(ts.vectors[0].name) + "\",";
+    s += "\"dimension\": " + std::to_string(ts.vectors[0].dimension) + ",";
+  }
+  std::string fn;
+  for (auto &f : ts.fields) fn += f.name + ",";
+  s += "\"field_names\": \"" + gjson::escape(fn) + "\"}";
+  *json_out = dup_malloc(s);
+  *json_len = (int)s.size();
+  return 0;
+}
+
+int GammaTestDocRoundtrip(const char *buf, int len, char **out,
+                          int *out_len) {
+  gfb::Doc doc;
+  if (!doc.parse(buf, (size_t)len)) return -1;
+  std::string s = doc.serialize();
+  gfb::Doc doc2;
+  if (!doc2.parse(s.data(), s.size())) return -2;
+  if (doc2.fields.size() != doc.fields.size()) return -3;
+  for (size_t i = 0; i < doc.fields.size(); i++) {
+    if (doc.fields[i].name != doc2.fields[i].name) return -4;
+    if (doc.fields[i].value != doc2.fields[i].value) return -5;
+    if (doc.fields[i].data_type != doc2.fields[i].data_type) return -6;
+  }
+  *out = (char *)malloc(s.size());
+  memcpy(*out, s.data(), s.size());
+  *out_len = (int)s.size();
+  return 0;
+}
+
+} /* extern "C" */

@@ -700,9 +700,10 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
   hipEventCreate(&e1);
   hipEventCreate(&e2);
   hipEventRecord(e0, s);
-  GAMMA_CHECK(coarse_assign(q_dev, nq, nprobe, metric_ip, q_norms_dev, s,
-                            scratch_probes_.as<int64_t>(),
-                            scratch_pdists_.as<float>()));
+  if (coarse_assign(q_dev, nq, nprobe, metric_ip, q_norms_dev, s,
+                    scratch_probes_.as<int64_t>(),
+                    scratch_pdists_.as<float>()) != 0)
+    return -1;
   hipEventRecord(e1, s);
   if (params_.kind == IndexKind::IVFPQ) {
     GAMMA_CHECK(gk::ivfpq_scan(s, nq, d_, M_, nprobe, k2, q_dev,

@@ -219,10 +219,15 @@ class Engine {
   int build_index(std::string *err);
 
   /* the hot path: batched vector search.
-   * metric: 0 default, 1 L2, 2 IP. Returns 0 ok, -2 killed, <0 error. */
+   * metric: 0 default, 1 L2, 2 IP. Returns 0 ok, -2 killed, <0 error.
+   * xq == nullptr -> use the cached device-resident queries (bench path,
+   * nq must equal the cached count). */
   int search(int nq, const float *xq, int k, int nprobe, int recall_num,
              int metric, bool brute_force, const std::string &request_id,
              float *out_dists, int64_t *out_ids, bool l2_sqrt = false);
+  /* upload queries once; later search(nq, nullptr, ...) reuses them */
+  int cache_queries(int nq, const float *xq);
+  int cached_nq() const { return cached_nq_; }
 
   int dump(std::string *err);
   int load(std::string *err);
@@ -266,6 +271,7 @@ class Engine {
   int64_t indexed_count_ = 0;
   bool table_created_ = false;
   hipStream_t stream_ = nullptr;
+  int cached_nq_ = 0;
   mutable std::shared_mutex rw_; /* search shared / add+build exclusive */
   DeviceBuf q_dev_, q_norms_dev_, keys_dev_, out_d_dev_, out_i_dev_;
   DeviceBuf flat_dots_, flat_keys_;

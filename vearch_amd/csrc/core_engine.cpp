@@ -262,9 +262,13 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
   if (k2 > 1536) return -1;
 
   tm.rec(0);
-  if (q_dev_.reserve((size_t)nq * dim_ * 4)) return -1;
-  GAMMA_CHECK(hipMemcpyAsync(q_dev_.get(), xq, (size_t)nq * dim_ * 4,
-                             hipMemcpyHostToDevice, s));
+  if (xq != nullptr) {
+    if (q_dev_.reserve((size_t)nq * dim_ * 4)) return -1;
+    GAMMA_CHECK(hipMemcpyAsync(q_dev_.get(), xq, (size_t)nq * dim_ * 4,
+                               hipMemcpyHostToDevice, s));
+  } else if (nq != cached_nq_) {
+    return -1; /* cache_queries() first */
+  }
   if (q_norms_dev_.reserve((size_t)nq * 4)) return -1;
   GAMMA_CHECK(gk::row_norms(s, q_dev_.as<float>(), nq, dim_,
                             q_norms_dev_.as<float>()));
@@ -328,6 +332,15 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
   last_timing[3] = tm.ms(2, 3) * 1000.0;
   last_timing[4] = tm.ms(3, 4) * 1000.0;
   last_timing[5] = tm.ms(0, 4) * 1000.0;
+  return 0;
+}
+
+int Engine::cache_queries(int nq, const float *xq) {
+  std::unique_lock<std::shared_mutex> g(rw_);
+  if (q_dev_.reserve((size_t)nq * dim_ * 4)) return -1;
+  GAMMA_CHECK(hipMemcpy(q_dev_.get(), xq, (size_t)nq * dim_ * 4,
+                        hipMemcpyHostToDevice));
+  cached_nq_ = nq;
   return 0;
 }
 

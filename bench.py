@@ -1,0 +1,324 @@
+#!/usr/bin/env python3
+"""bench.py — BASELINE.json's headline metric on the MI355X engine.
+
+Default (no flags): the north-star workload on 1 GPU —
+IVFPQ d=128, m=32, nbits=8, N=10M, nlist=4096, nprobe=32, k=10 with
+exact re-rank (recall_num=200), synthetic clustered float32 (seed 42),
+queries = perturbed DB samples. One "step" = one batched search of nq
+queries with the queries already resident in HBM (GammaCacheQueries).
+
+Multi-GPU (launched by the driver via torch.distributed.run): one engine
+(= one Vearch partition) per rank/GPU, per-GPU N fixed (weak scaling);
+per step every rank searches its partition and ONE RCCL all-gather over
+xGMI moves the per-partition top-k to rank 0, which merges with the
+router's semantics (client.go:1497/1558).
+
+Prints ONE JSON line from rank 0 (driver contract), including the
+roofline of the dominant kernel (the ADC list scan) and the CPU-oracle
+baseline (kind "port") timed on this host.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+# pin this rank's GPU before torch / libgamma load anything
+_local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+if "WORLD_SIZE" in os.environ and int(os.environ["WORLD_SIZE"]) > 1:
+    os.environ.setdefault("HIP_VISIBLE_DEVICES", str(_local_rank))
+
+import numpy as np  # noqa: E402
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+WORKLOADS = {
+    # the config BASELINE.json's metric is quoted on (configs[3])
+    "ivfpq_d128_n10m_nprobe32": dict(
+        kind="IVFPQ", d=128, n=10_000_000, nlist=4096, m=32, nprobe=32,
+        nq=10_000, k=10, rerank=200, train_n=160_000),
+    # configs[1]: FLAT d=128 N=1M nq=10k (parity/secondary line)
+    "flat_d128_n1m": dict(
+        kind="FLAT", d=128, n=1_000_000, nlist=0, m=0, nprobe=0,
+        nq=10_000, k=10, rerank=0, train_n=0),
+    # configs[2]: IVFFLAT d=128 N=10M nlist=4096 nprobe=32
+    "ivfflat_d128_n10m": dict(
+        kind="IVFFLAT", d=128, n=10_000_000, nlist=4096, m=0, nprobe=32,
+        nq=10_000, k=10, rerank=0, train_n=160_000),
+}
+
+HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def gen_data(n, d, seed, ncl=10000, sigma=0.10, chunk=1_000_000):
+    """Clustered-Gaussian DB (BASELINE.md protocol), chunked to bound RAM."""
+    rng = np.random.default_rng(seed)
+    centers = rng.random((ncl, d), dtype=np.float32)
+    out = np.empty((n, d), dtype=np.float32)
+    for s in range(0, n, chunk):
+        e = min(s + chunk, n)
+        asg = rng.integers(0, ncl, size=e - s)
+        out[s:e] = centers[asg]
+        out[s:e] += sigma * rng.standard_normal((e - s, d),
+                                                dtype=np.float32)
+    return out
+
+
+def build_engine(cfg, base, rank):
+    from vearch_amd import GammaEngine
+    eng = GammaEngine(path=f"/tmp/gamma_bench_r{rank}")
+    if cfg["kind"] == "FLAT":
+        params = '{"metric_type": "L2"}'
+    else:
+        params = (
+            '{"ncentroids": %d, %s"metric_type": "L2", '
+            '"training_threshold": %d, "bucket_max_size": 12800000}'
+            % (cfg["nlist"],
+               f'"nsubvector": {cfg["m"]}, ' if cfg["kind"] == "IVFPQ"
+               else "",
+               cfg["train_n"]))
+    eng.create_table(cfg["d"], cfg["kind"], params)
+    t0 = time.time()
+    step = 1_000_000
+    for s in range(0, base.shape[0], step):
+        eng.add(base[s:s + step])
+    log(f"rank{rank}: added {base.shape[0]} vecs in {time.time()-t0:.1f}s")
+    t0 = time.time()
+    eng.build_index()
+    log(f"rank{rank}: build_index in {time.time()-t0:.1f}s")
+    return eng
+
+
+def compute_recall(eng, cfg, base, queries, nq_gt=512):
+    """recall@k vs exact brute-force over THIS partition (engine FLAT
+    path, bit-exact — parity-tested against the oracle)."""
+    q = queries[:nq_gt]
+    gd, gi = eng.raw_search(q, cfg["k"], nprobe=cfg["nprobe"],
+                            rerank=cfg["rerank"])
+    from vearch_amd import proto  # noqa: F401
+    res = eng.search_pb(q, topn=100, brute=1)
+    hits = 0
+    for t in range(len(q)):
+        gt = {int(it["fields"]["_id"]) for it in res[t]["items"][:cfg["k"]]}
+        got = set(int(x) for x in gi[t] if x >= 0)
+        hits += len(gt & got)
+    return hits / (len(q) * cfg["k"])
+
+
+def cpu_baseline(eng, cfg, queries, budget_s=20.0):
+    """The CPU oracle (oracle/ref_scan.c, OpenMP 'port') on the same
+    trained model + lists, on a bounded query sample."""
+    import oracle as orc
+    from oracle.gamma_oracle import RefLib
+    if cfg["kind"] != "IVFPQ":
+        return None
+    d, nlist, M = cfg["d"], cfg["nlist"], cfg["m"]
+    ox = orc.OracleIVFPQ(d, nlist, M, metric="L2")
+    cent, books = eng.debug_model(nlist, d, M)
+    ox.centroids, ox.codebooks = cent, books
+    ids_all, codes_all, offsets = [], [], [0]
+    for ln in range(nlist):
+        li, lc = eng.debug_list(ln, M)
+        ids_all.append(li)
+        codes_all.append(lc)
+        offsets.append(offsets[-1] + len(li))
+    ox.ids = np.concatenate(ids_all)
+    ox.codes = (np.concatenate(codes_all) if ids_all else
+                np.empty((0, M), np.uint8))
+    ox.offsets = np.array(offsets, dtype=np.int64)
+    cores = RefLib.lib().oracle_num_threads()
+    # calibrate sample size: run 32 queries, scale to ~budget_s
+    sample = 32
+    t0 = time.time()
+    ox.search(queries[:sample], cfg["k"], cfg["nprobe"])
+    dt = time.time() - t0
+    per_q = dt / sample
+    sample = int(max(32, min(len(queries), budget_s / max(per_q, 1e-6))))
+    t0 = time.time()
+    ox.search(queries[:sample], cfg["k"], cfg["nprobe"])
+    dt = time.time() - t0
+    return {
+        "value": round(sample / dt, 3),
+        "unit": "QPS",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"{sample} queries of the same workload "
+                  f"(~{dt:.1f}s CPU, OpenMP over {cores} cores)",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--workload", default="ivfpq_d128_n10m_nprobe32",
+                    choices=sorted(WORKLOADS))
+    ap.add_argument("--n", type=int, default=0, help="override DB size")
+    ap.add_argument("--nq", type=int, default=0)
+    ap.add_argument("--no-recall", action="store_true")
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    cfg = dict(WORKLOADS[args.workload])
+    if args.n:
+        cfg["n"] = args.n
+        cfg["train_n"] = min(cfg["train_n"], cfg["n"])
+    if args.nq:
+        cfg["nq"] = args.nq
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+        tdist.init_process_group("nccl")
+        torch.cuda.set_device(0)  # HIP_VISIBLE_DEVICES pins the GPU
+        dist = tdist
+
+    import torch
+
+    # data: each rank owns its partition (disjoint synthetic shards)
+    t0 = time.time()
+    base = gen_data(cfg["n"], cfg["d"], seed=42 + rank)
+    rng = np.random.default_rng(43)
+    idx = rng.integers(0, cfg["n"], size=cfg["nq"])
+    queries = (base[idx] + 0.05 * rng.standard_normal(
+        (cfg["nq"], cfg["d"]), dtype=np.float32)) if rank == 0 else None
+    if world > 1:
+        # same queries everywhere: broadcast from rank 0
+        qt = torch.from_numpy(queries if rank == 0 else
+                              np.empty((cfg["nq"], cfg["d"]), np.float32))
+        dist.broadcast(qt, src=0)
+        queries = qt.numpy()
+    log(f"data gen in {time.time()-t0:.1f}s")
+
+    eng = build_engine(cfg, base, rank)
+    nq = eng.cache_queries(queries)
+
+    recall = None
+    if rank == 0 and not args.no_recall and world == 1:
+        t0 = time.time()
+        recall = compute_recall(eng, cfg, base, queries)
+        log(f"recall@{cfg['k']} = {recall:.4f} ({time.time()-t0:.1f}s)")
+
+    def step():
+        return eng.search_cached(nq, cfg["k"], nprobe=cfg["nprobe"],
+                                 rerank=cfg["rerank"])
+
+    def gather_and_merge(dists, ids):
+        if world == 1:
+            return dists, ids
+        # globalize ids: rank-local docid + rank * n; ONE all-gather over
+        # xGMI of the packed (dist-key, id) u64 blocks (~nq*k*8 B/rank)
+        gids = np.where(ids >= 0, ids + rank * cfg["n"], -1)
+        payload = torch.from_numpy(np.stack(
+            [dists.view(np.int32).astype(np.int64), gids], axis=2)).cuda()
+        out = [torch.empty_like(payload) for _ in range(world)]
+        dist.all_gather(out, payload)
+        if rank == 0:
+            from vearch_amd.merge import merge_topk
+            dl = [o[:, :, 0].cpu().numpy().astype(np.int32)
+                  .view(np.float32) for o in out]
+            il = [o[:, :, 1].cpu().numpy() for o in out]
+            return merge_topk(dl, il, cfg["k"])
+        return None, None
+
+    # warmup
+    for _ in range(args.warmup):
+        gather_and_merge(*step())
+
+    scan_ms = []
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t_start = time.time()
+    for _ in range(args.steps):
+        gather_and_merge(*step())
+        scan_ms.append(eng.last_timing()["scan_us"] / 1000.0)
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.time() - t_start
+    if world > 1:
+        el = torch.tensor([elapsed]).cuda()
+        dist.all_reduce(el, op=dist.ReduceOp.MAX)
+        elapsed = float(el.item())
+
+    if rank == 0:
+        qps = cfg["nq"] * args.steps / elapsed
+        timing = eng.last_timing()
+        # roofline of the dominant kernel (ADC list scan / flat scan)
+        if cfg["kind"] == "IVFPQ":
+            per_q_bytes = cfg["nprobe"] * (cfg["n"] / cfg["nlist"]) * \
+                (cfg["m"] + 8)
+        elif cfg["kind"] == "IVFFLAT":
+            per_q_bytes = cfg["nprobe"] * (cfg["n"] / cfg["nlist"]) * \
+                cfg["d"] * 4
+        else:
+            per_q_bytes = cfg["n"] * cfg["d"] * 4
+        t_scan_s = float(np.mean(scan_ms)) / 1000.0 if scan_ms else None
+        roofline = None
+        if t_scan_s and t_scan_s > 0:
+            achieved = cfg["nq"] * per_q_bytes / t_scan_s / 1e9
+            roofline = {
+                "bound": "hbm",
+                "achieved": round(achieved, 1),
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": round(achieved / HBM_PEAK_GBS, 4),
+                "traffic": None,
+            }
+        cpu = None
+        if not args.no_cpu_baseline and world == 1:
+            try:
+                cpu = cpu_baseline(eng, cfg, queries)
+            except Exception as ex:  # noqa: BLE001
+                log(f"cpu_baseline failed: {ex}")
+        line = {
+            "metric": "QPS @ recall@10, IVFPQ d=128 N=10M nprobe=32",
+            "value": round(qps, 2),
+            "unit": "QPS",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f32",
+            "data": "synthetic",
+            "config": {
+                "workload": args.workload,
+                "kind": cfg["kind"],
+                "d": cfg["d"],
+                "n_per_gpu": cfg["n"],
+                "nlist": cfg["nlist"],
+                "m": cfg["m"],
+                "nprobe": cfg["nprobe"],
+                "nq_per_step": cfg["nq"],
+                "topk": cfg["k"],
+                "recall_num": cfg["rerank"],
+                "recall_at_10": recall,
+                "parallelism": f"dp{world} (1 partition/GPU, RCCL "
+                               "all-gather top-k merge)",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu,
+            "stage_us_last_step": timing,
+        }
+        print(json.dumps(line), flush=True)
+
+    eng.close()
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,353 @@
+"""GPU parity tests (run on a real MI355X via gpurun): pin each stage of
+the HIP engine against the CPU oracle (oracle/ref_scan.c canonical
+arithmetic), restating the reference's own gates:
+  - FLAT: bit-exact (ids AND fp32 distances) — test_vector_index_flat.py
+    gate is exactness.
+  - IVFPQ: bit-exact vs oracle given the same trained model + probes;
+    recall floors from test_vector_index_ivfpq.py:106-111.
+  - deletes/updates, kill switch, dump/load.
+"""
+import ctypes
+import os
+
+import numpy as np
+import pytest
+
+import oracle as orc
+from oracle.gamma_oracle import RefLib, _c, _fp
+
+pytestmark = pytest.mark.gpu
+
+
+def make_engine(tmp, **kw):
+    from vearch_amd import GammaEngine
+    return GammaEngine(path=str(tmp), **kw)
+
+
+@pytest.fixture(scope="module")
+def data():
+    base = orc.gen_clustered(20000, 64, seed=42, ncl=200)
+    q = orc.gen_queries(base, 64, seed=1)
+    return base, q
+
+
+# ---------------------------------------------------------------- FLAT
+def test_flat_bitexact_small(tmp_path, data):
+    base, q = data
+    eng = make_engine(tmp_path)
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}')
+    eng.add(base)
+    gd, gi = eng.raw_search(q, 10)
+    od, oi = orc.flat_search(base, q, 10, "L2")
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    eng.close()
+
+
+def test_flat_bitexact_ip(tmp_path, data):
+    base, q = data
+    eng = make_engine(tmp_path)
+    eng.create_table(64, "FLAT", '{"metric_type": "InnerProduct"}')
+    eng.add(base)
+    gd, gi = eng.raw_search(q, 10)
+    od, oi = orc.flat_search(base, q, 10, "InnerProduct")
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    eng.close()
+
+
+def test_flat_gemm_path_bitexact(tmp_path):
+    """nq >= 512 and N >= 200k takes the chunked MFMA GEMM + select path;
+    results must still be bit-exact after the canonical re-rank."""
+    base = orc.gen_clustered(220000, 32, seed=5, ncl=500)
+    q = orc.gen_queries(base, 600, seed=6)
+    eng = make_engine("/tmp/gamma_flat_gemm")
+    eng.create_table(32, "FLAT", '{"metric_type": "L2"}')
+    eng.add(base)
+    gd, gi = eng.raw_search(q, 10)
+    od, oi = orc.flat_search(base, q, 10, "L2")
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    eng.close()
+
+
+def test_flat_deletes(tmp_path, data):
+    base, q = data
+    eng = make_engine(tmp_path)
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}')
+    eng.add(base)
+    for vid in range(0, 20000, 7):
+        eng.delete_doc(str(vid))
+    bm = np.zeros((20000 + 7) // 8, dtype=np.uint8)
+    for vid in range(0, 20000, 7):
+        bm[vid >> 3] |= 1 << (vid & 7)
+    gd, gi = eng.raw_search(q, 10)
+    od, oi = orc.flat_search(base, q, 10, "L2", del_bitmap=bm)
+    assert np.array_equal(gi, oi)
+    assert not any(v % 7 == 0 for v in gi.ravel().tolist() if v >= 0)
+    eng.close()
+
+
+# ---------------------------------------------------------------- IVFPQ
+def _oracle_from_engine(eng, d, nlist, M, metric="L2"):
+    ox = orc.OracleIVFPQ(d, nlist, M, metric=metric)
+    cent, books = eng.debug_model(nlist, d, M)
+    ox.centroids, ox.codebooks = cent, books
+    ids_all, codes_all, offsets = [], [], [0]
+    for ln in range(nlist):
+        li, lc = eng.debug_list(ln, M)
+        ids_all.append(li)
+        codes_all.append(lc)
+        offsets.append(offsets[-1] + len(li))
+    ox.ids = np.concatenate(ids_all)
+    ox.codes = np.concatenate(codes_all)
+    ox.offsets = np.array(offsets, dtype=np.int64)
+    return ox
+
+
+@pytest.fixture(scope="module")
+def ivfpq_engine(data):
+    base, q = data
+    eng = make_engine("/tmp/gamma_ivfpq_mod")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 64, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 8000}')
+    eng.add(base)
+    eng.build_index()
+    yield eng
+    eng.close()
+
+
+def test_ivfpq_bucket_population(ivfpq_engine):
+    total = 0
+    for ln in range(64):
+        li, _ = ivfpq_engine.debug_list(ln, 16)
+        total += len(li)
+    assert total == 20000
+
+
+def test_ivfpq_adc_bitexact_vs_oracle(data, ivfpq_engine):
+    base, q = data
+    eng = ivfpq_engine
+    gd, gi = eng.raw_search(q, 10, nprobe=16)
+    ox = _oracle_from_engine(eng, 64, 64, 16)
+    _, probes = eng.debug_coarse_assign(q, 16)
+    od, oi = ox.search(q, 10, 16, probes=probes)
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+
+
+def test_ivfpq_coarse_assign_close_to_oracle(data, ivfpq_engine):
+    """GEMM-form distances round differently; require same probe sets
+    wherever the oracle's distance gap exceeds fp32 noise."""
+    base, q = data
+    eng = ivfpq_engine
+    gdist, glists = eng.debug_coarse_assign(q, 8)
+    ox = _oracle_from_engine(eng, 64, 64, 16)
+    odist, olists = ox.coarse_assign(q, 8)
+    assert np.allclose(gdist, odist, rtol=1e-4, atol=1e-4)
+    agree = (glists == olists).mean()
+    assert agree > 0.95, f"probe agreement too low: {agree}"
+
+
+def test_ivfpq_rerank_exact_distances(data, ivfpq_engine):
+    base, q = data
+    eng = ivfpq_engine
+    gd, gi = eng.raw_search(q, 10, nprobe=16, rerank=100)
+    lib = RefLib.lib()
+    lib.oracle_l2sqr.restype = ctypes.c_float
+    for t in range(q.shape[0]):
+        for j in range(10):
+            if gi[t, j] < 0:
+                continue
+            want = lib.oracle_l2sqr(_fp(_c(q[t], np.float32)),
+                                    _fp(_c(base[gi[t, j]], np.float32)), 64)
+            assert gd[t, j] == want, (t, j)
+        # sorted ascending, ties by id
+        row = [(gd[t, j], gi[t, j]) for j in range(10) if gi[t, j] >= 0]
+        assert row == sorted(row)
+
+
+def test_ivfpq_recall_floors(data, ivfpq_engine):
+    """Reference gates (test_vector_index_ivfpq.py:106-111) with the
+    rerank leg on: R@1>=0.6, R@10>=0.9."""
+    base, q = data
+    eng = ivfpq_engine
+    _, gti = orc.flat_topk_f64(base, q, 100)
+    gd, gi = eng.raw_search(q, 10, nprobe=16, rerank=100)
+    r1 = orc.recall_at(gti, gi[:, :1], 1)
+    r10 = orc.recall_at(gti, gi, 10)
+    assert r1 >= 0.6, f"recall@1={r1}"
+    assert r10 >= 0.9, f"recall@10={r10}"
+
+
+def test_ivfpq_delete_and_update(data, ivfpq_engine):
+    base, q = data
+    eng = ivfpq_engine
+    gd0, gi0 = eng.raw_search(q[:4], 5, nprobe=64)
+    victim = int(gi0[0, 0])
+    eng.delete_doc(str(victim))
+    gd1, gi1 = eng.raw_search(q[:4], 5, nprobe=64)
+    assert victim not in gi1[0].tolist()
+    # re-add via the Doc path: becomes a new docid, old stays dead
+    eng.add_doc(str(victim), base[victim])
+    gd2, gi2 = eng.raw_search(q[:4], 5, nprobe=64)
+    new_id = eng.num_docs() - 1
+    assert new_id in gi2[0].tolist()
+
+
+# ---------------------------------------------------------------- IVFFLAT
+def test_ivfflat_bitexact_vs_oracle(data):
+    base, q = data
+    eng = make_engine("/tmp/gamma_ivfflat")
+    eng.create_table(
+        64, "IVFFLAT",
+        '{"ncentroids": 64, "metric_type": "L2", '
+        '"training_threshold": 8000}')
+    eng.add(base)
+    eng.build_index()
+    gd, gi = eng.raw_search(q, 10, nprobe=16)
+    # oracle with the engine's lists + probes
+    cent, _ = eng.debug_model(64, 64, 0)
+    ids_all, vec_all, offsets = [], [], [0]
+    for ln in range(64):
+        li, lc = eng.debug_list(ln, 64 * 4)
+        ids_all.append(li)
+        vec_all.append(lc.view(np.float32).reshape(-1, 64))
+        offsets.append(offsets[-1] + len(li))
+    ids = np.concatenate(ids_all)
+    vecs = np.ascontiguousarray(np.concatenate(vec_all), dtype=np.float32)
+    offsets = np.array(offsets, dtype=np.int64)
+    _, probes = eng.debug_coarse_assign(q, 16)
+    lib = RefLib.lib()
+    from oracle.gamma_oracle import _ip64, _up8
+    od = np.empty((64, 10), dtype=np.float32)
+    oi = np.empty((64, 10), dtype=np.int64)
+    lib.oracle_ivfflat_search(
+        64, 64, 64, _fp(_c(q, np.float32)), _ip64(offsets), _ip64(ids),
+        _fp(vecs), 16, _ip64(_c(probes, np.int64)), _up8(None), 0, 10,
+        _fp(od), _ip64(oi))
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    eng.close()
+
+
+# ----------------------------------------------------------- C-ABI Search
+def test_search_pb_end_to_end(data, ivfpq_engine):
+    base, q = data
+    eng = ivfpq_engine
+    res = eng.search_pb(q[:8], topn=5,
+                        index_params='{"nprobe": 16, "recall_num": 50}')
+    gd, gi = eng.raw_search(q[:8], 5, nprobe=16, rerank=50)
+    assert len(res) == 8
+    for t in range(8):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        scores = [it["score"] for it in res[t]["items"]]
+        want = [i for i in gi[t].tolist() if i >= 0]
+        assert ids == want
+        assert np.allclose(scores, gd[t][:len(scores)])
+
+
+def test_search_pb_l2_sqrt(data, ivfpq_engine):
+    base, q = data
+    eng = ivfpq_engine
+    res = eng.search_pb(q[:4], topn=3, index_params='{"nprobe": 16}',
+                        l2_sqrt=True)
+    gd, gi = eng.raw_search(q[:4], 3, nprobe=16)
+    for t in range(4):
+        scores = [it["score"] for it in res[t]["items"]]
+        assert np.allclose(scores, np.sqrt(gd[t][:len(scores)]), rtol=1e-6)
+
+
+def test_kill_switch(data, ivfpq_engine):
+    from vearch_amd import clear_kill, set_kill
+    base, q = data
+    eng = ivfpq_engine
+    set_kill("killme", 1)
+    with pytest.raises((InterruptedError, RuntimeError)):
+        eng.search_pb(q[:4], topn=3, request_id="killme", partition_id=1)
+    clear_kill("killme", 1)
+    res = eng.search_pb(q[:4], topn=3, request_id="killme", partition_id=1)
+    assert len(res) == 4
+
+
+def test_query_and_get_doc(data, ivfpq_engine):
+    eng = ivfpq_engine
+    import ctypes as c
+    from vearch_amd.engine import lib
+    out = c.c_char_p()
+    n = c.c_int()
+    rc = lib().GetDocByID(eng.h, b"123", 3, c.byref(out), c.byref(n))
+    assert rc == 0
+    buf = c.string_at(out, n.value)
+    from vearch_amd.fbsenc import DATA_VECTOR
+    # parse with our own fbs reader via the C++ roundtrip hook
+    L = lib()
+    L.GammaTestDocRoundtrip.argtypes = [c.c_char_p, c.c_int,
+                                        c.POINTER(c.c_char_p),
+                                        c.POINTER(c.c_int)]
+    out2 = c.c_char_p()
+    n2 = c.c_int()
+    assert L.GammaTestDocRoundtrip(buf, len(buf), c.byref(out2),
+                                   c.byref(n2)) == 0
+
+
+def test_dump_load_roundtrip(data):
+    base, q = data
+    path = "/tmp/gamma_dumpload"
+    os.makedirs(path, exist_ok=True)
+    eng = make_engine(path)
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 32, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 4000}')
+    eng.add(base[:8000])
+    eng.build_index()
+    gd0, gi0 = eng.raw_search(q, 10, nprobe=32)
+    eng.dump()
+    eng.close()
+    eng2 = make_engine(path)
+    eng2.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 32, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 4000}')
+    eng2.load()
+    assert eng2.num_docs() == 8000
+    gd1, gi1 = eng2.raw_search(q, 10, nprobe=32)
+    assert np.array_equal(gi0, gi1)
+    assert np.array_equal(gd0, gd1)
+    eng2.close()
+
+
+def test_untrained_falls_back_to_flat(data):
+    base, q = data
+    eng = make_engine("/tmp/gamma_untrained")
+    eng.create_table(64, "IVFPQ",
+                     '{"ncentroids": 64, "nsubvector": 16, '
+                     '"metric_type": "L2"}')
+    eng.add(base[:3000])
+    # no build_index: Search must brute-force (engine.cc:286-288 analog)
+    gd, gi = eng.raw_search(q, 10)
+    od, oi = orc.flat_search(base[:3000], q, 10, "L2")
+    assert np.array_equal(gi, oi)
+    eng.close()
+
+
+def test_ivfpq_north_star_shape_recall():
+    """d=128 m=32 (the north-star shape) at small N: recall gate >=0.9
+    with rerank; BASELINE's >=0.95 is checked in bench at full size."""
+    base = orc.gen_clustered(50000, 128, seed=42, ncl=500)
+    q = orc.gen_queries(base, 64, seed=2)
+    _, gti = orc.flat_topk_f64(base, q, 100)
+    eng = make_engine("/tmp/gamma_ns")
+    eng.create_table(
+        128, "IVFPQ",
+        '{"ncentroids": 256, "nsubvector": 32, "metric_type": "L2", '
+        '"training_threshold": 20000}')
+    eng.add(base)
+    eng.build_index()
+    _, gi = eng.raw_search(q, 10, nprobe=32, rerank=200)
+    r10 = orc.recall_at(gti, gi, 10)
+    assert r10 >= 0.9, f"recall@10={r10}"
+    eng.close()

@@ -1,0 +1,68 @@
+"""Multi-partition path on CPU: world_size=2 over gloo exercises the same
+all-gather + router-merge (client.go:1497/1558 semantics) that bench.py
+runs over RCCL on the 8-GPU node. Partition search is played by the CPU
+oracle; the collective and merge code are the real ones."""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+import oracle as orc
+from vearch_amd.merge import merge_topk
+
+K = 10
+
+
+def _worker(rank, world, port, out_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    base = orc.gen_clustered(4000, 32, seed=42, ncl=50)
+    q = orc.gen_queries(base, 16, seed=3)
+    # round-robin sharding: rank owns rows where i % world == rank
+    my_rows = np.arange(rank, 4000, world)
+    my_base = base[my_rows]
+    d_loc, i_loc = orc.flat_search(my_base, q, K, "L2")
+    # globalize local ids
+    gi = np.where(i_loc >= 0, my_rows[np.clip(i_loc, 0, None)], -1)
+    payload = torch.from_numpy(
+        np.concatenate([d_loc.astype(np.float32).reshape(16, K, 1),
+                        gi.astype(np.float32).reshape(16, K, 1)], axis=2))
+    gathered = [torch.empty_like(payload) for _ in range(world)]
+    dist.all_gather(gathered, payload)
+    if rank == 0:
+        dl = [g[:, :, 0].numpy() for g in gathered]
+        il = [g[:, :, 1].numpy().astype(np.int64) for g in gathered]
+        md, mi = merge_topk(dl, il, K)
+        od, oi = orc.flat_search(base, q, K, "L2")
+        out_q.put((np.array_equal(mi, oi), np.allclose(md, od)))
+    dist.destroy_process_group()
+
+
+def test_two_partition_merge_equals_global():
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    port = 29611
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, out_q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    ok_ids, ok_dists = out_q.get(timeout=300)
+    for p in procs:
+        p.join(timeout=120)
+    assert ok_ids, "merged ids != global FLAT ids"
+    assert ok_dists, "merged dists != global FLAT dists"
+
+
+def test_merge_topk_ties_and_padding():
+    d1 = np.array([[0.5, 1.0, -1.0]], dtype=np.float32)
+    i1 = np.array([[7, 3, -1]], dtype=np.int64)
+    d2 = np.array([[0.5, 2.0, -1.0]], dtype=np.float32)
+    i2 = np.array([[4, 9, -1]], dtype=np.int64)
+    md, mi = merge_topk([d1, d2], [i1, i2], 4)
+    assert mi[0].tolist() == [4, 7, 3, 9]  # tie 0.5 -> lower id first
+    md, mi = merge_topk([d1, d2], [i1, i2], 6)
+    assert mi[0].tolist() == [4, 7, 3, 9, -1, -1]

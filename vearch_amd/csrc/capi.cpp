@@ -252,7 +252,8 @@ struct CStatus Search(void *engine, const char *request_str, int req_len,
   std::vector<int64_t> ids((size_t)nq * k);
   int rc = e->search(nq, (const float *)vq.value.data(), k, nprobe,
                      recall_num, metric, req.brute == 1, req.request_id,
-                     dists.data(), ids.data(), req.l2_sqrt);
+                     req.partition_id, dists.data(), ids.data(),
+                     req.l2_sqrt);
   if (rc == -2) return err_status(-2, "request killed");
   if (rc != 0) return err_status(1, "search failed");
 
@@ -412,7 +413,7 @@ int GammaRawSearch(void *engine, int nq, const float *xq, int k, int nprobe,
                    int64_t *out_ids) {
   if (!engine) return -1;
   return static_cast<Engine *>(engine)->search(
-      nq, xq, k, nprobe, rerank, metric, false, "", out_dists, out_ids);
+      nq, xq, k, nprobe, rerank, metric, false, "", 0, out_dists, out_ids);
 }
 
 int GammaCacheQueries(void *engine, int nq, const float *xq) {
@@ -425,7 +426,7 @@ int GammaRawSearchCached(void *engine, int nq, int k, int nprobe,
                          int64_t *out_ids) {
   if (!engine) return -1;
   return static_cast<Engine *>(engine)->search(
-      nq, nullptr, k, nprobe, rerank, metric, false, "", out_dists,
+      nq, nullptr, k, nprobe, rerank, metric, false, "", 0, out_dists,
       out_ids);
 }
 

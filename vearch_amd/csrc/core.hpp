@@ -224,7 +224,8 @@ class Engine {
    * nq must equal the cached count). */
   int search(int nq, const float *xq, int k, int nprobe, int recall_num,
              int metric, bool brute_force, const std::string &request_id,
-             float *out_dists, int64_t *out_ids, bool l2_sqrt = false);
+             int partition_id, float *out_dists, int64_t *out_ids,
+             bool l2_sqrt = false);
   /* upload queries once; later search(nq, nullptr, ...) reuses them */
   int cache_queries(int nq, const float *xq);
   int cached_nq() const { return cached_nq_; }

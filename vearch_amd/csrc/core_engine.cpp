@@ -228,11 +228,11 @@ int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
 
 int Engine::search(int nq, const float *xq, int k, int nprobe,
                    int recall_num, int metric, bool brute_force,
-                   const std::string &request_id, float *out_dists,
-                   int64_t *out_ids, bool l2_sqrt) {
+                   const std::string &request_id, int partition_id,
+                   float *out_dists, int64_t *out_ids, bool l2_sqrt) {
   if (!table_created_ || nq <= 0 || k <= 0) return -1;
   std::shared_lock<std::shared_mutex> g(rw_);
-  const int pid = 0; /* partition id is carried by the kill key */
+  const int pid = partition_id;
   bool ip = metric == 0 ? params_.metric_ip : (metric == 2);
   hipStream_t s = stream_;
 

@@ -1,0 +1,151 @@
+"""Protobuf wire-format encoder/decoder for the vearchpb messages the test
+and bench harness exchanges with libgamma.so, playing the role of the Go
+caller (internal/proto/router_grpc.proto). No protoc involved."""
+import struct
+
+
+def _varint(v):
+    out = bytearray()
+    while v >= 0x80:
+        out.append((v & 0x7F) | 0x80)
+        v >>= 7
+    out.append(v)
+    return bytes(out)
+
+
+def _tag(field, wt):
+    return _varint((field << 3) | wt)
+
+
+def _ld(field, data):
+    return _tag(field, 2) + _varint(len(data)) + data
+
+
+def _vint(field, v):
+    return b"" if v == 0 else _tag(field, 0) + _varint(v)
+
+
+def _vdouble(field, v):
+    return _tag(field, 1) + struct.pack("<d", v)
+
+
+def encode_head(request_id="", partition_id=None):
+    params = b""
+    if request_id:
+        params += _ld(7, _ld(1, b"request_id") + _ld(2, request_id.encode()))
+    if partition_id is not None:
+        params += _ld(
+            7, _ld(1, b"partition_id") + _ld(2, str(partition_id).encode()))
+    return params
+
+
+def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
+                          request_id="req1", partition_id=1,
+                          index_params="", min_score=None, max_score=None,
+                          brute=0, fields=("_id",), l2_sqrt=False,
+                          is_vector_value=False):
+    """vearchpb.SearchRequest (router_grpc.proto:168-192)."""
+    out = b""
+    out += _ld(1, encode_head(request_id, partition_id))
+    out += _vint(2, req_num)
+    out += _vint(3, topn)
+    out += _vint(4, brute)
+    vq = _ld(1, vec_name.encode()) + _ld(2, queries_f32_bytes)
+    if min_score is not None:
+        vq += _vdouble(3, min_score)
+    if max_score is not None:
+        vq += _vdouble(4, max_score)
+    out += _ld(5, vq)
+    for f in fields:
+        out += _ld(6, f.encode())
+    if index_params:
+        out += _ld(9, index_params.encode())
+    if l2_sqrt:
+        out += _vint(11, 1)
+    if is_vector_value:
+        out += _vint(12, 1)
+    return out
+
+
+def encode_query_request(document_ids, fields=("_id",), request_id="req1",
+                         partition_id=1, is_vector_value=False):
+    out = b""
+    out += _ld(1, encode_head(request_id, partition_id))
+    for d in document_ids:
+        out += _ld(2, d.encode())
+    for f in fields:
+        out += _ld(7, f.encode())
+    if is_vector_value:
+        out += _vint(8, 1)
+    return out
+
+
+# ------------------------------------------------------------- decoding
+def _read_varint(buf, i):
+    v = 0
+    shift = 0
+    while True:
+        b = buf[i]
+        i += 1
+        v |= (b & 0x7F) << shift
+        if not b & 0x80:
+            return v, i
+        shift += 7
+
+
+def _fields(buf):
+    i = 0
+    n = len(buf)
+    while i < n:
+        key, i = _read_varint(buf, i)
+        f, wt = key >> 3, key & 7
+        if wt == 0:
+            v, i = _read_varint(buf, i)
+        elif wt == 1:
+            v = struct.unpack("<d", buf[i:i + 8])[0]
+            i += 8
+        elif wt == 2:
+            ln, i = _read_varint(buf, i)
+            v = buf[i:i + ln]
+            i += ln
+        elif wt == 5:
+            v = struct.unpack("<f", buf[i:i + 4])[0]
+            i += 4
+        else:
+            raise ValueError(f"wiretype {wt}")
+        yield f, wt, v
+
+
+def decode_search_response(buf):
+    """-> list per query: {'max_score','total','items':[{'score','fields':
+    {name: bytes}}]} (router_grpc.proto:195-216 SearchResult subset)."""
+    results = []
+    for f, wt, v in _fields(buf):
+        if f != 2:
+            continue
+        res = {"max_score": None, "total": None, "msg": "", "items": []}
+        for rf, rwt, rv in _fields(v):
+            if rf == 2:
+                res["max_score"] = rv
+            elif rf == 5:
+                for sf, _, sv in _fields(rv):
+                    if sf == 1:
+                        res["total"] = sv
+            elif rf == 6:
+                res["msg"] = rv.decode(errors="replace")
+            elif rf == 7:
+                item = {"score": None, "fields": {}}
+                for itf, _, itv in _fields(rv):
+                    if itf == 1:
+                        item["score"] = itv
+                    elif itf == 2:
+                        fd = {"name": None, "value": b""}
+                        for ff, _, fvv in _fields(itv):
+                            if ff == 1:
+                                fd["name"] = fvv.decode()
+                            elif ff == 3:
+                                fd["value"] = fvv
+                        item["fields"][fd["name"]] = fd["value"]
+                res["items"].append(item)
+        results.append(res)
+    return results

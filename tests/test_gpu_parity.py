@@ -182,9 +182,17 @@ def test_ivfpq_recall_floors(data, ivfpq_engine):
     assert r10 >= 0.9, f"recall@10={r10}"
 
 
-def test_ivfpq_delete_and_update(data, ivfpq_engine):
+def test_ivfpq_delete_and_update(data):
+    """Own engine: mutates state (the shared ivfpq_engine stays clean so
+    later pb tests can equate _id strings with docids)."""
     base, q = data
-    eng = ivfpq_engine
+    eng = make_engine("/tmp/gamma_ivfpq_mut")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 64, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 8000}')
+    eng.add(base)
+    eng.build_index()
     gd0, gi0 = eng.raw_search(q[:4], 5, nprobe=64)
     victim = int(gi0[0, 0])
     eng.delete_doc(str(victim))
@@ -195,6 +203,7 @@ def test_ivfpq_delete_and_update(data, ivfpq_engine):
     gd2, gi2 = eng.raw_search(q[:4], 5, nprobe=64)
     new_id = eng.num_docs() - 1
     assert new_id in gi2[0].tolist()
+    eng.close()
 
 
 # ---------------------------------------------------------------- IVFFLAT

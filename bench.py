@@ -69,9 +69,9 @@ def gen_data(n, d, seed, ncl=10000, sigma=0.10, chunk=1_000_000):
     return out
 
 
-def build_engine(cfg, base, rank):
+def build_engine(cfg, base, rank, path=None):
     from vearch_amd import GammaEngine
-    eng = GammaEngine(path=f"/tmp/gamma_bench_r{rank}")
+    eng = GammaEngine(path=path or f"/tmp/gamma_bench_r{rank}")
     if cfg["kind"] == "FLAT":
         params = '{"metric_type": "L2"}'
     else:
@@ -94,7 +94,7 @@ def build_engine(cfg, base, rank):
     return eng
 
 
-def compute_recall(eng, cfg, base, queries, nq_gt=512):
+def compute_recall(eng, cfg, queries, nq_gt=512):
     """recall@k vs exact brute-force over THIS partition (engine FLAT
     path, bit-exact — parity-tested against the oracle)."""
     q = queries[:nq_gt]
@@ -163,6 +163,11 @@ def main():
     ap.add_argument("--nq", type=int, default=0)
     ap.add_argument("--no-recall", action="store_true")
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--index-dir", default="",
+                    help="reuse a built index: load from DIR if a dump "
+                         "exists there, else build and dump to DIR")
+    ap.add_argument("--prepare-only", action="store_true",
+                    help="with --index-dir: build+dump, skip the bench")
     args = ap.parse_args()
 
     cfg = dict(WORKLOADS[args.workload])
@@ -184,28 +189,62 @@ def main():
 
     import torch
 
-    # data: each rank owns its partition (disjoint synthetic shards)
-    t0 = time.time()
-    base = gen_data(cfg["n"], cfg["d"], seed=42 + rank)
-    rng = np.random.default_rng(43)
-    idx = rng.integers(0, cfg["n"], size=cfg["nq"])
-    queries = (base[idx] + 0.05 * rng.standard_normal(
-        (cfg["nq"], cfg["d"]), dtype=np.float32)) if rank == 0 else None
-    if world > 1:
-        # same queries everywhere: broadcast from rank 0
-        qt = torch.from_numpy(queries if rank == 0 else
-                              np.empty((cfg["nq"], cfg["d"]), np.float32))
-        dist.broadcast(qt, src=0)
-        queries = qt.numpy()
-    log(f"data gen in {time.time()-t0:.1f}s")
+    idir = args.index_dir
+    have_dump = bool(idir) and os.path.exists(
+        os.path.join(idir, "gamma.dump")) and os.path.exists(
+        os.path.join(idir, "queries.npy"))
 
-    eng = build_engine(cfg, base, rank)
+    if have_dump:
+        from vearch_amd import GammaEngine
+        t0 = time.time()
+        eng = GammaEngine(path=idir)
+        if cfg["kind"] == "FLAT":
+            params = '{"metric_type": "L2"}'
+        else:
+            params = (
+                '{"ncentroids": %d, %s"metric_type": "L2", '
+                '"training_threshold": %d, "bucket_max_size": 12800000}'
+                % (cfg["nlist"],
+                   f'"nsubvector": {cfg["m"]}, ' if cfg["kind"] == "IVFPQ"
+                   else "", cfg["train_n"]))
+        eng.create_table(cfg["d"], cfg["kind"], params)
+        eng.load()
+        queries = np.load(os.path.join(idir, "queries.npy"))
+        cfg["nq"] = queries.shape[0]
+        log(f"loaded index ({eng.num_docs()} docs) + queries in "
+            f"{time.time()-t0:.1f}s")
+    else:
+        # data: each rank owns its partition (disjoint synthetic shards)
+        t0 = time.time()
+        base = gen_data(cfg["n"], cfg["d"], seed=42 + rank)
+        rng = np.random.default_rng(43)
+        idx = rng.integers(0, cfg["n"], size=cfg["nq"])
+        queries = (base[idx] + 0.05 * rng.standard_normal(
+            (cfg["nq"], cfg["d"]), dtype=np.float32)) if rank == 0 else None
+        if world > 1:
+            # same queries everywhere: broadcast from rank 0
+            qt = torch.from_numpy(queries if rank == 0 else
+                                  np.empty((cfg["nq"], cfg["d"]),
+                                           np.float32))
+            dist.broadcast(qt, src=0)
+            queries = qt.numpy()
+        log(f"data gen in {time.time()-t0:.1f}s")
+        if idir:
+            os.makedirs(idir, exist_ok=True)
+        eng = build_engine(cfg, base, rank, path=idir or None)
+        if idir:
+            eng.dump()
+            np.save(os.path.join(idir, "queries.npy"), queries)
+            log("index dumped")
+            if args.prepare_only:
+                eng.close()
+                return
     nq = eng.cache_queries(queries)
 
     recall = None
     if rank == 0 and not args.no_recall and world == 1:
         t0 = time.time()
-        recall = compute_recall(eng, cfg, base, queries)
+        recall = compute_recall(eng, cfg, queries)
         log(f"recall@{cfg['k']} = {recall:.4f} ({time.time()-t0:.1f}s)")
 
     def step():

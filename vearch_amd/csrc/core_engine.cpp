@@ -259,7 +259,7 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
    * (search_preassigned, ivfpq.cc:765-776) */
   int k2 = std::max(k, recall_num);
   bool rerank = recall_num > 0;
-  if (k2 > 1536) return -1;
+  if (k2 > 1024) return -1;
 
   tm.rec(0);
   if (xq != nullptr) {
@@ -284,7 +284,7 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
   if (use_flat) {
     /* FLAT margin so GEMM-order rounding cannot evict a true top-k
      * member before the canonical re-rank (DESIGN.md numerics note) */
-    int kf = std::min<int64_t>((int64_t)k2 + 64, 1536);
+    int kf = std::min<int64_t>((int64_t)k2 + 64, 1088);
     kf = (int)std::min<int64_t>(kf, std::max<int64_t>(raw_.size(), 1));
     if (keys_dev_.reserve((size_t)nq * kf * 8)) return -1;
     if (flat_search_keys(q_dev_.as<float>(), nq, kf,

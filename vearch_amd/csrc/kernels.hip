@@ -106,7 +106,7 @@ k_select_from_dots(int nq, int64_t ncols, int64_t col_base, int64_t ld,
   sel.init(sortbuf, res, state, k2);
   if (seeded) sel.seed(state_keys + (int64_t)q * k2, k2);
 
-  const int64_t chunk = (int64_t)blockDim.x * GAMMA_SEL_CHUNK;
+  const int64_t chunk = (int64_t)blockDim.x * 2;
   for (int64_t c0 = 0; c0 < ncols; c0 += chunk) {
     int64_t cend = min(c0 + chunk, ncols);
     for (int64_t c = c0 + threadIdx.x; c < cend; c += blockDim.x) {
@@ -116,7 +116,7 @@ k_select_from_dots(int nq, int64_t ncols, int64_t col_base, int64_t ld,
       float dist = l2 ? (qn + bnorms[id] - 2.0f * dot) : dot;
       sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
     }
-    sel.maybe_flush();
+    sel.maybe_flush(2 * blockDim.x);
   }
   sel.finish();
   for (int i = threadIdx.x; i < k2; i += blockDim.x)
@@ -278,7 +278,7 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
           sel.push(gamma_make_key<IP>(dis, (uint32_t)id));
         }
       }
-      sel.maybe_flush();
+      sel.maybe_flush(blockDim.x);
     }
     __syncthreads();  /* lut rebuilt next list: scan readers done */
   }
@@ -376,9 +376,9 @@ k_ivfflat_scan(int nq, int d, int nprobe, int k2,
       /* rotate the pushing lane so buffers fill evenly */
       if (live && lane == (int)(j & 63))
         sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
-      if ((it & 63) == 63) sel.maybe_flush();
+      if ((it & 15) == 15) sel.maybe_flush(blockDim.x);
     }
-    sel.maybe_flush();
+    sel.maybe_flush(blockDim.x);
   }
   sel.finish();
   for (int i = threadIdx.x; i < k2; i += blockDim.x)
@@ -460,7 +460,7 @@ k_flat_stream(int nq, int64_t n, int d, int k2,
     }
     if (live && lane == (int)(j & 63))
       sel.push(gamma_make_key<IP>(dist, (uint32_t)j));
-    if ((it & 63) == 63) sel.maybe_flush();
+    if ((it & 15) == 15) sel.maybe_flush(blockDim.x);
   }
   sel.finish();
   for (int i = threadIdx.x; i < k2; i += blockDim.x)

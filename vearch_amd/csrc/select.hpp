@@ -74,25 +74,30 @@ __device__ inline void gamma_bitonic_sort(uint64_t *buf, int n) {
 }
 
 #define GAMMA_SORT_CAP 2048   /* LDS sort scratch, u64 x 2048 = 16 KB */
-#define GAMMA_SEL_BUF 16      /* per-thread candidate registers */
-#define GAMMA_SEL_CHUNK 8     /* max pushes per thread between flush checks */
+#define GAMMA_SEL_MAX_K 1088  /* cap - margin must stay >= 0 */
 
+/* Candidates append into the shared LDS buffer through one LDS atomic;
+ * all selector state is scalar registers or LDS (no per-thread arrays —
+ * runtime-indexed register arrays land in scratch memory, §5.4 rule 20,
+ * which made v1 of this selector 60x slower).
+ *
+ * Invariant: the caller calls maybe_flush(margin) (all threads together)
+ * often enough that at most `margin` pushes can happen block-wide between
+ * two checks; the buffer then never overflows its cap = SORT_CAP - k. */
 struct GammaSelector {
   uint64_t *sortbuf;  /* uint64_t[GAMMA_SORT_CAP] in LDS */
   uint64_t *res;      /* uint64_t[k] in LDS, sorted ascending after flush */
-  int *state;         /* int[2] in LDS: [0]=dump counter, [1]=flags */
-  int k;              /* 1 <= k <= GAMMA_SORT_CAP/2 */
+  int *state;         /* int[1] in LDS: append counter */
+  int k, cap;
   uint64_t thresh;
-  uint64_t buf[GAMMA_SEL_BUF];
-  int cnt;
 
   __device__ void init(uint64_t *sortbuf_, uint64_t *res_, int *state_,
                        int k_) {
     sortbuf = sortbuf_; res = res_; state = state_; k = k_;
+    cap = GAMMA_SORT_CAP - k;
     thresh = GAMMA_KEY_EMPTY;
-    cnt = 0;
     for (int i = threadIdx.x; i < k; i += blockDim.x) res[i] = GAMMA_KEY_EMPTY;
-    if (threadIdx.x == 0) { state[0] = 0; state[1] = 0; }
+    if (threadIdx.x == 0) state[0] = 0;
     __syncthreads();
   }
 
@@ -100,20 +105,21 @@ struct GammaSelector {
   __device__ void seed(const uint64_t *keys, int n) {
     for (int i = threadIdx.x; i < k; i += blockDim.x)
       res[i] = (i < n) ? keys[i] : GAMMA_KEY_EMPTY;
+    __syncthreads();
     flush_();
   }
 
   __device__ __forceinline__ void push(uint64_t key) {
     if (key < thresh) {
-      buf[cnt++] = key;
-      if (cnt > GAMMA_SEL_BUF - GAMMA_SEL_CHUNK) atomicOr(&state[1], 1);
+      int idx = atomicAdd(&state[0], 1);
+      sortbuf[idx] = key; /* in range by the maybe_flush invariant */
     }
   }
 
-  /* all threads must arrive; flushes when any thread is near-full */
-  __device__ __forceinline__ void maybe_flush() {
+  /* all threads must arrive together */
+  __device__ __forceinline__ void maybe_flush(int margin) {
     __syncthreads();
-    if (state[1]) flush_();
+    if (state[0] > cap - margin) flush_();
   }
 
   __device__ void finish() {
@@ -122,39 +128,22 @@ struct GammaSelector {
   }
 
  private:
-  /* Drain every thread's buffer into sortbuf (in rounds if needed),
-   * merge with res, keep top-k, refresh threshold. */
   __device__ void flush_() {
     const int nt = blockDim.x, tid = threadIdx.x;
-    const int cap = GAMMA_SORT_CAP - k;
-    for (;;) {
-      if (tid == 0) { state[0] = 0; state[1] = 0; }
-      __syncthreads();
-      int base = (cnt > 0) ? atomicAdd(&state[0], cnt) : 0;
-      int fit = 0;
-      if (cnt > 0 && base < cap) fit = min(cnt, cap - base);
-      if (cnt > fit) atomicOr(&state[1], 1); /* leftovers -> another round */
-      __syncthreads();
-      int total = min(state[0], cap);
-      int more = state[1];
-      __syncthreads();               /* everyone read state before reuse */
-      /* pad candidate region, then append res at [total, total+k) */
-      int n = 1;
-      while (n < total + k) n <<= 1;
-      for (int i = tid; i < n; i += nt) sortbuf[i] = GAMMA_KEY_EMPTY;
-      __syncthreads();
-      for (int i = 0; i < fit; i++) sortbuf[base + i] = buf[i];
-      for (int i = fit; i < cnt; i++) buf[i - fit] = buf[i];
-      cnt -= fit;
-      for (int i = tid; i < k; i += nt) sortbuf[total + i] = res[i];
-      gamma_bitonic_sort(sortbuf, n);
-      for (int i = tid; i < k; i += nt) res[i] = sortbuf[i];
-      __syncthreads();
-      thresh = res[k - 1];
-      if (!more) break;
-    }
-    if (threadIdx.x == 0) { state[0] = 0; state[1] = 0; }
+    int total = state[0];
+    if (total > cap) total = cap; /* belt & braces */
+    /* append res at [total, total+k), pad to pow2, sort, keep k */
+    int n = 1;
+    while (n < total + k) n <<= 1;
     __syncthreads();
+    for (int i = total + k + tid; i < n; i += nt)
+      sortbuf[i] = GAMMA_KEY_EMPTY;
+    for (int i = tid; i < k; i += nt) sortbuf[total + i] = res[i];
+    gamma_bitonic_sort(sortbuf, n);
+    for (int i = tid; i < k; i += nt) res[i] = sortbuf[i];
+    if (tid == 0) state[0] = 0;
+    __syncthreads();
+    thresh = res[k - 1];
   }
 };
 

@@ -180,7 +180,12 @@ hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
  * Scan (h:923-953): skip bit-63-deleted ids and bitmap-deleted docs,
  * dis = dis0 + sum_m T[m][code_m] in ascending m (plain adds — matches
  * oracle_adc_scan_list bit-for-bit). */
-template <bool IP>
+/* MW = M/4 compile-time (0 = generic runtime-M path). The templated path
+ * stages GAMMA_ADC_C codes per thread in registers with all global loads
+ * issued before any use, so one barrier interval carries
+ * blockDim*C codes' worth of HBM latency instead of blockDim's. */
+#define GAMMA_ADC_C 4
+template <bool IP, int MW>
 __global__ void __launch_bounds__(WG)
 k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
              const float *__restrict__ queries,
@@ -258,27 +263,68 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
 
     const int64_t *ids = bk.ids;
     const uint8_t *codes = (const uint8_t *)bk.data;
-    const int mwords = M >> 2;
-    for (long long j0 = 0; j0 < bk.size; j0 += blockDim.x) {
-      long long j = j0 + threadIdx.x;
-      if (j < bk.size) {
-        int64_t id = ids[j];
-        if (!((uint64_t)id >> 63) &&
-            !gamma_bitmap_test(bitmap, (uint64_t)id)) {
-          const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
-          float dis = dis0;
-          const float *tab = lut;
-          for (int mw = 0; mw < mwords; mw++) {
-            uint32_t w = cw[mw];
-            dis += tab[w & 255u];         tab += ksub;
-            dis += tab[(w >> 8) & 255u];  tab += ksub;
-            dis += tab[(w >> 16) & 255u]; tab += ksub;
-            dis += tab[w >> 24];          tab += ksub;
+    if (MW > 0) {
+      const int C = GAMMA_ADC_C;
+      for (long long j0 = 0; j0 < bk.size;
+           j0 += (long long)blockDim.x * C) {
+        long long jb = j0 + (long long)threadIdx.x * C;
+        uint32_t w[C][MW ? MW : 1];
+        int64_t idv[C];
+#pragma unroll
+        for (int c = 0; c < C; c++) {
+          long long j = jb + c;
+          if (j < bk.size) {
+            idv[c] = ids[j];
+            const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
+#pragma unroll
+            for (int mw = 0; mw < MW; mw++) w[c][mw] = cw[mw];
+          } else {
+            idv[c] = -1; /* bit 63 set -> skipped below */
           }
-          sel.push(gamma_make_key<IP>(dis, (uint32_t)id));
         }
+#pragma unroll
+        for (int c = 0; c < C; c++) {
+          int64_t id = idv[c];
+          if (!((uint64_t)id >> 63) &&
+              !gamma_bitmap_test(bitmap, (uint64_t)id)) {
+            float dis = dis0;
+            const float *tab = lut;
+#pragma unroll
+            for (int mw = 0; mw < MW; mw++) {
+              uint32_t wv = w[c][mw];
+              dis += tab[wv & 255u];         tab += ksub;
+              dis += tab[(wv >> 8) & 255u];  tab += ksub;
+              dis += tab[(wv >> 16) & 255u]; tab += ksub;
+              dis += tab[wv >> 24];          tab += ksub;
+            }
+            sel.push(gamma_make_key<IP>(dis, (uint32_t)id));
+          }
+        }
+        sel.maybe_flush(blockDim.x * C);
       }
-      sel.maybe_flush(blockDim.x);
+    } else {
+      const int mwords = M >> 2;
+      for (long long j0 = 0; j0 < bk.size; j0 += blockDim.x) {
+        long long j = j0 + threadIdx.x;
+        if (j < bk.size) {
+          int64_t id = ids[j];
+          if (!((uint64_t)id >> 63) &&
+              !gamma_bitmap_test(bitmap, (uint64_t)id)) {
+            const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
+            float dis = dis0;
+            const float *tab = lut;
+            for (int mw = 0; mw < mwords; mw++) {
+              uint32_t wv = cw[mw];
+              dis += tab[wv & 255u];         tab += ksub;
+              dis += tab[(wv >> 8) & 255u];  tab += ksub;
+              dis += tab[(wv >> 16) & 255u]; tab += ksub;
+              dis += tab[wv >> 24];          tab += ksub;
+            }
+            sel.push(gamma_make_key<IP>(dis, (uint32_t)id));
+          }
+        }
+        sel.maybe_flush(blockDim.x);
+      }
     }
     __syncthreads();  /* lut rebuilt next list: scan readers done */
   }
@@ -296,14 +342,27 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
   size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
                 (GAMMA_SORT_CAP + k2) * 8 + (2 * d + 1) * 4 + 4 * sizeof(int);
   if (smem > 160 * 1024) return hipErrorInvalidValue;
-  if (ip)
-    k_ivfpq_scan<true><<<dim3(nq), dim3(WG), smem, s>>>(
-        nq, d, M, nprobe, k2, queries, centroids, codebooks, buckets, nlist,
-        probes, bitmap, out_keys);
-  else
-    k_ivfpq_scan<false><<<dim3(nq), dim3(WG), smem, s>>>(
-        nq, d, M, nprobe, k2, queries, centroids, codebooks, buckets, nlist,
-        probes, bitmap, out_keys);
+  /* batched path needs flush margin blockDim*C inside the selector cap */
+  bool fast = (k2 + WG * GAMMA_ADC_C) <= GAMMA_SORT_CAP &&
+              (M == 16 || M == 32 || M == 64);
+  dim3 g(nq), b(WG);
+#define GAMMA_LAUNCH_SCAN(IPV, MWV)                                       \
+  k_ivfpq_scan<IPV, MWV><<<g, b, smem, s>>>(nq, d, M, nprobe, k2,         \
+                                            queries, centroids,           \
+                                            codebooks, buckets, nlist,    \
+                                            probes, bitmap, out_keys)
+  if (ip) {
+    if (!fast) GAMMA_LAUNCH_SCAN(true, 0);
+    else if (M == 16) GAMMA_LAUNCH_SCAN(true, 4);
+    else if (M == 32) GAMMA_LAUNCH_SCAN(true, 8);
+    else GAMMA_LAUNCH_SCAN(true, 16);
+  } else {
+    if (!fast) GAMMA_LAUNCH_SCAN(false, 0);
+    else if (M == 16) GAMMA_LAUNCH_SCAN(false, 4);
+    else if (M == 32) GAMMA_LAUNCH_SCAN(false, 8);
+    else GAMMA_LAUNCH_SCAN(false, 16);
+  }
+#undef GAMMA_LAUNCH_SCAN
   return hipGetLastError();
 }
 

@@ -1,0 +1,228 @@
+/*
+ * adc_bench.hip — standalone ablation bench for the IVFPQ ADC scan kernel
+ * (north-star shape: d=128, M=32, nlist=4096, N=10M, nprobe=32, nq
+ * configurable). Builds synthetic buckets directly on the GPU and times
+ * kernel variants; used to find where the cycles go (§5 common-mistake 8:
+ * ablate before optimizing). Not part of the product path.
+ *
+ * Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/adc_bench.hip
+ *        -I vearch_amd/csrc -o /tmp/adc_bench
+ */
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include <stdio.h>
+#include <stdlib.h>
+
+#include "../vearch_amd/csrc/kernels.h"
+#include "../vearch_amd/csrc/select.hpp"
+
+#define CHECK(x)                                                      \
+  do {                                                                \
+    hipError_t e = (x);                                               \
+    if (e != hipSuccess) {                                            \
+      printf("HIP error %s at line %d\n", hipGetErrorString(e),       \
+             __LINE__);                                               \
+      exit(1);                                                        \
+    }                                                                 \
+  } while (0)
+
+/* variant bits */
+#define V_NOPUSH 1   /* skip selector (keep dis live via asm) */
+#define V_NOLUT 2    /* skip per-list LUT build (use list 0's) */
+#define V_NOSCAN 4   /* skip the code scan (LUT build only) */
+
+template <int MW, int C, int VAR>
+__global__ void __launch_bounds__(256)
+k_scan_var(int nq, int d, int M, int nprobe, int k2,
+           const float *__restrict__ queries,
+           const float *__restrict__ centroids,
+           const float *__restrict__ codebooks,
+           const GammaBucketDev *__restrict__ buckets, int nlist,
+           const int64_t *__restrict__ probes,
+           uint64_t *__restrict__ out_keys) {
+  extern __shared__ char smem[];
+  const int ksub = 256;
+  const int dsub = d / M;
+  float *lut = (float *)smem;
+  uint64_t *sortbuf = (uint64_t *)(smem + ((size_t)M * ksub * 4 + 7) / 8 * 8);
+  uint64_t *res = sortbuf + GAMMA_SORT_CAP;
+  float *qs = (float *)(res + k2);
+  float *rs = qs + d;
+  int *state = (int *)(rs + d);
+
+  const int q = blockIdx.x;
+  if (q >= nq) return;
+  const float *qg = queries + (int64_t)q * d;
+  for (int i = threadIdx.x; i < d; i += blockDim.x) qs[i] = qg[i];
+
+  GammaSelector sel;
+  sel.init(sortbuf, res, state, k2);
+
+  for (int p = 0; p < nprobe; p++) {
+    int64_t ln = probes[(int64_t)q * nprobe + p];
+    if (ln < 0 || ln >= nlist) continue;
+    GammaBucketDev bk = buckets[ln];
+    if (bk.size <= 0) continue;
+    const float *cent = centroids + (size_t)ln * d;
+
+    if (!(VAR & V_NOLUT) || p == 0) {
+      for (int i = threadIdx.x; i < d; i += blockDim.x)
+        rs[i] = qs[i] - cent[i];
+      __syncthreads();
+      for (int e = threadIdx.x; e < M * ksub; e += blockDim.x) {
+        int m = e >> 8, j = e & 255;
+        const float *cw = codebooks + ((size_t)m * ksub + j) * dsub;
+        const float *rm = rs + m * dsub;
+        float acc = 0.0f;
+        for (int t = 0; t < dsub; t++) {
+          float diff = rm[t] - cw[t];
+          acc = fmaf(diff, diff, acc);
+        }
+        lut[e] = acc;
+      }
+      __syncthreads();
+    }
+    if (VAR & V_NOSCAN) continue;
+
+    const int64_t *ids = bk.ids;
+    const uint8_t *codes = (const uint8_t *)bk.data;
+    for (long long j0 = 0; j0 < bk.size; j0 += (long long)blockDim.x * C) {
+      long long jb = j0 + (long long)threadIdx.x * C;
+      uint32_t w[C][MW];
+      int64_t idv[C];
+#pragma unroll
+      for (int c = 0; c < C; c++) {
+        long long j = jb + c;
+        if (j < bk.size) {
+          idv[c] = ids[j];
+          const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
+#pragma unroll
+          for (int mw = 0; mw < MW; mw++) w[c][mw] = cw[mw];
+        } else {
+          idv[c] = -1;
+        }
+      }
+#pragma unroll
+      for (int c = 0; c < C; c++) {
+        int64_t id = idv[c];
+        if (!((uint64_t)id >> 63)) {
+          float dis = 0.0f;
+          const float *tab = lut;
+#pragma unroll
+          for (int mw = 0; mw < MW; mw++) {
+            uint32_t wv = w[c][mw];
+            dis += tab[wv & 255u];         tab += ksub;
+            dis += tab[(wv >> 8) & 255u];  tab += ksub;
+            dis += tab[(wv >> 16) & 255u]; tab += ksub;
+            dis += tab[wv >> 24];          tab += ksub;
+          }
+          if (VAR & V_NOPUSH) {
+            asm volatile("" ::"v"(dis)); /* keep the work live */
+          } else {
+            sel.push(gamma_make_key<false>(dis, (uint32_t)id));
+          }
+        }
+      }
+      if (!(VAR & V_NOPUSH)) sel.maybe_flush(blockDim.x * C);
+    }
+    __syncthreads();
+  }
+  sel.finish();
+  for (int i = threadIdx.x; i < k2; i += blockDim.x)
+    out_keys[(int64_t)q * k2 + i] = res[i];
+}
+
+__global__ void k_fill(uint8_t *codes, int64_t *ids, int64_t n, int M,
+                       uint64_t seed) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  ids[i] = i;
+  uint64_t s = seed + i * 2654435761ull;
+  for (int m = 0; m < M; m++) {
+    s = s * 6364136223846793005ull + 1442695040888963407ull;
+    codes[i * M + m] = (uint8_t)(s >> 33);
+  }
+}
+
+int main(int argc, char **argv) {
+  int nq = argc > 1 ? atoi(argv[1]) : 10000;
+  const int d = 128, M = 32, MW = 8, nlist = 4096, nprobe = 32, k2 = 200;
+  const int64_t N = 10000000;
+  const int64_t per = N / nlist;
+
+  float *queries, *centroids, *codebooks;
+  CHECK(hipMalloc(&queries, (size_t)nq * d * 4));
+  CHECK(hipMalloc(&centroids, (size_t)nlist * d * 4));
+  CHECK(hipMalloc(&codebooks, (size_t)M * 256 * (d / M) * 4));
+  /* one big slab for codes+ids, sliced into buckets */
+  uint8_t *codes;
+  int64_t *ids;
+  CHECK(hipMalloc(&codes, (size_t)N * M));
+  CHECK(hipMalloc(&ids, (size_t)N * 8));
+  k_fill<<<dim3((uint32_t)((N + 255) / 256)), dim3(256)>>>(codes, ids, N, M,
+                                                           42);
+  std::vector<GammaBucketDev> hb(nlist);
+  for (int i = 0; i < nlist; i++) {
+    hb[i].ids = ids + (size_t)i * per;
+    hb[i].data = codes + (size_t)i * per * M;
+    hb[i].size = per;
+  }
+  GammaBucketDev *buckets;
+  CHECK(hipMalloc(&buckets, nlist * sizeof(GammaBucketDev)));
+  CHECK(hipMemcpy(buckets, hb.data(), nlist * sizeof(GammaBucketDev),
+                  hipMemcpyHostToDevice));
+  int64_t *probes;
+  CHECK(hipMalloc(&probes, (size_t)nq * nprobe * 8));
+  std::vector<int64_t> hp((size_t)nq * nprobe);
+  srand(7);
+  for (size_t i = 0; i < hp.size(); i++) hp[i] = rand() % nlist;
+  CHECK(hipMemcpy(probes, hp.data(), hp.size() * 8,
+                  hipMemcpyHostToDevice));
+  uint64_t *out;
+  CHECK(hipMalloc(&out, (size_t)nq * k2 * 8));
+  CHECK(hipMemset(queries, 1, (size_t)nq * d * 4));
+  CHECK(hipMemset(centroids, 2, (size_t)nlist * d * 4));
+  CHECK(hipMemset(codebooks, 3, (size_t)M * 256 * (d / M) * 4));
+
+  size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
+                (GAMMA_SORT_CAP + k2) * 8 + (2 * d + 1) * 4 + 16;
+  double bytes_per_q = (double)nprobe * per * (M + 8);
+
+  auto run = [&](const char *name, auto kern, int reps) {
+    /* warmup */
+    kern<<<dim3(nq), dim3(256), smem>>>(nq, d, M, nprobe, k2, queries,
+                                        centroids, codebooks, buckets,
+                                        nlist, probes, out);
+    CHECK(hipGetLastError());
+    CHECK(hipDeviceSynchronize());
+    hipEvent_t a, b;
+    hipEventCreate(&a);
+    hipEventCreate(&b);
+    hipEventRecord(a);
+    for (int r = 0; r < reps; r++)
+      kern<<<dim3(nq), dim3(256), smem>>>(nq, d, M, nprobe, k2, queries,
+                                          centroids, codebooks, buckets,
+                                          nlist, probes, out);
+    hipEventRecord(b);
+    CHECK(hipEventSynchronize(b));
+    float ms;
+    hipEventElapsedTime(&ms, a, b);
+    ms /= reps;
+    printf("%-28s %8.2f ms  %8.1f GB/s  %9.0f QPS\n", name, ms,
+           nq * bytes_per_q / ms / 1e6, nq / ms * 1000.0);
+    hipEventDestroy(a);
+    hipEventDestroy(b);
+  };
+
+  printf("nq=%d N=%lld nlist=%d nprobe=%d M=%d k2=%d smem=%zu\n", nq,
+         (long long)N, nlist, nprobe, M, k2, smem);
+  run("full C=4", k_scan_var<MW, 4, 0>, 3);
+  run("full C=8", k_scan_var<MW, 8, 0>, 3);
+  run("full C=2", k_scan_var<MW, 2, 0>, 3);
+  run("nopush C=4", k_scan_var<MW, 4, V_NOPUSH>, 3);
+  run("nolut C=4", k_scan_var<MW, 4, V_NOLUT>, 3);
+  run("lutonly", k_scan_var<MW, 4, V_NOSCAN | V_NOPUSH>, 3);
+  run("nopush+nolut C=4", k_scan_var<MW, 4, V_NOPUSH | V_NOLUT>, 3);
+  run("nopush+nolut C=8", k_scan_var<MW, 8, V_NOPUSH | V_NOLUT>, 3);
+  return 0;
+}

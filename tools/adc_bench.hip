@@ -30,6 +30,8 @@
 #define V_NOPUSH 1   /* skip selector (keep dis live via asm) */
 #define V_NOLUT 2    /* skip per-list LUT build (use list 0's) */
 #define V_NOSCAN 4   /* skip the code scan (LUT build only) */
+#define V_PCT1 8     /* decomposed tables: T = A_q + B_list (ivfpq.h:254) */
+#define V_BITMAP 16  /* test the delete bitmap per code */
 
 template <int MW, int C, int VAR>
 __global__ void __launch_bounds__(256)
@@ -39,7 +41,10 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
            const float *__restrict__ codebooks,
            const GammaBucketDev *__restrict__ buckets, int nlist,
            const int64_t *__restrict__ probes,
-           uint64_t *__restrict__ out_keys) {
+           uint64_t *__restrict__ out_keys,
+           const float *__restrict__ Atab,   /* nq x M*ksub */
+           const float *__restrict__ Btab,   /* nlist x M*ksub */
+           const uint32_t *__restrict__ bitmap) {
   extern __shared__ char smem[];
   const int ksub = 256;
   const int dsub = d / M;
@@ -65,7 +70,13 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
     if (bk.size <= 0) continue;
     const float *cent = centroids + (size_t)ln * d;
 
-    if (!(VAR & V_NOLUT) || p == 0) {
+    if (VAR & V_PCT1) {
+      const float *Aq = Atab + (size_t)q * M * ksub;
+      const float *Bl = Btab + (size_t)ln * M * ksub;
+      for (int e = threadIdx.x; e < M * ksub; e += blockDim.x)
+        lut[e] = Aq[e] + Bl[e];
+      __syncthreads();
+    } else if (!(VAR & V_NOLUT) || p == 0) {
       for (int i = threadIdx.x; i < d; i += blockDim.x)
         rs[i] = qs[i] - cent[i];
       __syncthreads();
@@ -105,7 +116,9 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
 #pragma unroll
       for (int c = 0; c < C; c++) {
         int64_t id = idv[c];
-        if (!((uint64_t)id >> 63)) {
+        if (!((uint64_t)id >> 63) &&
+            (!(VAR & V_BITMAP) ||
+             !gamma_bitmap_test(bitmap, (uint64_t)id))) {
           float dis = 0.0f;
           const float *tab = lut;
 #pragma unroll
@@ -180,6 +193,14 @@ int main(int argc, char **argv) {
                   hipMemcpyHostToDevice));
   uint64_t *out;
   CHECK(hipMalloc(&out, (size_t)nq * k2 * 8));
+  float *Atab, *Btab;
+  CHECK(hipMalloc(&Atab, (size_t)nq * M * 256 * 4));
+  CHECK(hipMalloc(&Btab, (size_t)nlist * M * 256 * 4));
+  CHECK(hipMemset(Atab, 0, (size_t)nq * M * 256 * 4));
+  CHECK(hipMemset(Btab, 0, (size_t)nlist * M * 256 * 4));
+  uint32_t *bitmap;
+  CHECK(hipMalloc(&bitmap, (size_t)(N + 31) / 32 * 4));
+  CHECK(hipMemset(bitmap, 0, (size_t)(N + 31) / 32 * 4));
   CHECK(hipMemset(queries, 1, (size_t)nq * d * 4));
   CHECK(hipMemset(centroids, 2, (size_t)nlist * d * 4));
   CHECK(hipMemset(codebooks, 3, (size_t)M * 256 * (d / M) * 4));
@@ -192,7 +213,8 @@ int main(int argc, char **argv) {
     /* warmup */
     kern<<<dim3(nq), dim3(256), smem>>>(nq, d, M, nprobe, k2, queries,
                                         centroids, codebooks, buckets,
-                                        nlist, probes, out);
+                                        nlist, probes, out, Atab, Btab,
+                                        bitmap);
     CHECK(hipGetLastError());
     CHECK(hipDeviceSynchronize());
     hipEvent_t a, b;
@@ -202,7 +224,8 @@ int main(int argc, char **argv) {
     for (int r = 0; r < reps; r++)
       kern<<<dim3(nq), dim3(256), smem>>>(nq, d, M, nprobe, k2, queries,
                                           centroids, codebooks, buckets,
-                                          nlist, probes, out);
+                                          nlist, probes, out, Atab, Btab,
+                                          bitmap);
     hipEventRecord(b);
     CHECK(hipEventSynchronize(b));
     float ms;
@@ -224,5 +247,10 @@ int main(int argc, char **argv) {
   run("lutonly", k_scan_var<MW, 4, V_NOSCAN | V_NOPUSH>, 3);
   run("nopush+nolut C=4", k_scan_var<MW, 4, V_NOPUSH | V_NOLUT>, 3);
   run("nopush+nolut C=8", k_scan_var<MW, 8, V_NOPUSH | V_NOLUT>, 3);
+  run("pct1 C=4", k_scan_var<MW, 4, V_PCT1>, 3);
+  run("pct1 C=2", k_scan_var<MW, 2, V_PCT1>, 3);
+  run("pct1 C=8", k_scan_var<MW, 8, V_PCT1>, 3);
+  run("pct1+bitmap C=4", k_scan_var<MW, 4, V_PCT1 | V_BITMAP>, 3);
+  run("pct1+nopush C=4", k_scan_var<MW, 4, V_PCT1 | V_NOPUSH>, 3);
   return 0;
 }

@@ -147,7 +147,7 @@ struct GammaSelector {
   }
 };
 
-/* delete-bitmap test (1 = deleted), u32 words */
+/* delete-bitmap test (1 = deleted), u32 words; null bitmap = none */
 __device__ __forceinline__ bool gamma_bitmap_test(const uint32_t *bm,
                                                   uint64_t id) {
   return bm && ((bm[id >> 5] >> (id & 31)) & 1u);

@@ -32,6 +32,8 @@
 #define V_NOSCAN 4   /* skip the code scan (LUT build only) */
 #define V_PCT1 8     /* decomposed tables: T = A_q + B_list (ivfpq.h:254) */
 #define V_BITMAP 16  /* test the delete bitmap per code */
+#define V_PCT1V 32   /* pct1 with float4-vectorized table build */
+#define V_FASTCMP 64 /* float-key compare before the full push */
 
 template <int MW, int C, int VAR>
 __global__ void __launch_bounds__(256)
@@ -70,7 +72,18 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
     if (bk.size <= 0) continue;
     const float *cent = centroids + (size_t)ln * d;
 
-    if (VAR & V_PCT1) {
+    if (VAR & V_PCT1V) {
+      const float4 *Aq =
+          (const float4 *)(Atab + (size_t)q * M * ksub);
+      const float4 *Bl =
+          (const float4 *)(Btab + (size_t)ln * M * ksub);
+      float4 *lut4 = (float4 *)lut;
+      for (int e = threadIdx.x; e < (M * ksub) / 4; e += blockDim.x) {
+        float4 a = Aq[e], b = Bl[e];
+        lut4[e] = make_float4(a.x + b.x, a.y + b.y, a.z + b.z, a.w + b.w);
+      }
+      __syncthreads();
+    } else if (VAR & V_PCT1) {
       const float *Aq = Atab + (size_t)q * M * ksub;
       const float *Bl = Btab + (size_t)ln * M * ksub;
       for (int e = threadIdx.x; e < M * ksub; e += blockDim.x)
@@ -131,6 +144,14 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
           }
           if (VAR & V_NOPUSH) {
             asm volatile("" ::"v"(dis)); /* keep the work live */
+          } else if (VAR & V_FASTCMP) {
+            uint32_t dk = gamma_f32_key(dis);
+            if (dk < (uint32_t)(sel.thresh >> 32) ||
+                (dk == (uint32_t)(sel.thresh >> 32) &&
+                 (uint32_t)id < (uint32_t)sel.thresh)) {
+              int idx = atomicAdd(&sel.state[0], 1);
+              sel.sortbuf[idx] = ((uint64_t)dk << 32) | (uint32_t)id;
+            }
           } else {
             sel.push(gamma_make_key<false>(dis, (uint32_t)id));
           }
@@ -252,5 +273,11 @@ int main(int argc, char **argv) {
   run("pct1 C=8", k_scan_var<MW, 8, V_PCT1>, 3);
   run("pct1+bitmap C=4", k_scan_var<MW, 4, V_PCT1 | V_BITMAP>, 3);
   run("pct1+nopush C=4", k_scan_var<MW, 4, V_PCT1 | V_NOPUSH>, 3);
+  run("pct1v C=4", k_scan_var<MW, 4, V_PCT1V>, 3);
+  run("pct1v C=2", k_scan_var<MW, 2, V_PCT1V>, 3);
+  run("pct1v+nopush C=2", k_scan_var<MW, 2, V_PCT1V | V_NOPUSH>, 3);
+  run("pct1v+fast C=2", k_scan_var<MW, 2, V_PCT1V | V_FASTCMP>, 3);
+  run("pct1v+fast C=4", k_scan_var<MW, 4, V_PCT1V | V_FASTCMP>, 3);
+  run("pct1v+fast+bm C=2", k_scan_var<MW, 2, V_PCT1V | V_FASTCMP | V_BITMAP>, 3);
   return 0;
 }

@@ -65,6 +65,12 @@ class RefLib:
                 f32p]
             lib.oracle_adc_table_ip.argtypes = [
                 ctypes.c_int, ctypes.c_int, ctypes.c_int, f32p, f32p, f32p]
+            lib.oracle_ivfpq_search_pct1.argtypes = [
+                ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                ctypes.c_int, f32p, f32p, f32p, i64p, i64p, u8p,
+                ctypes.c_int, i64p, f32p, u8p, ctypes.c_int, f32p, i64p]
+            lib.oracle_l2_gemm_form.restype = ctypes.c_float
+            lib.oracle_l2_gemm_form.argtypes = [f32p, f32p, ctypes.c_int]
             lib.oracle_num_threads.restype = ctypes.c_int
             cls._lib = lib
         return cls._lib
@@ -260,6 +266,48 @@ class OracleIVFPQ:
             _fp(_c(self.centroids, np.float32)), nprobe, self.metric_ip,
             _fp(dists), _ip64(lists))
         return dists, lists
+
+    def coarse_gemm_dists(self, q, probes):
+        """GEMM-form coarse distances fmaf(-2,dot,qn+cn) — the engine's
+        probe-distance arithmetic, canonical fmaf order."""
+        lib = RefLib.lib()
+        q = _c(q, np.float32)
+        cent = _c(self.centroids, np.float32)
+        out = np.empty(probes.shape, dtype=np.float32)
+        for i in range(q.shape[0]):
+            for p in range(probes.shape[1]):
+                ln = probes[i, p]
+                if ln < 0:
+                    out[i, p] = -1.0
+                    continue
+                out[i, p] = lib.oracle_l2_gemm_form(
+                    _fp(q[i]), _fp(cent[ln]), self.d)
+        return out
+
+    def search_pct1(self, q, k, nprobe, probes=None, probe_dists=None,
+                    del_bitmap=None):
+        """L2 search with decomposed tables (use_precomputed_table=1
+        semantics) — the mode the HIP engine runs; bit-exact vs the GPU
+        given the same probes + probe_dists."""
+        assert self.metric_ip == 0
+        q = _c(q, np.float32)
+        nq = q.shape[0]
+        if probes is None:
+            _, probes = self.coarse_assign(q, nprobe)
+        probes = _c(probes, np.int64)
+        if probe_dists is None:
+            probe_dists = self.coarse_gemm_dists(q, probes)
+        probe_dists = _c(probe_dists, np.float32)
+        dists = np.empty((nq, k), dtype=np.float32)
+        ids = np.empty((nq, k), dtype=np.int64)
+        RefLib.lib().oracle_ivfpq_search_pct1(
+            nq, self.d, self.M, self.ksub, self.nlist, _fp(q),
+            _fp(_c(self.centroids, np.float32)),
+            _fp(_c(self.codebooks, np.float32)),
+            _ip64(_c(self.offsets, np.int64)), _ip64(_c(self.ids, np.int64)),
+            _up8(_c(self.codes, np.uint8)), nprobe, _ip64(probes),
+            _fp(probe_dists), _up8(del_bitmap), k, _fp(dists), _ip64(ids))
+        return dists, ids
 
     def search(self, q, k, nprobe, probes=None, del_bitmap=None):
         q = _c(q, np.float32)

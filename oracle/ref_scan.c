@@ -265,6 +265,108 @@ EXPORT void oracle_ivfpq_search(int nq, int d, int M, int ksub, int nlist,
   }
 }
 
+/* ---- decomposed ADC tables (use_precomputed_table=1 semantics,
+ * gamma_index_ivfpq.h:254-262): T[m][j] = A_q[m][j] + B_list[m][j] with
+ *   A[m][j] = fmaf(-2, q_m . cw, ||cw||^2)   (query-level)
+ *   B[m][j] = 2 * (c_list,m . cw)            (train-time table)
+ *   dis0    = fmaf(-2, q . c, ||q||^2 + ||c||^2)  (the coarse distance)
+ * All dot products / norms are sequential fmaf chains — the exact
+ * arithmetic of the HIP kernels (MFMA f32 is a k-ordered fmaf chain,
+ * cdna_hip_programming.md §3). ---- */
+
+EXPORT void oracle_pct1_a_table(int d, int M, int ksub, const float *q,
+                                const float *codebooks, float *out_a) {
+  int dsub = d / M;
+  for (int m = 0; m < M; m++) {
+    const float *qm = q + m * dsub;
+    for (int j = 0; j < ksub; j++) {
+      const float *cw = codebooks + ((size_t)m * ksub + j) * dsub;
+      float cwn = 0.0f, dot = 0.0f;
+      for (int t = 0; t < dsub; t++) cwn = fmaf(cw[t], cw[t], cwn);
+      for (int t = 0; t < dsub; t++) dot = fmaf(qm[t], cw[t], dot);
+      out_a[(size_t)m * ksub + j] = fmaf(-2.0f, dot, cwn);
+    }
+  }
+}
+
+EXPORT void oracle_pct1_b_table(int d, int M, int ksub,
+                                const float *centroid,
+                                const float *codebooks, float *out_b) {
+  int dsub = d / M;
+  for (int m = 0; m < M; m++) {
+    const float *cm = centroid + m * dsub;
+    for (int j = 0; j < ksub; j++) {
+      const float *cw = codebooks + ((size_t)m * ksub + j) * dsub;
+      float dot = 0.0f;
+      for (int t = 0; t < dsub; t++) dot = fmaf(cm[t], cw[t], dot);
+      out_b[(size_t)m * ksub + j] = 2.0f * dot;
+    }
+  }
+}
+
+/* GEMM-form coarse L2 distance (= the HIP engine's probe distance):
+ * fmaf(-2, dot, qn + cn), sequential dots/norms. */
+EXPORT float oracle_l2_gemm_form(const float *q, const float *c, int d) {
+  float qn = 0.0f, cn = 0.0f, dot = 0.0f;
+  for (int t = 0; t < d; t++) qn = fmaf(q[t], q[t], qn);
+  for (int t = 0; t < d; t++) cn = fmaf(c[t], c[t], cn);
+  for (int t = 0; t < d; t++) dot = fmaf(q[t], c[t], dot);
+  return fmaf(-2.0f, dot, qn + cn);
+}
+
+/* IVFPQ L2 search with decomposed tables; probe_dists supplies dis0
+ * (pass oracle_l2_gemm_form values, or the engine's probe distances for
+ * bit-exact pinning). */
+EXPORT void oracle_ivfpq_search_pct1(
+    int nq, int d, int M, int ksub, int nlist, const float *queries,
+    const float *centroids, const float *codebooks,
+    const int64_t *list_offsets, const int64_t *ids, const uint8_t *codes,
+    int nprobe, const int64_t *probes, const float *probe_dists,
+    const uint8_t *del_bitmap, int k, float *out_dists, int64_t *out_ids) {
+  float *btabs = (float *)malloc((size_t)nlist * M * ksub * 4);
+  unsigned char *bdone = (unsigned char *)calloc((size_t)nlist, 1);
+#ifdef _OPENMP
+#pragma omp parallel for schedule(dynamic)
+#endif
+  for (int ln = 0; ln < nlist; ln++) {
+    /* build only lists that are probed by someone: cheap enough to do all */
+    oracle_pct1_b_table(d, M, ksub, centroids + (size_t)ln * d, codebooks,
+                        btabs + (size_t)ln * M * ksub);
+    bdone[ln] = 1;
+  }
+#ifdef _OPENMP
+#pragma omp parallel for schedule(dynamic)
+#endif
+  for (int qi = 0; qi < nq; qi++) {
+    const float *q = queries + (size_t)qi * d;
+    float *atab = (float *)malloc((size_t)M * ksub * 4);
+    float *table = (float *)malloc((size_t)M * ksub * 4);
+    uint64_t *heap = (uint64_t *)malloc((size_t)k * 8);
+    int size = 0;
+    oracle_pct1_a_table(d, M, ksub, q, codebooks, atab);
+    for (int p = 0; p < nprobe; p++) {
+      int64_t ln = probes[(size_t)qi * nprobe + p];
+      if (ln < 0 || ln >= nlist) continue;
+      int64_t beg = list_offsets[ln], end = list_offsets[ln + 1];
+      if (end <= beg) continue;
+      float dis0 = probe_dists[(size_t)qi * nprobe + p];
+      const float *btab = btabs + (size_t)ln * M * ksub;
+      for (size_t e = 0; e < (size_t)M * ksub; e++)
+        table[e] = atab[e] + btab[e];
+      oracle_adc_scan_list(end - beg, M, ksub, codes + (size_t)beg * M,
+                           ids + beg, table, dis0, del_bitmap, 0, k, heap,
+                           &size);
+    }
+    emit_sorted(heap, size, k, 0, out_dists + (size_t)qi * k,
+                out_ids + (size_t)qi * k);
+    free(atab);
+    free(table);
+    free(heap);
+  }
+  free(btabs);
+  free(bdone);
+}
+
 /* ---- IVFFLAT search given assignments (gamma_index_ivfflat.h:36-91) ----
  * list vectors stored as fp32 d-dim codes. */
 EXPORT void oracle_ivfflat_search(int nq, int d, int nlist,

@@ -132,8 +132,8 @@ def test_ivfpq_adc_bitexact_vs_oracle(data, ivfpq_engine):
     eng = ivfpq_engine
     gd, gi = eng.raw_search(q, 10, nprobe=16)
     ox = _oracle_from_engine(eng, 64, 64, 16)
-    _, probes = eng.debug_coarse_assign(q, 16)
-    od, oi = ox.search(q, 10, 16, probes=probes)
+    pdists, probes = eng.debug_coarse_assign(q, 16)
+    od, oi = ox.search_pct1(q, 10, 16, probes=probes, probe_dists=pdists)
     assert np.array_equal(gi, oi)
     assert np.array_equal(gd, od)
 

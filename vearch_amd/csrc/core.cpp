@@ -193,6 +193,7 @@ int Bitmap::ensure(int64_t nbits, hipStream_t s) {
 }
 int Bitmap::set(int64_t vid, hipStream_t s) {
   if (ensure(vid + 1, s)) return -1;
+  if (!((host_[vid >> 5] >> (vid & 31)) & 1u)) set_count_++;
   host_[vid >> 5] |= 1u << (vid & 31);
   /* mirror the single word to device */
   if (hipMemcpy((uint32_t *)dev_.get() + (vid >> 5), &host_[vid >> 5], 4,
@@ -222,6 +223,8 @@ int Bitmap::load(FILE *f, hipStream_t s) {
   if (words && fread(tmp.data(), 4, words, f) != (size_t)words) return -1;
   if (ensure(words * 32, s)) return -1;
   std::copy(tmp.begin(), tmp.end(), host_.begin());
+  set_count_ = 0;
+  for (uint32_t w : host_) set_count_ += __builtin_popcount(w);
   if (words)
     hipMemcpy(dev_.get(), host_.data(), (size_t)words * 4,
               hipMemcpyHostToDevice);
@@ -455,6 +458,11 @@ int IVFIndex::train(const float *xt, int64_t n, hipStream_t s,
     if (codebooks_.reserve((size_t)M_ * ksub_ * dsub_ * 4)) return -1;
     hipMemcpy(codebooks_.get(), books.data(), (size_t)M_ * ksub_ * dsub_ * 4,
               hipMemcpyHostToDevice);
+    if (btable_.reserve((size_t)nlist_ * M_ * ksub_ * 4)) return -1;
+    if (gk::pq_tables_b(s, d_, M_, nlist_, centroids_.as<float>(),
+                        codebooks_.as<float>(),
+                        btable_.as<float>()) != hipSuccess)
+      return -1;
   }
   trained_ = true;
   return 0;
@@ -706,9 +714,19 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
     return -1;
   hipEventRecord(e1, s);
   if (params_.kind == IndexKind::IVFPQ) {
+    const float *atab = nullptr;
+    if (!metric_ip) {
+      if (scratch_atab_.reserve((size_t)nq * M_ * ksub_ * 4)) return -1;
+      GAMMA_CHECK(gk::pq_tables_a(s, nq, d_, M_, q_dev,
+                                  codebooks_.as<float>(),
+                                  scratch_atab_.as<float>()));
+      atab = scratch_atab_.as<float>();
+    }
     GAMMA_CHECK(gk::ivfpq_scan(s, nq, d_, M_, nprobe, k2, q_dev,
                                centroids_.as<float>(),
-                               codebooks_.as<float>(),
+                               codebooks_.as<float>(), atab,
+                               btable_.as<float>(),
+                               scratch_pdists_.as<float>(),
                                dev_buckets_.as<GammaBucketDev>(), nlist_,
                                scratch_probes_.as<int64_t>(), bitmap_dev,
                                metric_ip, out_keys_dev));
@@ -817,6 +835,11 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
     if (codebooks_.reserve(books.size() * 4)) return -1;
     hipMemcpy(codebooks_.get(), books.data(), books.size() * 4,
               hipMemcpyHostToDevice);
+    if (btable_.reserve((size_t)nlist_ * M_ * ksub_ * 4)) return -1;
+    if (gk::pq_tables_b(s, d_, M_, nlist_, centroids_.as<float>(),
+                        codebooks_.as<float>(),
+                        btable_.as<float>()) != hipSuccess)
+      return -1;
   }
   const size_t entry =
       params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;

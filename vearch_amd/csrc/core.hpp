@@ -105,6 +105,7 @@ class Bitmap {
  public:
   int set(int64_t vid, hipStream_t s); /* mark deleted */
   bool test(int64_t vid) const;
+  bool any() const { return set_count_ > 0; }
   int ensure(int64_t nbits, hipStream_t s);
   const uint32_t *dev() const { return dev_.as<uint32_t>(); }
   int64_t popcount() const;
@@ -115,6 +116,7 @@ class Bitmap {
   std::vector<uint32_t> host_;
   DeviceBuf dev_;
   int64_t bits_ = 0;
+  int64_t set_count_ = 0;
 };
 
 enum class IndexKind { FLAT, IVFPQ, IVFFLAT };
@@ -180,6 +182,8 @@ class IVFIndex {
   int d_ = 0, M_ = 0, ksub_ = 256, dsub_ = 0, code_size_ = 0, nlist_ = 0;
   int64_t ntotal_ = 0;
   DeviceBuf centroids_, cent_norms_, codebooks_;
+  DeviceBuf btable_;       /* pct1 B table: nlist x M x ksub f32 */
+  DeviceBuf scratch_atab_; /* pct1 A table: nq x M x ksub f32 */
   struct Bucket {
     std::unique_ptr<DeviceBuf> ids, data;
     long long size = 0, cap = 0;

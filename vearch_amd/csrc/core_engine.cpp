@@ -199,10 +199,10 @@ int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
                              hipStream_t s, uint64_t *out_keys_dev) {
   const int64_t n = raw_.size();
   if (nq < 512 || n < 200000) {
-    GAMMA_CHECK(gk::flat_stream_scan(s, nq, n, dim_, k2, q_dev,
-                                     raw_.dev_seg_table(),
-                                     RawStore::SEG_SHIFT, bitmap_.dev(), ip,
-                                     out_keys_dev));
+    GAMMA_CHECK(gk::flat_stream_scan(
+        s, nq, n, dim_, k2, q_dev, raw_.dev_seg_table(),
+        RawStore::SEG_SHIFT, bitmap_.any() ? bitmap_.dev() : nullptr, ip,
+        out_keys_dev));
     return 0;
   }
   /* chunked MFMA GEMM + seeded select (nq large): per segment-run chunks */
@@ -215,11 +215,11 @@ int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
     int64_t take = std::min(run, chunk);
     GAMMA_CHECK(gk::dots_mfma(s, q_dev, nq, seg, take, dim_,
                               flat_dots_.as<float>()));
-    GAMMA_CHECK(gk::select_from_dots(s, nq, take, v0, take,
-                                     flat_dots_.as<float>(), q_norms_dev,
-                                     raw_.dev_norms(), !ip, ip,
-                                     bitmap_.dev(), k2, out_keys_dev,
-                                     seeded));
+    GAMMA_CHECK(gk::select_from_dots(
+        s, nq, take, v0, take, flat_dots_.as<float>(), q_norms_dev,
+        raw_.dev_norms(), !ip, ip,
+        bitmap_.any() ? bitmap_.dev() : nullptr, k2, out_keys_dev,
+        seeded));
     seeded = true;
     v0 += take;
   }
@@ -293,7 +293,8 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
       return -1;
     k2 = kf;
   } else {
-    if (index_->search(q_dev_.as<float>(), nq, k2, nprobe, bitmap_.dev(),
+    if (index_->search(q_dev_.as<float>(), nq, k2, nprobe,
+                       bitmap_.any() ? bitmap_.dev() : nullptr,
                        ip, s, keys_dev_.as<uint64_t>(),
                        q_norms_dev_.as<float>(), &t_assign, &t_scan))
       return -1;

@@ -38,12 +38,27 @@ hipError_t argmin_rows(hipStream_t s, int64_t nrows, int ncols,
                        const float *dots, const float *qnorms,
                        const float *bnorms, bool l2, int32_t *out);
 
-/* IVFPQ fused search: one workgroup per query; per probed list builds the
- * residual ADC table in LDS (gamma_index_ivfpq.h:243-249) and scans the
- * list (h:923-953). out_keys: nq x k2. */
+/* Decomposed ADC tables (use_precomputed_table=1 semantics,
+ * gamma_index_ivfpq.h:254-262):
+ *   btab[ln][m][j] = 2 * (c_ln,m . cw_mj)        (train-time)
+ *   atab[q][m][j]  = fmaf(-2, q_m . cw, ||cw||^2) (per search batch)
+ * Sequential-fmaf arithmetic matching oracle_pct1_*_table. */
+hipError_t pq_tables_b(hipStream_t s, int d, int M, int nlist,
+                       const float *centroids, const float *codebooks,
+                       float *btab);
+hipError_t pq_tables_a(hipStream_t s, int nq, int d, int M,
+                       const float *queries, const float *codebooks,
+                       float *atab);
+
+/* IVFPQ fused search: one workgroup per query; per probed list stages
+ * T = A_q + B_list in LDS (L2; dis0 = the coarse probe distance) or the
+ * query-level IP table (h:164-167), then scans the list (h:923-953).
+ * out_keys: nq x k2. */
 hipError_t ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
                       int k2, const float *queries, const float *centroids,
-                      const float *codebooks, const GammaBucketDev *buckets,
+                      const float *codebooks, const float *atab,
+                      const float *btab, const float *probe_dists,
+                      const GammaBucketDev *buckets,
                       int nlist, const int64_t *probes,
                       const uint32_t *bitmap, bool ip, uint64_t *out_keys);
 

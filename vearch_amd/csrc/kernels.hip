@@ -113,7 +113,7 @@ k_select_from_dots(int nq, int64_t ncols, int64_t col_base, int64_t ld,
       int64_t id = col_base + c;
       if (gamma_bitmap_test(bitmap, (uint64_t)id)) continue;
       float dot = row[c];
-      float dist = l2 ? (qn + bnorms[id] - 2.0f * dot) : dot;
+      float dist = l2 ? fmaf(-2.0f, dot, qn + bnorms[id]) : dot;
       sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
     }
     sel.maybe_flush(2 * blockDim.x);
@@ -154,7 +154,7 @@ __global__ void k_argmin_rows(int64_t nrows, int ncols,
   float best = INFINITY;
   int bestj = 0;
   for (int j = 0; j < ncols; j++) {
-    float v = l2 ? (qn + bnorms[j] - 2.0f * row[j]) : -row[j];
+    float v = l2 ? fmaf(-2.0f, row[j], qn + bnorms[j]) : -row[j];
     if (v < best) { best = v; bestj = j; }
   }
   out[i] = bestj;
@@ -184,13 +184,16 @@ hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
  * stages GAMMA_ADC_C codes per thread in registers with all global loads
  * issued before any use, so one barrier interval carries
  * blockDim*C codes' worth of HBM latency instead of blockDim's. */
-#define GAMMA_ADC_C 4
+#define GAMMA_ADC_C 2
 template <bool IP, int MW>
 __global__ void __launch_bounds__(WG)
 k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
              const float *__restrict__ queries,
              const float *__restrict__ centroids,
              const float *__restrict__ codebooks,
+             const float *__restrict__ atab,
+             const float *__restrict__ btab,
+             const float *__restrict__ probe_dists,
              const GammaBucketDev *__restrict__ buckets, int nlist,
              const int64_t *__restrict__ probes,
              const uint32_t *__restrict__ bitmap,
@@ -234,6 +237,7 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
     if (bk.size <= 0) continue;
     const float *cent = centroids + (size_t)ln * d;
 
+    float dis0;
     if (IP) {
       if (threadIdx.x == 0) {  /* dis0 = dot(q, c), canonical order */
         float acc = 0.0f;
@@ -241,25 +245,21 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
         dis0s[0] = acc;
       }
       __syncthreads();
+      dis0 = dis0s[0];
     } else {
-      /* residual + per-list table */
-      for (int i = threadIdx.x; i < d; i += blockDim.x)
-        rs[i] = qs[i] - cent[i];
-      __syncthreads();
-      for (int e = threadIdx.x; e < M * ksub; e += blockDim.x) {
-        int m = e >> 8, j = e & 255;
-        const float *cw = codebooks + ((size_t)m * ksub + j) * dsub;
-        const float *rm = rs + m * dsub;
-        float acc = 0.0f;
-        for (int t = 0; t < dsub; t++) {
-          float diff = rm[t] - cw[t];
-          acc = fmaf(diff, diff, acc);
-        }
-        lut[e] = acc;
+      /* T = A_q + B_list, float4-vectorized (M*ksub % 4 == 0);
+       * dis0 = the coarse probe distance (ivfpq.h:255 dis0=coarse_dis) */
+      const float4 *Aq =
+          (const float4 *)(atab + (size_t)q * M * ksub);
+      const float4 *Bl = (const float4 *)(btab + (size_t)ln * M * ksub);
+      float4 *lut4 = (float4 *)lut;
+      for (int e = threadIdx.x; e < (M * ksub) >> 2; e += blockDim.x) {
+        float4 a = Aq[e], b = Bl[e];
+        lut4[e] = make_float4(a.x + b.x, a.y + b.y, a.z + b.z, a.w + b.w);
       }
       __syncthreads();
+      dis0 = probe_dists[(int64_t)q * nprobe + p];
     }
-    const float dis0 = IP ? dis0s[0] : 0.0f;
 
     const int64_t *ids = bk.ids;
     const uint8_t *codes = (const uint8_t *)bk.data;
@@ -336,6 +336,8 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
 hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
                           int k2, const float *queries,
                           const float *centroids, const float *codebooks,
+                          const float *atab, const float *btab,
+                          const float *probe_dists,
                           const GammaBucketDev *buckets, int nlist,
                           const int64_t *probes, const uint32_t *bitmap,
                           bool ip, uint64_t *out_keys) {
@@ -349,7 +351,8 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
 #define GAMMA_LAUNCH_SCAN(IPV, MWV)                                       \
   k_ivfpq_scan<IPV, MWV><<<g, b, smem, s>>>(nq, d, M, nprobe, k2,         \
                                             queries, centroids,           \
-                                            codebooks, buckets, nlist,    \
+                                            codebooks, atab, btab,        \
+                                            probe_dists, buckets, nlist,  \
                                             probes, bitmap, out_keys)
   if (ip) {
     if (!fast) GAMMA_LAUNCH_SCAN(true, 0);
@@ -649,6 +652,62 @@ hipError_t gk::unpack_keys(hipStream_t s, int64_t n, const uint64_t *keys,
   else
     k_unpack<false><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(
         n, keys, out_dists, out_ids);
+  return hipGetLastError();
+}
+
+/* ---------------------------------------------------- pct1 ADC tables */
+__global__ void k_pq_btable(int d, int M, int nlist,
+                            const float *__restrict__ centroids,
+                            const float *__restrict__ codebooks,
+                            float *__restrict__ btab) {
+  int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)nlist * M * 256;
+  if (e >= total) return;
+  int dsub = d / M;
+  int j = (int)(e & 255);
+  int m = (int)((e >> 8) % M);
+  int64_t ln = e / ((int64_t)M * 256);
+  const float *cm = centroids + ln * d + m * dsub;
+  const float *cw = codebooks + ((size_t)m * 256 + j) * dsub;
+  float dot = 0.0f;
+  for (int t = 0; t < dsub; t++) dot = fmaf(cm[t], cw[t], dot);
+  btab[e] = 2.0f * dot;
+}
+
+hipError_t gk::pq_tables_b(hipStream_t s, int d, int M, int nlist,
+                           const float *centroids, const float *codebooks,
+                           float *btab) {
+  int64_t total = (int64_t)nlist * M * 256;
+  k_pq_btable<<<dim3((uint32_t)((total + WG - 1) / WG)), dim3(WG), 0, s>>>(
+      d, M, nlist, centroids, codebooks, btab);
+  return hipGetLastError();
+}
+
+__global__ void k_pq_atable(int nq, int d, int M,
+                            const float *__restrict__ queries,
+                            const float *__restrict__ codebooks,
+                            float *__restrict__ atab) {
+  int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)nq * M * 256;
+  if (e >= total) return;
+  int dsub = d / M;
+  int j = (int)(e & 255);
+  int m = (int)((e >> 8) % M);
+  int64_t q = e / ((int64_t)M * 256);
+  const float *qm = queries + q * d + m * dsub;
+  const float *cw = codebooks + ((size_t)m * 256 + j) * dsub;
+  float cwn = 0.0f, dot = 0.0f;
+  for (int t = 0; t < dsub; t++) cwn = fmaf(cw[t], cw[t], cwn);
+  for (int t = 0; t < dsub; t++) dot = fmaf(qm[t], cw[t], dot);
+  atab[e] = fmaf(-2.0f, dot, cwn);
+}
+
+hipError_t gk::pq_tables_a(hipStream_t s, int nq, int d, int M,
+                           const float *queries, const float *codebooks,
+                           float *atab) {
+  int64_t total = (int64_t)nq * M * 256;
+  k_pq_atable<<<dim3((uint32_t)((total + WG - 1) / WG)), dim3(WG), 0, s>>>(
+      nq, d, M, queries, codebooks, atab);
   return hipGetLastError();
 }
 

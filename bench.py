@@ -159,10 +159,12 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--workload", default="ivfpq_d128_n10m_nprobe32",
                     choices=sorted(WORKLOADS))
-    ap.add_argument("--n", type=int, default=0, help="override DB size")
-    ap.add_argument("--nq", type=int, default=0)
-    ap.add_argument("--no-recall", action="store_true")
-    ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--db-size", type=int, default=0,
+                    help="override DB size (N per GPU)")
+    ap.add_argument("--queries", type=int, default=0,
+                    help="override nq per step")
+    ap.add_argument("--skip-recall", action="store_true")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--index-dir", default="",
                     help="reuse a built index: load from DIR if a dump "
                          "exists there, else build and dump to DIR")
@@ -171,11 +173,11 @@ def main():
     args = ap.parse_args()
 
     cfg = dict(WORKLOADS[args.workload])
-    if args.n:
-        cfg["n"] = args.n
+    if args.db_size:
+        cfg["n"] = args.db_size
         cfg["train_n"] = min(cfg["train_n"], cfg["n"])
-    if args.nq:
-        cfg["nq"] = args.nq
+    if args.queries:
+        cfg["nq"] = args.queries
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -242,7 +244,7 @@ def main():
     nq = eng.cache_queries(queries)
 
     recall = None
-    if rank == 0 and not args.no_recall and world == 1:
+    if rank == 0 and not args.skip_recall and world == 1:
         t0 = time.time()
         recall = compute_recall(eng, cfg, queries)
         log(f"recall@{cfg['k']} = {recall:.4f} ({time.time()-t0:.1f}s)")
@@ -315,7 +317,7 @@ def main():
                 "traffic": None,
             }
         cpu = None
-        if not args.no_cpu_baseline and world == 1:
+        if not args.skip_cpu_baseline and world == 1:
             try:
                 cpu = cpu_baseline(eng, cfg, queries)
             except Exception as ex:  # noqa: BLE001

@@ -564,12 +564,24 @@ __global__ void k_rerank(int nq, int ncand, int d,
   const int64_t seg_mask = ((int64_t)1 << seg_shift) - 1;
   const float *v = segs[id >> seg_shift] + (size_t)(id & seg_mask) * d;
   const float *qv = queries + (int64_t)q * d;
+  /* float4 loads (d % 4 == 0), scalar fmaf chain in canonical order */
+  const float4 *v4 = (const float4 *)v;
+  const float4 *q4 = (const float4 *)qv;
   float acc = 0.0f;
-  for (int t = 0; t < d; t++) {
-    if (IP) acc = fmaf(qv[t], v[t], acc);
-    else {
-      float diff = qv[t] - v[t];
-      acc = fmaf(diff, diff, acc);
+  for (int t = 0; t < (d >> 2); t++) {
+    float4 a = q4[t], b = v4[t];
+    if (IP) {
+      acc = fmaf(a.x, b.x, acc);
+      acc = fmaf(a.y, b.y, acc);
+      acc = fmaf(a.z, b.z, acc);
+      acc = fmaf(a.w, b.w, acc);
+    } else {
+      float dx = a.x - b.x, dy = a.y - b.y, dz = a.z - b.z,
+            dw = a.w - b.w;
+      acc = fmaf(dx, dx, acc);
+      acc = fmaf(dy, dy, acc);
+      acc = fmaf(dz, dz, acc);
+      acc = fmaf(dw, dw, acc);
     }
   }
   keys_out[idx] = gamma_make_key<IP>(acc, id);

@@ -181,12 +181,14 @@ def main():
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
+    backend = os.environ.get("GAMMA_BENCH_BACKEND", "nccl")
     dist = None
     if world > 1:
         import torch
         import torch.distributed as tdist
-        tdist.init_process_group("nccl")
-        torch.cuda.set_device(0)  # HIP_VISIBLE_DEVICES pins the GPU
+        tdist.init_process_group(backend)
+        if backend == "nccl":
+            torch.cuda.set_device(0)  # HIP_VISIBLE_DEVICES pins the GPU
         dist = tdist
 
     import torch
@@ -243,6 +245,22 @@ def main():
                 return
     nq = eng.cache_queries(queries)
 
+    # actual scanned codes/query (clustered data probes the dense lists,
+    # so this exceeds nprobe*N/nlist; the roofline stays on algorithmic
+    # bytes per BASELINE.md)
+    eff_codes = None
+    if rank == 0 and cfg["kind"] in ("IVFPQ", "IVFFLAT"):
+        try:
+            from vearch_amd.engine import lib as _lib
+            _, pl = eng.debug_coarse_assign(queries[:256], cfg["nprobe"])
+            sizes = {int(ln): _lib().GammaDebugGetList(
+                eng.h, int(ln), None, None)
+                for ln in np.unique(pl) if ln >= 0}
+            eff_codes = float(np.mean(
+                [sum(sizes.get(int(x), 0) for x in row) for row in pl]))
+        except Exception:
+            pass
+
     recall = None
     if rank == 0 and not args.skip_recall and world == 1:
         t0 = time.time()
@@ -260,7 +278,9 @@ def main():
         # xGMI of the packed (dist-key, id) u64 blocks (~nq*k*8 B/rank)
         gids = np.where(ids >= 0, ids + rank * cfg["n"], -1)
         payload = torch.from_numpy(np.stack(
-            [dists.view(np.int32).astype(np.int64), gids], axis=2)).cuda()
+            [dists.view(np.int32).astype(np.int64), gids], axis=2))
+        if backend == "nccl":
+            payload = payload.cuda()
         out = [torch.empty_like(payload) for _ in range(world)]
         dist.all_gather(out, payload)
         if rank == 0:
@@ -288,7 +308,9 @@ def main():
         dist.barrier()
     elapsed = time.time() - t_start
     if world > 1:
-        el = torch.tensor([elapsed]).cuda()
+        el = torch.tensor([elapsed])
+        if backend == "nccl":
+            el = el.cuda()
         dist.all_reduce(el, op=dist.ReduceOp.MAX)
         elapsed = float(el.item())
 
@@ -347,6 +369,10 @@ def main():
                 "topk": cfg["k"],
                 "recall_num": cfg["rerank"],
                 "recall_at_10": recall,
+                "scanned_codes_per_query": eff_codes,
+                "algorithmic_codes_per_query":
+                    cfg["nprobe"] * cfg["n"] / cfg["nlist"]
+                    if cfg["nlist"] else None,
                 "parallelism": f"dp{world} (1 partition/GPU, RCCL "
                                "all-gather top-k merge)",
             },

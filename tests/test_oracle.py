@@ -200,3 +200,58 @@ def test_empty_and_ragged_inputs():
     probes = np.full((4, 3), -1, dtype=np.int64)  # key<0 skipped (cc:639)
     sd, si = ix.search(q, 5, nprobe=3, probes=probes)
     assert (si == -1).all()
+
+
+def test_pct1_tables_match_numpy():
+    """Decomposed tables (use_precomputed_table=1, ivfpq.h:254-262):
+    A + B must equal the direct residual table to fp32 rounding, and the
+    pct1 search must agree with the direct-mode search on recall."""
+    import ctypes
+    from oracle.gamma_oracle import _fp, _c
+    d, M, ksub, dsub = 32, 8, 256, 4
+    rng = np.random.default_rng(5)
+    q = rng.random(d, dtype=np.float32)
+    cent = rng.random(d, dtype=np.float32)
+    books = rng.standard_normal((M, ksub, dsub)).astype(np.float32) * 0.2
+    lib = RefLib.lib()
+    A = np.empty((M, ksub), dtype=np.float32)
+    B = np.empty((M, ksub), dtype=np.float32)
+    lib.oracle_pct1_a_table.argtypes = lib.oracle_adc_table_ip.argtypes
+    lib.oracle_pct1_b_table.argtypes = lib.oracle_adc_table_ip.argtypes
+    lib.oracle_pct1_a_table(d, M, ksub, _fp(_c(q, np.float32)),
+                            _fp(_c(books, np.float32)), _fp(A))
+    lib.oracle_pct1_b_table(d, M, ksub, _fp(_c(cent, np.float32)),
+                            _fp(_c(books, np.float32)), _fp(B))
+    # numpy f64 check: dis0 + A + B ~ ||(q-c)_m - cw||^2
+    qf, cf, bf = (x.astype(np.float64) for x in (q, cent, books))
+    lib.oracle_l2_gemm_form.restype = ctypes.c_float
+    dis0 = lib.oracle_l2_gemm_form(_fp(_c(q, np.float32)),
+                                   _fp(_c(cent, np.float32)), d)
+    for m in (0, 3, 7):
+        for j in (0, 100, 255):
+            want = (((qf - cf)[m * dsub:(m + 1) * dsub] - bf[m, j]) ** 2
+                    ).sum() + ((qf - cf) ** 2).sum() \
+                - np.dot(qf - cf, qf - cf)
+            got = dis0 + A[m, j] + B[m, j] - ((qf - cf) ** 2).sum()
+            # compare the per-subquantizer term alone
+            term = dis0 + A[m, j] + B[m, j]
+            direct = (((qf - cf)[m * dsub:(m + 1) * dsub] - bf[m, j]) ** 2
+                      ).sum() + ((qf - cf) ** 2).sum() \
+                - (((qf - cf)[m * dsub:(m + 1) * dsub]) ** 2).sum()
+            assert abs(term - direct) < 1e-4, (m, j, term, direct)
+
+
+def test_pct1_search_recall_equivalent():
+    base = gen_clustered(8000, 32, seed=11, ncl=60)
+    q = gen_queries(base, 32, seed=12)
+    ix = OracleIVFPQ(32, 32, 8)
+    ix.train(base[:4000])
+    ix.add(base)
+    d0, i0 = ix.search(q, 10, nprobe=8)
+    d1, i1 = ix.search_pct1(q, 10, 8)
+    # same probes; fp32 rounding differs -> allow small rank churn
+    overlap = np.mean([len(set(a) & set(b)) / 10
+                       for a, b in zip(i0.tolist(), i1.tolist())])
+    assert overlap >= 0.9, overlap
+    _, gti = flat_topk_f64(base, q, 10)
+    assert abs(recall_at(gti, i0, 10) - recall_at(gti, i1, 10)) < 0.05

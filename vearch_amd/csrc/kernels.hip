@@ -399,7 +399,6 @@ k_ivfflat_scan(int nq, int d, int nprobe, int k2,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nwave = blockDim.x >> 6;
-  const int epl = (d + 63) >> 6; /* elements per lane */
 
   for (int p = 0; p < nprobe; p++) {
     int64_t ln = probes[(int64_t)q * nprobe + p];
@@ -418,16 +417,20 @@ k_ivfflat_scan(int nq, int d, int nprobe, int k2,
         id = (uint64_t)ids[j];
         live = !(id >> 63) && !gamma_bitmap_test(bitmap, id);
         if (live) {
-          const float *v = vecs + (size_t)j * d;
+          /* float2 per lane: a wave reads 512 B coalesced per pass at
+           * d=128 (selection only; the canonical re-rank follows) */
+          const float2 *v2 = (const float2 *)(vecs + (size_t)j * d);
+          const float2 *q2 = (const float2 *)qs;
           float acc = 0.0f;
-          for (int e = 0; e < epl; e++) {
-            int t = lane + (e << 6);
-            if (t < d) {
-              if (IP) acc = fmaf(qs[t], v[t], acc);
-              else {
-                float diff = qs[t] - v[t];
-                acc = fmaf(diff, diff, acc);
-              }
+          for (int e = lane; e < (d >> 1); e += 64) {
+            float2 a = q2[e], b = v2[e];
+            if (IP) {
+              acc = fmaf(a.x, b.x, acc);
+              acc = fmaf(a.y, b.y, acc);
+            } else {
+              float dx = a.x - b.x, dy = a.y - b.y;
+              acc = fmaf(dx, dx, acc);
+              acc = fmaf(dy, dy, acc);
             }
           }
           for (int off = 32; off > 0; off >>= 1)
@@ -492,7 +495,6 @@ k_flat_stream(int nq, int64_t n, int d, int k2,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nwave = blockDim.x >> 6;
-  const int epl = (d + 63) >> 6;
   const int64_t seg_mask = ((int64_t)1 << seg_shift) - 1;
 
   long long iters = (n + nwave - 1) / nwave;
@@ -503,16 +505,19 @@ k_flat_stream(int nq, int64_t n, int d, int k2,
     if (j < n) {
       live = !gamma_bitmap_test(bitmap, (uint64_t)j);
       if (live) {
-        const float *v = segs[j >> seg_shift] + (size_t)(j & seg_mask) * d;
+        const float2 *v2 = (const float2 *)(
+            segs[j >> seg_shift] + (size_t)(j & seg_mask) * d);
+        const float2 *q2 = (const float2 *)qs;
         float acc = 0.0f;
-        for (int e = 0; e < epl; e++) {
-          int t = lane + (e << 6);
-          if (t < d) {
-            if (IP) acc = fmaf(qs[t], v[t], acc);
-            else {
-              float diff = qs[t] - v[t];
-              acc = fmaf(diff, diff, acc);
-            }
+        for (int e = lane; e < (d >> 1); e += 64) {
+          float2 a = q2[e], b = v2[e];
+          if (IP) {
+            acc = fmaf(a.x, b.x, acc);
+            acc = fmaf(a.y, b.y, acc);
+          } else {
+            float dx = a.x - b.x, dy = a.y - b.y;
+            acc = fmaf(dx, dx, acc);
+            acc = fmaf(dy, dy, acc);
           }
         }
         for (int off = 32; off > 0; off >>= 1)

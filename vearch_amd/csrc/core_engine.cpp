@@ -231,7 +231,11 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
                    const std::string &request_id, int partition_id,
                    float *out_dists, int64_t *out_ids, bool l2_sqrt) {
   if (!table_created_ || nq <= 0 || k <= 0) return -1;
-  std::shared_lock<std::shared_mutex> g(rw_);
+  /* Exclusive: concurrent cgo Search calls (engine.cc allows them) are
+   * serialized here because they share the engine's scratch device
+   * buffers and stream — GPU throughput comes from batching inside one
+   * call, not from concurrent kernels. Adds still exclude searches. */
+  std::unique_lock<std::shared_mutex> g(rw_);
   const int pid = partition_id;
   bool ip = metric == 0 ? params_.metric_ip : (metric == 2);
   hipStream_t s = stream_;

@@ -75,11 +75,101 @@ k_dots_mfma(const float *__restrict__ Q, int nq,
   }
 }
 
+/* LDS-tiled variant: 128x128 output tile per 4-wave workgroup, BK=32,
+ * +1-float row pad for conflict-free column-slice ds_reads (guide §2
+ * standard fix). Each wave owns a 64x64 sub-tile = 4x4 fragments of
+ * 16x16. The MFMA accumulation stays a k-ordered f32 fmaf chain (k tiles
+ * processed in order; zero-padded tails add exact 0s), so the dot values
+ * are bit-identical to the simple kernel and the oracle's sequential
+ * chain. */
+#define GT_BM 128
+#define GT_BK 32
+__global__ void __launch_bounds__(WG)
+k_dots_mfma_tiled(const float *__restrict__ Q, int nq,
+                  const float *__restrict__ B, int64_t n, int d,
+                  float *__restrict__ out) {
+  __shared__ float As[GT_BM][GT_BK + 1];
+  __shared__ float Bs[GT_BM][GT_BK + 1];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64; /* wave row offset in tile: 0/64 */
+  const int wc = (wave & 1) * 64;  /* wave col offset in tile: 0/64 */
+  const int row0 = blockIdx.y * GT_BM;
+  const int64_t col0 = (int64_t)blockIdx.x * GT_BM;
+
+  f32x4 acc[4][4] = {};
+  /* each thread stages 16 floats of A and B per K tile: 4 rows x float4 */
+  const int ld_row = threadIdx.x >> 3;        /* 0..31 */
+  const int ld_col = (threadIdx.x & 7) * 4;   /* 0,4,..,28 */
+
+  for (int k0 = 0; k0 < d; k0 += GT_BK) {
+#pragma unroll
+    for (int rr = 0; rr < 4; rr++) {
+      int r = ld_row + rr * 32;
+      int a_row = row0 + r;
+      int64_t b_row = col0 + r;
+      int kk = k0 + ld_col;
+      /* d % 4 == 0, so a 4-group never straddles the d boundary */
+      float4 av = make_float4(0.f, 0.f, 0.f, 0.f), bv = av;
+      if (kk < d) {
+        if (a_row < nq)
+          av = *(const float4 *)(Q + (int64_t)a_row * d + kk);
+        if (b_row < n) bv = *(const float4 *)(B + b_row * d + kk);
+      }
+      As[r][ld_col + 0] = av.x;
+      As[r][ld_col + 1] = av.y;
+      As[r][ld_col + 2] = av.z;
+      As[r][ld_col + 3] = av.w;
+      Bs[r][ld_col + 0] = bv.x;
+      Bs[r][ld_col + 1] = bv.y;
+      Bs[r][ld_col + 2] = bv.z;
+      Bs[r][ld_col + 3] = bv.w;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < GT_BK; kk += 4) {
+      float a[4], b[4];
+#pragma unroll
+      for (int f = 0; f < 4; f++) {
+        a[f] = As[wr + f * 16 + (lane & 15)][kk + (lane >> 4)];
+        b[f] = Bs[wc + f * 16 + (lane & 15)][kk + (lane >> 4)];
+      }
+#pragma unroll
+      for (int fi = 0; fi < 4; fi++)
+#pragma unroll
+        for (int fj = 0; fj < 4; fj++)
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a[fi], b[fj], acc[fi][fj], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  /* D mapping: col=lane&15, row=(lane>>4)*4+r (guide §3) */
+#pragma unroll
+  for (int fi = 0; fi < 4; fi++) {
+#pragma unroll
+    for (int fj = 0; fj < 4; fj++) {
+      int64_t col = col0 + wc + fj * 16 + (lane & 15);
+      if (col >= n) continue;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int qi = row0 + wr + fi * 16 + (lane >> 4) * 4 + r;
+        if (qi < nq) out[(int64_t)qi * n + col] = acc[fi][fj][r];
+      }
+    }
+  }
+}
+
 hipError_t gk::dots_mfma(hipStream_t s, const float *Q, int nq,
                          const float *B, int64_t n, int d, float *out) {
   if (nq == 0 || n == 0) return hipSuccess;
-  dim3 grid((uint32_t)((n + 15) / 16), (uint32_t)((nq + 63) / 64));
-  k_dots_mfma<<<grid, dim3(WG), 0, s>>>(Q, nq, B, n, d, out);
+  if (nq >= 64 && n >= 64) {
+    dim3 grid((uint32_t)((n + GT_BM - 1) / GT_BM),
+              (uint32_t)((nq + GT_BM - 1) / GT_BM));
+    k_dots_mfma_tiled<<<grid, dim3(WG), 0, s>>>(Q, nq, B, n, d, out);
+  } else {
+    dim3 grid((uint32_t)((n + 15) / 16), (uint32_t)((nq + 63) / 64));
+    k_dots_mfma<<<grid, dim3(WG), 0, s>>>(Q, nq, B, n, d, out);
+  }
   return hipGetLastError();
 }
 

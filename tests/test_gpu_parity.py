@@ -360,3 +360,66 @@ def test_ivfpq_north_star_shape_recall():
     r10 = orc.recall_at(gti, gi, 10)
     assert r10 >= 0.9, f"recall@10={r10}"
     eng.close()
+
+
+def test_ivfpq_ip_bitexact_vs_oracle(data):
+    """InnerProduct IVFPQ: spherical k-means + query-level IP table +
+    dis0 = q.c (gamma_index_ivfpq.h:164-167, 223-236) — bit-exact vs the
+    oracle on the same model/probes."""
+    base, q = data
+    eng = make_engine("/tmp/gamma_ivfpq_ip")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 64, "nsubvector": 16, '
+        '"metric_type": "InnerProduct", "training_threshold": 8000}')
+    eng.add(base)
+    eng.build_index()
+    gd, gi = eng.raw_search(q, 10, nprobe=16)
+    ox = _oracle_from_engine(eng, 64, 64, 16, metric="InnerProduct")
+    _, probes = eng.debug_coarse_assign(q, 16)
+    od, oi = ox.search(q, 10, 16, probes=probes)
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    # IP similarity is descending
+    for t in range(q.shape[0]):
+        row = [gd[t, j] for j in range(10) if gi[t, j] >= 0]
+        assert row == sorted(row, reverse=True)
+    eng.close()
+
+
+def test_edge_cases(data):
+    base, q = data
+    eng = make_engine("/tmp/gamma_edge")
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}')
+    # search an empty engine: all -1
+    gd, gi = eng.raw_search(q[:4], 5)
+    assert (gi == -1).all()
+    # k > n
+    eng.add(base[:3])
+    gd, gi = eng.raw_search(q[:4], 8)
+    assert (gi[:, 3:] == -1).all()
+    assert set(gi[0, :3].tolist()) == {0, 1, 2}
+    eng.close()
+    # nprobe > nlist clamps (ivfpq.cc:572-580 semantics)
+    eng2 = make_engine("/tmp/gamma_edge2")
+    eng2.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 16, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 2000}')
+    eng2.add(base[:4000])
+    eng2.build_index()
+    gd, gi = eng2.raw_search(q[:4], 5, nprobe=999)
+    assert (gi[:, :5] >= 0).all()
+    eng2.close()
+
+
+def test_delete_all_then_search(data):
+    base, q = data
+    eng = make_engine("/tmp/gamma_delall")
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}')
+    eng.add(base[:100])
+    for vid in range(100):
+        eng.delete_doc(str(vid))
+    gd, gi = eng.raw_search(q[:4], 5)
+    assert (gi == -1).all()
+    eng.close()

@@ -699,6 +699,7 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                      double *t_scan_ms) {
   if (!trained_) return -1;
   nprobe = std::min(nprobe, nlist_);
+  if (nprobe > 1024) nprobe = 1024; /* selector cap (select.hpp) */
   if (scratch_probes_.reserve((size_t)nq * nprobe * 8)) return -1;
   if (scratch_pdists_.reserve((size_t)nq * nprobe * 4)) return -1;
   if (update_dev_buckets(s)) return -1;

@@ -489,6 +489,8 @@ k_ivfflat_scan(int nq, int d, int nprobe, int k2,
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nwave = blockDim.x >> 6;
+  const int half = lane >> 5;   /* two vectors per wave: one per half */
+  const int hlane = lane & 31;
 
   for (int p = 0; p < nprobe; p++) {
     int64_t ln = probes[(int64_t)q * nprobe + p];
@@ -497,9 +499,10 @@ k_ivfflat_scan(int nq, int d, int nprobe, int k2,
     const int64_t *ids = bk.ids;
     const float *vecs = (const float *)bk.data;
 
-    long long iters = (bk.size + nwave - 1) / nwave;
+    const int vpp = nwave * 2; /* vectors per WG pass */
+    long long iters = (bk.size + vpp - 1) / vpp;
     for (long long it = 0; it < iters; it++) {
-      long long j = it * nwave + wave;
+      long long j = it * vpp + wave * 2 + half;
       bool live = false;
       uint64_t id = 0;
       float dist = 0.0f;
@@ -507,29 +510,35 @@ k_ivfflat_scan(int nq, int d, int nprobe, int k2,
         id = (uint64_t)ids[j];
         live = !(id >> 63) && !gamma_bitmap_test(bitmap, id);
         if (live) {
-          /* float2 per lane: a wave reads 512 B coalesced per pass at
-           * d=128 (selection only; the canonical re-rank follows) */
-          const float2 *v2 = (const float2 *)(vecs + (size_t)j * d);
-          const float2 *q2 = (const float2 *)qs;
+          /* float4 per half-wave lane: 32 x 16 B = one 512-B coalesced
+           * read per vector at d=128 (selection only; the canonical
+           * re-rank recomputes final distances) */
+          const float4 *v4 = (const float4 *)(vecs + (size_t)j * d);
+          const float4 *q4 = (const float4 *)qs;
           float acc = 0.0f;
-          for (int e = lane; e < (d >> 1); e += 64) {
-            float2 a = q2[e], b = v2[e];
+          for (int e = hlane; e < (d >> 2); e += 32) {
+            float4 a = q4[e], b = v4[e];
             if (IP) {
               acc = fmaf(a.x, b.x, acc);
               acc = fmaf(a.y, b.y, acc);
+              acc = fmaf(a.z, b.z, acc);
+              acc = fmaf(a.w, b.w, acc);
             } else {
-              float dx = a.x - b.x, dy = a.y - b.y;
+              float dx = a.x - b.x, dy = a.y - b.y, dz = a.z - b.z,
+                    dw = a.w - b.w;
               acc = fmaf(dx, dx, acc);
               acc = fmaf(dy, dy, acc);
+              acc = fmaf(dz, dz, acc);
+              acc = fmaf(dw, dw, acc);
             }
           }
-          for (int off = 32; off > 0; off >>= 1)
-            acc += __shfl_xor(acc, off, 64);
+          for (int off = 16; off > 0; off >>= 1)
+            acc += __shfl_xor(acc, off, 64); /* within each 32-half */
           dist = acc;
         }
       }
-      /* rotate the pushing lane so buffers fill evenly */
-      if (live && lane == (int)(j & 63))
+      /* rotate the pushing hlane so buffers fill evenly */
+      if (live && hlane == (int)(it & 31))
         sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
       if ((it & 15) == 15) sel.maybe_flush(blockDim.x);
     }

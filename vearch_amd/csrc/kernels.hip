@@ -718,7 +718,8 @@ __global__ void k_sort_rows(int nq, int ncand, int k,
   const int q = blockIdx.x;
   if (q >= nq) return;
   int n = 1;
-  while (n < ncand) n <<= 1;
+  while (n < ncand || n < k) n <<= 1; /* k <= n so the output reads are
+                                       * always inside the padded region */
   for (int i = threadIdx.x; i < n; i += blockDim.x)
     buf[i] = (i < ncand) ? keys[(int64_t)q * ncand + i] : GAMMA_KEY_EMPTY;
   gamma_bitonic_sort(buf, n);
@@ -734,7 +735,7 @@ hipError_t gk::sort_rows(hipStream_t s, int nq, int ncand, int k,
                          const uint64_t *keys, bool ip, float *out_dists,
                          int64_t *out_ids) {
   int n = 1;
-  while (n < ncand) n <<= 1;
+  while (n < ncand || n < k) n <<= 1;
   size_t smem = (size_t)n * 8;
   if (smem > 160 * 1024) return hipErrorInvalidValue;
   if (ip)

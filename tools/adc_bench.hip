@@ -34,6 +34,7 @@
 #define V_BITMAP 16  /* test the delete bitmap per code */
 #define V_PCT1V 32   /* pct1 with float4-vectorized table build */
 #define V_FASTCMP 64 /* float-key compare before the full push */
+#define V_U32ID 128  /* 4-byte ids (bit 31 delete) instead of int64 */
 
 template <int MW, int C, int VAR>
 __global__ void __launch_bounds__(256)
@@ -109,6 +110,7 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
     if (VAR & V_NOSCAN) continue;
 
     const int64_t *ids = bk.ids;
+    const uint32_t *ids32 = (const uint32_t *)bk.ids;
     const uint8_t *codes = (const uint8_t *)bk.data;
     for (long long j0 = 0; j0 < bk.size; j0 += (long long)blockDim.x * C) {
       long long jb = j0 + (long long)threadIdx.x * C;
@@ -118,7 +120,9 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
       for (int c = 0; c < C; c++) {
         long long j = jb + c;
         if (j < bk.size) {
-          idv[c] = ids[j];
+          idv[c] = (VAR & V_U32ID)
+                       ? (int64_t)(int32_t)ids32[j]
+                       : ids[j];
           const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
 #pragma unroll
           for (int mw = 0; mw < MW; mw++) w[c][mw] = cw[mw];
@@ -279,5 +283,9 @@ int main(int argc, char **argv) {
   run("pct1v+fast C=2", k_scan_var<MW, 2, V_PCT1V | V_FASTCMP>, 3);
   run("pct1v+fast C=4", k_scan_var<MW, 4, V_PCT1V | V_FASTCMP>, 3);
   run("pct1v+fast+bm C=2", k_scan_var<MW, 2, V_PCT1V | V_FASTCMP | V_BITMAP>, 3);
+  run("pct1v+u32 C=2", k_scan_var<MW, 2, V_PCT1V | V_U32ID>, 3);
+  run("pct1v+u32 C=4", k_scan_var<MW, 4, V_PCT1V | V_U32ID>, 3);
+  run("pct1v+u32+nopush C=2",
+      k_scan_var<MW, 2, V_PCT1V | V_U32ID | V_NOPUSH>, 3);
   return 0;
 }

@@ -537,7 +537,7 @@ int IVFIndex::update_dev_buckets(hipStream_t s) {
   if (!dev_buckets_dirty_) return 0;
   std::vector<GammaBucketDev> h(nlist_);
   for (int i = 0; i < nlist_; i++) {
-    h[i].ids = buckets_[i].ids ? buckets_[i].ids->as<int64_t>() : nullptr;
+    h[i].ids = buckets_[i].ids ? buckets_[i].ids->as<uint32_t>() : nullptr;
     h[i].data = buckets_[i].data ? buckets_[i].data->get() : nullptr;
     h[i].size = buckets_[i].size;
   }
@@ -606,13 +606,13 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
     }
 
     /* group by bucket (AddKeys analog, realtime_mem_data.cc) */
-    std::map<int32_t, std::pair<std::vector<int64_t>, std::vector<uint8_t>>>
+    std::map<int32_t, std::pair<std::vector<uint32_t>, std::vector<uint8_t>>>
         groups;
     for (int64_t i = 0; i < cn; i++) {
       int32_t b = asg_h[i];
       if (b < 0 || b >= nlist_) b = (int32_t)(vids[c0 + i] % nlist_);
       auto &g = groups[b];
-      g.first.push_back(vids[c0 + i]);
+      g.first.push_back((uint32_t)vids[c0 + i]); /* vid < 2^31 enforced */
       size_t off = g.second.size();
       g.second.resize(off + entry);
       memcpy(g.second.data() + off, payload_h + (size_t)i * entry, entry);
@@ -630,10 +630,10 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
                                 bk.size + add_n));
         auto nids = std::make_unique<DeviceBuf>();
         auto ndata = std::make_unique<DeviceBuf>();
-        if (nids->reserve((size_t)ncap * 8)) return -1;
+        if (nids->reserve((size_t)ncap * 4)) return -1;
         if (ndata->reserve((size_t)ncap * entry)) return -1;
         if (bk.size > 0) {
-          hipMemcpy(nids->get(), bk.ids->get(), (size_t)bk.size * 8,
+          hipMemcpy(nids->get(), bk.ids->get(), (size_t)bk.size * 4,
                     hipMemcpyDeviceToDevice);
           hipMemcpy(ndata->get(), bk.data->get(), (size_t)bk.size * entry,
                     hipMemcpyDeviceToDevice);
@@ -643,8 +643,8 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
         bk.cap = ncap;
         dev_buckets_dirty_ = true;
       }
-      hipMemcpy(bk.ids->as<int64_t>() + bk.size, kv.second.first.data(),
-                (size_t)add_n * 8, hipMemcpyHostToDevice);
+      hipMemcpy(bk.ids->as<uint32_t>() + bk.size, kv.second.first.data(),
+                (size_t)add_n * 4, hipMemcpyHostToDevice);
       hipMemcpy((uint8_t *)bk.data->get() + (size_t)bk.size * entry,
                 kv.second.second.data(), (size_t)add_n * entry,
                 hipMemcpyHostToDevice);
@@ -662,8 +662,9 @@ int IVFIndex::del(int64_t vid, hipStream_t s) {
   auto it = vid_loc_.find(vid);
   if (it == vid_loc_.end()) return 0;
   Bucket &bk = buckets_[it->second.first];
-  int64_t marked = vid | (int64_t)((uint64_t)1 << 63); /* kDelIdxMask */
-  if (hipMemcpy(bk.ids->as<int64_t>() + it->second.second, &marked, 8,
+  /* device form: bit 31 = the kDelIdxMask bit-63 mark */
+  uint32_t marked = (uint32_t)vid | 0x80000000u;
+  if (hipMemcpy(bk.ids->as<uint32_t>() + it->second.second, &marked, 4,
                 hipMemcpyHostToDevice) != hipSuccess)
     return -1;
   return 0;
@@ -762,9 +763,16 @@ int IVFIndex::copy_list_to_host(int64_t ln, int64_t *ids, uint8_t *codes,
   const size_t entry =
       params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;
   if (bk.size == 0) return 0;
-  if (ids)
-    GAMMA_CHECK(hipMemcpy(ids, bk.ids->get(), (size_t)bk.size * 8,
+  if (ids) {
+    std::vector<uint32_t> tmp(bk.size);
+    GAMMA_CHECK(hipMemcpy(tmp.data(), bk.ids->get(), (size_t)bk.size * 4,
                           hipMemcpyDeviceToHost));
+    for (long long j = 0; j < bk.size; j++)
+      ids[j] = (tmp[j] & 0x80000000u)
+                   ? ((int64_t)(tmp[j] & 0x7fffffffu) |
+                      (int64_t)((uint64_t)1 << 63))
+                   : (int64_t)tmp[j];
+  }
   if (codes)
     GAMMA_CHECK(hipMemcpy(codes, bk.data->get(), (size_t)bk.size * entry,
                           hipMemcpyDeviceToHost));
@@ -807,8 +815,7 @@ int IVFIndex::dump(FILE *f, hipStream_t s) const {
     if (sz > 0) {
       std::vector<int64_t> ids(sz);
       std::vector<uint8_t> data((size_t)sz * entry);
-      hipMemcpy(ids.data(), buckets_[i].ids->get(), (size_t)sz * 8,
-                hipMemcpyDeviceToHost);
+      copy_list_to_host(i, ids.data(), nullptr, s);
       hipMemcpy(data.data(), buckets_[i].data->get(), (size_t)sz * entry,
                 hipMemcpyDeviceToHost);
       fwrite(ids.data(), 8, sz, f);
@@ -858,9 +865,15 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
       if (fread(data.data(), 1, data.size(), f) != data.size()) return -1;
       bk.ids = std::make_unique<DeviceBuf>();
       bk.data = std::make_unique<DeviceBuf>();
-      if (bk.ids->reserve((size_t)sz * 8)) return -1;
+      if (bk.ids->reserve((size_t)sz * 4)) return -1;
       if (bk.data->reserve(data.size())) return -1;
-      hipMemcpy(bk.ids->get(), ids.data(), (size_t)sz * 8,
+      std::vector<uint32_t> ids32(sz);
+      for (long long j = 0; j < sz; j++) {
+        uint64_t v = (uint64_t)ids[j];
+        ids32[j] = (v >> 63) ? ((uint32_t)(v & 0x7fffffffu) | 0x80000000u)
+                             : (uint32_t)v;
+      }
+      hipMemcpy(bk.ids->get(), ids32.data(), (size_t)sz * 4,
                 hipMemcpyHostToDevice);
       hipMemcpy(bk.data->get(), data.data(), data.size(),
                 hipMemcpyHostToDevice);

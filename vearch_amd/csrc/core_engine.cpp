@@ -122,7 +122,7 @@ int Engine::bulk_add(int64_t n, const float *vecs) {
   if (!table_created_ || n <= 0) return -1;
   std::unique_lock<std::shared_mutex> g(rw_);
   int64_t base = max_docid_;
-  if ((uint64_t)(base + n) > 0xffffffffull) return -1; /* u32 vid space */
+  if ((uint64_t)(base + n) > 0x7fffffffull) return -1; /* 31-bit vids */
   docid2pkey_.reserve(base + n);
   char buf[24];
   for (int64_t i = 0; i < n; i++) {

@@ -4,11 +4,15 @@
 #include <stdint.h>
 
 /* One IVF bucket as the device sees it (RT list storage, SURVEY §8 a7;
- * semantics of realtime_mem_data.cc:57-68: append-only SoA, ids with
- * bit-63 delete mask). For IVFPQ `data` is uint8 codes (code_size per
- * entry); for IVFFLAT it is fp32 vectors (d floats per entry). */
+ * semantics of realtime_mem_data.cc:57-68: append-only SoA with a
+ * per-entry delete mark). Device ids are compressed to u32 with bit 31
+ * as the delete mark (docids < 2^31 per partition); every host-visible
+ * surface (debug API, Dump format, oracle) keeps the reference's int64 +
+ * bit-63 form — SURVEY §8d notes the 4-byte id option and the roofline
+ * stays on the algorithmic 8-byte ids. For IVFPQ `data` is uint8 codes
+ * (code_size per entry); for IVFFLAT fp32 vectors (d floats). */
 struct GammaBucketDev {
-  const int64_t *ids;
+  const uint32_t *ids;
   const void *data;
   long long size;
 };

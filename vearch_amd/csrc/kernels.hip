@@ -351,7 +351,7 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
       dis0 = probe_dists[(int64_t)q * nprobe + p];
     }
 
-    const int64_t *ids = bk.ids;
+    const uint32_t *ids = bk.ids;
     const uint8_t *codes = (const uint8_t *)bk.data;
     if (MW > 0) {
       const int C = GAMMA_ADC_C;
@@ -364,7 +364,7 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
         for (int c = 0; c < C; c++) {
           long long j = jb + c;
           if (j < bk.size) {
-            idv[c] = ids[j];
+            idv[c] = (int64_t)(int32_t)ids[j]; /* bit31 -> negative */
             const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
 #pragma unroll
             for (int mw = 0; mw < MW; mw++) w[c][mw] = cw[mw];
@@ -397,7 +397,7 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
       for (long long j0 = 0; j0 < bk.size; j0 += blockDim.x) {
         long long j = j0 + threadIdx.x;
         if (j < bk.size) {
-          int64_t id = ids[j];
+          int64_t id = (int64_t)(int32_t)ids[j];
           if (!((uint64_t)id >> 63) &&
               !gamma_bitmap_test(bitmap, (uint64_t)id)) {
             const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
@@ -496,7 +496,7 @@ k_ivfflat_scan(int nq, int d, int nprobe, int k2,
     int64_t ln = probes[(int64_t)q * nprobe + p];
     if (ln < 0 || ln >= nlist) continue;
     GammaBucketDev bk = buckets[ln];
-    const int64_t *ids = bk.ids;
+    const uint32_t *ids = bk.ids;
     const float *vecs = (const float *)bk.data;
 
     const int vpp = nwave * 2; /* vectors per WG pass */
@@ -507,7 +507,7 @@ k_ivfflat_scan(int nq, int d, int nprobe, int k2,
       uint64_t id = 0;
       float dist = 0.0f;
       if (j < bk.size) {
-        id = (uint64_t)ids[j];
+        id = (uint64_t)(int64_t)(int32_t)ids[j];
         live = !(id >> 63) && !gamma_bitmap_test(bitmap, id);
         if (live) {
           /* float4 per half-wave lane: 32 x 16 B = one 512-B coalesced

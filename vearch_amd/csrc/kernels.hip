@@ -88,69 +88,51 @@ __global__ void __launch_bounds__(WG)
 k_dots_mfma_tiled(const float *__restrict__ Q, int nq,
                   const float *__restrict__ B, int64_t n, int d,
                   float *__restrict__ out) {
-  /* double-buffered: register-stage tile t+1's global loads while the
-   * MFMAs consume tile t from LDS; one barrier per K tile (guide §5 T14
-   * write-after-compute form). */
-  __shared__ float As[2][GT_BM][GT_BK + 1];
-  __shared__ float Bs[2][GT_BM][GT_BK + 1];
+  __shared__ float As[GT_BM][GT_BK + 1];
+  __shared__ float Bs[GT_BM][GT_BK + 1];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wr = (wave >> 1) * 64;
-  const int wc = (wave & 1) * 64;
+  const int wr = (wave >> 1) * 64; /* wave row offset in tile: 0/64 */
+  const int wc = (wave & 1) * 64;  /* wave col offset in tile: 0/64 */
   const int row0 = blockIdx.y * GT_BM;
   const int64_t col0 = (int64_t)blockIdx.x * GT_BM;
 
   f32x4 acc[4][4] = {};
-  const int ld_row = threadIdx.x >> 3;      /* 0..31 */
-  const int ld_col = (threadIdx.x & 7) * 4; /* 0,4,..,28 */
-  const int a_row = row0 + ld_row;          /* +rr*32 below */
-  const int64_t b_row = col0 + ld_row;
+  /* each thread stages 16 floats of A and B per K tile: 4 rows x float4 */
+  const int ld_row = threadIdx.x >> 3;        /* 0..31 */
+  const int ld_col = (threadIdx.x & 7) * 4;   /* 0,4,..,28 */
 
-  float4 sa[4], sb[4];
-  auto load_tile = [&](int k0) {
-#pragma unroll
-    for (int rr = 0; rr < 4; rr++) {
-      int kk = k0 + ld_col;
-      sa[rr] = make_float4(0.f, 0.f, 0.f, 0.f);
-      sb[rr] = sa[rr];
-      if (kk < d) { /* d % 4 == 0: groups never straddle */
-        if (a_row + rr * 32 < nq)
-          sa[rr] = *(const float4 *)(Q + (int64_t)(a_row + rr * 32) * d +
-                                     kk);
-        if (b_row + rr * 32 < n)
-          sb[rr] = *(const float4 *)(B + (b_row + rr * 32) * d + kk);
-      }
-    }
-  };
-  auto write_tile = [&](int buf) {
+  for (int k0 = 0; k0 < d; k0 += GT_BK) {
 #pragma unroll
     for (int rr = 0; rr < 4; rr++) {
       int r = ld_row + rr * 32;
-      As[buf][r][ld_col + 0] = sa[rr].x;
-      As[buf][r][ld_col + 1] = sa[rr].y;
-      As[buf][r][ld_col + 2] = sa[rr].z;
-      As[buf][r][ld_col + 3] = sa[rr].w;
-      Bs[buf][r][ld_col + 0] = sb[rr].x;
-      Bs[buf][r][ld_col + 1] = sb[rr].y;
-      Bs[buf][r][ld_col + 2] = sb[rr].z;
-      Bs[buf][r][ld_col + 3] = sb[rr].w;
+      int a_row = row0 + r;
+      int64_t b_row = col0 + r;
+      int kk = k0 + ld_col;
+      /* d % 4 == 0, so a 4-group never straddles the d boundary */
+      float4 av = make_float4(0.f, 0.f, 0.f, 0.f), bv = av;
+      if (kk < d) {
+        if (a_row < nq)
+          av = *(const float4 *)(Q + (int64_t)a_row * d + kk);
+        if (b_row < n) bv = *(const float4 *)(B + b_row * d + kk);
+      }
+      As[r][ld_col + 0] = av.x;
+      As[r][ld_col + 1] = av.y;
+      As[r][ld_col + 2] = av.z;
+      As[r][ld_col + 3] = av.w;
+      Bs[r][ld_col + 0] = bv.x;
+      Bs[r][ld_col + 1] = bv.y;
+      Bs[r][ld_col + 2] = bv.z;
+      Bs[r][ld_col + 3] = bv.w;
     }
-  };
-
-  const int nt = (d + GT_BK - 1) / GT_BK;
-  load_tile(0);
-  write_tile(0);
-  __syncthreads();
-  for (int t = 0; t < nt; t++) {
-    if (t + 1 < nt) load_tile((t + 1) * GT_BK); /* loads in flight */
-    const int cur = t & 1;
+    __syncthreads();
 #pragma unroll
     for (int kk = 0; kk < GT_BK; kk += 4) {
       float a[4], b[4];
 #pragma unroll
       for (int f = 0; f < 4; f++) {
-        a[f] = As[cur][wr + f * 16 + (lane & 15)][kk + (lane >> 4)];
-        b[f] = Bs[cur][wc + f * 16 + (lane & 15)][kk + (lane >> 4)];
+        a[f] = As[wr + f * 16 + (lane & 15)][kk + (lane >> 4)];
+        b[f] = Bs[wc + f * 16 + (lane & 15)][kk + (lane >> 4)];
       }
 #pragma unroll
       for (int fi = 0; fi < 4; fi++)
@@ -159,10 +141,7 @@ k_dots_mfma_tiled(const float *__restrict__ Q, int nq,
           acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(
               a[fi], b[fj], acc[fi][fj], 0, 0, 0);
     }
-    if (t + 1 < nt) {
-      write_tile(cur ^ 1); /* other buffer: no read hazard */
-      __syncthreads();
-    }
+    __syncthreads();
   }
   /* D mapping: col=lane&15, row=(lane>>4)*4+r (guide §3) */
 #pragma unroll

@@ -423,3 +423,91 @@ def test_delete_all_then_search(data):
     gd, gi = eng.raw_search(q[:4], 5)
     assert (gi == -1).all()
     eng.close()
+
+
+def test_filtered_search_term_and_range(data):
+    """scalar filters -> device exclusion bitmap. Term and
+    range filters through the real protobuf Search; parity vs the oracle
+    FLAT scan with an equivalent bitmap."""
+    import struct
+    from vearch_amd import fbsenc
+    base, q = data
+    eng = make_engine("/tmp/gamma_filter")
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}',
+                     scalar_fields=[("tag", fbsenc.DATA_STRING),
+                                    ("num", fbsenc.DATA_INT)])
+    n = 3000
+    for vid in range(n):
+        eng.add_doc(str(vid), base[vid],
+                    fields=[("tag", b"grp%d" % (vid % 4),
+                             fbsenc.DATA_STRING),
+                            ("num", struct.pack("<i", vid % 100),
+                             fbsenc.DATA_INT)])
+    # term filter tag == grp1
+    res = eng.search_pb(q[:8], topn=10,
+                        term_filters=[("tag", b"grp1")])
+    bm = np.zeros((n + 7) // 8, dtype=np.uint8)
+    for vid in range(n):
+        if vid % 4 != 1:
+            bm[vid >> 3] |= 1 << (vid & 7)
+    od, oi = orc.flat_search(base[:n], q[:8], 10, "L2", del_bitmap=bm)
+    for t in range(8):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert ids == [i for i in oi[t].tolist() if i >= 0]
+        assert all(i % 4 == 1 for i in ids)
+    # term filter with two \x01-separated values (grp1 or grp2)
+    res = eng.search_pb(q[:4], topn=10,
+                        term_filters=[("tag", b"grp1\x01grp2")])
+    for t in range(4):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert all(i % 4 in (1, 2) for i in ids)
+    # range filter 10 <= num <= 20
+    res = eng.search_pb(
+        q[:8], topn=10,
+        range_filters=[("num", struct.pack("<i", 10),
+                        struct.pack("<i", 20), True, True)])
+    bm2 = np.zeros((n + 7) // 8, dtype=np.uint8)
+    for vid in range(n):
+        if not (10 <= vid % 100 <= 20):
+            bm2[vid >> 3] |= 1 << (vid & 7)
+    od2, oi2 = orc.flat_search(base[:n], q[:8], 10, "L2", del_bitmap=bm2)
+    for t in range(8):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert ids == [i for i in oi2[t].tolist() if i >= 0]
+    # combined AND + unknown field error
+    res = eng.search_pb(
+        q[:4], topn=10, term_filters=[("tag", b"grp1")],
+        range_filters=[("num", struct.pack("<i", 10),
+                        struct.pack("<i", 40), True, False)])
+    for t in range(4):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert all(i % 4 == 1 and 10 <= i % 100 < 40 for i in ids)
+    with pytest.raises(RuntimeError):
+        eng.search_pb(q[:2], topn=5, term_filters=[("nope", b"x")])
+    eng.close()
+
+
+def test_filtered_search_ivfpq(data, ivfpq_engine):
+    """filters work on the IVFPQ path too (same bitmap arg)."""
+    base, q = data
+    eng = make_engine("/tmp/gamma_filter_pq")
+    from vearch_amd import fbsenc
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 64, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 8000}',
+        scalar_fields=[("tag", fbsenc.DATA_STRING)])
+    import struct
+    for vid in range(20000):
+        eng.add_doc(str(vid), base[vid],
+                    fields=[("tag", b"odd" if vid % 2 else b"even",
+                             fbsenc.DATA_STRING)])
+    eng.build_index()
+    res = eng.search_pb(q[:8], topn=10,
+                        index_params='{"nprobe": 64, "recall_num": 100}',
+                        term_filters=[("tag", b"odd")])
+    for t in range(8):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert len(ids) > 0
+        assert all(i % 2 == 1 for i in ids)
+    eng.close()

@@ -223,10 +223,6 @@ struct CStatus Search(void *engine, const char *request_str, int req_len,
     return err_status(1, "no vector query (scalar-only search via Query)");
   if (req.vec_fields.size() > 1)
     return err_status(1, "multi-vector ranking not supported this round");
-  if (req.n_filters > 0)
-    return err_status(1,
-                      "scalar filtered search not supported this round "
-                      "(SURVEY 8f-2)");
   if (req.topn <= 0) return err_status(1, "limit[topN] is zero");
 
   const gpb::VectorQuery &vq = req.vec_fields[0];
@@ -247,14 +243,29 @@ struct CStatus Search(void *engine, const char *request_str, int req_len,
         metric = strcasecmp(mt.c_str(), "L2") == 0 ? 1 : 2;
     }
   }
+  /* scalar filters -> exclusion bitmap consumed by the scan kernels
+   * (linear predicate pass, scalar indexes are a later
+   * row). filter_operator OR is not supported yet. */
+  std::vector<vgamma::TermFilterSpec> terms;
+  for (auto &t : req.term_filters)
+    terms.push_back({t.field, t.value, t.is_union});
+  std::vector<vgamma::RangeFilterSpec> ranges;
+  for (auto &t : req.range_filters)
+    ranges.push_back({t.field, t.lower, t.upper, t.include_lower,
+                      t.include_upper});
+  if (req.op == 1 && req.n_filters > 1)
+    return err_status(1, "filter operator OR not supported this round");
+
   int k = req.topn + req.offset;
   std::vector<float> dists((size_t)nq * k);
   std::vector<int64_t> ids((size_t)nq * k);
+  std::string ferr;
   int rc = e->search(nq, (const float *)vq.value.data(), k, nprobe,
                      recall_num, metric, req.brute == 1, req.request_id,
                      req.partition_id, dists.data(), ids.data(),
-                     req.l2_sqrt);
+                     req.l2_sqrt, &terms, &ranges, &ferr);
   if (rc == -2) return err_status(-2, "request killed");
+  if (rc == -3) return err_status(1, ferr);
   if (rc != 0) return err_status(1, "search failed");
 
   /* score-range filter (SearchCondition::IsSimilarScoreValid,

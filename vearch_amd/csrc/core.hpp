@@ -202,6 +202,19 @@ struct FieldMeta {
   int data_type = 0; /* gamma_api DataType */
 };
 
+/* scalar filters (vearchpb Term/RangeFilter). Evaluated on
+ * the host columns into an exclusion bitmap the scan kernels consume —
+ * scalar INDEX acceleration (roaring/inverted) is a later row; this is
+ * a linear predicate pass per request. */
+struct TermFilterSpec {
+  std::string field, value; /* value: terms separated by \x01 */
+  int is_union = 0;
+};
+struct RangeFilterSpec {
+  std::string field, lower, upper; /* raw binary of the field type */
+  bool inc_l = false, inc_u = false;
+};
+
 /* The engine: one per Init() (one Vearch partition). Single vector field
  * round 1 (vector_manager multi-field merging is a later row). */
 class Engine {
@@ -229,7 +242,10 @@ class Engine {
   int search(int nq, const float *xq, int k, int nprobe, int recall_num,
              int metric, bool brute_force, const std::string &request_id,
              int partition_id, float *out_dists, int64_t *out_ids,
-             bool l2_sqrt = false);
+             bool l2_sqrt = false,
+             const std::vector<TermFilterSpec> *term_filters = nullptr,
+             const std::vector<RangeFilterSpec> *range_filters = nullptr,
+             std::string *filter_err = nullptr);
   /* upload queries once; later search(nq, nullptr, ...) reuses them */
   int cache_queries(int nq, const float *xq);
   int cached_nq() const { return cached_nq_; }
@@ -258,7 +274,11 @@ class Engine {
  private:
   int flat_search_keys(const float *q_dev, int nq, int k2,
                        const float *q_norms_dev, bool ip, hipStream_t s,
-                       uint64_t *out_keys_dev);
+                       uint64_t *out_keys_dev, const uint32_t *bm);
+  /* 1 = excluded (deleted or fails a filter); returns device ptr via out */
+  int build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
+                           const std::vector<RangeFilterSpec> &ranges,
+                           const uint32_t **dev_out, std::string *err);
   std::string path_, log_dir_, space_name_;
   std::string index_type_ = "IVFPQ";
   std::string vec_name_;
@@ -279,7 +299,8 @@ class Engine {
   int cached_nq_ = 0;
   mutable std::shared_mutex rw_; /* search shared / add+build exclusive */
   DeviceBuf q_dev_, q_norms_dev_, keys_dev_, out_d_dev_, out_i_dev_;
-  DeviceBuf flat_dots_, flat_keys_;
+  DeviceBuf flat_dots_, flat_keys_, filt_dev_;
+  std::vector<uint32_t> filt_host_;
 };
 
 }  // namespace vgamma

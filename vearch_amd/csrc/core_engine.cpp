@@ -196,13 +196,13 @@ int Engine::build_index(std::string *err) {
 
 int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
                              const float *q_norms_dev, bool ip,
-                             hipStream_t s, uint64_t *out_keys_dev) {
+                             hipStream_t s, uint64_t *out_keys_dev,
+                             const uint32_t *bm) {
   const int64_t n = raw_.size();
   if (nq < 512 || n < 200000) {
     GAMMA_CHECK(gk::flat_stream_scan(
         s, nq, n, dim_, k2, q_dev, raw_.dev_seg_table(),
-        RawStore::SEG_SHIFT, bitmap_.any() ? bitmap_.dev() : nullptr, ip,
-        out_keys_dev));
+        RawStore::SEG_SHIFT, bm, ip, out_keys_dev));
     return 0;
   }
   /* chunked MFMA GEMM + seeded select (nq large): per segment-run chunks */
@@ -217,19 +217,148 @@ int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
                               flat_dots_.as<float>()));
     GAMMA_CHECK(gk::select_from_dots(
         s, nq, take, v0, take, flat_dots_.as<float>(), q_norms_dev,
-        raw_.dev_norms(), !ip, ip,
-        bitmap_.any() ? bitmap_.dev() : nullptr, k2, out_keys_dev,
-        seeded));
+        raw_.dev_norms(), !ip, ip, bm, k2, out_keys_dev, seeded));
     seeded = true;
     v0 += take;
   }
   return 0;
 }
 
+namespace {
+/* decode a raw-binary scalar by gamma DataType for range compares */
+static bool decode_num(int dt, const std::string &b, double *out) {
+  if (dt == 0 && b.size() >= 4) { /* INT */
+    int32_t v; memcpy(&v, b.data(), 4); *out = v; return true;
+  }
+  if ((dt == 1 || dt == 7) && b.size() >= 8) { /* LONG / DATE */
+    int64_t v; memcpy(&v, b.data(), 8); *out = (double)v; return true;
+  }
+  if (dt == 2 && b.size() >= 4) { /* FLOAT */
+    float v; memcpy(&v, b.data(), 4); *out = v; return true;
+  }
+  if (dt == 3 && b.size() >= 8) { /* DOUBLE */
+    double v; memcpy(&v, b.data(), 8); *out = v; return true;
+  }
+  if (dt == 6 && b.size() >= 1) { /* BOOL */
+    *out = b[0] != 0; return true;
+  }
+  return false;
+}
+
+static bool term_match(int dt, const std::string &doc_val,
+                       const std::string &filter_val) {
+  /* filter value may hold several terms separated by \x01; STRINGARRAY
+   * doc values are \x01-separated too (c_api/api_data/doc.cc:102) */
+  auto split = [](const std::string &s) {
+    std::vector<std::string> out;
+    size_t p = 0;
+    while (p <= s.size()) {
+      size_t q = s.find('\x01', p);
+      if (q == std::string::npos) { out.push_back(s.substr(p)); break; }
+      out.push_back(s.substr(p, q - p));
+      p = q + 1;
+    }
+    return out;
+  };
+  std::vector<std::string> terms = split(filter_val);
+  if (dt == 8) { /* STRINGARRAY: any element matches any term */
+    for (auto &el : split(doc_val))
+      for (auto &t : terms)
+        if (!el.empty() && el == t) return true;
+    return false;
+  }
+  for (auto &t : terms)
+    if (doc_val == t) return true;
+  return false;
+}
+}  // namespace
+
+int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
+                                 const std::vector<RangeFilterSpec> &ranges,
+                                 const uint32_t **dev_out,
+                                 std::string *err) {
+  /* resolve fields once */
+  struct TF { int dt; const std::vector<std::string> *col;
+              const TermFilterSpec *f; };
+  struct RF { int dt; const std::vector<std::string> *col;
+              const RangeFilterSpec *f; double lo, hi; };
+  std::vector<TF> tfs;
+  std::vector<RF> rfs;
+  for (auto &t : terms) {
+    auto it = field_vals_.find(t.field);
+    if (it == field_vals_.end()) {
+      if (err) *err = "unknown filter field " + t.field;
+      return -1;
+    }
+    int dt = 0;
+    for (auto &fm : fields_)
+      if (fm.name == t.field) dt = fm.data_type;
+    tfs.push_back({dt, &it->second, &t});
+  }
+  for (auto &r : ranges) {
+    auto it = field_vals_.find(r.field);
+    if (it == field_vals_.end()) {
+      if (err) *err = "unknown filter field " + r.field;
+      return -1;
+    }
+    int dt = 0;
+    for (auto &fm : fields_)
+      if (fm.name == r.field) dt = fm.data_type;
+    RF rf{dt, &it->second, &r, 0, 0};
+    if (dt != 4 && dt != 8) { /* numeric */
+      if (!decode_num(dt, r.lower, &rf.lo) ||
+          !decode_num(dt, r.upper, &rf.hi)) {
+        if (err) *err = "bad range filter value for field " + r.field;
+        return -1;
+      }
+    }
+    rfs.push_back(rf);
+  }
+  int64_t n = max_docid_;
+  int64_t words = (n + 31) / 32;
+  filt_host_.assign((size_t)words, 0);
+  for (int64_t id = 0; id < n; id++) {
+    bool excl = bitmap_.test(id);
+    static const std::string kNone;
+    for (auto &t : tfs) {
+      if (excl) break;
+      const std::string &v =
+          id < (int64_t)t.col->size() ? (*t.col)[id] : kNone;
+      if (!term_match(t.dt, v, t.f->value)) excl = true;
+    }
+    for (auto &r : rfs) {
+      if (excl) break;
+      const std::string &v =
+          id < (int64_t)r.col->size() ? (*r.col)[id] : kNone;
+      if (r.dt == 4) { /* STRING: lexicographic */
+        bool ok_l = r.f->inc_l ? (v >= r.f->lower) : (v > r.f->lower);
+        bool ok_u = r.f->inc_u ? (v <= r.f->upper) : (v < r.f->upper);
+        if (!(ok_l && ok_u)) excl = true;
+      } else {
+        double x;
+        if (!decode_num(r.dt, v, &x)) { excl = true; continue; }
+        bool ok_l = r.f->inc_l ? (x >= r.lo) : (x > r.lo);
+        bool ok_u = r.f->inc_u ? (x <= r.hi) : (x < r.hi);
+        if (!(ok_l && ok_u)) excl = true;
+      }
+    }
+    if (excl) filt_host_[id >> 5] |= 1u << (id & 31);
+  }
+  if (filt_dev_.reserve((size_t)std::max<int64_t>(words, 1) * 4)) return -1;
+  GAMMA_CHECK(hipMemcpyAsync(filt_dev_.get(), filt_host_.data(),
+                             (size_t)words * 4, hipMemcpyHostToDevice,
+                             stream_));
+  *dev_out = filt_dev_.as<uint32_t>();
+  return 0;
+}
+
 int Engine::search(int nq, const float *xq, int k, int nprobe,
                    int recall_num, int metric, bool brute_force,
                    const std::string &request_id, int partition_id,
-                   float *out_dists, int64_t *out_ids, bool l2_sqrt) {
+                   float *out_dists, int64_t *out_ids, bool l2_sqrt,
+                   const std::vector<TermFilterSpec> *term_filters,
+                   const std::vector<RangeFilterSpec> *range_filters,
+                   std::string *filter_err) {
   if (!table_created_ || nq <= 0 || k <= 0) return -1;
   /* Exclusive: concurrent cgo Search calls (engine.cc allows them) are
    * serialized here because they share the engine's scratch device
@@ -281,6 +410,18 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
   if (out_i_dev_.reserve((size_t)nq * k * 8)) return -1;
   tm.rec(1);
 
+  const uint32_t *bm = bitmap_.any() ? bitmap_.dev() : nullptr;
+  bool have_filters = (term_filters && !term_filters->empty()) ||
+                      (range_filters && !range_filters->empty());
+  if (have_filters) {
+    static const std::vector<TermFilterSpec> kNoT;
+    static const std::vector<RangeFilterSpec> kNoR;
+    if (build_filter_bitmap_(term_filters ? *term_filters : kNoT,
+                             range_filters ? *range_filters : kNoR, &bm,
+                             filter_err))
+      return -3;
+  }
+
   bool use_flat = brute_force || params_.kind == IndexKind::FLAT ||
                   !index_ || !index_->trained();
   double t_assign = 0, t_scan = 0;
@@ -293,12 +434,11 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
     if (keys_dev_.reserve((size_t)nq * kf * 8)) return -1;
     if (flat_search_keys(q_dev_.as<float>(), nq, kf,
                          q_norms_dev_.as<float>(), ip, s,
-                         keys_dev_.as<uint64_t>()))
+                         keys_dev_.as<uint64_t>(), bm))
       return -1;
     k2 = kf;
   } else {
-    if (index_->search(q_dev_.as<float>(), nq, k2, nprobe,
-                       bitmap_.any() ? bitmap_.dev() : nullptr,
+    if (index_->search(q_dev_.as<float>(), nq, k2, nprobe, bm,
                        ip, s, keys_dev_.as<uint64_t>(),
                        q_norms_dev_.as<float>(), &t_assign, &t_scan))
       return -1;

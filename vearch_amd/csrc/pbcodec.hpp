@@ -133,13 +133,27 @@ struct VectorQuery { /* router_grpc.proto:128-135 */
   std::string format, index_type;
 };
 
+struct TermFilter { /* router_grpc.proto:110-114 */
+  std::string field;
+  std::string value; /* one or more terms separated by \x01 */
+  int is_union = 0;
+};
+struct RangeFilter { /* router_grpc.proto:116-123; values = raw binary of
+                        the field type (NumericToSortableStr convention,
+                        table/inverted_index.cc:263) */
+  std::string field, lower, upper;
+  bool include_lower = false, include_upper = false;
+};
+
 struct SearchRequest { /* router_grpc.proto:168-192 */
   std::string request_id;
   int partition_id = 0;
   int req_num = 0, topn = 0, brute = 0;
   std::vector<VectorQuery> vec_fields;
   std::vector<std::string> fields;
-  int n_filters = 0; /* range+term (filters are out of scope, §8f-2) */
+  std::vector<RangeFilter> range_filters;
+  std::vector<TermFilter> term_filters;
+  int n_filters = 0;
   std::string index_params, ranker;
   int multi_vector_rank = 0;
   bool l2_sqrt = false, trace = false, is_vector_value = false;
@@ -207,7 +221,47 @@ struct SearchRequest { /* router_grpc.proto:168-192 */
         }
         case 6: { std::string s; if (!r.bytes(s)) return false;
                   fields.push_back(std::move(s)); break; }
-        case 7: case 8: { if (!r.skip(wt)) return false; n_filters++; break; }
+        case 7: { /* range_filters */
+          Reader fr(nullptr, 0);
+          if (!r.sub(fr)) return false;
+          RangeFilter rf;
+          uint32_t ff, fwt;
+          while (!fr.done()) {
+            if (!fr.key(ff, fwt)) return false;
+            switch (ff) {
+              case 1: if (!fr.bytes(rf.field)) return false; break;
+              case 2: if (!fr.bytes(rf.lower)) return false; break;
+              case 3: if (!fr.bytes(rf.upper)) return false; break;
+              case 4: { uint64_t v; if (!fr.varint(v)) return false;
+                        rf.include_lower = v != 0; break; }
+              case 5: { uint64_t v; if (!fr.varint(v)) return false;
+                        rf.include_upper = v != 0; break; }
+              default: if (!fr.skip(fwt)) return false;
+            }
+          }
+          range_filters.push_back(std::move(rf));
+          n_filters++;
+          break;
+        }
+        case 8: { /* term_filters */
+          Reader fr(nullptr, 0);
+          if (!r.sub(fr)) return false;
+          TermFilter tf;
+          uint32_t ff, fwt;
+          while (!fr.done()) {
+            if (!fr.key(ff, fwt)) return false;
+            switch (ff) {
+              case 1: if (!fr.bytes(tf.field)) return false; break;
+              case 2: if (!fr.bytes(tf.value)) return false; break;
+              case 3: { uint64_t v; if (!fr.varint(v)) return false;
+                        tf.is_union = (int)v; break; }
+              default: if (!fr.skip(fwt)) return false;
+            }
+          }
+          term_filters.push_back(std::move(tf));
+          n_filters++;
+          break;
+        }
         case 9: if (!r.bytes(index_params)) return false; break;
         case 10: { uint64_t v; if (!r.varint(v)) return false;
                    multi_vector_rank = (int)v; break; }

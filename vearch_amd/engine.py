@@ -197,7 +197,8 @@ class GammaEngine:
 
     def search_pb(self, queries, topn, index_params="", fields=("_id",),
                   request_id="req1", partition_id=1, brute=0,
-                  min_score=None, max_score=None, l2_sqrt=False):
+                  min_score=None, max_score=None, l2_sqrt=False,
+                  term_filters=(), range_filters=()):
         """The real C-ABI Search with protobuf marshalling (reader.go
         path)."""
         from . import proto
@@ -206,7 +207,8 @@ class GammaEngine:
             self.vec_name, q.tobytes(), topn, q.shape[0],
             request_id=request_id, partition_id=partition_id,
             index_params=index_params, brute=brute, fields=fields,
-            min_score=min_score, max_score=max_score, l2_sqrt=l2_sqrt)
+            min_score=min_score, max_score=max_score, l2_sqrt=l2_sqrt,
+            term_filters=term_filters, range_filters=range_filters)
         out = ctypes.c_char_p()
         n = ctypes.c_int()
         st = lib().Search(self.h, req, len(req), ctypes.byref(out),

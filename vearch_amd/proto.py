@@ -43,8 +43,12 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
                           request_id="req1", partition_id=1,
                           index_params="", min_score=None, max_score=None,
                           brute=0, fields=("_id",), l2_sqrt=False,
-                          is_vector_value=False):
-    """vearchpb.SearchRequest (router_grpc.proto:168-192)."""
+                          is_vector_value=False, term_filters=(),
+                          range_filters=()):
+    """vearchpb.SearchRequest (router_grpc.proto:168-192).
+    term_filters: (field, value_bytes) or (field, value_bytes, is_union).
+    range_filters: (field, lower_bytes, upper_bytes, incl_lower,
+    incl_upper) — values are the raw binary of the field type."""
     out = b""
     out += _ld(1, encode_head(request_id, partition_id))
     out += _vint(2, req_num)
@@ -58,6 +62,17 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
     out += _ld(5, vq)
     for f in fields:
         out += _ld(6, f.encode())
+    for rf in range_filters:
+        field, lower, upper, il, iu = rf
+        body = _ld(1, field.encode()) + _ld(2, lower) + _ld(3, upper)
+        body += _vint(4, 1 if il else 0) + _vint(5, 1 if iu else 0)
+        out += _ld(7, body)
+    for tf in term_filters:
+        field, value = tf[0], tf[1]
+        body = _ld(1, field.encode()) + _ld(2, value)
+        if len(tf) > 2:
+            body += _vint(3, tf[2])
+        out += _ld(8, body)
     if index_params:
         out += _ld(9, index_params.encode())
     if l2_sqrt:

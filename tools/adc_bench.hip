@@ -36,8 +36,8 @@
 #define V_FASTCMP 64 /* float-key compare before the full push */
 #define V_U32ID 128  /* 4-byte ids (bit 31 delete) instead of int64 */
 
-template <int MW, int C, int VAR>
-__global__ void __launch_bounds__(256)
+template <int MW, int C, int VAR, int BS = 256>
+__global__ void __launch_bounds__(BS)
 k_scan_var(int nq, int d, int M, int nprobe, int k2,
            const float *__restrict__ queries,
            const float *__restrict__ centroids,
@@ -109,8 +109,7 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
     }
     if (VAR & V_NOSCAN) continue;
 
-    const int64_t *ids = bk.ids;
-    const uint32_t *ids32 = (const uint32_t *)bk.ids;
+    const uint32_t *ids32 = bk.ids;
     const uint8_t *codes = (const uint8_t *)bk.data;
     for (long long j0 = 0; j0 < bk.size; j0 += (long long)blockDim.x * C) {
       long long jb = j0 + (long long)threadIdx.x * C;
@@ -120,9 +119,7 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
       for (int c = 0; c < C; c++) {
         long long j = jb + c;
         if (j < bk.size) {
-          idv[c] = (VAR & V_U32ID)
-                       ? (int64_t)(int32_t)ids32[j]
-                       : ids[j];
+          idv[c] = (int64_t)(int32_t)ids32[j];
           const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
 #pragma unroll
           for (int mw = 0; mw < MW; mw++) w[c][mw] = cw[mw];
@@ -170,11 +167,11 @@ k_scan_var(int nq, int d, int M, int nprobe, int k2,
     out_keys[(int64_t)q * k2 + i] = res[i];
 }
 
-__global__ void k_fill(uint8_t *codes, int64_t *ids, int64_t n, int M,
+__global__ void k_fill(uint8_t *codes, uint32_t *ids, int64_t n, int M,
                        uint64_t seed) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  ids[i] = i;
+  ids[i] = (uint32_t)i;
   uint64_t s = seed + i * 2654435761ull;
   for (int m = 0; m < M; m++) {
     s = s * 6364136223846793005ull + 1442695040888963407ull;
@@ -194,9 +191,9 @@ int main(int argc, char **argv) {
   CHECK(hipMalloc(&codebooks, (size_t)M * 256 * (d / M) * 4));
   /* one big slab for codes+ids, sliced into buckets */
   uint8_t *codes;
-  int64_t *ids;
+  uint32_t *ids;
   CHECK(hipMalloc(&codes, (size_t)N * M));
-  CHECK(hipMalloc(&ids, (size_t)N * 8));
+  CHECK(hipMalloc(&ids, (size_t)N * 4));
   k_fill<<<dim3((uint32_t)((N + 255) / 256)), dim3(256)>>>(codes, ids, N, M,
                                                            42);
   std::vector<GammaBucketDev> hb(nlist);
@@ -234,12 +231,12 @@ int main(int argc, char **argv) {
                 (GAMMA_SORT_CAP + k2) * 8 + (2 * d + 1) * 4 + 16;
   double bytes_per_q = (double)nprobe * per * (M + 8);
 
-  auto run = [&](const char *name, auto kern, int reps) {
+  auto run_bs = [&](const char *name, auto kern, int reps, int bs) {
     /* warmup */
-    kern<<<dim3(nq), dim3(256), smem>>>(nq, d, M, nprobe, k2, queries,
-                                        centroids, codebooks, buckets,
-                                        nlist, probes, out, Atab, Btab,
-                                        bitmap);
+    kern<<<dim3(nq), dim3(bs), smem>>>(nq, d, M, nprobe, k2, queries,
+                                       centroids, codebooks, buckets,
+                                       nlist, probes, out, Atab, Btab,
+                                       bitmap);
     CHECK(hipGetLastError());
     CHECK(hipDeviceSynchronize());
     hipEvent_t a, b;
@@ -247,10 +244,10 @@ int main(int argc, char **argv) {
     hipEventCreate(&b);
     hipEventRecord(a);
     for (int r = 0; r < reps; r++)
-      kern<<<dim3(nq), dim3(256), smem>>>(nq, d, M, nprobe, k2, queries,
-                                          centroids, codebooks, buckets,
-                                          nlist, probes, out, Atab, Btab,
-                                          bitmap);
+      kern<<<dim3(nq), dim3(bs), smem>>>(nq, d, M, nprobe, k2, queries,
+                                         centroids, codebooks, buckets,
+                                         nlist, probes, out, Atab, Btab,
+                                         bitmap);
     hipEventRecord(b);
     CHECK(hipEventSynchronize(b));
     float ms;
@@ -260,6 +257,9 @@ int main(int argc, char **argv) {
            nq * bytes_per_q / ms / 1e6, nq / ms * 1000.0);
     hipEventDestroy(a);
     hipEventDestroy(b);
+  };
+  auto run = [&](const char *name, auto kern, int reps) {
+    run_bs(name, kern, reps, 256);
   };
 
   printf("nq=%d N=%lld nlist=%d nprobe=%d M=%d k2=%d smem=%zu\n", nq,
@@ -287,5 +287,13 @@ int main(int argc, char **argv) {
   run("pct1v+u32 C=4", k_scan_var<MW, 4, V_PCT1V | V_U32ID>, 3);
   run("pct1v+u32+nopush C=2",
       k_scan_var<MW, 2, V_PCT1V | V_U32ID | V_NOPUSH>, 3);
+  run_bs("pct1v+u32 C=2 BS512",
+         k_scan_var<MW, 2, V_PCT1V | V_U32ID, 512>, 3, 512);
+  run_bs("pct1v+u32 C=1 BS512",
+         k_scan_var<MW, 1, V_PCT1V | V_U32ID, 512>, 3, 512);
+  run_bs("pct1v+u32 C=4 BS512",
+         k_scan_var<MW, 4, V_PCT1V | V_U32ID, 512>, 3, 512);
+  run_bs("pct1v+u32+nopush BS512",
+         k_scan_var<MW, 2, V_PCT1V | V_U32ID | V_NOPUSH, 512>, 3, 512);
   return 0;
 }

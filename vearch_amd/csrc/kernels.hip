@@ -275,8 +275,8 @@ hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
  * issued before any use, so one barrier interval carries
  * blockDim*C codes' worth of HBM latency instead of blockDim's. */
 #define GAMMA_ADC_C 2
-template <bool IP, int MW>
-__global__ void __launch_bounds__(WG)
+template <bool IP, int MW, int BS>
+__global__ void __launch_bounds__(BS)
 k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
              const float *__restrict__ queries,
              const float *__restrict__ centroids,
@@ -434,26 +434,27 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
   size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
                 (GAMMA_SORT_CAP + k2) * 8 + (2 * d + 1) * 4 + 4 * sizeof(int);
   if (smem > 160 * 1024) return hipErrorInvalidValue;
-  /* batched path needs flush margin blockDim*C inside the selector cap */
-  bool fast = (k2 + WG * GAMMA_ADC_C) <= GAMMA_SORT_CAP &&
+  /* batched path needs flush margin blockDim*C inside the selector cap.
+   * 512-thread blocks put 24 waves on a CU at the same LDS/WG (the ADC
+   * gather stream wants ~4 waves/SIMD — microarch §LDS). */
+  const int BS = 512;
+  bool fast = (k2 + BS * GAMMA_ADC_C) <= GAMMA_SORT_CAP &&
               (M == 16 || M == 32 || M == 64);
-  dim3 g(nq), b(WG);
-#define GAMMA_LAUNCH_SCAN(IPV, MWV)                                       \
-  k_ivfpq_scan<IPV, MWV><<<g, b, smem, s>>>(nq, d, M, nprobe, k2,         \
-                                            queries, centroids,           \
-                                            codebooks, atab, btab,        \
-                                            probe_dists, buckets, nlist,  \
-                                            probes, bitmap, out_keys)
+  dim3 g(nq);
+#define GAMMA_LAUNCH_SCAN(IPV, MWV, BSV)                                  \
+  k_ivfpq_scan<IPV, MWV, BSV><<<g, dim3(BSV), smem, s>>>(                 \
+      nq, d, M, nprobe, k2, queries, centroids, codebooks, atab, btab,    \
+      probe_dists, buckets, nlist, probes, bitmap, out_keys)
   if (ip) {
-    if (!fast) GAMMA_LAUNCH_SCAN(true, 0);
-    else if (M == 16) GAMMA_LAUNCH_SCAN(true, 4);
-    else if (M == 32) GAMMA_LAUNCH_SCAN(true, 8);
-    else GAMMA_LAUNCH_SCAN(true, 16);
+    if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG);
+    else if (M == 16) GAMMA_LAUNCH_SCAN(true, 4, 512);
+    else if (M == 32) GAMMA_LAUNCH_SCAN(true, 8, 512);
+    else GAMMA_LAUNCH_SCAN(true, 16, 512);
   } else {
-    if (!fast) GAMMA_LAUNCH_SCAN(false, 0);
-    else if (M == 16) GAMMA_LAUNCH_SCAN(false, 4);
-    else if (M == 32) GAMMA_LAUNCH_SCAN(false, 8);
-    else GAMMA_LAUNCH_SCAN(false, 16);
+    if (!fast) GAMMA_LAUNCH_SCAN(false, 0, WG);
+    else if (M == 16) GAMMA_LAUNCH_SCAN(false, 4, 512);
+    else if (M == 32) GAMMA_LAUNCH_SCAN(false, 8, 512);
+    else GAMMA_LAUNCH_SCAN(false, 16, 512);
   }
 #undef GAMMA_LAUNCH_SCAN
   return hipGetLastError();

@@ -511,3 +511,67 @@ def test_filtered_search_ivfpq(data, ivfpq_engine):
         assert len(ids) > 0
         assert all(i % 2 == 1 for i in ids)
     eng.close()
+
+
+def test_query_pb_by_ids_and_filters(data):
+    from vearch_amd import fbsenc
+    base, q = data
+    eng = make_engine("/tmp/gamma_query")
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}',
+                     scalar_fields=[("tag", fbsenc.DATA_STRING)])
+    for vid in range(200):
+        eng.add_doc(str(vid), base[vid],
+                    fields=[("tag", b"a" if vid % 2 else b"b",
+                             fbsenc.DATA_STRING)])
+    res = eng.query_pb(document_ids=["5", "7", "nope"], fields=("_id", "tag"))
+    ids = [int(it["fields"]["_id"]) for it in res[0]["items"]]
+    assert ids == [5, 7]
+    assert res[0]["items"][0]["fields"]["tag"] == b"a"
+    # filtered browse with limit (Engine::Query filter path)
+    res = eng.query_pb(term_filters=[("tag", b"b")], limit=10,
+                       fields=("_id", "tag"))
+    ids = [int(it["fields"]["_id"]) for it in res[0]["items"]]
+    assert len(ids) == 10
+    assert all(i % 2 == 0 for i in ids)
+    eng.close()
+
+
+def test_concurrent_search_and_mutation(data):
+    """The reference allows Search from arbitrary cgo threads while a
+    background thread mutates (engine.cc:1108-1127); the engine
+    serializes internally — results must stay sane under contention."""
+    import threading
+    base, q = data
+    eng = make_engine("/tmp/gamma_conc")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 64, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 8000}')
+    eng.add(base)
+    eng.build_index()
+    errors = []
+
+    def searcher(tid):
+        try:
+            for _ in range(20):
+                gd, gi = eng.raw_search(q[tid:tid + 8], 5, nprobe=16)
+                live = gi[gi >= 0]
+                assert (live < 30000).all()
+        except Exception as ex:  # noqa: BLE001
+            errors.append(ex)
+
+    threads = [threading.Thread(target=searcher, args=(t,))
+               for t in range(4)]
+    for t in threads:
+        t.start()
+    for vid in range(0, 2000, 3):
+        eng.delete_doc(str(vid))
+    for t in threads:
+        t.join(timeout=120)
+    assert not errors, errors
+    # after the dust settles: deleted ids gone
+    gd, gi = eng.raw_search(q[:8], 10, nprobe=64)
+    live = gi[gi >= 0]
+    bad = [v for v in live.tolist() if v < 2000 and v % 3 == 0]
+    assert not bad
+    eng.close()

@@ -317,14 +317,30 @@ struct CStatus Query(void *engine, const char *request_str, int req_len,
   gpb::QueryRequest req;
   if (!req.parse(request_str, req_len))
     return err_status(1, "parse query request failed");
-  if (req.n_filters > 0)
-    return err_status(1, "filtered query not supported this round");
+  std::vector<int64_t> docids;
+  if (!req.document_ids.empty()) {
+    for (auto &pk : req.document_ids) {
+      int64_t id = e->docid_of(pk);
+      if (id >= 0 && !e->bitmap().test(id)) docids.push_back(id);
+    }
+  } else if (req.n_filters > 0) {
+    /* browse by scalar predicate (Engine::Query filter path) */
+    std::vector<vgamma::TermFilterSpec> terms;
+    for (auto &t : req.term_filters)
+      terms.push_back({t.field, t.value, t.is_union});
+    std::vector<vgamma::RangeFilterSpec> ranges;
+    for (auto &t : req.range_filters)
+      ranges.push_back({t.field, t.lower, t.upper, t.include_lower,
+                        t.include_upper});
+    std::string ferr;
+    if (e->filter_docids(terms, ranges, 0,
+                         req.limit > 0 ? req.limit : 50, &docids, &ferr))
+      return err_status(1, ferr);
+  }
   std::vector<gpb::SearchResult> results(1);
   gpb::SearchResult &res = results[0];
-  res.total = (int)req.document_ids.size();
-  for (auto &pk : req.document_ids) {
-    int64_t id = e->docid_of(pk);
-    if (id < 0 || e->bitmap().test(id)) continue;
+  res.total = (int)docids.size();
+  for (int64_t id : docids) {
     gpb::ResultItem item;
     item.score = 0;
     item.fields.push_back({"_id", e->pkey_of(id)});

@@ -253,6 +253,12 @@ class Engine {
   int dump(std::string *err);
   int load(std::string *err);
 
+  /* doc browse by scalar predicate (Engine::Query filter path,
+   * engine.cc:420+): docids passing all filters, offset/limit applied */
+  int filter_docids(const std::vector<TermFilterSpec> &terms,
+                    const std::vector<RangeFilterSpec> &ranges, int offset,
+                    int limit, std::vector<int64_t> *out, std::string *err);
+
   int64_t num_docs() const { return max_docid_; }
   IVFIndex *index() { return index_.get(); }
   RawStore &raw() { return raw_; }

@@ -480,6 +480,23 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
   return 0;
 }
 
+int Engine::filter_docids(const std::vector<TermFilterSpec> &terms,
+                          const std::vector<RangeFilterSpec> &ranges,
+                          int offset, int limit, std::vector<int64_t> *out,
+                          std::string *err) {
+  std::unique_lock<std::shared_mutex> g(rw_);
+  const uint32_t *unused = nullptr;
+  if (build_filter_bitmap_(terms, ranges, &unused, err)) return -1;
+  int64_t skipped = 0;
+  for (int64_t id = 0; id < max_docid_; id++) {
+    if ((filt_host_[id >> 5] >> (id & 31)) & 1u) continue;
+    if (skipped++ < offset) continue;
+    out->push_back(id);
+    if (limit > 0 && (int)out->size() >= limit) break;
+  }
+  return 0;
+}
+
 int Engine::cache_queries(int nq, const float *xq) {
   std::unique_lock<std::shared_mutex> g(rw_);
   if (q_dev_.reserve((size_t)nq * dim_ * 4)) return -1;

@@ -294,10 +294,9 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
   float *lut = (float *)smem;                       /* M*ksub */
   uint64_t *sortbuf = (uint64_t *)(smem + ((size_t)M * ksub * 4 + 7) / 8 * 8);
   uint64_t *res = sortbuf + GAMMA_SORT_CAP;
-  float *qs = (float *)(res + k2);                  /* d */
-  float *rs = qs + d;                               /* d residual */
-  float *dis0s = rs + d;                            /* 1 */
-  int *state = (int *)(dis0s + 1) + 1;              /* int[2], 8B aligned-ish */
+  float *qs = (float *)(res + k2);                  /* d (IP table build) */
+  float *dis0s = qs + d;                            /* 1 */
+  int *state = (int *)(dis0s + 1) + 1;              /* int[2] */
 
   const int q = blockIdx.x;
   if (q >= nq) return;
@@ -432,7 +431,7 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
                           const int64_t *probes, const uint32_t *bitmap,
                           bool ip, uint64_t *out_keys) {
   size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
-                (GAMMA_SORT_CAP + k2) * 8 + (2 * d + 1) * 4 + 4 * sizeof(int);
+                (GAMMA_SORT_CAP + k2) * 8 + (d + 1) * 4 + 4 * sizeof(int);
   if (smem > 160 * 1024) return hipErrorInvalidValue;
   /* batched path needs flush margin blockDim*C inside the selector cap.
    * 512-thread blocks put 24 waves on a CU at the same LDS/WG (the ADC

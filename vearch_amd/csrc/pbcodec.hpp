@@ -288,6 +288,8 @@ struct QueryRequest { /* router_grpc.proto:146-166 */
   int partition_id = 0;
   std::vector<std::string> document_ids;
   std::vector<std::string> fields;
+  std::vector<RangeFilter> range_filters;
+  std::vector<TermFilter> term_filters;
   bool is_vector_value = false;
   int limit = 0;
   int n_filters = 0;
@@ -323,7 +325,47 @@ struct QueryRequest { /* router_grpc.proto:146-166 */
         }
         case 2: { std::string s; if (!r.bytes(s)) return false;
                   document_ids.push_back(std::move(s)); break; }
-        case 5: case 6: if (!r.skip(wt)) return false; else n_filters++; break;
+        case 5: { /* range_filters */
+          Reader fr(nullptr, 0);
+          if (!r.sub(fr)) return false;
+          RangeFilter rf;
+          uint32_t ff, fwt;
+          while (!fr.done()) {
+            if (!fr.key(ff, fwt)) return false;
+            switch (ff) {
+              case 1: if (!fr.bytes(rf.field)) return false; break;
+              case 2: if (!fr.bytes(rf.lower)) return false; break;
+              case 3: if (!fr.bytes(rf.upper)) return false; break;
+              case 4: { uint64_t v; if (!fr.varint(v)) return false;
+                        rf.include_lower = v != 0; break; }
+              case 5: { uint64_t v; if (!fr.varint(v)) return false;
+                        rf.include_upper = v != 0; break; }
+              default: if (!fr.skip(fwt)) return false;
+            }
+          }
+          range_filters.push_back(std::move(rf));
+          n_filters++;
+          break;
+        }
+        case 6: { /* term_filters */
+          Reader fr(nullptr, 0);
+          if (!r.sub(fr)) return false;
+          TermFilter tf;
+          uint32_t ff, fwt;
+          while (!fr.done()) {
+            if (!fr.key(ff, fwt)) return false;
+            switch (ff) {
+              case 1: if (!fr.bytes(tf.field)) return false; break;
+              case 2: if (!fr.bytes(tf.value)) return false; break;
+              case 3: { uint64_t v; if (!fr.varint(v)) return false;
+                        tf.is_union = (int)v; break; }
+              default: if (!fr.skip(fwt)) return false;
+            }
+          }
+          term_filters.push_back(std::move(tf));
+          n_filters++;
+          break;
+        }
         case 7: { std::string s; if (!r.bytes(s)) return false;
                   fields.push_back(std::move(s)); break; }
         case 8: { uint64_t v; if (!r.varint(v)) return false;

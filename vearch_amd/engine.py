@@ -219,6 +219,21 @@ class GammaEngine:
         buf = ctypes.string_at(out, n.value)
         return proto.decode_search_response(buf)
 
+    def query_pb(self, document_ids=(), fields=("_id",), term_filters=(),
+                 range_filters=(), limit=0):
+        """The C-ABI Query (doc fetch / filtered browse)."""
+        from . import proto
+        req = proto.encode_query_request(
+            list(document_ids), fields=fields, term_filters=term_filters,
+            range_filters=range_filters, limit=limit)
+        out = ctypes.c_char_p()
+        n = ctypes.c_int()
+        st = lib().Query(self.h, req, len(req), ctypes.byref(out),
+                         ctypes.byref(n))
+        _check(st, "Query")
+        buf = ctypes.string_at(out, n.value)
+        return proto.decode_search_response(buf)
+
     def last_timing(self):
         t = (ctypes.c_double * 6)()
         lib().GammaLastSearchTiming(self.h, t)

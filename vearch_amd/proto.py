@@ -83,15 +83,26 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
 
 
 def encode_query_request(document_ids, fields=("_id",), request_id="req1",
-                         partition_id=1, is_vector_value=False):
+                         partition_id=1, is_vector_value=False,
+                         term_filters=(), range_filters=(), limit=0):
     out = b""
     out += _ld(1, encode_head(request_id, partition_id))
     for d in document_ids:
         out += _ld(2, d.encode())
+    for rf in range_filters:
+        field, lower, upper, il, iu = rf
+        body = _ld(1, field.encode()) + _ld(2, lower) + _ld(3, upper)
+        body += _vint(4, 1 if il else 0) + _vint(5, 1 if iu else 0)
+        out += _ld(5, body)
+    for tf in term_filters:
+        body = _ld(1, tf[0].encode()) + _ld(2, tf[1])
+        out += _ld(6, body)
     for f in fields:
         out += _ld(7, f.encode())
     if is_vector_value:
         out += _vint(8, 1)
+    if limit:
+        out += _vint(9, limit)
     return out
 
 

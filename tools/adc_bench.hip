@@ -181,7 +181,8 @@ __global__ void k_fill(uint8_t *codes, uint32_t *ids, int64_t n, int M,
 
 int main(int argc, char **argv) {
   int nq = argc > 1 ? atoi(argv[1]) : 10000;
-  const int d = 128, M = 32, MW = 8, nlist = 4096, nprobe = 32, k2 = 200;
+  const int nprobe = argc > 2 ? atoi(argv[2]) : 32;
+  const int d = 128, M = 32, MW = 8, nlist = 4096, k2 = 200;
   const int64_t N = 10000000;
   const int64_t per = N / nlist;
 

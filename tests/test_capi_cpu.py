@@ -148,3 +148,44 @@ def test_search_response_decoder():
     assert out[0]["max_score"] == 1.5
     assert out[0]["items"][0]["score"] == 1.5
     assert out[0]["items"][0]["fields"]["_id"] == b"k1"
+
+
+def test_codec_fuzz_no_crash(L):
+    """Mutated/truncated buffers must never crash the parsers — they
+    either fail cleanly or produce bounded output (the Go side can feed
+    arbitrary bytes across the C ABI)."""
+    import random
+
+    rng = random.Random(1234)
+    q = np.arange(64, dtype=np.float32)
+    pb = bytearray(proto.encode_search_request(
+        "emb", q.tobytes(), topn=5, req_num=1,
+        index_params='{"nprobe": 8}', term_filters=[("t", b"x")],
+        range_filters=[("r", b"\x01\x00\x00\x00", b"\x05\x00\x00\x00",
+                        True, True)]))
+    fb = bytearray(fbsenc.build_table(
+        "s", [("tag", fbsenc.DATA_STRING)], "v", 32, "IVFPQ", "{}"))
+    doc = bytearray(fbsenc.build_doc(
+        [("_id", b"k", fbsenc.DATA_STRING),
+         ("v", np.arange(32, dtype=np.float32).tobytes(),
+          fbsenc.DATA_VECTOR)]))
+    L.GammaTestParseSearchRequest.argtypes = [
+        ctypes.c_char_p, ctypes.c_int, ctypes.POINTER(ctypes.c_char_p),
+        ctypes.POINTER(ctypes.c_int)]
+    L.GammaTestParseTable.argtypes = L.GammaTestParseSearchRequest.argtypes
+    L.GammaTestDocRoundtrip.argtypes = L.GammaTestParseSearchRequest.argtypes
+
+    def hammer(fn, base):
+        for _ in range(400):
+            b = bytearray(base)
+            for _ in range(rng.randint(1, 6)):
+                b[rng.randrange(len(b))] = rng.randrange(256)
+            cut = rng.randint(0, len(b))
+            buf = bytes(b[:cut]) if rng.random() < 0.3 else bytes(b)
+            out = ctypes.c_char_p()
+            n = ctypes.c_int()
+            fn(buf, len(buf), ctypes.byref(out), ctypes.byref(n))
+
+    hammer(L.GammaTestParseSearchRequest, pb)
+    hammer(L.GammaTestParseTable, fb)
+    hammer(L.GammaTestDocRoundtrip, doc)

@@ -272,10 +272,13 @@ struct CStatus Search(void *engine, const char *request_str, int req_len,
    * gamma_common_data.h:94) applied on the final top-k */
   bool has_range = vq.has_min || vq.has_max;
 
-  std::vector<gpb::SearchResult> results(nq);
+  /* the response carries req_num results (response.cc:257 loop); the
+   * engine computed nq = bytes/(4d) of them */
+  int nres = req.req_num > 0 ? std::min(req.req_num, nq) : nq;
+  std::vector<gpb::SearchResult> results(nres);
   int64_t total = e->num_docs() - e->bitmap().popcount();
   bool want_vec = req.is_vector_value;
-  for (int i = 0; i < nq; i++) {
+  for (int i = 0; i < nres; i++) {
     gpb::SearchResult &res = results[i];
     res.total = (int)total;
     for (int j = req.offset; j < k; j++) {

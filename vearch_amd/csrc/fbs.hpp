@@ -28,7 +28,7 @@ class Reader {
 
   /* absolute position of field `id` inside table at `tpos`, 0 if absent */
   uint32_t field(uint32_t tpos, int id) const {
-    if (tpos + 4 > n_) return 0;
+    if ((size_t)tpos + 4 > n_) return 0;
     int32_t soff;
     memcpy(&soff, b_ + tpos, 4);
     int64_t vpos = (int64_t)tpos - soff;
@@ -47,26 +47,32 @@ class Reader {
     if (!fpos) return "";
     uint32_t s = indirect(fpos);
     uint32_t len = rd32(s);
-    if (s + 4 + len > n_) return "";
+    if ((size_t)s + 4 + (size_t)len > n_) return "";
     return std::string((const char *)b_ + s + 4, len);
   }
   uint32_t vec_len(uint32_t fpos) const {
     if (!fpos) return 0;
-    return rd32(indirect(fpos));
+    uint32_t len = rd32(indirect(fpos));
+    /* bounds cap: a corrupt buffer cannot claim more elements than
+     * bytes (elements are at least 1 byte / 4-byte offsets) */
+    return (size_t)len > n_ ? 0 : len;
   }
   /* absolute position of vector element i (elem_size bytes or offsets) */
   uint32_t vec_elem(uint32_t fpos, uint32_t i, uint32_t elem_size) const {
     uint32_t v = indirect(fpos);
     return v + 4 + i * elem_size;
   }
-  /* table-vector element: follow the stored uoffset */
+  /* table-vector element: follow the stored uoffset (0 if out of
+   * bounds — field() then rejects it) */
   uint32_t vec_table(uint32_t fpos, uint32_t i) const {
     uint32_t e = vec_elem(fpos, i, 4);
-    return indirect(e);
+    if ((size_t)e + 4 > n_) return 0;
+    uint32_t t = indirect(e);
+    return (size_t)t + 4 > n_ ? 0 : t;
   }
   template <class T>
   T scalar(uint32_t fpos, T deflt) const {
-    if (!fpos || fpos + sizeof(T) > n_) return deflt;
+    if (!fpos || (size_t)fpos + sizeof(T) > n_) return deflt;
     T v;
     memcpy(&v, b_ + fpos, sizeof(T));
     return v;
@@ -76,19 +82,19 @@ class Reader {
     if (!fpos) return nullptr;
     uint32_t v = indirect(fpos);
     *len = rd32(v);
-    if (v + 4 + *len > n_) { *len = 0; return nullptr; }
+    if ((size_t)v + 4 + (size_t)*len > n_) { *len = 0; return nullptr; }
     return b_ + v + 4;
   }
 
  private:
   uint16_t rd16(uint32_t p) const {
     uint16_t v = 0;
-    if (p + 2 <= n_) memcpy(&v, b_ + p, 2);
+    if ((size_t)p + 2 <= n_) memcpy(&v, b_ + p, 2);
     return v;
   }
   uint32_t rd32(uint32_t p) const {
     uint32_t v = 0;
-    if (p + 4 <= n_) memcpy(&v, b_ + p, 4);
+    if ((size_t)p + 4 <= n_) memcpy(&v, b_ + p, 4);
     return v;
   }
   const uint8_t *b_;
@@ -175,10 +181,10 @@ struct TableSchema { /* table.fbs Table: name(0), fields(1),
 
   static std::string str_direct(const void *buf, size_t n, uint32_t s) {
     const uint8_t *b = (const uint8_t *)buf;
-    if (s + 4 > n) return "";
+    if ((size_t)s + 4 > n) return "";
     uint32_t len;
     memcpy(&len, b + s, 4);
-    if (s + 4 + len > n) return "";
+    if ((size_t)s + 4 + (size_t)len > n) return "";
     return std::string((const char *)b + s + 4, len);
   }
 };

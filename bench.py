@@ -45,6 +45,11 @@ WORKLOADS = {
     "ivfflat_d128_n10m": dict(
         kind="IVFFLAT", d=128, n=10_000_000, nlist=4096, m=0, nprobe=32,
         nq=10_000, k=10, rerank=0, train_n=160_000),
+    # configs[4] shape (d=768, m=96) at single-GPU scale; the 8-GPU
+    # N=100M version is this sharded 8 ways by the driver's SCALE run
+    "ivfpq_d768_n2m": dict(
+        kind="IVFPQ", d=768, n=2_000_000, nlist=4096, m=96, nprobe=32,
+        nq=2_000, k=10, rerank=200, train_n=160_000),
 }
 
 HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)

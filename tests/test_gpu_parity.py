@@ -575,3 +575,35 @@ def test_concurrent_search_and_mutation(data):
     bad = [v for v in live.tolist() if v < 2000 and v % 3 == 0]
     assert not bad
     eng.close()
+
+
+def test_ivfpq_d768_m96_bitexact():
+    """Config-5 shape (d=768, m=96, dsub=8): the wide-M template path
+    stays bit-exact vs the oracle."""
+    base = orc.gen_clustered(12000, 768, seed=9, ncl=64)
+    q = orc.gen_queries(base, 16, seed=10)
+    eng = make_engine("/tmp/gamma_d768")
+    eng.create_table(
+        768, "IVFPQ",
+        '{"ncentroids": 32, "nsubvector": 96, "metric_type": "L2", '
+        '"training_threshold": 6000}')
+    eng.add(base)
+    eng.build_index()
+    gd, gi = eng.raw_search(q, 10, nprobe=8)
+    ox = orc.OracleIVFPQ(768, 32, 96, metric="L2")
+    cent, books = eng.debug_model(32, 768, 96)
+    ox.centroids, ox.codebooks = cent, books
+    ids_all, codes_all, offsets = [], [], [0]
+    for ln in range(32):
+        li, lc = eng.debug_list(ln, 96)
+        ids_all.append(li)
+        codes_all.append(lc)
+        offsets.append(offsets[-1] + len(li))
+    ox.ids = np.concatenate(ids_all)
+    ox.codes = np.concatenate(codes_all)
+    ox.offsets = np.array(offsets, dtype=np.int64)
+    pdists, probes = eng.debug_coarse_assign(q, 8)
+    od, oi = ox.search_pct1(q, 10, 8, probes=probes, probe_dists=pdists)
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    eng.close()

@@ -438,7 +438,7 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
    * gather stream wants ~4 waves/SIMD — microarch §LDS). */
   const int BS = 512;
   bool fast = (k2 + BS * GAMMA_ADC_C) <= GAMMA_SORT_CAP &&
-              (M == 16 || M == 32 || M == 64);
+              (M == 16 || M == 32 || M == 64 || M == 96);
   dim3 g(nq);
 #define GAMMA_LAUNCH_SCAN(IPV, MWV, BSV)                                  \
   k_ivfpq_scan<IPV, MWV, BSV><<<g, dim3(BSV), smem, s>>>(                 \
@@ -448,12 +448,14 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
     if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG);
     else if (M == 16) GAMMA_LAUNCH_SCAN(true, 4, 512);
     else if (M == 32) GAMMA_LAUNCH_SCAN(true, 8, 512);
-    else GAMMA_LAUNCH_SCAN(true, 16, 512);
+    else if (M == 64) GAMMA_LAUNCH_SCAN(true, 16, 512);
+    else GAMMA_LAUNCH_SCAN(true, 24, 512);
   } else {
     if (!fast) GAMMA_LAUNCH_SCAN(false, 0, WG);
     else if (M == 16) GAMMA_LAUNCH_SCAN(false, 4, 512);
     else if (M == 32) GAMMA_LAUNCH_SCAN(false, 8, 512);
-    else GAMMA_LAUNCH_SCAN(false, 16, 512);
+    else if (M == 64) GAMMA_LAUNCH_SCAN(false, 16, 512);
+    else GAMMA_LAUNCH_SCAN(false, 24, 512);
   }
 #undef GAMMA_LAUNCH_SCAN
   return hipGetLastError();

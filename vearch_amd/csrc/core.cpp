@@ -705,10 +705,12 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
   if (scratch_pdists_.reserve((size_t)nq * nprobe * 4)) return -1;
   if (update_dev_buckets(s)) return -1;
 
-  hipEvent_t e0, e1, e2;
-  hipEventCreate(&e0);
-  hipEventCreate(&e1);
-  hipEventCreate(&e2);
+  struct Ev3 { /* RAII so error paths cannot leak events */
+    hipEvent_t e[3];
+    Ev3() { for (auto &x : e) (void)hipEventCreate(&x); }
+    ~Ev3() { for (auto &x : e) (void)hipEventDestroy(x); }
+  } ev;
+  hipEvent_t e0 = ev.e[0], e1 = ev.e[1], e2 = ev.e[2];
   hipEventRecord(e0, s);
   if (coarse_assign(q_dev, nq, nprobe, metric_ip, q_norms_dev, s,
                     scratch_probes_.as<int64_t>(),
@@ -745,9 +747,6 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
   if (t_assign_ms) *t_assign_ms = ms;
   hipEventElapsedTime(&ms, e1, e2);
   if (t_scan_ms) *t_scan_ms = ms;
-  hipEventDestroy(e0);
-  hipEventDestroy(e1);
-  hipEventDestroy(e2);
   return 0;
 }
 

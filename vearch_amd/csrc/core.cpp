@@ -648,8 +648,12 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
       hipMemcpy((uint8_t *)bk.data->get() + (size_t)bk.size * entry,
                 kv.second.second.data(), (size_t)add_n * entry,
                 hipMemcpyHostToDevice);
-      for (int64_t i = 0; i < add_n; i++)
-        vid_loc_[kv.second.first[i]] = {kv.first, bk.size + i};
+      for (int64_t i = 0; i < add_n; i++) {
+        int64_t vid = (int64_t)kv.second.first[i];
+        if (vid >= (int64_t)vid_loc_.size())
+          vid_loc_.resize((size_t)vid + 1024, -1);
+        vid_loc_[vid] = ((int64_t)kv.first << 40) | (bk.size + i);
+      }
       bk.size += add_n;
       dev_buckets_dirty_ = true;
     }
@@ -659,12 +663,14 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
 }
 
 int IVFIndex::del(int64_t vid, hipStream_t s) {
-  auto it = vid_loc_.find(vid);
-  if (it == vid_loc_.end()) return 0;
-  Bucket &bk = buckets_[it->second.first];
+  if (vid < 0 || vid >= (int64_t)vid_loc_.size() || vid_loc_[vid] < 0)
+    return 0;
+  int32_t bno = (int32_t)(vid_loc_[vid] >> 40);
+  long long pos = vid_loc_[vid] & (((int64_t)1 << 40) - 1);
+  Bucket &bk = buckets_[bno];
   /* device form: bit 31 = the kDelIdxMask bit-63 mark */
   uint32_t marked = (uint32_t)vid | 0x80000000u;
-  if (hipMemcpy(bk.ids->as<uint32_t>() + it->second.second, &marked, 4,
+  if (hipMemcpy(bk.ids->as<uint32_t>() + pos, &marked, 4,
                 hipMemcpyHostToDevice) != hipSuccess)
     return -1;
   return 0;
@@ -879,7 +885,11 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
       bk.size = bk.cap = sz;
       for (long long j = 0; j < sz; j++) {
         int64_t vid = ids[j];
-        if (!((uint64_t)vid >> 63)) vid_loc_[vid] = {i, j};
+        if (!((uint64_t)vid >> 63)) {
+          if (vid >= (int64_t)vid_loc_.size())
+            vid_loc_.resize((size_t)vid + 1024, -1);
+          vid_loc_[vid] = ((int64_t)i << 40) | j;
+        }
       }
     }
   }

@@ -191,7 +191,8 @@ class IVFIndex {
   std::vector<Bucket> buckets_;
   DeviceBuf dev_buckets_; /* GammaBucketDev[nlist] */
   bool dev_buckets_dirty_ = true;
-  std::unordered_map<int64_t, std::pair<int32_t, long long>> vid_loc_;
+  /* vid -> (bucket<<40 | pos); -1 = absent. Dense (vids are 0..N). */
+  std::vector<int64_t> vid_loc_;
   /* scratch for search (grow-only) */
   DeviceBuf scratch_dots_, scratch_keys_, scratch_probes_, scratch_pdists_;
   DeviceBuf scratch_i32_, scratch_f32_;

@@ -168,6 +168,9 @@ def main():
                     help="override DB size (N per GPU)")
     ap.add_argument("--queries", type=int, default=0,
                     help="override nq per step")
+    ap.add_argument("--nprobe", type=int, default=0)
+    ap.add_argument("--rerank", type=int, default=-1,
+                    help="recall_num (-1 = workload default)")
     ap.add_argument("--skip-recall", action="store_true")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--index-dir", default="",
@@ -183,6 +186,10 @@ def main():
         cfg["train_n"] = min(cfg["train_n"], cfg["n"])
     if args.queries:
         cfg["nq"] = args.queries
+    if args.nprobe:
+        cfg["nprobe"] = args.nprobe
+    if args.rerank >= 0:
+        cfg["rerank"] = args.rerank
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))

@@ -607,3 +607,42 @@ def test_ivfpq_d768_m96_bitexact():
     assert np.array_equal(gi, oi)
     assert np.array_equal(gd, od)
     eng.close()
+
+
+def test_ivfflat_ip_bitexact(data):
+    """IVFFLAT + InnerProduct (spherical k-means assign, descending
+    similarity) — canonical re-rank makes returned scores bit-exact."""
+    base, q = data
+    eng = make_engine("/tmp/gamma_ivfflat_ip")
+    eng.create_table(
+        64, "IVFFLAT",
+        '{"ncentroids": 64, "metric_type": "InnerProduct", '
+        '"training_threshold": 8000}')
+    eng.add(base)
+    eng.build_index()
+    gd, gi = eng.raw_search(q, 10, nprobe=16)
+    cent, _ = eng.debug_model(64, 64, 0)
+    ids_all, vec_all, offsets = [], [], [0]
+    for ln in range(64):
+        li, lc = eng.debug_list(ln, 64 * 4)
+        ids_all.append(li)
+        vec_all.append(lc.view(np.float32).reshape(-1, 64))
+        offsets.append(offsets[-1] + len(li))
+    ids = np.concatenate(ids_all)
+    vecs = np.ascontiguousarray(np.concatenate(vec_all), dtype=np.float32)
+    offsets = np.array(offsets, dtype=np.int64)
+    _, probes = eng.debug_coarse_assign(q, 16)
+    from oracle.gamma_oracle import _ip64, _up8
+    lib = RefLib.lib()
+    od = np.empty((64, 10), dtype=np.float32)
+    oi = np.empty((64, 10), dtype=np.int64)
+    lib.oracle_ivfflat_search(
+        64, 64, 64, _fp(_c(q, np.float32)), _ip64(offsets), _ip64(ids),
+        _fp(vecs), 16, _ip64(_c(probes, np.int64)), _up8(None), 1, 10,
+        _fp(od), _ip64(oi))
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    for t in range(q.shape[0]):
+        row = [gd[t, j] for j in range(10) if gi[t, j] >= 0]
+        assert row == sorted(row, reverse=True)
+    eng.close()

@@ -8,10 +8,14 @@ queries = perturbed DB samples. One "step" = one batched search of nq
 queries with the queries already resident in HBM (GammaCacheQueries).
 
 Multi-GPU (launched by the driver via torch.distributed.run): one engine
-(= one Vearch partition) per rank/GPU, per-GPU N fixed (weak scaling);
-per step every rank searches its partition and ONE RCCL all-gather over
-xGMI moves the per-partition top-k to rank 0, which merges with the
-router's semantics (client.go:1497/1558).
+(= one Vearch partition) per rank/GPU; the N-vector table is sharded
+round-robin over the partitions exactly as Vearch splits a space
+(STRONG scaling — the metric "N=10M ... 1/2/4/8 MI355X" holds N fixed,
+and BASELINE's >=6x-at-8-GPUs target is aggregate QPS on the same
+table). Per step every rank searches its shard and ONE RCCL all-gather
+over xGMI moves the per-partition top-k (packed (dist,id) keys) to
+rank 0, which merges with the router's semantics (client.go:1497/1558)
+by one sort over world*k keys.
 
 Prints ONE JSON line from rank 0 (driver contract), including the
 roofline of the dominant kernel (the ADC list scan) and the CPU-oracle
@@ -206,6 +210,8 @@ def main():
     import torch
 
     idir = args.index_dir
+    if idir and world > 1:
+        idir = os.path.join(idir, f"r{rank}")
     have_dump = bool(idir) and os.path.exists(
         os.path.join(idir, "gamma.dump")) and os.path.exists(
         os.path.join(idir, "queries.npy"))
@@ -230,21 +236,19 @@ def main():
         log(f"loaded index ({eng.num_docs()} docs) + queries in "
             f"{time.time()-t0:.1f}s")
     else:
-        # data: each rank owns its partition (disjoint synthetic shards)
+        # one table of N vectors, sharded round-robin over partitions
+        # (shard-local docid * world + rank = global docid); every rank
+        # derives the identical query set from the same seeds
         t0 = time.time()
-        base = gen_data(cfg["n"], cfg["d"], seed=42 + rank)
+        base_full = gen_data(cfg["n"], cfg["d"], seed=42)
         rng = np.random.default_rng(43)
         idx = rng.integers(0, cfg["n"], size=cfg["nq"])
-        queries = (base[idx] + 0.05 * rng.standard_normal(
-            (cfg["nq"], cfg["d"]), dtype=np.float32)) if rank == 0 else None
-        if world > 1:
-            # same queries everywhere: broadcast from rank 0
-            qt = torch.from_numpy(queries if rank == 0 else
-                                  np.empty((cfg["nq"], cfg["d"]),
-                                           np.float32))
-            dist.broadcast(qt, src=0)
-            queries = qt.numpy()
-        log(f"data gen in {time.time()-t0:.1f}s")
+        queries = base_full[idx] + 0.05 * rng.standard_normal(
+            (cfg["nq"], cfg["d"]), dtype=np.float32)
+        base = np.ascontiguousarray(base_full[rank::world])             if world > 1 else base_full
+        del base_full
+        log(f"data gen in {time.time()-t0:.1f}s "
+            f"(shard {base.shape[0]} of {cfg['n']})")
         if idir:
             os.makedirs(idir, exist_ok=True)
         eng = build_engine(cfg, base, rank, path=idir or None)
@@ -283,24 +287,25 @@ def main():
         return eng.search_cached(nq, cfg["k"], nprobe=cfg["nprobe"],
                                  rerank=cfg["rerank"])
 
+    from vearch_amd.merge import pack_keys_signed, unpack_keys_signed
+
     def gather_and_merge(dists, ids):
         if world == 1:
             return dists, ids
-        # globalize ids: rank-local docid + rank * n; ONE all-gather over
-        # xGMI of the packed (dist-key, id) u64 blocks (~nq*k*8 B/rank)
-        gids = np.where(ids >= 0, ids + rank * cfg["n"], -1)
-        payload = torch.from_numpy(np.stack(
-            [dists.view(np.int32).astype(np.int64), gids], axis=2))
+        # globalize round-robin shard ids, pack (dist,id) into signed
+        # int64 keys, ONE all-gather over xGMI (~nq*k*8 B per rank),
+        # merge = one torch sort over world*k keys (client.go:1497
+        # semantics; parity-tested vs merge_topk)
+        gids = np.where(ids >= 0, ids * world + rank, -1)
+        keys = torch.from_numpy(pack_keys_signed(dists, gids))
         if backend == "nccl":
-            payload = payload.cuda()
-        out = [torch.empty_like(payload) for _ in range(world)]
-        dist.all_gather(out, payload)
+            keys = keys.cuda()
+        out = [torch.empty_like(keys) for _ in range(world)]
+        dist.all_gather(out, keys)
         if rank == 0:
-            from vearch_amd.merge import merge_topk
-            dl = [o[:, :, 0].cpu().numpy().astype(np.int32)
-                  .view(np.float32) for o in out]
-            il = [o[:, :, 1].cpu().numpy() for o in out]
-            return merge_topk(dl, il, cfg["k"])
+            merged = torch.sort(torch.cat(out, dim=1),
+                                dim=1).values[:, :cfg["k"]]
+            return unpack_keys_signed(merged.cpu().numpy())
         return None, None
 
     # warmup
@@ -365,7 +370,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "strong",
             "vs_baseline": None,
             "dtype": "f32",
             "data": "synthetic",
@@ -373,7 +378,8 @@ def main():
                 "workload": args.workload,
                 "kind": cfg["kind"],
                 "d": cfg["d"],
-                "n_per_gpu": cfg["n"],
+                "n_total": cfg["n"],
+                "n_per_gpu": cfg["n"] // world,
                 "nlist": cfg["nlist"],
                 "m": cfg["m"],
                 "nprobe": cfg["nprobe"],

@@ -66,3 +66,27 @@ def test_merge_topk_ties_and_padding():
     assert mi[0].tolist() == [4, 7, 3, 9]  # tie 0.5 -> lower id first
     md, mi = merge_topk([d1, d2], [i1, i2], 6)
     assert mi[0].tolist() == [4, 7, 3, 9, -1, -1]
+
+
+def test_packed_key_merge_matches_reference_merge():
+    """The transport form bench.py uses (signed-int64 packed keys sorted
+    by torch) must merge identically to merge_topk."""
+    import numpy as np
+    from vearch_amd.merge import (merge_topk, pack_keys_signed,
+                                  unpack_keys_signed)
+    rng = np.random.default_rng(3)
+    for descending in (False, True):
+        dl = [rng.standard_normal((50, 7)).astype(np.float32)
+              for _ in range(3)]
+        il = [rng.integers(0, 10**6, (50, 7)).astype(np.int64)
+              for _ in range(3)]
+        il[1][:, 5:] = -1  # padding slots
+        want_d, want_i = merge_topk(dl, il, 9, descending=descending)
+        keys = np.concatenate(
+            [pack_keys_signed(d, i, descending) for d, i in zip(dl, il)],
+            axis=1)
+        t = torch.from_numpy(keys)
+        merged = torch.sort(t, dim=1).values[:, :9].numpy()
+        got_d, got_i = unpack_keys_signed(merged, descending)
+        assert np.array_equal(got_i, want_i)
+        assert np.array_equal(got_d, want_d)

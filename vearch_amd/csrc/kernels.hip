@@ -789,7 +789,19 @@ __global__ void k_pq_btable(int d, int M, int nlist,
   const float *cm = centroids + ln * d + m * dsub;
   const float *cw = codebooks + ((size_t)m * 256 + j) * dsub;
   float dot = 0.0f;
-  for (int t = 0; t < dsub; t++) dot = fmaf(cm[t], cw[t], dot);
+  if ((dsub & 3) == 0) {
+#pragma unroll 2
+    for (int t = 0; t < dsub; t += 4) {
+      float4 c4 = *(const float4 *)(cw + t);
+      float4 m4 = *(const float4 *)(cm + t);
+      dot = fmaf(m4.x, c4.x, dot);
+      dot = fmaf(m4.y, c4.y, dot);
+      dot = fmaf(m4.z, c4.z, dot);
+      dot = fmaf(m4.w, c4.w, dot);
+    }
+  } else {
+    for (int t = 0; t < dsub; t++) dot = fmaf(cm[t], cw[t], dot);
+  }
   btab[e] = 2.0f * dot;
 }
 
@@ -816,8 +828,24 @@ __global__ void k_pq_atable(int nq, int d, int M,
   const float *qm = queries + q * d + m * dsub;
   const float *cw = codebooks + ((size_t)m * 256 + j) * dsub;
   float cwn = 0.0f, dot = 0.0f;
-  for (int t = 0; t < dsub; t++) cwn = fmaf(cw[t], cw[t], cwn);
-  for (int t = 0; t < dsub; t++) dot = fmaf(qm[t], cw[t], dot);
+  if ((dsub & 3) == 0) { /* float4 loads, same fmaf order */
+#pragma unroll 2
+    for (int t = 0; t < dsub; t += 4) {
+      float4 c4 = *(const float4 *)(cw + t);
+      float4 q4 = *(const float4 *)(qm + t);
+      cwn = fmaf(c4.x, c4.x, cwn);
+      cwn = fmaf(c4.y, c4.y, cwn);
+      cwn = fmaf(c4.z, c4.z, cwn);
+      cwn = fmaf(c4.w, c4.w, cwn);
+      dot = fmaf(q4.x, c4.x, dot);
+      dot = fmaf(q4.y, c4.y, dot);
+      dot = fmaf(q4.z, c4.z, dot);
+      dot = fmaf(q4.w, c4.w, dot);
+    }
+  } else {
+    for (int t = 0; t < dsub; t++) cwn = fmaf(cw[t], cw[t], cwn);
+    for (int t = 0; t < dsub; t++) dot = fmaf(qm[t], cw[t], dot);
+  }
   atab[e] = fmaf(-2.0f, dot, cwn);
 }
 

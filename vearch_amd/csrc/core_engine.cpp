@@ -616,6 +616,19 @@ int Engine::load(std::string *err) {
   fread(&dim, 4, 1, f);
   fread(&tt, 4, 1, f);
   fread(&maxdoc, 8, 1, f);
+  if (table_created_ && dim != dim_) {
+    fclose(f);
+    if (err)
+      *err = "dump dimension " + std::to_string(dim) +
+             " != table dimension " + std::to_string(dim_);
+    return -1;
+  }
+  if (table_created_ && idx_type != index_type_) {
+    fclose(f);
+    if (err)
+      *err = "dump index type " + idx_type + " != table " + index_type_;
+    return -1;
+  }
   int nfields = 0;
   fread(&nfields, 4, 1, f);
   std::vector<FieldMeta> fms(nfields);

@@ -646,3 +646,48 @@ def test_ivfflat_ip_bitexact(data):
         row = [gd[t, j] for j in range(10) if gi[t, j] >= 0]
         assert row == sorted(row, reverse=True)
     eng.close()
+
+
+def test_flat_gemm_path_ip_bitexact():
+    """IP ordering through the chunked GEMM + seeded select path
+    (nq >= 512, N >= 200k)."""
+    base = orc.gen_clustered(210000, 32, seed=15, ncl=300)
+    q = orc.gen_queries(base, 520, seed=16)
+    eng = make_engine("/tmp/gamma_flat_gemm_ip")
+    eng.create_table(32, "FLAT", '{"metric_type": "InnerProduct"}')
+    eng.add(base)
+    gd, gi = eng.raw_search(q, 10)
+    od, oi = orc.flat_search(base, q, 10, "InnerProduct")
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    eng.close()
+
+
+def test_cached_queries_equivalent(data, ivfpq_engine):
+    """GammaCacheQueries + GammaRawSearchCached (the bench hot path) must
+    equal the host-pointer search exactly."""
+    base, q = data
+    eng = ivfpq_engine
+    gd0, gi0 = eng.raw_search(q, 10, nprobe=16, rerank=100)
+    nq = eng.cache_queries(q)
+    gd1, gi1 = eng.search_cached(nq, 10, nprobe=16, rerank=100)
+    assert np.array_equal(gi0, gi1)
+    assert np.array_equal(gd0, gd1)
+
+
+def test_metric_override_per_query(data, ivfpq_engine):
+    """Per-request metric override (retrieval params metric_type,
+    ivfpq.cc:247-259): IP search on an L2-trained index runs the IP
+    table path and orders descending."""
+    base, q = data
+    eng = ivfpq_engine
+    gd, gi = eng.raw_search(q[:8], 5, nprobe=16, metric=2)
+    for t in range(8):
+        row = [gd[t, j] for j in range(5) if gi[t, j] >= 0]
+        assert row == sorted(row, reverse=True)
+    # exact IP values after canonical rerank? no rerank requested ->
+    # ADC-IP values; check ordering + plausibility vs exact IP top-1
+    odf, oif = orc.flat_search(base, q[:8], 50, "InnerProduct")
+    overlap = np.mean([len(set(gi[t].tolist()) & set(oif[t].tolist())) / 5
+                       for t in range(8)])
+    assert overlap >= 0.4, overlap

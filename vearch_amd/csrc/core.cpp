@@ -45,9 +45,17 @@ int DeviceBuf::reserve(size_t bytes) {
   return 0;
 }
 void DeviceBuf::free() {
-  if (p_) hipFree(p_);
+  if (p_) (void)hipFree(p_);
   p_ = nullptr;
   bytes_ = 0;
+}
+
+void DeviceBuf::steal(DeviceBuf &other) {
+  free();
+  p_ = other.p_;
+  bytes_ = other.bytes_;
+  other.p_ = nullptr;
+  other.bytes_ = 0;
 }
 
 /* ---------------------------------------------------------------- RawStore */
@@ -80,16 +88,9 @@ int RawStore::ensure_capacity(int64_t n_new, hipStream_t s) {
     DeviceBuf nb;
     if (nb.reserve((size_t)cap * 4)) return -1;
     if (norms_cap_ > 0)
-      hipMemcpy(nb.get(), norms_.get(), (size_t)n_ * 4,
-                hipMemcpyDeviceToDevice);
-    /* steal */
-    norms_.free();
-    if (norms_.reserve(0)) return -1;
-    /* move nb into norms_: DeviceBuf lacks move; emulate */
-    static_assert(sizeof(DeviceBuf) == sizeof(void *) + sizeof(size_t),
-                  "layout");
-    memcpy((void *)&norms_, (void *)&nb, sizeof(DeviceBuf));
-    memset((void *)&nb, 0, sizeof(DeviceBuf));
+      (void)hipMemcpy(nb.get(), norms_.get(), (size_t)n_ * 4,
+                      hipMemcpyDeviceToDevice);
+    norms_.steal(nb);
     norms_cap_ = cap;
   }
   return 0;
@@ -182,12 +183,11 @@ int Bitmap::ensure(int64_t nbits, hipStream_t s) {
   host_.resize(words, 0);
   DeviceBuf nd;
   if (nd.reserve((size_t)words * 4)) return -1;
-  hipMemset(nd.get(), 0, (size_t)words * 4);
+  (void)hipMemset(nd.get(), 0, (size_t)words * 4);
   if (old_words)
-    hipMemcpy(nd.get(), dev_.get(), old_words * 4, hipMemcpyDeviceToDevice);
-  dev_.free();
-  memcpy((void *)&dev_, (void *)&nd, sizeof(DeviceBuf));
-  memset((void *)&nd, 0, sizeof(DeviceBuf));
+    (void)hipMemcpy(nd.get(), dev_.get(), old_words * 4,
+                    hipMemcpyDeviceToDevice);
+  dev_.steal(nd);
   bits_ = words * 32;
   return 0;
 }

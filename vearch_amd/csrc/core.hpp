@@ -58,6 +58,8 @@ class DeviceBuf {
   DeviceBuf &operator=(const DeviceBuf &) = delete;
   int reserve(size_t bytes);         /* grow-only */
   void free();
+  /* take other's allocation (frees our own); other becomes empty */
+  void steal(DeviceBuf &other);
   void *get() const { return p_; }
   template <class T> T *as() const { return (T *)p_; }
   size_t bytes() const { return bytes_; }

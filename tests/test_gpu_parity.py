@@ -691,3 +691,19 @@ def test_metric_override_per_query(data, ivfpq_engine):
     overlap = np.mean([len(set(gi[t].tolist()) & set(oif[t].tolist())) / 5
                        for t in range(8)])
     assert overlap >= 0.4, overlap
+
+
+def test_search_deterministic_across_runs(data, ivfpq_engine):
+    """The selector's LDS append order is nondeterministic; the exact
+    (dist,id) total order must make results run-invariant anyway."""
+    base, q = data
+    eng = ivfpq_engine
+    runs = [eng.raw_search(q, 10, nprobe=16, rerank=100)
+            for _ in range(3)]
+    for gd, gi in runs[1:]:
+        assert np.array_equal(gi, runs[0][1])
+        assert np.array_equal(gd, runs[0][0])
+    runs = [eng.raw_search(q, 10, nprobe=16) for _ in range(3)]
+    for gd, gi in runs[1:]:
+        assert np.array_equal(gi, runs[0][1])
+        assert np.array_equal(gd, runs[0][0])

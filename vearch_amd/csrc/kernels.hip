@@ -261,16 +261,15 @@ hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
 }
 
 /* ------------------------------------------------------- IVFPQ fused scan
- * One workgroup per query (grid fills 256 CUs at nq >= ~512; at the
- * north-star nq=10k it is ~40 rounds of blocks per CU).
- * L2 by-residual (use_precomputed_table=0, ivfpq.cc:196): per probed list
- * build T[m][j] = ||r_m - cw_mj||^2 in LDS, dis0 = 0.
- * IP: one query-level table (h:164-167), dis0 = dot(q, c_list) (h:223-236,
- * canonical order on one thread).
- * Scan (h:923-953): skip bit-63-deleted ids and bitmap-deleted docs,
+ * One workgroup per query.
+ * L2: per probed list stage T = A_q + B_list in LDS (the decomposed
+ * use_precomputed_table=1 tables, ivfpq.h:254-262) with dis0 = the
+ * coarse probe distance; IP: one query-level table (h:164-167) with
+ * dis0 = dot(q, c_list) in canonical order.
+ * Scan (h:923-953): skip delete-marked ids and bitmap-deleted docs,
  * dis = dis0 + sum_m T[m][code_m] in ascending m (plain adds — matches
- * oracle_adc_scan_list bit-for-bit). */
-/* MW = M/4 compile-time (0 = generic runtime-M path). The templated path
+ * oracle bit-for-bit, see oracle_ivfpq_search_pct1).
+ * MW = M/4 compile-time (0 = generic runtime-M path). The templated path
  * stages GAMMA_ADC_C codes per thread in registers with all global loads
  * issued before any use, so one barrier interval carries
  * blockDim*C codes' worth of HBM latency instead of blockDim's. */
@@ -357,7 +356,7 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
       for (long long j0 = 0; j0 < bk.size;
            j0 += (long long)blockDim.x * C) {
         long long jb = j0 + (long long)threadIdx.x * C;
-        uint32_t w[C][MW ? MW : 1];
+        uint32_t w[C][MW ? MW : 1]; /* compile-time bounds -> registers */
         int64_t idv[C];
 #pragma unroll
         for (int c = 0; c < C; c++) {

@@ -287,19 +287,20 @@ def main():
         return eng.search_cached(nq, cfg["k"], nprobe=cfg["nprobe"],
                                  rerank=cfg["rerank"])
 
-    from vearch_amd.merge import pack_keys_signed, unpack_keys_signed
+    from vearch_amd.merge import pack_keys_signed_torch, unpack_keys_signed
 
     def gather_and_merge(dists, ids):
         if world == 1:
             return dists, ids
-        # globalize round-robin shard ids, pack (dist,id) into signed
-        # int64 keys, ONE all-gather over xGMI (~nq*k*8 B per rank),
-        # merge = one torch sort over world*k keys (client.go:1497
-        # semantics; parity-tested vs merge_topk)
-        gids = np.where(ids >= 0, ids * world + rank, -1)
-        keys = torch.from_numpy(pack_keys_signed(dists, gids))
+        # globalize round-robin shard ids + pack (dist,id) into signed
+        # int64 keys on the GPU, ONE all-gather over xGMI (~nq*k*8 B per
+        # rank), merge = one torch sort over world*k keys
+        # (client.go:1497 semantics; parity-tested vs merge_topk)
+        d_t = torch.from_numpy(dists)
+        i_t = torch.from_numpy(ids)
         if backend == "nccl":
-            keys = keys.cuda()
+            d_t, i_t = d_t.cuda(), i_t.cuda()
+        keys = pack_keys_signed_torch(d_t, i_t, world, rank)
         out = [torch.empty_like(keys) for _ in range(world)]
         dist.all_gather(out, keys)
         if rank == 0:

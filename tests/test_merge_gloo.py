@@ -90,3 +90,19 @@ def test_packed_key_merge_matches_reference_merge():
         got_d, got_i = unpack_keys_signed(merged, descending)
         assert np.array_equal(got_i, want_i)
         assert np.array_equal(got_d, want_d)
+
+
+def test_torch_pack_matches_numpy_pack():
+    import numpy as np
+    from vearch_amd.merge import pack_keys_signed, pack_keys_signed_torch
+    rng = np.random.default_rng(7)
+    d = rng.standard_normal((20, 5)).astype(np.float32)
+    i = rng.integers(0, 10**6, (20, 5)).astype(np.int64)
+    i[3, 2:] = -1
+    for world, rank in ((1, 0), (4, 3)):
+        gids = np.where(i >= 0, i * world + rank, -1)
+        want = pack_keys_signed(d, gids)
+        got = pack_keys_signed_torch(torch.from_numpy(d),
+                                     torch.from_numpy(i), world,
+                                     rank).numpy()
+        assert np.array_equal(want, got)

@@ -682,20 +682,32 @@ int IVFIndex::coarse_assign(const float *q_dev, int nq, int nprobe, bool ip,
   const int64_t sub = 16384;
   if (scratch_dots_.reserve((size_t)std::min<int64_t>(nq, sub) * nlist_ * 4))
     return -1;
-  if (scratch_keys_.reserve((size_t)nq * nprobe * 8)) return -1;
+  const bool full_sort = nlist_ <= 8192;
+  if (!full_sort && scratch_keys_.reserve((size_t)nq * nprobe * 8))
+    return -1;
   for (int64_t r0 = 0; r0 < nq; r0 += sub) {
     int64_t rn = std::min<int64_t>(sub, nq - r0);
     GAMMA_CHECK(gk::dots_mfma(s, q_dev + (size_t)r0 * d_, (int)rn,
                               centroids_.as<float>(), nlist_, d_,
                               scratch_dots_.as<float>()));
-    GAMMA_CHECK(gk::select_from_dots(
-        s, (int)rn, nlist_, 0, nlist_, scratch_dots_.as<float>(),
-        q_norms_dev + r0, cent_norms_.as<float>(), !ip, ip, nullptr, nprobe,
-        scratch_keys_.as<uint64_t>() + (size_t)r0 * nprobe, false));
+    if (full_sort) {
+      GAMMA_CHECK(gk::select_rows_full(
+          s, (int)rn, nlist_, nlist_, scratch_dots_.as<float>(),
+          q_norms_dev + r0, cent_norms_.as<float>(), !ip, ip, nprobe,
+          probe_dists_dev + (size_t)r0 * nprobe,
+          probes_dev + (size_t)r0 * nprobe));
+    } else {
+      GAMMA_CHECK(gk::select_from_dots(
+          s, (int)rn, nlist_, 0, nlist_, scratch_dots_.as<float>(),
+          q_norms_dev + r0, cent_norms_.as<float>(), !ip, ip, nullptr,
+          nprobe, scratch_keys_.as<uint64_t>() + (size_t)r0 * nprobe,
+          false));
+    }
   }
-  GAMMA_CHECK(gk::unpack_keys(s, (int64_t)nq * nprobe,
-                              scratch_keys_.as<uint64_t>(), ip,
-                              probe_dists_dev, probes_dev));
+  if (!full_sort)
+    GAMMA_CHECK(gk::unpack_keys(s, (int64_t)nq * nprobe,
+                                scratch_keys_.as<uint64_t>(), ip,
+                                probe_dists_dev, probes_dev));
   return 0;
 }
 

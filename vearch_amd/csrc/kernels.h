@@ -37,6 +37,16 @@ hipError_t select_from_dots(hipStream_t s, int nq, int64_t ncols,
                             bool l2, bool ip_order, const uint32_t *bitmap,
                             int k2, uint64_t *state_keys, bool seeded);
 
+/* Full-sort row select for short rows (ncols <= 8192, no bitmap):
+ * one workgroup bitonic-sorts the whole row of (dist,id) keys and emits
+ * the top-k2 as (dists, ids) — used for the coarse top-nprobe
+ * (quantizer->search, ivfpq.cc:595). Same key order as
+ * select_from_dots. */
+hipError_t select_rows_full(hipStream_t s, int nq, int ncols, int64_t ld,
+                            const float *dots, const float *qnorms,
+                            const float *bnorms, bool l2, bool ip_order,
+                            int k2, float *out_dists, int64_t *out_ids);
+
 /* argmin over each row (ties -> lowest col). out int32[nrows]. */
 hipError_t argmin_rows(hipStream_t s, int64_t nrows, int ncols,
                        const float *dots, const float *qnorms,

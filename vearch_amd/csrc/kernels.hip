@@ -231,6 +231,61 @@ hipError_t gk::select_from_dots(hipStream_t s, int nq, int64_t ncols,
   return hipGetLastError();
 }
 
+/* ----------------------------------------------- full-sort row select */
+template <bool IP>
+__global__ void __launch_bounds__(512)
+k_select_rows_full(int nq, int ncols, int64_t ld,
+                   const float *__restrict__ dots,
+                   const float *__restrict__ qnorms,
+                   const float *__restrict__ bnorms, int l2, int k2,
+                   float *__restrict__ out_dists,
+                   int64_t *__restrict__ out_ids) {
+  extern __shared__ char smem[];
+  uint64_t *buf = (uint64_t *)smem;
+  const int q = blockIdx.x;
+  if (q >= nq) return;
+  int n = 1;
+  while (n < ncols || n < k2) n <<= 1;
+  const float *row = dots + (int64_t)q * ld;
+  const float qn = l2 ? qnorms[q] : 0.0f;
+  for (int c = threadIdx.x; c < n; c += blockDim.x) {
+    uint64_t key = GAMMA_KEY_EMPTY;
+    if (c < ncols) {
+      float dot = row[c];
+      float dist = l2 ? fmaf(-2.0f, dot, qn + bnorms[c]) : dot;
+      key = gamma_make_key<IP>(dist, (uint32_t)c);
+    }
+    buf[c] = key;
+  }
+  gamma_bitonic_sort(buf, n);
+  for (int i = threadIdx.x; i < k2; i += blockDim.x) {
+    uint64_t key = buf[i];
+    out_dists[(int64_t)q * k2 + i] =
+        (key == GAMMA_KEY_EMPTY) ? -1.0f : gamma_key_dist<IP>(key);
+    out_ids[(int64_t)q * k2 + i] = gamma_key_id(key);
+  }
+}
+
+hipError_t gk::select_rows_full(hipStream_t s, int nq, int ncols,
+                                int64_t ld, const float *dots,
+                                const float *qnorms, const float *bnorms,
+                                bool l2, bool ip_order, int k2,
+                                float *out_dists, int64_t *out_ids) {
+  int n = 1;
+  while (n < ncols || n < k2) n <<= 1;
+  size_t smem = (size_t)n * 8;
+  if (smem > 64 * 1024) return hipErrorInvalidValue;
+  if (ip_order)
+    k_select_rows_full<true><<<dim3(nq), dim3(512), smem, s>>>(
+        nq, ncols, ld, dots, qnorms, bnorms, l2 ? 1 : 0, k2, out_dists,
+        out_ids);
+  else
+    k_select_rows_full<false><<<dim3(nq), dim3(512), smem, s>>>(
+        nq, ncols, ld, dots, qnorms, bnorms, l2 ? 1 : 0, k2, out_dists,
+        out_ids);
+  return hipGetLastError();
+}
+
 /* --------------------------------------------------------------- argmin */
 __global__ void k_argmin_rows(int64_t nrows, int ncols,
                               const float *__restrict__ dots,

@@ -67,3 +67,22 @@ def unpack_keys_signed(skey, descending=False):
     u = np.where(neg, ~dk, dk & np.uint32(0x7FFFFFFF)).astype(np.uint32)
     dists = np.where(empty, np.float32(-1.0), u.view(np.float32))
     return dists, ids
+
+
+def pack_keys_signed_torch(dists_t, ids_t, world=1, rank=0,
+                           descending=False):
+    """torch version of pack_keys_signed with round-robin id
+    globalization fused (runs on GPU in the multi-rank bench so packing
+    never gates the step). dists_t float32, ids_t int64 tensors."""
+    import torch
+    u = dists_t.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    neg = (u >> 31) != 0
+    dk = torch.where(neg, (~u) & 0xFFFFFFFF, u | 0x80000000)
+    if descending:
+        dk = 0xFFFFFFFF - dk
+    gids = ids_t * world + rank
+    skey = ((dk - (1 << 31)) << 32) | (gids & 0xFFFFFFFF)
+    return torch.where(ids_t < 0,
+                       torch.tensor(torch.iinfo(torch.int64).max,
+                                    dtype=torch.int64,
+                                    device=dists_t.device), skey)

@@ -713,11 +713,22 @@ int IVFIndex::coarse_assign(const float *q_dev, int nq, int nprobe, bool ip,
   return 0;
 }
 
+int IVFIndex::probe_split(int nq, int k2, int nprobe) const {
+  /* small batches underfill the 256 CUs with one WG per query: split
+   * probes across S sub-workgroups (merged by sort_rows; S*k2 <= 2048
+   * keeps the merge a single row sort) */
+  if (params_.kind != IndexKind::IVFPQ) return 1;
+  int S = 1;
+  while (nq * S < 1024 && S < nprobe && (int64_t)(2 * S) * k2 <= 2048)
+    S *= 2;
+  return S;
+}
+
 int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                      const uint32_t *bitmap_dev, bool metric_ip,
                      hipStream_t s, uint64_t *out_keys_dev,
                      const float *q_norms_dev, double *t_assign_ms,
-                     double *t_scan_ms) {
+                     double *t_scan_ms, int S) {
   if (!trained_) return -1;
   nprobe = std::min(nprobe, nlist_);
   if (nprobe > 1024) nprobe = 1024; /* selector cap (select.hpp) */
@@ -746,7 +757,7 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                                   scratch_atab_.as<float>()));
       atab = scratch_atab_.as<float>();
     }
-    GAMMA_CHECK(gk::ivfpq_scan(s, nq, d_, M_, nprobe, k2, q_dev,
+    GAMMA_CHECK(gk::ivfpq_scan(s, nq, S, d_, M_, nprobe, k2, q_dev,
                                centroids_.as<float>(),
                                codebooks_.as<float>(), atab,
                                btable_.as<float>(),

@@ -151,10 +151,13 @@ class IVFIndex {
           hipStream_t s);
   int del(int64_t vid, hipStream_t s); /* set bit 63 in the bucket slot */
   /* search: writes keys (nq x k2) into out_keys (device) */
+  /* S = probe-split (out_keys must hold nq*S*k2 keys); see
+   * probe_split() for the small-batch policy */
   int search(const float *q_dev, int nq, int k2, int nprobe,
              const uint32_t *bitmap_dev, bool metric_ip, hipStream_t s,
              uint64_t *out_keys_dev, const float *q_norms_dev,
-             double *t_assign_ms, double *t_scan_ms);
+             double *t_assign_ms, double *t_scan_ms, int S = 1);
+  int probe_split(int nq, int k2, int nprobe) const;
   int coarse_assign(const float *q_dev, int nq, int nprobe, bool ip,
                     const float *q_norms_dev, hipStream_t s,
                     int64_t *probes_dev, float *probe_dists_dev);

@@ -438,12 +438,17 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
       return -1;
     k2 = kf;
   } else {
+    int S = index_->probe_split(nq, k2, nprobe > 0 ? nprobe : 1024);
+    if (S > 1 && keys_dev_.reserve((size_t)nq * S * k2 * 8)) return -1;
     if (index_->search(q_dev_.as<float>(), nq, k2, nprobe, bm,
                        ip, s, keys_dev_.as<uint64_t>(),
-                       q_norms_dev_.as<float>(), &t_assign, &t_scan))
+                       q_norms_dev_.as<float>(), &t_assign, &t_scan, S))
       return -1;
+    k2 = k2 * S; /* sub-block partials merge in the sort below */
     /* ADC distances already match the oracle bit-for-bit; canonical
      * re-rank only when the caller asked for the exact rerank leg */
+    /* (sub-block partials are unsorted relative to each other, but
+     * sort_rows below always sorts — no extra pass needed for S>1) */
     need_canonical_rerank =
         rerank || params_.kind == IndexKind::IVFFLAT;
   }

@@ -68,7 +68,10 @@ hipError_t pq_tables_a(hipStream_t s, int nq, int d, int M,
  * T = A_q + B_list in LDS (L2; dis0 = the coarse probe distance) or the
  * query-level IP table (h:164-167), then scans the list (h:923-953).
  * out_keys: nq x k2. */
-hipError_t ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
+/* S = probe-split factor: S sub-workgroups per query (out_keys is
+ * nq x S x k2; merge with sort_rows). S>1 serves small batches. */
+hipError_t ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
+                      int nprobe,
                       int k2, const float *queries, const float *centroids,
                       const float *codebooks, const float *atab,
                       const float *btab, const float *probe_dists,

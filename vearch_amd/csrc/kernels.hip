@@ -331,7 +331,7 @@ hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
 #define GAMMA_ADC_C 2
 template <bool IP, int MW, int BS>
 __global__ void __launch_bounds__(BS)
-k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
+k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
              const float *__restrict__ queries,
              const float *__restrict__ centroids,
              const float *__restrict__ codebooks,
@@ -352,7 +352,11 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
   float *dis0s = qs + d;                            /* 1 */
   int *state = (int *)(dis0s + 1) + 1;              /* int[2] */
 
-  const int q = blockIdx.x;
+  /* probe-split: S sub-workgroups per query, sub-block handles probes
+   * p ≡ sub (mod S); partials merged by sort_rows afterwards. S>1 is
+   * the small-batch/serving path (a lone query still fills S CUs). */
+  const int q = blockIdx.x / S;
+  const int sub = blockIdx.x - q * S;
   if (q >= nq) return;
   const float *qg = queries + (int64_t)q * d;
   for (int i = threadIdx.x; i < d; i += blockDim.x) qs[i] = qg[i];
@@ -373,7 +377,7 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
     __syncthreads();
   }
 
-  for (int p = 0; p < nprobe; p++) {
+  for (int p = sub; p < nprobe; p += S) {
     int64_t ln = probes[(int64_t)q * nprobe + p];
     if (ln < 0 || ln >= nlist) continue;
     GammaBucketDev bk = buckets[ln];
@@ -473,10 +477,11 @@ k_ivfpq_scan(int nq, int d, int M, int nprobe, int k2,
   }
   sel.finish();
   for (int i = threadIdx.x; i < k2; i += blockDim.x)
-    out_keys[(int64_t)q * k2 + i] = res[i];
+    out_keys[(int64_t)blockIdx.x * k2 + i] = res[i];
 }
 
-hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
+hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
+                          int nprobe,
                           int k2, const float *queries,
                           const float *centroids, const float *codebooks,
                           const float *atab, const float *btab,
@@ -493,11 +498,11 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int d, int M, int nprobe,
   const int BS = 512;
   bool fast = (k2 + BS * GAMMA_ADC_C) <= GAMMA_SORT_CAP &&
               (M == 16 || M == 32 || M == 64 || M == 96);
-  dim3 g(nq);
+  dim3 g((uint32_t)nq * (uint32_t)S);
 #define GAMMA_LAUNCH_SCAN(IPV, MWV, BSV)                                  \
   k_ivfpq_scan<IPV, MWV, BSV><<<g, dim3(BSV), smem, s>>>(                 \
-      nq, d, M, nprobe, k2, queries, centroids, codebooks, atab, btab,    \
-      probe_dists, buckets, nlist, probes, bitmap, out_keys)
+      nq, S, d, M, nprobe, k2, queries, centroids, codebooks, atab,       \
+      btab, probe_dists, buckets, nlist, probes, bitmap, out_keys)
   if (ip) {
     if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG);
     else if (M == 16) GAMMA_LAUNCH_SCAN(true, 4, 512);

@@ -5,6 +5,7 @@
  */
 #include <math.h>
 #include <stdlib.h>
+#include <unistd.h>
 #include <string.h>
 
 #include <string>
@@ -36,6 +37,21 @@ static CStatus err_status(int code, const std::string &msg) {
 }
 
 static int g_memory_limit_mb = 0;
+
+/* host RSS in MB (the reference's MemoryManager watermark check,
+ * gamma_api.cc:180 / memory/memoryManager.cc) */
+static long rss_mb() {
+  FILE *f = fopen("/proc/self/statm", "r");
+  if (!f) return 0;
+  long pages = 0, rss = 0;
+  if (fscanf(f, "%ld %ld", &pages, &rss) != 2) rss = 0;
+  fclose(f);
+  return rss * (sysconf(_SC_PAGESIZE) / 1024) / 1024;
+}
+
+static bool memory_exceeded() {
+  return g_memory_limit_mb > 0 && rss_mb() > g_memory_limit_mb;
+}
 
 extern "C" {
 
@@ -91,6 +107,7 @@ struct CStatus CreateTable(void *engine, const char *table_str, int len) {
 
 int AddOrUpdateDoc(void *engine, const char *doc_str, int len) {
   if (!engine) return -1;
+  if (memory_exceeded()) return -1;
   auto *e = static_cast<Engine *>(engine);
   gfb::Doc doc;
   if (!doc.parse(doc_str, (size_t)len)) return -1;
@@ -215,6 +232,8 @@ int Load(void *engine) {
 struct CStatus Search(void *engine, const char *request_str, int req_len,
                       char **response_str, int *res_len) {
   if (!engine) return err_status(1, "null engine");
+  if (memory_exceeded())
+    return err_status(-2, "memory limit exceeded"); /* reader.go:170 */
   auto *e = static_cast<Engine *>(engine);
   gpb::SearchRequest req;
   if (!req.parse(request_str, req_len))

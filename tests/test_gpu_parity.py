@@ -707,3 +707,36 @@ def test_search_deterministic_across_runs(data, ivfpq_engine):
     for gd, gi in runs[1:]:
         assert np.array_equal(gi, runs[0][1])
         assert np.array_equal(gd, runs[0][0])
+
+
+def test_kill_mid_flight_machinery(data, ivfpq_engine):
+    """SetKillStatus during an in-flight search: the armed device flag
+    stops the scan between lists (ivfpq.h:927 analog). Timing-dependent,
+    so accept either a killed or a completed search — but the engine
+    must stay healthy afterwards."""
+    import threading
+    from vearch_amd import clear_kill, set_kill
+    base, q = data
+    eng = ivfpq_engine
+    outcome = {}
+
+    def searcher():
+        try:
+            outcome["res"] = eng.search_pb(q, topn=10,
+                                           index_params='{"nprobe": 64}',
+                                           request_id="midkill",
+                                           partition_id=2)
+        except InterruptedError:
+            outcome["killed"] = True
+        except RuntimeError as ex:
+            outcome["err"] = ex
+
+    t = threading.Thread(target=searcher)
+    t.start()
+    set_kill("midkill", 2)
+    t.join(timeout=60)
+    clear_kill("midkill", 2)
+    assert "err" not in outcome, outcome
+    # engine healthy afterwards
+    res = eng.search_pb(q[:4], topn=5, index_params='{"nprobe": 16}')
+    assert len(res) == 4 and res[0]["items"]

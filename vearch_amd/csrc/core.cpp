@@ -23,6 +23,29 @@ KillRegistry &KillRegistry::inst() {
 void KillRegistry::set(const std::string &rid, int pid) {
   std::lock_guard<std::mutex> g(mu_);
   map_[{rid, pid}] = 1;
+  for (auto &kv : armed_) {
+    if (kv.second.first == rid && kv.second.second == pid) {
+      /* a dedicated non-blocking stream: the legacy null stream would
+       * wait for the very scan kernel we are trying to interrupt */
+      static hipStream_t kstream = nullptr;
+      if (!kstream)
+        (void)hipStreamCreateWithFlags(&kstream, hipStreamNonBlocking);
+      static int one = 1;
+      (void)hipMemcpyAsync(kv.first, &one, 4, hipMemcpyHostToDevice,
+                           kstream);
+      (void)hipStreamSynchronize(kstream);
+    }
+  }
+}
+
+void KillRegistry::arm(const std::string &rid, int pid, int *dev_flag) {
+  std::lock_guard<std::mutex> g(mu_);
+  armed_[dev_flag] = {rid, pid};
+}
+
+void KillRegistry::disarm(int *dev_flag) {
+  std::lock_guard<std::mutex> g(mu_);
+  armed_.erase(dev_flag);
 }
 void KillRegistry::del(const std::string &rid, int pid) {
   std::lock_guard<std::mutex> g(mu_);
@@ -728,7 +751,7 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                      const uint32_t *bitmap_dev, bool metric_ip,
                      hipStream_t s, uint64_t *out_keys_dev,
                      const float *q_norms_dev, double *t_assign_ms,
-                     double *t_scan_ms, int S) {
+                     double *t_scan_ms, int S, const int *kill_flag_dev) {
   if (!trained_) return -1;
   nprobe = std::min(nprobe, nlist_);
   if (nprobe > 1024) nprobe = 1024; /* selector cap (select.hpp) */
@@ -764,7 +787,7 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                                scratch_pdists_.as<float>(),
                                dev_buckets_.as<GammaBucketDev>(), nlist_,
                                scratch_probes_.as<int64_t>(), bitmap_dev,
-                               metric_ip, out_keys_dev));
+                               metric_ip, out_keys_dev, kill_flag_dev));
   } else {
     GAMMA_CHECK(gk::ivfflat_scan(s, nq, d_, nprobe, k2, q_dev,
                                  dev_buckets_.as<GammaBucketDev>(), nlist_,

@@ -44,10 +44,16 @@ class KillRegistry {
   void set(const std::string &rid, int pid);
   void del(const std::string &rid, int pid);
   bool killed(const std::string &rid, int pid);
+  /* in-flight searches register a device int; SetKillStatus writes 1 to
+   * it so running scan kernels stop between lists (ivfpq.h:927
+   * is_killed_every analog) */
+  void arm(const std::string &rid, int pid, int *dev_flag);
+  void disarm(int *dev_flag);
 
  private:
   std::mutex mu_;
   std::map<std::pair<std::string, int>, int> map_;
+  std::map<int *, std::pair<std::string, int>> armed_;
 };
 
 class DeviceBuf {
@@ -156,7 +162,8 @@ class IVFIndex {
   int search(const float *q_dev, int nq, int k2, int nprobe,
              const uint32_t *bitmap_dev, bool metric_ip, hipStream_t s,
              uint64_t *out_keys_dev, const float *q_norms_dev,
-             double *t_assign_ms, double *t_scan_ms, int S = 1);
+             double *t_assign_ms, double *t_scan_ms, int S = 1,
+             const int *kill_flag_dev = nullptr);
   int probe_split(int nq, int k2, int nprobe) const;
   int coarse_assign(const float *q_dev, int nq, int nprobe, bool ip,
                     const float *q_norms_dev, hipStream_t s,
@@ -311,6 +318,7 @@ class Engine {
   int cached_nq_ = 0;
   mutable std::shared_mutex rw_; /* search shared / add+build exclusive */
   DeviceBuf q_dev_, q_norms_dev_, keys_dev_, out_d_dev_, out_i_dev_;
+  DeviceBuf kill_flag_;
   DeviceBuf flat_dots_, flat_keys_, filt_dev_;
   std::vector<uint32_t> filt_host_;
 };

@@ -440,10 +440,21 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
   } else {
     int S = index_->probe_split(nq, k2, nprobe > 0 ? nprobe : 1024);
     if (S > 1 && keys_dev_.reserve((size_t)nq * S * k2 * 8)) return -1;
-    if (index_->search(q_dev_.as<float>(), nq, k2, nprobe, bm,
-                       ip, s, keys_dev_.as<uint64_t>(),
-                       q_norms_dev_.as<float>(), &t_assign, &t_scan, S))
-      return -1;
+    /* arm the in-flight kill flag so SetKillStatus stops the scan
+     * between lists (request_context.h:83 semantics) */
+    const int *kf = nullptr;
+    if (!request_id.empty()) {
+      if (kill_flag_.reserve(4)) return -1;
+      GAMMA_CHECK(hipMemsetAsync(kill_flag_.get(), 0, 4, s));
+      KillRegistry::inst().arm(request_id, pid, kill_flag_.as<int>());
+      kf = kill_flag_.as<int>();
+    }
+    int rc_idx = index_->search(q_dev_.as<float>(), nq, k2, nprobe, bm,
+                                ip, s, keys_dev_.as<uint64_t>(),
+                                q_norms_dev_.as<float>(), &t_assign,
+                                &t_scan, S, kf);
+    if (kf) KillRegistry::inst().disarm(kill_flag_.as<int>());
+    if (rc_idx) return -1;
     k2 = k2 * S; /* sub-block partials merge in the sort below */
     /* ADC distances already match the oracle bit-for-bit; canonical
      * re-rank only when the caller asked for the exact rerank leg */

@@ -77,7 +77,8 @@ hipError_t ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
                       const float *btab, const float *probe_dists,
                       const GammaBucketDev *buckets,
                       int nlist, const int64_t *probes,
-                      const uint32_t *bitmap, bool ip, uint64_t *out_keys);
+                      const uint32_t *bitmap, bool ip, uint64_t *out_keys,
+                      const int *kill_flag);
 
 /* IVFFLAT fused search (gamma_index_ivfflat.h:36-91). */
 hipError_t ivfflat_scan(hipStream_t s, int nq, int d, int nprobe, int k2,

@@ -341,7 +341,8 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
              const GammaBucketDev *__restrict__ buckets, int nlist,
              const int64_t *__restrict__ probes,
              const uint32_t *__restrict__ bitmap,
-             uint64_t *__restrict__ out_keys) {
+             uint64_t *__restrict__ out_keys,
+             const int *__restrict__ kill_flag) {
   extern __shared__ char smem[];
   const int ksub = 256;
   const int dsub = d / M;
@@ -377,7 +378,15 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
     __syncthreads();
   }
 
+  /* in-flight kill (is_killed_every<1024> analog, ivfpq.h:927): poll a
+   * device flag between lists with an agent-scope load (L2-served, so a
+   * host write during the kernel is visible — plain loads can stay
+   * L1-stale, microarch §Workgroup dispatch). */
   for (int p = sub; p < nprobe; p += S) {
+    if (kill_flag &&
+        __hip_atomic_load(kill_flag, __ATOMIC_RELAXED,
+                          __HIP_MEMORY_SCOPE_AGENT))
+      break;
     int64_t ln = probes[(int64_t)q * nprobe + p];
     if (ln < 0 || ln >= nlist) continue;
     GammaBucketDev bk = buckets[ln];
@@ -488,7 +497,8 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
                           const float *probe_dists,
                           const GammaBucketDev *buckets, int nlist,
                           const int64_t *probes, const uint32_t *bitmap,
-                          bool ip, uint64_t *out_keys) {
+                          bool ip, uint64_t *out_keys,
+                          const int *kill_flag) {
   size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
                 (GAMMA_SORT_CAP + k2) * 8 + (d + 1) * 4 + 4 * sizeof(int);
   if (smem > 160 * 1024) return hipErrorInvalidValue;
@@ -502,7 +512,8 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
 #define GAMMA_LAUNCH_SCAN(IPV, MWV, BSV)                                  \
   k_ivfpq_scan<IPV, MWV, BSV><<<g, dim3(BSV), smem, s>>>(                 \
       nq, S, d, M, nprobe, k2, queries, centroids, codebooks, atab,       \
-      btab, probe_dists, buckets, nlist, probes, bitmap, out_keys)
+      btab, probe_dists, buckets, nlist, probes, bitmap, out_keys,        \
+      kill_flag)
   if (ip) {
     if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG);
     else if (M == 16) GAMMA_LAUNCH_SCAN(true, 4, 512);

@@ -740,3 +740,31 @@ def test_kill_mid_flight_machinery(data, ivfpq_engine):
     # engine healthy afterwards
     res = eng.search_pb(q[:4], topn=5, index_params='{"nprobe": 16}')
     assert len(res) == 4 and res[0]["items"]
+
+
+def test_rebuild_index(data):
+    """RebuildIndex(drop=1) retrains and re-adds everything; results stay
+    parity-clean afterwards."""
+    import ctypes as c
+    from vearch_amd.engine import lib
+    base, q = data
+    eng = make_engine("/tmp/gamma_rebuild")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 32, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 4000}')
+    eng.add(base[:9000])
+    eng.build_index()
+    gd0, gi0 = eng.raw_search(q, 10, nprobe=32, rerank=100)
+    # grow the corpus, then rebuild from scratch
+    eng.add(base[9000:12000])
+    lib().RebuildIndex.argtypes = [c.c_void_p, c.c_int, c.c_int, c.c_int]
+    assert lib().RebuildIndex(eng.h, 1, 0, 0) == 0
+    gd1, gi1 = eng.raw_search(q, 10, nprobe=32, rerank=100)
+    # rebuilt model differs, but recall must hold on the larger corpus
+    _, gti = orc.flat_topk_f64(base[:12000], q, 10)
+    assert orc.recall_at(gti, gi1, 10) >= 0.9
+    # and the total indexed count covers everything
+    total = sum(len(eng.debug_list(ln, 16)[0]) for ln in range(32))
+    assert total == 12000
+    eng.close()

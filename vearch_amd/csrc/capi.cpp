@@ -207,10 +207,14 @@ int BuildIndex(void *engine) {
 
 int RebuildIndex(void *engine, int drop_before_rebuild, int limit_cpu,
                  int describe) {
-  (void)drop_before_rebuild;
-  (void)limit_cpu;
+  (void)limit_cpu; /* GPU engine has no CPU throttle */
   (void)describe;
-  return BuildIndex(engine);
+  if (!engine) return -1;
+  std::string err;
+  int rc = static_cast<Engine *>(engine)->rebuild_index(
+      drop_before_rebuild != 0, &err);
+  if (rc) fprintf(stderr, "[gamma] RebuildIndex: %s\n", err.c_str());
+  return rc;
 }
 
 int Dump(void *engine) {

@@ -247,6 +247,9 @@ class Engine {
   int bulk_add(int64_t n, const float *vecs);
   int delete_doc(const std::string &p_key);
   int build_index(std::string *err);
+  /* RebuildIndex (gamma_api.h:103): drop the trained model + lists and
+   * retrain from the current raw vectors. */
+  int rebuild_index(bool drop_before_rebuild, std::string *err);
 
   /* the hot path: batched vector search.
    * metric: 0 default, 1 L2, 2 IP. Returns 0 ok, -2 killed, <0 error.

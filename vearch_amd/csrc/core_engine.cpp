@@ -194,6 +194,26 @@ int Engine::build_index(std::string *err) {
   return 0;
 }
 
+int Engine::rebuild_index(bool drop_before_rebuild, std::string *err) {
+  if (!table_created_) {
+    if (err) *err = "no table";
+    return -1;
+  }
+  if (params_.kind == IndexKind::FLAT) return 0;
+  {
+    std::unique_lock<std::shared_mutex> g(rw_);
+    if (drop_before_rebuild || index_->trained()) {
+      index_ = std::make_unique<IVFIndex>();
+      if (index_->init(dim_, params_)) {
+        if (err) *err = "index re-init failed";
+        return -1;
+      }
+      indexed_count_ = 0;
+    }
+  }
+  return build_index(err);
+}
+
 int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
                              const float *q_norms_dev, bool ip,
                              hipStream_t s, uint64_t *out_keys_dev,

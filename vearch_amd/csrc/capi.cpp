@@ -494,19 +494,19 @@ int GammaDebugCoarseAssign(void *engine, int nq, const float *xq,
   vgamma::DeviceBuf qd, qn, probes, pdists;
   int d = e->dimension();
   if (qd.reserve((size_t)nq * d * 4)) return -1;
-  hipMemcpy(qd.get(), xq, (size_t)nq * d * 4, hipMemcpyHostToDevice);
+  (void)hipMemcpy(qd.get(), xq, (size_t)nq * d * 4, hipMemcpyHostToDevice);
   if (qn.reserve((size_t)nq * 4)) return -1;
-  gk::row_norms(s, qd.as<float>(), nq, d, qn.as<float>());
+  (void)gk::row_norms(s, qd.as<float>(), nq, d, qn.as<float>());
   if (probes.reserve((size_t)nq * nprobe * 8)) return -1;
   if (pdists.reserve((size_t)nq * nprobe * 4)) return -1;
   if (ix->coarse_assign(qd.as<float>(), nq, nprobe,
                         e->metric_ip_default(), qn.as<float>(), s,
                         probes.as<int64_t>(), pdists.as<float>()))
     return -1;
-  hipStreamSynchronize(s);
-  hipMemcpy(out_lists, probes.get(), (size_t)nq * nprobe * 8,
+  (void)hipStreamSynchronize(s);
+  (void)hipMemcpy(out_lists, probes.get(), (size_t)nq * nprobe * 8,
             hipMemcpyDeviceToHost);
-  hipMemcpy(out_dists, pdists.get(), (size_t)nq * nprobe * 4,
+  (void)hipMemcpy(out_dists, pdists.get(), (size_t)nq * nprobe * 4,
             hipMemcpyDeviceToHost);
   return 0;
 }

@@ -62,7 +62,7 @@ int DeviceBuf::reserve(size_t bytes) {
   if (bytes <= bytes_) return 0;
   void *np = nullptr;
   if (hipMalloc(&np, bytes) != hipSuccess) return -1;
-  if (p_) hipFree(p_);
+  if (p_) (void)hipFree(p_);
   p_ = np;
   bytes_ = bytes;
   return 0;
@@ -249,7 +249,7 @@ int Bitmap::load(FILE *f, hipStream_t s) {
   set_count_ = 0;
   for (uint32_t w : host_) set_count_ += __builtin_popcount(w);
   if (words)
-    hipMemcpy(dev_.get(), host_.data(), (size_t)words * 4,
+    (void)hipMemcpy(dev_.get(), host_.data(), (size_t)words * 4,
               hipMemcpyHostToDevice);
   return 0;
 }
@@ -435,10 +435,11 @@ int IVFIndex::train(const float *xt, int64_t n, hipStream_t s,
     /* residuals of the training set under the final centroids */
     DeviceBuf xd, dots, asg, resid;
     if (xd.reserve((size_t)n * d_ * 4)) return -1;
-    hipMemcpy(xd.get(), xt, (size_t)n * d_ * 4, hipMemcpyHostToDevice);
+    GAMMA_CHECK(hipMemcpy(xd.get(), xt, (size_t)n * d_ * 4,
+                          hipMemcpyHostToDevice));
     DeviceBuf xnorm;
     xnorm.reserve((size_t)n * 4);
-    gk::row_norms(s, xd.as<float>(), n, d_, xnorm.as<float>());
+    GAMMA_CHECK(gk::row_norms(s, xd.as<float>(), n, d_, xnorm.as<float>()));
     const int64_t chunk = 16384;
     dots.reserve((size_t)std::min(n, chunk) * nlist_ * 4);
     asg.reserve((size_t)n * 4);
@@ -459,8 +460,8 @@ int IVFIndex::train(const float *xt, int64_t n, hipStream_t s,
                       asg.as<int32_t>(), resid.as<float>()) != hipSuccess)
       return -1;
     std::vector<float> resid_h((size_t)n * d_);
-    hipMemcpy(resid_h.data(), resid.get(), (size_t)n * d_ * 4,
-              hipMemcpyDeviceToHost);
+    GAMMA_CHECK(hipMemcpy(resid_h.data(), resid.get(), (size_t)n * d_ * 4,
+                          hipMemcpyDeviceToHost));
 
     /* per-subspace k-means, ksub=256, niter=25 (faiss PQ default) */
     std::vector<float> books((size_t)M_ * ksub_ * dsub_);
@@ -479,8 +480,9 @@ int IVFIndex::train(const float *xt, int64_t n, hipStream_t s,
              (size_t)ksub_ * dsub_ * 4);
     }
     if (codebooks_.reserve((size_t)M_ * ksub_ * dsub_ * 4)) return -1;
-    hipMemcpy(codebooks_.get(), books.data(), (size_t)M_ * ksub_ * dsub_ * 4,
-              hipMemcpyHostToDevice);
+    GAMMA_CHECK(hipMemcpy(codebooks_.get(), books.data(),
+                          (size_t)M_ * ksub_ * dsub_ * 4,
+                          hipMemcpyHostToDevice));
     if (btable_.reserve((size_t)nlist_ * M_ * ksub_ * 4)) return -1;
     if (gk::pq_tables_b(s, d_, M_, nlist_, centroids_.as<float>(),
                         codebooks_.as<float>(),
@@ -498,7 +500,7 @@ int IVFIndex::pq_subspace_kmeans_(const float *sub_host, int64_t n,
   const int niter = 25;
   DeviceBuf xd, cbd, codes;
   if (xd.reserve((size_t)n * dsub_ * 4)) return -1;
-  hipMemcpy(xd.get(), sub_host, (size_t)n * dsub_ * 4,
+  (void)hipMemcpy(xd.get(), sub_host, (size_t)n * dsub_ * 4,
             hipMemcpyHostToDevice);
   if (cbd.reserve((size_t)ksub_ * dsub_ * 4)) return -1;
   if (codes.reserve((size_t)n)) return -1;
@@ -517,12 +519,12 @@ int IVFIndex::pq_subspace_kmeans_(const float *sub_host, int64_t n,
   std::vector<double> sums((size_t)ksub_ * dsub_);
   std::vector<int64_t> counts(ksub_);
   for (int it = 0; it < niter; it++) {
-    hipMemcpy(cbd.get(), cb.data(), (size_t)ksub_ * dsub_ * 4,
+    (void)hipMemcpy(cbd.get(), cb.data(), (size_t)ksub_ * dsub_ * 4,
               hipMemcpyHostToDevice);
     if (gk::pq_encode(s, n, dsub_, 1, ksub_, xd.as<float>(),
                       cbd.as<float>(), codes.as<uint8_t>()) != hipSuccess)
       return -1;
-    hipMemcpy(codes_h.data(), codes.get(), (size_t)n, hipMemcpyDeviceToHost);
+    (void)hipMemcpy(codes_h.data(), codes.get(), (size_t)n, hipMemcpyDeviceToHost);
     std::fill(sums.begin(), sums.end(), 0.0);
     std::fill(counts.begin(), counts.end(), 0);
     for (int64_t i = 0; i < n; i++) {
@@ -587,10 +589,10 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
   for (int64_t c0 = 0; c0 < n; c0 += chunk) {
     int64_t cn = std::min(chunk, n - c0);
     if (xd.reserve((size_t)cn * d_ * 4)) return -1;
-    hipMemcpy(xd.get(), x_host + (size_t)c0 * d_, (size_t)cn * d_ * 4,
+    (void)hipMemcpy(xd.get(), x_host + (size_t)c0 * d_, (size_t)cn * d_ * 4,
               hipMemcpyHostToDevice);
     if (xnorm.reserve((size_t)cn * 4)) return -1;
-    gk::row_norms(s, xd.as<float>(), cn, d_, xnorm.as<float>());
+    (void)gk::row_norms(s, xd.as<float>(), cn, d_, xnorm.as<float>());
     const int64_t sub = 16384;
     if (dots.reserve((size_t)std::min(cn, sub) * nlist_ * 4)) return -1;
     if (asg.reserve((size_t)cn * 4)) return -1;
@@ -606,7 +608,7 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
                           asg.as<int32_t>() + r0) != hipSuccess)
         return -1;
     }
-    hipMemcpy(asg_h.data(), asg.get(), (size_t)cn * 4,
+    (void)hipMemcpy(asg_h.data(), asg.get(), (size_t)cn * 4,
               hipMemcpyDeviceToHost);
 
     const uint8_t *payload_h = nullptr;
@@ -621,7 +623,7 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
                         codes.as<uint8_t>()) != hipSuccess)
         return -1;
       codes_h.resize((size_t)cn * code_size_);
-      hipMemcpy(codes_h.data(), codes.get(), (size_t)cn * code_size_,
+      (void)hipMemcpy(codes_h.data(), codes.get(), (size_t)cn * code_size_,
                 hipMemcpyDeviceToHost);
       payload_h = codes_h.data();
     } else {
@@ -656,9 +658,9 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
         if (nids->reserve((size_t)ncap * 4)) return -1;
         if (ndata->reserve((size_t)ncap * entry)) return -1;
         if (bk.size > 0) {
-          hipMemcpy(nids->get(), bk.ids->get(), (size_t)bk.size * 4,
+          (void)hipMemcpy(nids->get(), bk.ids->get(), (size_t)bk.size * 4,
                     hipMemcpyDeviceToDevice);
-          hipMemcpy(ndata->get(), bk.data->get(), (size_t)bk.size * entry,
+          (void)hipMemcpy(ndata->get(), bk.data->get(), (size_t)bk.size * entry,
                     hipMemcpyDeviceToDevice);
         }
         bk.ids = std::move(nids);
@@ -666,9 +668,9 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
         bk.cap = ncap;
         dev_buckets_dirty_ = true;
       }
-      hipMemcpy(bk.ids->as<uint32_t>() + bk.size, kv.second.first.data(),
+      (void)hipMemcpy(bk.ids->as<uint32_t>() + bk.size, kv.second.first.data(),
                 (size_t)add_n * 4, hipMemcpyHostToDevice);
-      hipMemcpy((uint8_t *)bk.data->get() + (size_t)bk.size * entry,
+      (void)hipMemcpy((uint8_t *)bk.data->get() + (size_t)bk.size * entry,
                 kv.second.second.data(), (size_t)add_n * entry,
                 hipMemcpyHostToDevice);
       for (int64_t i = 0; i < add_n; i++) {
@@ -765,12 +767,12 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
     ~Ev3() { for (auto &x : e) (void)hipEventDestroy(x); }
   } ev;
   hipEvent_t e0 = ev.e[0], e1 = ev.e[1], e2 = ev.e[2];
-  hipEventRecord(e0, s);
+  (void)hipEventRecord(e0, s);
   if (coarse_assign(q_dev, nq, nprobe, metric_ip, q_norms_dev, s,
                     scratch_probes_.as<int64_t>(),
                     scratch_pdists_.as<float>()) != 0)
     return -1;
-  hipEventRecord(e1, s);
+  (void)hipEventRecord(e1, s);
   if (params_.kind == IndexKind::IVFPQ) {
     const float *atab = nullptr;
     if (!metric_ip) {
@@ -794,12 +796,12 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                                  scratch_probes_.as<int64_t>(), bitmap_dev,
                                  metric_ip, out_keys_dev));
   }
-  hipEventRecord(e2, s);
+  (void)hipEventRecord(e2, s);
   GAMMA_CHECK(hipEventSynchronize(e2));
   float ms = 0;
-  hipEventElapsedTime(&ms, e0, e1);
+  (void)hipEventElapsedTime(&ms, e0, e1);
   if (t_assign_ms) *t_assign_ms = ms;
-  hipEventElapsedTime(&ms, e1, e2);
+  (void)hipEventElapsedTime(&ms, e1, e2);
   if (t_scan_ms) *t_scan_ms = ms;
   return 0;
 }
@@ -850,12 +852,12 @@ int IVFIndex::dump(FILE *f, hipStream_t s) const {
   fwrite(&tr, 4, 1, f);
   if (!trained_) return 0;
   std::vector<float> cent((size_t)nlist_ * d_);
-  hipMemcpy(cent.data(), centroids_.get(), cent.size() * 4,
+  (void)hipMemcpy(cent.data(), centroids_.get(), cent.size() * 4,
             hipMemcpyDeviceToHost);
   fwrite(cent.data(), 4, cent.size(), f);
   if (params_.kind == IndexKind::IVFPQ) {
     std::vector<float> books((size_t)M_ * ksub_ * dsub_);
-    hipMemcpy(books.data(), codebooks_.get(), books.size() * 4,
+    (void)hipMemcpy(books.data(), codebooks_.get(), books.size() * 4,
               hipMemcpyDeviceToHost);
     fwrite(books.data(), 4, books.size(), f);
   }
@@ -869,7 +871,7 @@ int IVFIndex::dump(FILE *f, hipStream_t s) const {
       std::vector<int64_t> ids(sz);
       std::vector<uint8_t> data((size_t)sz * entry);
       copy_list_to_host(i, ids.data(), nullptr, s);
-      hipMemcpy(data.data(), buckets_[i].data->get(), (size_t)sz * entry,
+      (void)hipMemcpy(data.data(), buckets_[i].data->get(), (size_t)sz * entry,
                 hipMemcpyDeviceToHost);
       fwrite(ids.data(), 8, sz, f);
       fwrite(data.data(), 1, data.size(), f);
@@ -885,16 +887,16 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
   std::vector<float> cent((size_t)nlist_ * d_);
   if (fread(cent.data(), 4, cent.size(), f) != cent.size()) return -1;
   if (centroids_.reserve(cent.size() * 4)) return -1;
-  hipMemcpy(centroids_.get(), cent.data(), cent.size() * 4,
+  (void)hipMemcpy(centroids_.get(), cent.data(), cent.size() * 4,
             hipMemcpyHostToDevice);
   if (cent_norms_.reserve((size_t)nlist_ * 4)) return -1;
-  gk::row_norms(s, centroids_.as<float>(), nlist_, d_,
+  (void)gk::row_norms(s, centroids_.as<float>(), nlist_, d_,
                 cent_norms_.as<float>());
   if (params_.kind == IndexKind::IVFPQ) {
     std::vector<float> books((size_t)M_ * ksub_ * dsub_);
     if (fread(books.data(), 4, books.size(), f) != books.size()) return -1;
     if (codebooks_.reserve(books.size() * 4)) return -1;
-    hipMemcpy(codebooks_.get(), books.data(), books.size() * 4,
+    (void)hipMemcpy(codebooks_.get(), books.data(), books.size() * 4,
               hipMemcpyHostToDevice);
     if (btable_.reserve((size_t)nlist_ * M_ * ksub_ * 4)) return -1;
     if (gk::pq_tables_b(s, d_, M_, nlist_, centroids_.as<float>(),
@@ -926,9 +928,9 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
         ids32[j] = (v >> 63) ? ((uint32_t)(v & 0x7fffffffu) | 0x80000000u)
                              : (uint32_t)v;
       }
-      hipMemcpy(bk.ids->get(), ids32.data(), (size_t)sz * 4,
+      (void)hipMemcpy(bk.ids->get(), ids32.data(), (size_t)sz * 4,
                 hipMemcpyHostToDevice);
-      hipMemcpy(bk.data->get(), data.data(), data.size(),
+      (void)hipMemcpy(bk.data->get(), data.data(), data.size(),
                 hipMemcpyHostToDevice);
       bk.size = bk.cap = sz;
       for (long long j = 0; j < sz; j++) {

@@ -401,7 +401,7 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
     void rec(int i) { hipEventRecord(ev[i], s); }
     double ms(int a, int b) {
       float m = 0;
-      hipEventElapsedTime(&m, ev[a], ev[b]);
+      (void)hipEventElapsedTime(&m, ev[a], ev[b]);
       return m;
     }
   } tm(s);

@@ -35,6 +35,7 @@
 #define V_PCT1V 32   /* pct1 with float4-vectorized table build */
 #define V_FASTCMP 64 /* float-key compare before the full push */
 #define V_U32ID 128  /* 4-byte ids (bit 31 delete) instead of int64 */
+#define V_WSEL 256   /* wave-local selector (no cross-wave barriers) */
 
 template <int MW, int C, int VAR, int BS = 256>
 __global__ void __launch_bounds__(BS)
@@ -179,6 +180,128 @@ __global__ void k_fill(uint8_t *codes, uint32_t *ids, int64_t n, int M,
   }
 }
 
+__global__ void k_fillf(float *p, int64_t n, uint64_t seed) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint64_t s = (seed + i) * 2654435761ull;
+  s = s * 6364136223846793005ull + 1442695040888963407ull;
+  p[i] = (float)((s >> 40) & 0xFFFF) / 65536.0f - 0.5f;
+}
+
+/* Wave-selector scan: pct1v LUT + u32 ids + GammaWaveSelector. Each
+ * wave keeps its own top-k2 in a private LDS region; the only
+ * block-wide barriers are around the per-list LUT build. Final merge:
+ * compact nw*k2 partials, pad, one block bitonic. */
+template <int MW, int C, int BS>
+__global__ void __launch_bounds__(BS)
+k_scan_wsel(int nq, int d, int M, int nprobe, int k2,
+            const float *__restrict__ queries,
+            const float *__restrict__ centroids,
+            const float *__restrict__ codebooks,
+            const GammaBucketDev *__restrict__ buckets, int nlist,
+            const int64_t *__restrict__ probes,
+            uint64_t *__restrict__ out_keys,
+            const float *__restrict__ Atab,
+            const float *__restrict__ Btab,
+            const uint32_t *__restrict__ bitmap) {
+  extern __shared__ char smem[];
+  const int ksub = 256;
+  const int nw = BS / 64;
+  float *lut = (float *)smem;
+  uint64_t *wbase = (uint64_t *)(smem + ((size_t)M * ksub * 4 + 7) / 8 * 8);
+  int *cnts = (int *)(wbase + (size_t)nw * GAMMA_WSEL_CAP);
+
+  const int q = blockIdx.x;
+  if (q >= nq) return;
+  const int wave = threadIdx.x >> 6;
+
+  GammaWaveSelector wsel;
+  wsel.init(wbase + (size_t)wave * GAMMA_WSEL_CAP, cnts + wave, k2);
+
+  for (int p = 0; p < nprobe; p++) {
+    int64_t ln = probes[(int64_t)q * nprobe + p];
+    if (ln < 0 || ln >= nlist) continue;
+    GammaBucketDev bk = buckets[ln];
+    if (bk.size <= 0) continue;
+
+    __syncthreads(); /* previous list's LUT reads done before overwrite */
+    const float4 *Aq = (const float4 *)(Atab + (size_t)q * M * ksub);
+    const float4 *Bl = (const float4 *)(Btab + (size_t)ln * M * ksub);
+    float4 *lut4 = (float4 *)lut;
+    for (int e = threadIdx.x; e < (M * ksub) / 4; e += blockDim.x) {
+      float4 a = Aq[e], b = Bl[e];
+      lut4[e] = make_float4(a.x + b.x, a.y + b.y, a.z + b.z, a.w + b.w);
+    }
+    __syncthreads();
+
+    const uint32_t *ids32 = bk.ids;
+    const uint8_t *codes = (const uint8_t *)bk.data;
+    for (long long j0 = 0; j0 < bk.size; j0 += (long long)blockDim.x * C) {
+      long long jb = j0 + (long long)threadIdx.x * C;
+      uint32_t w[C][MW];
+      int64_t idv[C];
+#pragma unroll
+      for (int c = 0; c < C; c++) {
+        long long j = jb + c;
+        if (j < bk.size) {
+          idv[c] = (int64_t)(int32_t)ids32[j];
+          const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
+#pragma unroll
+          for (int mw = 0; mw < MW; mw++) w[c][mw] = cw[mw];
+        } else {
+          idv[c] = -1;
+        }
+      }
+#pragma unroll
+      for (int c = 0; c < C; c++) {
+        int64_t id = idv[c];
+        if (!((uint64_t)id >> 63)) {
+          float dis = 0.0f;
+          const float *tab = lut;
+#pragma unroll
+          for (int mw = 0; mw < MW; mw++) {
+            uint32_t wv = w[c][mw];
+            dis += tab[wv & 255u];         tab += ksub;
+            dis += tab[(wv >> 8) & 255u];  tab += ksub;
+            dis += tab[(wv >> 16) & 255u]; tab += ksub;
+            dis += tab[wv >> 24];          tab += ksub;
+          }
+          wsel.push(gamma_make_key<false>(dis, (uint32_t)id));
+        }
+      }
+      wsel.maybe_flush(64 * C); /* per-wave; no block barrier */
+    }
+  }
+  wsel.finish();
+  __syncthreads();
+
+  /* compact the nw sorted partials [w*CAP, w*CAP+k2) into [0, nw*k2):
+   * read-all-then-write (regions overlap), <=4 elems per thread */
+  const int tot = nw * k2;
+  const int tid = threadIdx.x;
+  uint64_t v0 = GAMMA_KEY_EMPTY, v1 = GAMMA_KEY_EMPTY,
+           v2 = GAMMA_KEY_EMPTY, v3 = GAMMA_KEY_EMPTY;
+  {
+    int e0 = tid, e1 = tid + BS, e2 = tid + 2 * BS, e3 = tid + 3 * BS;
+    if (e0 < tot) v0 = wbase[(e0 / k2) * GAMMA_WSEL_CAP + e0 % k2];
+    if (e1 < tot) v1 = wbase[(e1 / k2) * GAMMA_WSEL_CAP + e1 % k2];
+    if (e2 < tot) v2 = wbase[(e2 / k2) * GAMMA_WSEL_CAP + e2 % k2];
+    if (e3 < tot) v3 = wbase[(e3 / k2) * GAMMA_WSEL_CAP + e3 % k2];
+    __syncthreads();
+    if (e0 < tot) wbase[e0] = v0;
+    if (e1 < tot) wbase[e1] = v1;
+    if (e2 < tot) wbase[e2] = v2;
+    if (e3 < tot) wbase[e3] = v3;
+  }
+  int n2 = 1;
+  while (n2 < tot) n2 <<= 1;
+  __syncthreads();
+  for (int i = tot + tid; i < n2; i += BS) wbase[i] = GAMMA_KEY_EMPTY;
+  gamma_bitonic_sort(wbase, n2);
+  for (int i = tid; i < k2; i += BS)
+    out_keys[(int64_t)q * k2 + i] = wbase[i];
+}
+
 int main(int argc, char **argv) {
   int nq = argc > 1 ? atoi(argv[1]) : 10000;
   const int nprobe = argc > 2 ? atoi(argv[2]) : 32;
@@ -219,8 +342,10 @@ int main(int argc, char **argv) {
   float *Atab, *Btab;
   CHECK(hipMalloc(&Atab, (size_t)nq * M * 256 * 4));
   CHECK(hipMalloc(&Btab, (size_t)nlist * M * 256 * 4));
-  CHECK(hipMemset(Atab, 0, (size_t)nq * M * 256 * 4));
-  CHECK(hipMemset(Btab, 0, (size_t)nlist * M * 256 * 4));
+  k_fillf<<<dim3((uint32_t)(((size_t)nq * M * 256 + 255) / 256)),
+            dim3(256)>>>(Atab, (int64_t)nq * M * 256, 11);
+  k_fillf<<<dim3((uint32_t)(((size_t)nlist * M * 256 + 255) / 256)),
+            dim3(256)>>>(Btab, (int64_t)nlist * M * 256, 13);
   uint32_t *bitmap;
   CHECK(hipMalloc(&bitmap, (size_t)(N + 31) / 32 * 4));
   CHECK(hipMemset(bitmap, 0, (size_t)(N + 31) / 32 * 4));
@@ -262,6 +387,23 @@ int main(int argc, char **argv) {
   auto run = [&](const char *name, auto kern, int reps) {
     run_bs(name, kern, reps, 256);
   };
+  /* result validation: compare out vs the reference variant's output */
+  std::vector<uint64_t> h_ref((size_t)nq * k2), h_got((size_t)nq * k2);
+  auto snap = [&](std::vector<uint64_t> &dst) {
+    CHECK(hipMemcpy(dst.data(), out, dst.size() * 8,
+                    hipMemcpyDeviceToHost));
+  };
+  auto check = [&](const char *name) {
+    snap(h_got);
+    size_t bad = 0;
+    for (size_t i = 0; i < h_got.size(); i++)
+      if (h_got[i] != h_ref[i]) bad++;
+    if (bad)
+      printf("  MISMATCH %-22s %zu/%zu keys differ\n", name, bad,
+             h_got.size());
+    else
+      printf("  ok       %-22s output == reference\n", name);
+  };
 
   printf("nq=%d N=%lld nlist=%d nprobe=%d M=%d k2=%d smem=%zu\n", nq,
          (long long)N, nlist, nprobe, M, k2, smem);
@@ -290,11 +432,54 @@ int main(int argc, char **argv) {
       k_scan_var<MW, 2, V_PCT1V | V_U32ID | V_NOPUSH>, 3);
   run_bs("pct1v+u32 C=2 BS512",
          k_scan_var<MW, 2, V_PCT1V | V_U32ID, 512>, 3, 512);
+  snap(h_ref); /* engine config = the validation reference */
+  run("pct1v+fast C=2 chk", k_scan_var<MW, 2, V_PCT1V | V_FASTCMP>, 3);
+  check("pct1v+fast C=2");
   run_bs("pct1v+u32 C=1 BS512",
          k_scan_var<MW, 1, V_PCT1V | V_U32ID, 512>, 3, 512);
   run_bs("pct1v+u32 C=4 BS512",
          k_scan_var<MW, 4, V_PCT1V | V_U32ID, 512>, 3, 512);
   run_bs("pct1v+u32+nopush BS512",
          k_scan_var<MW, 2, V_PCT1V | V_U32ID | V_NOPUSH, 512>, 3, 512);
+
+  auto run_wsel = [&](const char *name, auto kern, int reps, int bs) {
+    int nw = bs / 64;
+    size_t wsmem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
+                   (size_t)nw * GAMMA_WSEL_CAP * 8 + nw * 4;
+    kern<<<dim3(nq), dim3(bs), wsmem>>>(nq, d, M, nprobe, k2, queries,
+                                        centroids, codebooks, buckets,
+                                        nlist, probes, out, Atab, Btab,
+                                        bitmap);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) {
+      printf("%-28s launch failed: %s (smem=%zu)\n", name,
+             hipGetErrorString(e), wsmem);
+      return;
+    }
+    CHECK(hipDeviceSynchronize());
+    check(name);
+    hipEvent_t a, b;
+    (void)hipEventCreate(&a);
+    (void)hipEventCreate(&b);
+    (void)hipEventRecord(a);
+    for (int r = 0; r < reps; r++)
+      kern<<<dim3(nq), dim3(bs), wsmem>>>(nq, d, M, nprobe, k2, queries,
+                                          centroids, codebooks, buckets,
+                                          nlist, probes, out, Atab, Btab,
+                                          bitmap);
+    (void)hipEventRecord(b);
+    CHECK(hipEventSynchronize(b));
+    float ms;
+    (void)hipEventElapsedTime(&ms, a, b);
+    ms /= reps;
+    printf("%-28s %8.2f ms  %8.1f GB/s  %9.0f QPS\n", name, ms,
+           nq * bytes_per_q / ms / 1e6, nq / ms * 1000.0);
+    (void)hipEventDestroy(a);
+    (void)hipEventDestroy(b);
+  };
+  run_wsel("wsel C=2 BS128", k_scan_wsel<MW, 2, 128>, 3, 128);
+  run_wsel("wsel C=2 BS256", k_scan_wsel<MW, 2, 256>, 3, 256);
+  run_wsel("wsel C=4 BS256", k_scan_wsel<MW, 4, 256>, 3, 256);
+  run_wsel("wsel C=1 BS256", k_scan_wsel<MW, 1, 256>, 3, 256);
   return 0;
 }

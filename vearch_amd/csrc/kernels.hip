@@ -326,9 +326,11 @@ hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
  * oracle bit-for-bit, see oracle_ivfpq_search_pct1).
  * MW = M/4 compile-time (0 = generic runtime-M path). The templated path
  * stages GAMMA_ADC_C codes per thread in registers with all global loads
- * issued before any use, so one barrier interval carries
- * blockDim*C codes' worth of HBM latency instead of blockDim's. */
-#define GAMMA_ADC_C 2
+ * issued before any use. C=1 measured fastest at BS=512 (9.06 ms vs
+ * 9.84 ms for C=2 on the uniform 10M/nprobe=32 microbench; 24 waves/CU
+ * already cover the load latency, and C=1 halves the register-staging
+ * pressure) — see tools/adc_bench.hip. */
+#define GAMMA_ADC_C 1
 template <bool IP, int MW, int BS>
 __global__ void __launch_bounds__(BS)
 k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,

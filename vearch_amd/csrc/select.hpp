@@ -152,3 +152,90 @@ __device__ __forceinline__ bool gamma_bitmap_test(const uint32_t *bm,
                                                   uint64_t id) {
   return bm && ((bm[id >> 5] >> (id & 31)) & 1u);
 }
+
+/* ------------------------------------------------------------------------
+ * MEASURED DEAD END — kept only for tools/adc_bench.hip and the record.
+ * On the 10M/nprobe=32 microbench the wave selector validates bit-exact
+ * against the block selector but runs 38-50% SLOWER (13.6-17.1 ms vs
+ * 9.1-9.8 ms): the per-wave bitonic flush does 8 elems/lane over 45
+ * substages, and 64-lane sorts amortize far worse than 512-thread ones;
+ * the block barriers it removes were not the bottleneck. Not used by any
+ * product kernel.
+ *
+ * Wave-local exact top-k selector: one 64-lane wave owns a private LDS
+ * region, so the scan needs NO cross-wave barriers (a wave is lockstep;
+ * __builtin_amdgcn_wave_barrier() pins the compiler's DS ordering at
+ * phase boundaries). The block merges the per-wave results once at the
+ * end.
+ * Region layout per wave (buf = uint64_t[GAMMA_WSEL_CAP], pow2):
+ *   [0, k)            running top-k, sorted ascending after a flush
+ *   [k, GAMMA_WSEL_CAP) append region (cap = GAMMA_WSEL_CAP - k slots)
+ * flush sorts [0, k+total) in place (res already in front — no copies,
+ * no overlap) and re-tightens the threshold.
+ * Invariant: maybe_flush(margin) runs at least every `margin`
+ * wave-pushes and k + margin <= GAMMA_WSEL_CAP. */
+#define GAMMA_WSEL_CAP 512
+
+struct GammaWaveSelector {
+  uint64_t *buf; /* uint64_t[GAMMA_WSEL_CAP] in LDS, per wave */
+  int *cnt;      /* int[1] in LDS, per wave */
+  int k, cap;
+  uint64_t thresh;
+
+  __device__ void init(uint64_t *buf_, int *cnt_, int k_) {
+    buf = buf_; cnt = cnt_; k = k_;
+    cap = GAMMA_WSEL_CAP - k;
+    thresh = GAMMA_KEY_EMPTY;
+    const int lane = threadIdx.x & 63;
+    for (int i = lane; i < k; i += 64) buf[i] = GAMMA_KEY_EMPTY;
+    if (lane == 0) *cnt = 0;
+    __builtin_amdgcn_wave_barrier();
+  }
+
+  __device__ __forceinline__ void push(uint64_t key) {
+    if (key < thresh) {
+      int idx = atomicAdd(cnt, 1);
+      buf[k + idx] = key; /* in range by the maybe_flush invariant */
+    }
+  }
+
+  __device__ __forceinline__ void maybe_flush(int margin) {
+    __builtin_amdgcn_wave_barrier();
+    if (*cnt > cap - margin) flush_();
+  }
+
+  __device__ void finish() {
+    __builtin_amdgcn_wave_barrier();
+    flush_();
+  }
+
+  /* wave-level bitonic sort of [0, k+total): lockstep lanes,
+   * wave_barrier pins DS ordering between stages */
+  __device__ void flush_() {
+    const int lane = threadIdx.x & 63;
+    int total = *cnt;
+    if (total > cap) total = cap;
+    int n = 1;
+    while (n < k + total) n <<= 1; /* n <= GAMMA_WSEL_CAP (pow2) */
+    __builtin_amdgcn_wave_barrier();
+    for (int i = k + total + lane; i < n; i += 64)
+      buf[i] = GAMMA_KEY_EMPTY;
+    __builtin_amdgcn_wave_barrier();
+    for (int len = 2; len <= n; len <<= 1) {
+      for (int inc = len >> 1; inc > 0; inc >>= 1) {
+        for (int i = lane; i < n; i += 64) {
+          int j = i ^ inc;
+          if (j > i) {
+            bool up = (i & len) == 0;
+            uint64_t a = buf[i], b = buf[j];
+            if ((a > b) == up) { buf[i] = b; buf[j] = a; }
+          }
+        }
+        __builtin_amdgcn_wave_barrier();
+      }
+    }
+    if (lane == 0) *cnt = 0;
+    __builtin_amdgcn_wave_barrier();
+    thresh = buf[k - 1];
+  }
+};

@@ -154,13 +154,14 @@ __device__ __forceinline__ bool gamma_bitmap_test(const uint32_t *bm,
 }
 
 /* ------------------------------------------------------------------------
- * MEASURED DEAD END — kept only for tools/adc_bench.hip and the record.
- * On the 10M/nprobe=32 microbench the wave selector validates bit-exact
- * against the block selector but runs 38-50% SLOWER (13.6-17.1 ms vs
- * 9.1-9.8 ms): the per-wave bitonic flush does 8 elems/lane over 45
- * substages, and 64-lane sorts amortize far worse than 512-thread ones;
- * the block barriers it removes were not the bottleneck. Not used by any
- * product kernel.
+ * Scale note: on the BIG ADC scan (10M codes, 275k candidates/query)
+ * this wave selector validates bit-exact but runs 38-50% SLOWER than the
+ * block selector (13.6-17.1 ms vs 9.1-9.8 ms, tools/adc_bench.hip): the
+ * 64-lane bitonic flush does 8 elems/lane over 45 substages and fires
+ * ~4x as often (cap 312 vs 1848). It wins where per-query data is SMALL
+ * (k_select_from_dots_wave: top-nprobe of nlist=4096 coarse distances,
+ * ~k2*ln(n/k2) pushes => ~1 flush total) because it drops every block
+ * barrier and packs 8 queries per workgroup.
  *
  * Wave-local exact top-k selector: one 64-lane wave owns a private LDS
  * region, so the scan needs NO cross-wave barriers (a wave is lockstep;

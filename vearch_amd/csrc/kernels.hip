@@ -798,7 +798,7 @@ hipError_t gk::flat_stream_scan(hipStream_t s, int nq, int64_t n, int d,
  * Canonical-order exact distances (matches oracle_l2sqr/oracle_ip).
  * Thread per (query, candidate); consecutive threads share the query so
  * its elements broadcast from L1. */
-template <bool IP>
+template <bool IP, int DV>
 __global__ void k_rerank(int nq, int ncand, int d,
                          const float *__restrict__ queries,
                          const float *const *__restrict__ segs,
@@ -813,11 +813,15 @@ __global__ void k_rerank(int nq, int ncand, int d,
   const int64_t seg_mask = ((int64_t)1 << seg_shift) - 1;
   const float *v = segs[id >> seg_shift] + (size_t)(id & seg_mask) * d;
   const float *qv = queries + (int64_t)q * d;
-  /* float4 loads (d % 4 == 0), scalar fmaf chain in canonical order */
+  /* float4 loads (d % 4 == 0), scalar fmaf chain in canonical order.
+   * DV != 0 makes the trip count compile-time so the loads pipeline
+   * ahead of the dependent fmaf chain. */
   const float4 *v4 = (const float4 *)v;
   const float4 *q4 = (const float4 *)qv;
   float acc = 0.0f;
-  for (int t = 0; t < (d >> 2); t++) {
+  const int nt4 = DV ? (DV >> 2) : (d >> 2);
+#pragma unroll 8
+  for (int t = 0; t < nt4; t++) {
     float4 a = q4[t], b = v4[t];
     if (IP) {
       acc = fmaf(a.x, b.x, acc);
@@ -843,12 +847,19 @@ hipError_t gk::rerank(hipStream_t s, int nq, int ncand, int d,
   int64_t total = (int64_t)nq * ncand;
   if (total == 0) return hipSuccess;
   int64_t blocks = (total + WG - 1) / WG;
-  if (ip)
-    k_rerank<true><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(
-        nq, ncand, d, queries, segs, seg_shift, keys_in, keys_out);
-  else
-    k_rerank<false><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(
-        nq, ncand, d, queries, segs, seg_shift, keys_in, keys_out);
+#define GAMMA_LAUNCH_RERANK(IPV, DVV)                                     \
+  k_rerank<IPV, DVV><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(         \
+      nq, ncand, d, queries, segs, seg_shift, keys_in, keys_out)
+  if (ip) {
+    if (d == 128) GAMMA_LAUNCH_RERANK(true, 128);
+    else if (d == 768) GAMMA_LAUNCH_RERANK(true, 768);
+    else GAMMA_LAUNCH_RERANK(true, 0);
+  } else {
+    if (d == 128) GAMMA_LAUNCH_RERANK(false, 128);
+    else if (d == 768) GAMMA_LAUNCH_RERANK(false, 768);
+    else GAMMA_LAUNCH_RERANK(false, 0);
+  }
+#undef GAMMA_LAUNCH_RERANK
   return hipGetLastError();
 }
 

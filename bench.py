@@ -41,6 +41,11 @@ WORKLOADS = {
     "ivfpq_d128_n10m_nprobe32": dict(
         kind="IVFPQ", d=128, n=10_000_000, nlist=4096, m=32, nprobe=32,
         nq=10_000, k=10, rerank=200, train_n=160_000),
+    # same, at the reference's default nlist=2048 (ivfpq.cc:112 — §8d
+    # asks for this secondary report; same 10M docs, coarser lists)
+    "ivfpq_d128_n10m_nlist2048": dict(
+        kind="IVFPQ", d=128, n=10_000_000, nlist=2048, m=32, nprobe=32,
+        nq=10_000, k=10, rerank=200, train_n=160_000),
     # configs[1]: FLAT d=128 N=1M nq=10k (parity/secondary line)
     "flat_d128_n1m": dict(
         kind="FLAT", d=128, n=1_000_000, nlist=0, m=0, nprobe=0,

@@ -538,8 +538,9 @@ def test_query_pb_by_ids_and_filters(data):
 
 def test_concurrent_search_and_mutation(data):
     """The reference allows Search from arbitrary cgo threads while a
-    background thread mutates (engine.cc:1108-1127); the engine
-    serializes internally — results must stay sane under contention."""
+    background thread mutates (engine.cc:1108-1127). Searches run
+    concurrently (read lock + per-search scratch/stream pool); mutators
+    take the write lock — results must stay sane under contention."""
     import threading
     base, q = data
     eng = make_engine("/tmp/gamma_conc")
@@ -574,6 +575,42 @@ def test_concurrent_search_and_mutation(data):
     live = gi[gi >= 0]
     bad = [v for v in live.tolist() if v < 2000 and v % 3 == 0]
     assert not bad
+    eng.close()
+
+
+def test_concurrent_searches_deterministic(data):
+    """4 threads searching the same batch concurrently (each on its own
+    SearchScratch stream) must all return exactly the sequential
+    result — concurrency must not perturb selection or rerank."""
+    import threading
+    base, q = data
+    eng = make_engine("/tmp/gamma_conc2")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 64, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 8000}')
+    eng.add(base)
+    eng.build_index()
+    want_d, want_i = eng.raw_search(q, 10, nprobe=16, rerank=50)
+    results = [None] * 6
+    errors = []
+
+    def searcher(tid):
+        try:
+            results[tid] = eng.raw_search(q, 10, nprobe=16, rerank=50)
+        except Exception as ex:  # noqa: BLE001
+            errors.append(ex)
+
+    threads = [threading.Thread(target=searcher, args=(t,))
+               for t in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    assert not errors, errors
+    for gd, gi in results:
+        assert np.array_equal(gi, want_i)
+        assert np.array_equal(gd, want_d)
     eng.close()
 
 

@@ -499,10 +499,16 @@ int GammaDebugCoarseAssign(void *engine, int nq, const float *xq,
   (void)gk::row_norms(s, qd.as<float>(), nq, d, qn.as<float>());
   if (probes.reserve((size_t)nq * nprobe * 8)) return -1;
   if (pdists.reserve((size_t)nq * nprobe * 4)) return -1;
+  vgamma::SearchScratch tmp_sc; /* debug path: throwaway scratch */
+  tmp_sc.stream = s;
   if (ix->coarse_assign(qd.as<float>(), nq, nprobe,
                         e->metric_ip_default(), qn.as<float>(), s,
-                        probes.as<int64_t>(), pdists.as<float>()))
+                        probes.as<int64_t>(), pdists.as<float>(),
+                        tmp_sc)) {
+    tmp_sc.stream = nullptr; /* not ours to destroy */
     return -1;
+  }
+  tmp_sc.stream = nullptr; /* not ours to destroy */
   (void)hipStreamSynchronize(s);
   (void)hipMemcpy(out_lists, probes.get(), (size_t)nq * nprobe * 8,
             hipMemcpyDeviceToHost);

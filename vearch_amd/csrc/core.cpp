@@ -560,6 +560,11 @@ int IVFIndex::pq_subspace_kmeans_(const float *sub_host, int64_t n,
 
 int IVFIndex::update_dev_buckets(hipStream_t s) {
   if (!dev_buckets_dirty_) return 0;
+  /* every mutator (add/load) re-uploads before releasing the write
+   * lock, so under concurrent read-locked searches this only fires on
+   * the first search of an empty index — serialize that one case */
+  std::lock_guard<std::mutex> lk(bk_mu_);
+  if (!dev_buckets_dirty_) return 0;
   std::vector<GammaBucketDev> h(nlist_);
   for (int i = 0; i < nlist_; i++) {
     h[i].ids = buckets_[i].ids ? buckets_[i].ids->as<uint32_t>() : nullptr;
@@ -703,37 +708,38 @@ int IVFIndex::del(int64_t vid, hipStream_t s) {
 
 int IVFIndex::coarse_assign(const float *q_dev, int nq, int nprobe, bool ip,
                             const float *q_norms_dev, hipStream_t s,
-                            int64_t *probes_dev, float *probe_dists_dev) {
+                            int64_t *probes_dev, float *probe_dists_dev,
+                            SearchScratch &sc) {
   const int64_t sub = 16384;
-  if (scratch_dots_.reserve((size_t)std::min<int64_t>(nq, sub) * nlist_ * 4))
+  if (sc.dots.reserve((size_t)std::min<int64_t>(nq, sub) * nlist_ * 4))
     return -1;
   /* measured: the chunked selector beats the full bitonic sort at
    * nlist>=2048 (fewer barriers); keep full sort for tiny nlist only */
   const bool full_sort = nlist_ <= 512;
-  if (!full_sort && scratch_keys_.reserve((size_t)nq * nprobe * 8))
+  if (!full_sort && sc.sel_keys.reserve((size_t)nq * nprobe * 8))
     return -1;
   for (int64_t r0 = 0; r0 < nq; r0 += sub) {
     int64_t rn = std::min<int64_t>(sub, nq - r0);
     GAMMA_CHECK(gk::dots_mfma(s, q_dev + (size_t)r0 * d_, (int)rn,
                               centroids_.as<float>(), nlist_, d_,
-                              scratch_dots_.as<float>()));
+                              sc.dots.as<float>()));
     if (full_sort) {
       GAMMA_CHECK(gk::select_rows_full(
-          s, (int)rn, nlist_, nlist_, scratch_dots_.as<float>(),
+          s, (int)rn, nlist_, nlist_, sc.dots.as<float>(),
           q_norms_dev + r0, cent_norms_.as<float>(), !ip, ip, nprobe,
           probe_dists_dev + (size_t)r0 * nprobe,
           probes_dev + (size_t)r0 * nprobe));
     } else {
       GAMMA_CHECK(gk::select_from_dots(
-          s, (int)rn, nlist_, 0, nlist_, scratch_dots_.as<float>(),
+          s, (int)rn, nlist_, 0, nlist_, sc.dots.as<float>(),
           q_norms_dev + r0, cent_norms_.as<float>(), !ip, ip, nullptr,
-          nprobe, scratch_keys_.as<uint64_t>() + (size_t)r0 * nprobe,
+          nprobe, sc.sel_keys.as<uint64_t>() + (size_t)r0 * nprobe,
           false));
     }
   }
   if (!full_sort)
     GAMMA_CHECK(gk::unpack_keys(s, (int64_t)nq * nprobe,
-                                scratch_keys_.as<uint64_t>(), ip,
+                                sc.sel_keys.as<uint64_t>(), ip,
                                 probe_dists_dev, probes_dev));
   return 0;
 }
@@ -753,12 +759,13 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                      const uint32_t *bitmap_dev, bool metric_ip,
                      hipStream_t s, uint64_t *out_keys_dev,
                      const float *q_norms_dev, double *t_assign_ms,
-                     double *t_scan_ms, int S, const int *kill_flag_dev) {
+                     double *t_scan_ms, SearchScratch &sc, int S,
+                     const int *kill_flag_dev) {
   if (!trained_) return -1;
   nprobe = std::min(nprobe, nlist_);
   if (nprobe > 1024) nprobe = 1024; /* selector cap (select.hpp) */
-  if (scratch_probes_.reserve((size_t)nq * nprobe * 8)) return -1;
-  if (scratch_pdists_.reserve((size_t)nq * nprobe * 4)) return -1;
+  if (sc.probes.reserve((size_t)nq * nprobe * 8)) return -1;
+  if (sc.pdists.reserve((size_t)nq * nprobe * 4)) return -1;
   if (update_dev_buckets(s)) return -1;
 
   struct Ev3 { /* RAII so error paths cannot leak events */
@@ -769,31 +776,31 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
   hipEvent_t e0 = ev.e[0], e1 = ev.e[1], e2 = ev.e[2];
   (void)hipEventRecord(e0, s);
   if (coarse_assign(q_dev, nq, nprobe, metric_ip, q_norms_dev, s,
-                    scratch_probes_.as<int64_t>(),
-                    scratch_pdists_.as<float>()) != 0)
+                    sc.probes.as<int64_t>(), sc.pdists.as<float>(),
+                    sc) != 0)
     return -1;
   (void)hipEventRecord(e1, s);
   if (params_.kind == IndexKind::IVFPQ) {
     const float *atab = nullptr;
     if (!metric_ip) {
-      if (scratch_atab_.reserve((size_t)nq * M_ * ksub_ * 4)) return -1;
+      if (sc.atab.reserve((size_t)nq * M_ * ksub_ * 4)) return -1;
       GAMMA_CHECK(gk::pq_tables_a(s, nq, d_, M_, q_dev,
                                   codebooks_.as<float>(),
-                                  scratch_atab_.as<float>()));
-      atab = scratch_atab_.as<float>();
+                                  sc.atab.as<float>()));
+      atab = sc.atab.as<float>();
     }
     GAMMA_CHECK(gk::ivfpq_scan(s, nq, S, d_, M_, nprobe, k2, q_dev,
                                centroids_.as<float>(),
                                codebooks_.as<float>(), atab,
                                btable_.as<float>(),
-                               scratch_pdists_.as<float>(),
+                               sc.pdists.as<float>(),
                                dev_buckets_.as<GammaBucketDev>(), nlist_,
-                               scratch_probes_.as<int64_t>(), bitmap_dev,
+                               sc.probes.as<int64_t>(), bitmap_dev,
                                metric_ip, out_keys_dev, kill_flag_dev));
   } else {
     GAMMA_CHECK(gk::ivfflat_scan(s, nq, d_, nprobe, k2, q_dev,
                                  dev_buckets_.as<GammaBucketDev>(), nlist_,
-                                 scratch_probes_.as<int64_t>(), bitmap_dev,
+                                 sc.probes.as<int64_t>(), bitmap_dev,
                                  metric_ip, out_keys_dev));
   }
   (void)hipEventRecord(e2, s);

@@ -17,6 +17,8 @@
 #include <memory>
 #include <mutex>
 #include <shared_mutex>
+#include <condition_variable>
+#include <mutex>
 #include <string>
 #include <unordered_map>
 #include <vector>
@@ -148,6 +150,26 @@ struct IndexParams {
  * (int64 ids with bit-63 delete mask, packed codes), append-only with
  * capacity extension; size published after the data copy so concurrent
  * scans see a consistent prefix. */
+/* Per-search scratch context: one HIP stream + every device buffer a
+ * search touches. A pool of these (Engine::kMaxConcurrentSearches)
+ * makes concurrent cgo Search calls truly concurrent (engine.cc
+ * 1108-1127 threading contract): searches share the engine
+ * read-locked, each on its own stream; Add/Build/Load take the write
+ * lock. Grow-only buffers, so steady-state searches never allocate. */
+struct SearchScratch {
+  hipStream_t stream = nullptr;
+  /* engine-side */
+  DeviceBuf q_dev, q_norms, keys, out_d, out_i, kill_flag, flat_dots,
+      filt_dev;
+  std::vector<uint32_t> filt_host;
+  /* index-side (coarse assign + ADC scan) */
+  DeviceBuf dots, sel_keys, probes, pdists, atab;
+  bool in_use = false;
+  ~SearchScratch() {
+    if (stream) (void)hipStreamDestroy(stream);
+  }
+};
+
 class IVFIndex {
  public:
   int init(int d, const IndexParams &p);
@@ -162,12 +184,13 @@ class IVFIndex {
   int search(const float *q_dev, int nq, int k2, int nprobe,
              const uint32_t *bitmap_dev, bool metric_ip, hipStream_t s,
              uint64_t *out_keys_dev, const float *q_norms_dev,
-             double *t_assign_ms, double *t_scan_ms, int S = 1,
-             const int *kill_flag_dev = nullptr);
+             double *t_assign_ms, double *t_scan_ms, SearchScratch &sc,
+             int S = 1, const int *kill_flag_dev = nullptr);
   int probe_split(int nq, int k2, int nprobe) const;
   int coarse_assign(const float *q_dev, int nq, int nprobe, bool ip,
                     const float *q_norms_dev, hipStream_t s,
-                    int64_t *probes_dev, float *probe_dists_dev);
+                    int64_t *probes_dev, float *probe_dists_dev,
+                    SearchScratch &sc);
   int64_t ntotal() const { return ntotal_; }
   const IndexParams &params() const { return params_; }
   int d() const { return d_; }
@@ -194,8 +217,7 @@ class IVFIndex {
   int d_ = 0, M_ = 0, ksub_ = 256, dsub_ = 0, code_size_ = 0, nlist_ = 0;
   int64_t ntotal_ = 0;
   DeviceBuf centroids_, cent_norms_, codebooks_;
-  DeviceBuf btable_;       /* pct1 B table: nlist x M x ksub f32 */
-  DeviceBuf scratch_atab_; /* pct1 A table: nq x M x ksub f32 */
+  DeviceBuf btable_; /* pct1 B table: nlist x M x ksub f32 */
   struct Bucket {
     std::unique_ptr<DeviceBuf> ids, data;
     long long size = 0, cap = 0;
@@ -205,9 +227,11 @@ class IVFIndex {
   bool dev_buckets_dirty_ = true;
   /* vid -> (bucket<<40 | pos); -1 = absent. Dense (vids are 0..N). */
   std::vector<int64_t> vid_loc_;
-  /* scratch for search (grow-only) */
-  DeviceBuf scratch_dots_, scratch_keys_, scratch_probes_, scratch_pdists_;
+  /* scratch for add/train (grow-only; search scratch lives in
+   * SearchScratch so searches can run concurrently) */
   DeviceBuf scratch_i32_, scratch_f32_;
+  std::mutex bk_mu_; /* guards the one lazy write under shared lock:
+                        first update_dev_buckets with no prior add */
 };
 
 struct FieldMeta {
@@ -296,11 +320,14 @@ class Engine {
  private:
   int flat_search_keys(const float *q_dev, int nq, int k2,
                        const float *q_norms_dev, bool ip, hipStream_t s,
-                       uint64_t *out_keys_dev, const uint32_t *bm);
-  /* 1 = excluded (deleted or fails a filter); returns device ptr via out */
+                       uint64_t *out_keys_dev, const uint32_t *bm,
+                       SearchScratch &sc);
+  /* 1 = excluded (deleted or fails a filter); dev_out == nullptr skips
+   * the device upload (filter_docids browse path) */
   int build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
                            const std::vector<RangeFilterSpec> &ranges,
-                           const uint32_t **dev_out, std::string *err);
+                           SearchScratch &sc, const uint32_t **dev_out,
+                           std::string *err);
   std::string path_, log_dir_, space_name_;
   std::string index_type_ = "IVFPQ";
   std::string vec_name_;
@@ -317,13 +344,25 @@ class Engine {
   int64_t max_docid_ = 0;
   int64_t indexed_count_ = 0;
   bool table_created_ = false;
-  hipStream_t stream_ = nullptr;
+  hipStream_t stream_ = nullptr; /* mutation stream (add/build/load) */
   int cached_nq_ = 0;
+  DeviceBuf cached_q_dev_; /* cache_queries() upload, read-only in search */
   mutable std::shared_mutex rw_; /* search shared / add+build exclusive */
-  DeviceBuf q_dev_, q_norms_dev_, keys_dev_, out_d_dev_, out_i_dev_;
-  DeviceBuf kill_flag_;
-  DeviceBuf flat_dots_, flat_keys_, filt_dev_;
-  std::vector<uint32_t> filt_host_;
+  /* search-context pool: concurrent Search calls each take one */
+  static constexpr int kMaxConcurrentSearches = 4;
+  std::vector<std::unique_ptr<SearchScratch>> pool_;
+  std::mutex pool_mu_;
+  std::condition_variable pool_cv_;
+  std::mutex timing_mu_;
+  SearchScratch *acquire_scratch_();
+  void release_scratch_(SearchScratch *sc);
+  struct ScratchGuard {
+    Engine *e;
+    SearchScratch *sc;
+    ~ScratchGuard() {
+      if (sc) e->release_scratch_(sc);
+    }
+  };
 };
 
 }  // namespace vgamma

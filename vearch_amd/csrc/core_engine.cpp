@@ -217,7 +217,7 @@ int Engine::rebuild_index(bool drop_before_rebuild, std::string *err) {
 int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
                              const float *q_norms_dev, bool ip,
                              hipStream_t s, uint64_t *out_keys_dev,
-                             const uint32_t *bm) {
+                             const uint32_t *bm, SearchScratch &sc) {
   const int64_t n = raw_.size();
   if (nq < 512 || n < 200000) {
     GAMMA_CHECK(gk::flat_stream_scan(
@@ -227,16 +227,16 @@ int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
   }
   /* chunked MFMA GEMM + seeded select (nq large): per segment-run chunks */
   const int64_t chunk = 65536;
-  if (flat_dots_.reserve((size_t)nq * chunk * 4)) return -1;
+  if (sc.flat_dots.reserve((size_t)nq * chunk * 4)) return -1;
   bool seeded = false;
   for (int64_t v0 = 0; v0 < n;) {
     int64_t run = 0;
     const float *seg = raw_.dev_run(v0, &run);
     int64_t take = std::min(run, chunk);
     GAMMA_CHECK(gk::dots_mfma(s, q_dev, nq, seg, take, dim_,
-                              flat_dots_.as<float>()));
+                              sc.flat_dots.as<float>()));
     GAMMA_CHECK(gk::select_from_dots(
-        s, nq, take, v0, take, flat_dots_.as<float>(), q_norms_dev,
+        s, nq, take, v0, take, sc.flat_dots.as<float>(), q_norms_dev,
         raw_.dev_norms(), !ip, ip, bm, k2, out_keys_dev, seeded));
     seeded = true;
     v0 += take;
@@ -295,6 +295,7 @@ static bool term_match(int dt, const std::string &doc_val,
 
 int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
                                  const std::vector<RangeFilterSpec> &ranges,
+                                 SearchScratch &sc,
                                  const uint32_t **dev_out,
                                  std::string *err) {
   /* resolve fields once */
@@ -336,7 +337,7 @@ int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
   }
   int64_t n = max_docid_;
   int64_t words = (n + 31) / 32;
-  filt_host_.assign((size_t)words, 0);
+  sc.filt_host.assign((size_t)words, 0);
   for (int64_t id = 0; id < n; id++) {
     bool excl = bitmap_.test(id);
     static const std::string kNone;
@@ -362,14 +363,46 @@ int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
         if (!(ok_l && ok_u)) excl = true;
       }
     }
-    if (excl) filt_host_[id >> 5] |= 1u << (id & 31);
+    if (excl) sc.filt_host[id >> 5] |= 1u << (id & 31);
   }
-  if (filt_dev_.reserve((size_t)std::max<int64_t>(words, 1) * 4)) return -1;
-  GAMMA_CHECK(hipMemcpyAsync(filt_dev_.get(), filt_host_.data(),
+  if (!dev_out) return 0; /* browse path: host bitmap only */
+  if (sc.filt_dev.reserve((size_t)std::max<int64_t>(words, 1) * 4))
+    return -1;
+  GAMMA_CHECK(hipMemcpyAsync(sc.filt_dev.get(), sc.filt_host.data(),
                              (size_t)words * 4, hipMemcpyHostToDevice,
-                             stream_));
-  *dev_out = filt_dev_.as<uint32_t>();
+                             sc.stream));
+  *dev_out = sc.filt_dev.as<uint32_t>();
   return 0;
+}
+
+SearchScratch *Engine::acquire_scratch_() {
+  std::unique_lock<std::mutex> lk(pool_mu_);
+  for (;;) {
+    for (auto &p : pool_)
+      if (!p->in_use) {
+        p->in_use = true;
+        return p.get();
+      }
+    if ((int)pool_.size() < kMaxConcurrentSearches) {
+      auto p = std::make_unique<SearchScratch>();
+      if (hipStreamCreate(&p->stream) != hipSuccess) return nullptr;
+      p->in_use = true;
+      pool_.push_back(std::move(p));
+      return pool_.back().get();
+    }
+    pool_cv_.wait(lk);
+  }
+}
+
+void Engine::release_scratch_(SearchScratch *sc) {
+  /* all async work on this context must be done before another caller
+   * (or a write-locked mutator) can touch shared device state */
+  (void)hipStreamSynchronize(sc->stream);
+  {
+    std::lock_guard<std::mutex> lk(pool_mu_);
+    sc->in_use = false;
+  }
+  pool_cv_.notify_one();
 }
 
 int Engine::search(int nq, const float *xq, int k, int nprobe,
@@ -380,14 +413,20 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
                    const std::vector<RangeFilterSpec> *range_filters,
                    std::string *filter_err) {
   if (!table_created_ || nq <= 0 || k <= 0) return -1;
-  /* Exclusive: concurrent cgo Search calls (engine.cc allows them) are
-   * serialized here because they share the engine's scratch device
-   * buffers and stream — GPU throughput comes from batching inside one
-   * call, not from concurrent kernels. Adds still exclude searches. */
-  std::unique_lock<std::shared_mutex> g(rw_);
+  /* Concurrent searches: each takes the read lock (Add/Build/Load are
+   * write-locked) plus one SearchScratch context from the pool — its
+   * own HIP stream and device buffers — so arbitrary cgo threads can
+   * search while a background thread indexes (engine.cc:1108-1127,
+   * SURVEY 8b). The ScratchGuard synchronizes the context's stream
+   * before release, so no async work survives the read lock. */
+  std::shared_lock<std::shared_mutex> g(rw_);
+  SearchScratch *scp = acquire_scratch_();
+  if (!scp) return -1;
+  ScratchGuard sg{this, scp};
+  SearchScratch &sc = *scp;
   const int pid = partition_id;
   bool ip = metric == 0 ? params_.metric_ip : (metric == 2);
-  hipStream_t s = stream_;
+  hipStream_t s = sc.stream;
 
   struct Timer {
     hipEvent_t ev[6];
@@ -415,19 +454,22 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
   if (k2 > 1024) return -1;
 
   tm.rec(0);
+  const float *qptr = nullptr;
   if (xq != nullptr) {
-    if (q_dev_.reserve((size_t)nq * dim_ * 4)) return -1;
-    GAMMA_CHECK(hipMemcpyAsync(q_dev_.get(), xq, (size_t)nq * dim_ * 4,
+    if (sc.q_dev.reserve((size_t)nq * dim_ * 4)) return -1;
+    GAMMA_CHECK(hipMemcpyAsync(sc.q_dev.get(), xq, (size_t)nq * dim_ * 4,
                                hipMemcpyHostToDevice, s));
+    qptr = sc.q_dev.as<float>();
   } else if (nq != cached_nq_) {
     return -1; /* cache_queries() first */
+  } else {
+    qptr = cached_q_dev_.as<float>();
   }
-  if (q_norms_dev_.reserve((size_t)nq * 4)) return -1;
-  GAMMA_CHECK(gk::row_norms(s, q_dev_.as<float>(), nq, dim_,
-                            q_norms_dev_.as<float>()));
-  if (keys_dev_.reserve((size_t)nq * k2 * 8)) return -1;
-  if (out_d_dev_.reserve((size_t)nq * k * 4)) return -1;
-  if (out_i_dev_.reserve((size_t)nq * k * 8)) return -1;
+  if (sc.q_norms.reserve((size_t)nq * 4)) return -1;
+  GAMMA_CHECK(gk::row_norms(s, qptr, nq, dim_, sc.q_norms.as<float>()));
+  if (sc.keys.reserve((size_t)nq * k2 * 8)) return -1;
+  if (sc.out_d.reserve((size_t)nq * k * 4)) return -1;
+  if (sc.out_i.reserve((size_t)nq * k * 8)) return -1;
   tm.rec(1);
 
   const uint32_t *bm = bitmap_.any() ? bitmap_.dev() : nullptr;
@@ -437,8 +479,8 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
     static const std::vector<TermFilterSpec> kNoT;
     static const std::vector<RangeFilterSpec> kNoR;
     if (build_filter_bitmap_(term_filters ? *term_filters : kNoT,
-                             range_filters ? *range_filters : kNoR, &bm,
-                             filter_err))
+                             range_filters ? *range_filters : kNoR, sc,
+                             &bm, filter_err))
       return -3;
   }
 
@@ -451,29 +493,28 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
      * member before the canonical re-rank (DESIGN.md numerics note) */
     int kf = std::min<int64_t>((int64_t)k2 + 64, 1088);
     kf = (int)std::min<int64_t>(kf, std::max<int64_t>(raw_.size(), 1));
-    if (keys_dev_.reserve((size_t)nq * kf * 8)) return -1;
-    if (flat_search_keys(q_dev_.as<float>(), nq, kf,
-                         q_norms_dev_.as<float>(), ip, s,
-                         keys_dev_.as<uint64_t>(), bm))
+    if (sc.keys.reserve((size_t)nq * kf * 8)) return -1;
+    if (flat_search_keys(qptr, nq, kf, sc.q_norms.as<float>(), ip, s,
+                         sc.keys.as<uint64_t>(), bm, sc))
       return -1;
     k2 = kf;
   } else {
     int S = index_->probe_split(nq, k2, nprobe > 0 ? nprobe : 1024);
-    if (S > 1 && keys_dev_.reserve((size_t)nq * S * k2 * 8)) return -1;
+    if (S > 1 && sc.keys.reserve((size_t)nq * S * k2 * 8)) return -1;
     /* arm the in-flight kill flag so SetKillStatus stops the scan
      * between lists (request_context.h:83 semantics) */
     const int *kf = nullptr;
     if (!request_id.empty()) {
-      if (kill_flag_.reserve(4)) return -1;
-      GAMMA_CHECK(hipMemsetAsync(kill_flag_.get(), 0, 4, s));
-      KillRegistry::inst().arm(request_id, pid, kill_flag_.as<int>());
-      kf = kill_flag_.as<int>();
+      if (sc.kill_flag.reserve(4)) return -1;
+      GAMMA_CHECK(hipMemsetAsync(sc.kill_flag.get(), 0, 4, s));
+      KillRegistry::inst().arm(request_id, pid, sc.kill_flag.as<int>());
+      kf = sc.kill_flag.as<int>();
     }
-    int rc_idx = index_->search(q_dev_.as<float>(), nq, k2, nprobe, bm,
-                                ip, s, keys_dev_.as<uint64_t>(),
-                                q_norms_dev_.as<float>(), &t_assign,
-                                &t_scan, S, kf);
-    if (kf) KillRegistry::inst().disarm(kill_flag_.as<int>());
+    int rc_idx = index_->search(qptr, nq, k2, nprobe, bm, ip, s,
+                                sc.keys.as<uint64_t>(),
+                                sc.q_norms.as<float>(), &t_assign,
+                                &t_scan, sc, S, kf);
+    if (kf) KillRegistry::inst().disarm(sc.kill_flag.as<int>());
     if (rc_idx) return -1;
     k2 = k2 * S; /* sub-block partials merge in the sort below */
     /* ADC distances already match the oracle bit-for-bit; canonical
@@ -487,18 +528,18 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
   if (KillRegistry::inst().killed(request_id, pid)) return -2;
 
   if (need_canonical_rerank) {
-    GAMMA_CHECK(gk::rerank(s, nq, k2, dim_, q_dev_.as<float>(),
-                           raw_.dev_seg_table(), RawStore::SEG_SHIFT, ip,
-                           keys_dev_.as<uint64_t>(),
-                           keys_dev_.as<uint64_t>()));
+    GAMMA_CHECK(gk::rerank(s, nq, k2, dim_, qptr, raw_.dev_seg_table(),
+                           RawStore::SEG_SHIFT, ip,
+                           sc.keys.as<uint64_t>(),
+                           sc.keys.as<uint64_t>()));
   }
-  GAMMA_CHECK(gk::sort_rows(s, nq, k2, k, keys_dev_.as<uint64_t>(), ip,
-                            out_d_dev_.as<float>(),
-                            out_i_dev_.as<int64_t>()));
+  GAMMA_CHECK(gk::sort_rows(s, nq, k2, k, sc.keys.as<uint64_t>(), ip,
+                            sc.out_d.as<float>(),
+                            sc.out_i.as<int64_t>()));
   tm.rec(3);
-  GAMMA_CHECK(hipMemcpyAsync(out_dists, out_d_dev_.get(),
+  GAMMA_CHECK(hipMemcpyAsync(out_dists, sc.out_d.get(),
                              (size_t)nq * k * 4, hipMemcpyDeviceToHost, s));
-  GAMMA_CHECK(hipMemcpyAsync(out_ids, out_i_dev_.get(), (size_t)nq * k * 8,
+  GAMMA_CHECK(hipMemcpyAsync(out_ids, sc.out_i.get(), (size_t)nq * k * 8,
                              hipMemcpyDeviceToHost, s));
   tm.rec(4);
   GAMMA_CHECK(hipStreamSynchronize(s));
@@ -507,12 +548,15 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
     for (int64_t i = 0; i < (int64_t)nq * k; i++)
       if (out_ids[i] >= 0) out_dists[i] = sqrtf(out_dists[i]);
   }
-  last_timing[0] = tm.ms(0, 1) * 1000.0;
-  last_timing[1] = (use_flat ? 0.0 : t_assign) * 1000.0;
-  last_timing[2] = (use_flat ? tm.ms(1, 2) : t_scan) * 1000.0;
-  last_timing[3] = tm.ms(2, 3) * 1000.0;
-  last_timing[4] = tm.ms(3, 4) * 1000.0;
-  last_timing[5] = tm.ms(0, 4) * 1000.0;
+  {
+    std::lock_guard<std::mutex> lk(timing_mu_);
+    last_timing[0] = tm.ms(0, 1) * 1000.0;
+    last_timing[1] = (use_flat ? 0.0 : t_assign) * 1000.0;
+    last_timing[2] = (use_flat ? tm.ms(1, 2) : t_scan) * 1000.0;
+    last_timing[3] = tm.ms(2, 3) * 1000.0;
+    last_timing[4] = tm.ms(3, 4) * 1000.0;
+    last_timing[5] = tm.ms(0, 4) * 1000.0;
+  }
   return 0;
 }
 
@@ -520,12 +564,14 @@ int Engine::filter_docids(const std::vector<TermFilterSpec> &terms,
                           const std::vector<RangeFilterSpec> &ranges,
                           int offset, int limit, std::vector<int64_t> *out,
                           std::string *err) {
-  std::unique_lock<std::shared_mutex> g(rw_);
-  const uint32_t *unused = nullptr;
-  if (build_filter_bitmap_(terms, ranges, &unused, err)) return -1;
+  std::shared_lock<std::shared_mutex> g(rw_);
+  SearchScratch *scp = acquire_scratch_();
+  if (!scp) return -1;
+  ScratchGuard sg{this, scp};
+  if (build_filter_bitmap_(terms, ranges, *scp, nullptr, err)) return -1;
   int64_t skipped = 0;
   for (int64_t id = 0; id < max_docid_; id++) {
-    if ((filt_host_[id >> 5] >> (id & 31)) & 1u) continue;
+    if ((scp->filt_host[id >> 5] >> (id & 31)) & 1u) continue;
     if (skipped++ < offset) continue;
     out->push_back(id);
     if (limit > 0 && (int)out->size() >= limit) break;
@@ -535,8 +581,8 @@ int Engine::filter_docids(const std::vector<TermFilterSpec> &terms,
 
 int Engine::cache_queries(int nq, const float *xq) {
   std::unique_lock<std::shared_mutex> g(rw_);
-  if (q_dev_.reserve((size_t)nq * dim_ * 4)) return -1;
-  GAMMA_CHECK(hipMemcpy(q_dev_.get(), xq, (size_t)nq * dim_ * 4,
+  if (cached_q_dev_.reserve((size_t)nq * dim_ * 4)) return -1;
+  GAMMA_CHECK(hipMemcpy(cached_q_dev_.get(), xq, (size_t)nq * dim_ * 4,
                         hipMemcpyHostToDevice));
   cached_nq_ = nq;
   return 0;

@@ -118,6 +118,10 @@ class Bitmap {
   bool any() const { return set_count_ > 0; }
   int ensure(int64_t nbits, hipStream_t s);
   const uint32_t *dev() const { return dev_.as<uint32_t>(); }
+  /* host shadow word (0 when out of range) for filter-bitmap merges */
+  uint32_t host_word(int64_t w) const {
+    return (w >= 0 && w < (int64_t)host_.size()) ? host_[w] : 0;
+  }
   int64_t popcount() const;
   int dump(FILE *f) const;
   int load(FILE *f, hipStream_t s);
@@ -234,6 +238,21 @@ class IVFIndex {
                         first update_dev_buckets with no prior add */
 };
 
+/* Per-field scalar index (reference: internal/engine/table +
+ * field_range_index roaring bitmaps, SURVEY 2 "Scalar table + indexes"
+ * and 8f-2): term posting lists + value-sorted arrays, appended
+ * incrementally as docs arrive, so a filtered search touches only the
+ * matching docids instead of decoding every doc's bytes per filter.
+ * Derived data: rebuilt lazily after load(), never dumped. */
+struct ScalarFieldIndex {
+  int64_t terms_upto = 0; /* postings cover docids [0, terms_upto) */
+  int64_t range_upto = 0; /* nvals/svals cover docids [0, range_upto) */
+  /* raw field bytes (or \x01-split elements for STRINGARRAY) -> ids */
+  std::unordered_map<std::string, std::vector<int64_t>> postings;
+  std::vector<std::pair<double, int64_t>> nvals;       /* numeric asc */
+  std::vector<std::pair<std::string, int64_t>> svals;  /* string asc */
+};
+
 struct FieldMeta {
   std::string name;
   int data_type = 0; /* gamma_api DataType */
@@ -335,6 +354,8 @@ class Engine {
   int training_threshold_ = 0;
   std::vector<FieldMeta> fields_;
   std::unordered_map<std::string, std::vector<std::string>> field_vals_;
+  std::unordered_map<std::string, ScalarFieldIndex> scalar_idx_;
+  std::mutex scalar_mu_; /* lazy index appends under the shared lock */
   std::unordered_map<std::string, int64_t> pkey2docid_;
   std::vector<std::string> docid2pkey_;
   RawStore raw_;

@@ -265,31 +265,18 @@ static bool decode_num(int dt, const std::string &b, double *out) {
   return false;
 }
 
-static bool term_match(int dt, const std::string &doc_val,
-                       const std::string &filter_val) {
-  /* filter value may hold several terms separated by \x01; STRINGARRAY
-   * doc values are \x01-separated too (c_api/api_data/doc.cc:102) */
-  auto split = [](const std::string &s) {
-    std::vector<std::string> out;
-    size_t p = 0;
-    while (p <= s.size()) {
-      size_t q = s.find('\x01', p);
-      if (q == std::string::npos) { out.push_back(s.substr(p)); break; }
-      out.push_back(s.substr(p, q - p));
-      p = q + 1;
-    }
-    return out;
-  };
-  std::vector<std::string> terms = split(filter_val);
-  if (dt == 8) { /* STRINGARRAY: any element matches any term */
-    for (auto &el : split(doc_val))
-      for (auto &t : terms)
-        if (!el.empty() && el == t) return true;
-    return false;
+/* \x01-separated multi-values: filter terms and STRINGARRAY doc
+ * values both use this encoding (c_api/api_data/doc.cc:102) */
+static std::vector<std::string> split_x01(const std::string &s) {
+  std::vector<std::string> out;
+  size_t p = 0;
+  while (p <= s.size()) {
+    size_t q = s.find('\x01', p);
+    if (q == std::string::npos) { out.push_back(s.substr(p)); break; }
+    out.push_back(s.substr(p, q - p));
+    p = q + 1;
   }
-  for (auto &t : terms)
-    if (doc_val == t) return true;
-  return false;
+  return out;
 }
 }  // namespace
 
@@ -298,73 +285,142 @@ int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
                                  SearchScratch &sc,
                                  const uint32_t **dev_out,
                                  std::string *err) {
-  /* resolve fields once */
-  struct TF { int dt; const std::vector<std::string> *col;
-              const TermFilterSpec *f; };
-  struct RF { int dt; const std::vector<std::string> *col;
-              const RangeFilterSpec *f; double lo, hi; };
-  std::vector<TF> tfs;
-  std::vector<RF> rfs;
-  for (auto &t : terms) {
-    auto it = field_vals_.find(t.field);
-    if (it == field_vals_.end()) {
-      if (err) *err = "unknown filter field " + t.field;
-      return -1;
-    }
-    int dt = 0;
-    for (auto &fm : fields_)
-      if (fm.name == t.field) dt = fm.data_type;
-    tfs.push_back({dt, &it->second, &t});
-  }
-  for (auto &r : ranges) {
-    auto it = field_vals_.find(r.field);
-    if (it == field_vals_.end()) {
-      if (err) *err = "unknown filter field " + r.field;
-      return -1;
-    }
-    int dt = 0;
-    for (auto &fm : fields_)
-      if (fm.name == r.field) dt = fm.data_type;
-    RF rf{dt, &it->second, &r, 0, 0};
-    if (dt != 4 && dt != 8) { /* numeric */
-      if (!decode_num(dt, r.lower, &rf.lo) ||
-          !decode_num(dt, r.upper, &rf.hi)) {
-        if (err) *err = "bad range filter value for field " + r.field;
+  const int64_t n = max_docid_;
+  const int64_t words = (n + 31) / 32;
+  /* resolve fields + lazily extend their scalar indexes. Semantics are
+   * identical to a per-doc scan of the raw column bytes: a doc is
+   * included iff it is live AND matches every filter; docs whose bytes
+   * fail numeric decode (or rows missing a value) fail that filter. */
+  struct Q { const ScalarFieldIndex *ix; int dt;
+             const TermFilterSpec *tf; const RangeFilterSpec *rf;
+             double lo, hi; };
+  std::vector<Q> qs;
+  {
+    std::lock_guard<std::mutex> lk(scalar_mu_);
+    auto dtype_of = [&](const std::string &f) {
+      for (auto &fm : fields_)
+        if (fm.name == f) return fm.data_type;
+      return 0;
+    };
+    auto extend = [&](const std::string &fname, int dt, bool want_terms,
+                      bool want_range) -> const ScalarFieldIndex * {
+      auto cit = field_vals_.find(fname);
+      if (cit == field_vals_.end()) return nullptr;
+      const std::vector<std::string> &col = cit->second;
+      ScalarFieldIndex &ix = scalar_idx_[fname];
+      static const std::string kNone;
+      if (want_terms && ix.terms_upto < n) {
+        for (int64_t id = ix.terms_upto; id < n; id++) {
+          const std::string &v = id < (int64_t)col.size() ? col[id] : kNone;
+          if (dt == 8) {
+            for (auto &el : split_x01(v))
+              if (!el.empty()) ix.postings[el].push_back(id);
+          } else {
+            ix.postings[v].push_back(id);
+          }
+        }
+        ix.terms_upto = n;
+      }
+      if (want_range && ix.range_upto < n) {
+        for (int64_t id = ix.range_upto; id < n; id++) {
+          const std::string &v = id < (int64_t)col.size() ? col[id] : kNone;
+          if (dt == 4) {
+            ix.svals.emplace_back(v, id);
+          } else {
+            double x;
+            if (decode_num(dt, v, &x)) ix.nvals.emplace_back(x, id);
+            /* decode failure: absent from nvals -> fails the range */
+          }
+        }
+        ix.range_upto = n;
+        std::sort(ix.svals.begin(), ix.svals.end());
+        std::sort(ix.nvals.begin(), ix.nvals.end());
+      }
+      return &ix;
+    };
+    for (auto &t : terms) {
+      int dt = dtype_of(t.field);
+      const ScalarFieldIndex *ix = extend(t.field, dt, true, false);
+      if (!ix) {
+        if (err) *err = "unknown filter field " + t.field;
         return -1;
       }
+      qs.push_back({ix, dt, &t, nullptr, 0, 0});
     }
-    rfs.push_back(rf);
-  }
-  int64_t n = max_docid_;
-  int64_t words = (n + 31) / 32;
-  sc.filt_host.assign((size_t)words, 0);
-  for (int64_t id = 0; id < n; id++) {
-    bool excl = bitmap_.test(id);
-    static const std::string kNone;
-    for (auto &t : tfs) {
-      if (excl) break;
-      const std::string &v =
-          id < (int64_t)t.col->size() ? (*t.col)[id] : kNone;
-      if (!term_match(t.dt, v, t.f->value)) excl = true;
-    }
-    for (auto &r : rfs) {
-      if (excl) break;
-      const std::string &v =
-          id < (int64_t)r.col->size() ? (*r.col)[id] : kNone;
-      if (r.dt == 4) { /* STRING: lexicographic */
-        bool ok_l = r.f->inc_l ? (v >= r.f->lower) : (v > r.f->lower);
-        bool ok_u = r.f->inc_u ? (v <= r.f->upper) : (v < r.f->upper);
-        if (!(ok_l && ok_u)) excl = true;
-      } else {
-        double x;
-        if (!decode_num(r.dt, v, &x)) { excl = true; continue; }
-        bool ok_l = r.f->inc_l ? (x >= r.lo) : (x > r.lo);
-        bool ok_u = r.f->inc_u ? (x <= r.hi) : (x < r.hi);
-        if (!(ok_l && ok_u)) excl = true;
+    for (auto &r : ranges) {
+      int dt = dtype_of(r.field);
+      const ScalarFieldIndex *ix = extend(r.field, dt, false, true);
+      if (!ix) {
+        if (err) *err = "unknown filter field " + r.field;
+        return -1;
       }
+      Q q{ix, dt, nullptr, &r, 0, 0};
+      if (dt != 4 && dt != 8) {
+        if (!decode_num(dt, r.lower, &q.lo) ||
+            !decode_num(dt, r.upper, &q.hi)) {
+          if (err) *err = "bad range filter value for field " + r.field;
+          return -1;
+        }
+      }
+      qs.push_back(q);
     }
-    if (excl) sc.filt_host[id >> 5] |= 1u << (id & 31);
   }
+
+  /* included = AND over filters of each filter's match set */
+  std::vector<uint32_t> inc((size_t)words, 0xffffffffu);
+  std::vector<uint32_t> mt;
+  for (auto &q : qs) {
+    mt.assign((size_t)words, 0);
+    if (q.tf) {
+      for (auto &t : split_x01(q.tf->value)) {
+        auto pit = q.ix->postings.find(t);
+        if (pit == q.ix->postings.end()) continue;
+        for (int64_t id : pit->second)
+          if (id < n) mt[id >> 5] |= 1u << (id & 31);
+      }
+    } else if (q.dt == 4) { /* lexicographic string range */
+      const auto &sv = q.ix->svals;
+      auto cmp = [](const std::pair<std::string, int64_t> &a,
+                    const std::string &b) { return a.first < b; };
+      auto cmp2 = [](const std::string &a,
+                     const std::pair<std::string, int64_t> &b) {
+        return a < b.first;
+      };
+      auto b = q.rf->inc_l
+                   ? std::lower_bound(sv.begin(), sv.end(), q.rf->lower, cmp)
+                   : std::upper_bound(sv.begin(), sv.end(), q.rf->lower,
+                                      cmp2);
+      auto e = q.rf->inc_u
+                   ? std::upper_bound(sv.begin(), sv.end(), q.rf->upper,
+                                      cmp2)
+                   : std::lower_bound(sv.begin(), sv.end(), q.rf->upper,
+                                      cmp);
+      for (; b < e; ++b)
+        if (b->second < n) mt[b->second >> 5] |= 1u << (b->second & 31);
+    } else if (q.dt != 8) { /* numeric range (range on STRINGARRAY
+                               matches nothing, as in the byte-scan) */
+      const auto &nv = q.ix->nvals;
+      auto cmp = [](const std::pair<double, int64_t> &a, double b) {
+        return a.first < b;
+      };
+      auto cmp2 = [](double a, const std::pair<double, int64_t> &b) {
+        return a < b.first;
+      };
+      auto b = q.rf->inc_l
+                   ? std::lower_bound(nv.begin(), nv.end(), q.lo, cmp)
+                   : std::upper_bound(nv.begin(), nv.end(), q.lo, cmp2);
+      auto e = q.rf->inc_u
+                   ? std::upper_bound(nv.begin(), nv.end(), q.hi, cmp2)
+                   : std::lower_bound(nv.begin(), nv.end(), q.hi, cmp);
+      for (; b < e; ++b)
+        if (b->second < n) mt[b->second >> 5] |= 1u << (b->second & 31);
+    }
+    for (int64_t w = 0; w < words; w++) inc[w] &= mt[w];
+  }
+
+  sc.filt_host.assign((size_t)words, 0);
+  for (int64_t w = 0; w < words; w++)
+    sc.filt_host[w] = ~inc[w] | bitmap_.host_word(w);
   if (!dev_out) return 0; /* browse path: host bitmap only */
   if (sc.filt_dev.reserve((size_t)std::max<int64_t>(words, 1) * 4))
     return -1;

@@ -487,6 +487,54 @@ def test_filtered_search_term_and_range(data):
     eng.close()
 
 
+def test_filter_index_incremental_and_stringarray(data):
+    """The scalar indexes append lazily: filter, add more docs (+ one
+    update and one delete), filter again — results must track the
+    mutations exactly. STRINGARRAY any-element matching included."""
+    import struct
+    from vearch_amd import fbsenc
+    base, q = data
+    eng = make_engine("/tmp/gamma_filter_inc")
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}',
+                     scalar_fields=[("tags", fbsenc.DATA_STRINGARRAY),
+                                    ("num", fbsenc.DATA_INT)])
+
+    def addd(vid, tags, num):
+        eng.add_doc(str(vid), base[vid],
+                    fields=[("tags", tags, fbsenc.DATA_STRINGARRAY),
+                            ("num", struct.pack("<i", num),
+                             fbsenc.DATA_INT)])
+
+    for vid in range(1000):
+        addd(vid, b"a\x01b" if vid % 2 == 0 else b"c", vid)
+    res = eng.search_pb(q[:4], topn=10, term_filters=[("tags", b"b")])
+    for t in range(4):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert ids and all(i % 2 == 0 for i in ids)
+    # incremental rows after the first (index-building) filter call
+    for vid in range(1000, 2000):
+        addd(vid, b"b", vid)
+    addd("7", b"zzz", 7)  # update: docid 7 re-keyed, old row dead
+    eng.delete_doc("1002")
+    res = eng.search_pb(q[:4], topn=300, term_filters=[("tags", b"b")])
+    got = set()
+    for t in range(4):
+        got |= {int(it["fields"]["_id"]) for it in res[t]["items"]}
+    assert all((v < 1000 and v % 2 == 0) or v >= 1000 for v in got)
+    assert 1002 not in got          # deleted
+    assert 7 not in got             # updated away from tag b
+    assert any(v >= 1000 for v in got)  # new rows visible
+    # numeric range over the grown column
+    res = eng.search_pb(
+        q[:4], topn=300,
+        range_filters=[("num", struct.pack("<i", 990),
+                        struct.pack("<i", 1010), True, True)])
+    for t in range(4):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert ids and all(990 <= i <= 1010 and i != 1002 for i in ids)
+    eng.close()
+
+
 def test_filtered_search_ivfpq(data, ivfpq_engine):
     """filters work on the IVFPQ path too (same bitmap arg)."""
     base, q = data

@@ -57,6 +57,12 @@ int GammaDebugCoarseAssign(void *engine, int nq, const float *xq, int nprobe,
 /* Debug: copy the trained model to host: centroids (nlist*d), pq codebooks
  * (M*ksub*dsub). Buffers may be NULL to skip. */
 int GammaDebugGetModel(void *engine, float *centroids, float *codebooks);
+/* OPQ debug hooks (oracle parity chains on the engine's own R and its
+ * GPU rotation so downstream ADC stays bit-exact):
+ * GetOPQ copies the d*d row-major rotation; ApplyOPQ returns the
+ * engine-rotated queries (nq*d in, nq*d out). -1 if no OPQ. */
+int GammaDebugGetOPQ(void *engine, float *R);
+int GammaDebugApplyOPQ(void *engine, const float *xq, int nq, float *out);
 
 /* Debug: fetch the contents of inverted list `list_no`: returns size, and
  * copies ids (int64, with bit-63 delete marks preserved, reference

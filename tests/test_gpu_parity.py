@@ -662,6 +662,65 @@ def test_concurrent_searches_deterministic(data):
     eng.close()
 
 
+def test_opq_train_search_parity(data):
+    """OPQ pre-rotation (ivfpq.cc:168-177 params, :362-364 train,
+    :585-588 search, :735 raw-space rerank): R is orthonormal, the
+    ADC path in rotated space stays bit-exact vs the oracle (fed the
+    engine-rotated queries), recall with the raw-space rerank holds the
+    reference floors, and dump/load round-trips R."""
+    base, q = data
+    os.makedirs("/tmp/gamma_opq", exist_ok=True)
+    eng = make_engine("/tmp/gamma_opq")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 64, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 8000, "opq": {"nsubvector": 16}}')
+    eng.add(base)
+    eng.build_index()
+    # R orthonormal
+    R = eng.debug_opq(64)
+    assert np.allclose(R @ R.T, np.eye(64), atol=2e-3)
+    # ADC bit-exact in rotated space: oracle gets the engine's own
+    # GPU-rotated queries + rotated-space model, probes from the GPU
+    q_rot = eng.debug_apply_opq(q)
+    gd, gi = eng.raw_search(q, 10, nprobe=16)
+    ox = _oracle_from_engine(eng, 64, 64, 16)
+    pdists, probes = eng.debug_coarse_assign(q_rot, 16)
+    od, oi = ox.search_pct1(q_rot, 10, 16, probes=probes,
+                            probe_dists=pdists)
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    # recall floors with the raw-space rerank leg
+    _, gti = orc.flat_topk_f64(base, q, 100)
+    rd, ri = eng.raw_search(q, 10, nprobe=16, rerank=100)
+    assert orc.recall_at(gti, ri, 10) >= 0.9
+    # rerank distances are raw-space exact
+    lib = RefLib.lib()
+    lib.oracle_l2sqr.restype = ctypes.c_float
+    for t in range(0, q.shape[0], 7):
+        for j in range(3):
+            if ri[t, j] < 0:
+                continue
+            want = lib.oracle_l2sqr(_fp(_c(q[t], np.float32)),
+                                    _fp(_c(base[ri[t, j]], np.float32)),
+                                    64)
+            assert rd[t, j] == want
+    # dump/load: identical results, R survives
+    eng.dump()
+    eng.close()
+    eng2 = make_engine("/tmp/gamma_opq")
+    eng2.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 64, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 8000, "opq": {"nsubvector": 16}}')
+    eng2.load()
+    R2 = eng2.debug_opq(64)
+    assert np.array_equal(R, R2)
+    gd2, gi2 = eng2.raw_search(q, 10, nprobe=16)
+    assert np.array_equal(gd, gd2) and np.array_equal(gi, gi2)
+    eng2.close()
+
+
 def test_ivfpq_d768_m96_bitexact():
     """Config-5 shape (d=768, m=96, dsub=8): the wide-M template path
     stays bit-exact vs the oracle."""

@@ -517,6 +517,32 @@ int GammaDebugCoarseAssign(void *engine, int nq, const float *xq,
   return 0;
 }
 
+int GammaDebugGetOPQ(void *engine, float *R) {
+  if (!engine) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  auto *ix = e->index();
+  if (!ix || !ix->has_opq() || ix->opq_R_host().empty()) return -1;
+  memcpy(R, ix->opq_R_host().data(), ix->opq_R_host().size() * 4);
+  return 0;
+}
+
+int GammaDebugApplyOPQ(void *engine, const float *xq, int nq, float *out) {
+  if (!engine || nq <= 0) return -1;
+  auto *e = static_cast<Engine *>(engine);
+  auto *ix = e->index();
+  if (!ix || !ix->has_opq()) return -1;
+  hipStream_t s = e->stream();
+  int d = e->dimension();
+  vgamma::DeviceBuf qd, qr;
+  if (qd.reserve((size_t)nq * d * 4) || qr.reserve((size_t)nq * d * 4))
+    return -1;
+  (void)hipMemcpy(qd.get(), xq, (size_t)nq * d * 4, hipMemcpyHostToDevice);
+  if (ix->rotate_dev(qd.as<float>(), nq, qr.as<float>(), s)) return -1;
+  (void)hipStreamSynchronize(s);
+  (void)hipMemcpy(out, qr.get(), (size_t)nq * d * 4, hipMemcpyDeviceToHost);
+  return 0;
+}
+
 int GammaDebugGetModel(void *engine, float *centroids, float *codebooks) {
   if (!engine) return -1;
   auto *e = static_cast<Engine *>(engine);

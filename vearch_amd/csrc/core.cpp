@@ -288,8 +288,19 @@ int IndexParams::parse(const std::string &json, std::string *err) {
     return -1;
   }
   if (v.has("opq")) {
-    if (err) *err = "opq not supported yet (SURVEY 8f-4)";
-    return -1;
+    /* {"opq": {"nsubvector": N}} (ivfpq.h:1202-1212) */
+    const gjson::Value *o = v.get("opq");
+    if (!o || o->type != gjson::Value::OBJ) {
+      if (err) *err = "invalid opq params";
+      return -1;
+    }
+    int on = 0;
+    if (!o->get_int("nsubvector", on) || on <= 0) {
+      if (err) *err = "invalid opq_nsubvector";
+      return -1;
+    }
+    has_opq = true;
+    opq_nsubvector = on;
   }
   return 0;
 }
@@ -298,6 +309,11 @@ int IndexParams::parse(const std::string &json, std::string *err) {
 int IVFIndex::init(int d, const IndexParams &p) {
   params_ = p;
   d_ = d;
+  if (p.has_opq) {
+    /* mirror the reference's validation (ivfpq.cc:169-174) */
+    if (p.kind != IndexKind::IVFPQ) return -1;
+    if (p.opq_nsubvector <= 0 || d % p.opq_nsubvector != 0) return -1;
+  }
   nlist_ = p.ncentroids;
   if (params_.kind == IndexKind::IVFPQ) {
     M_ = p.nsubvector > 0 ? p.nsubvector : d / 2; /* ivfpq.cc:122-124 */
@@ -415,8 +431,202 @@ int IVFIndex::kmeans_gpu(const float *x_host, int64_t n, int ncl, int niter,
   return 0;
 }
 
+int IVFIndex::rotate_dev(const float *x_dev, int64_t n, float *y_dev,
+                         hipStream_t s) const {
+  /* y_i = R x_i as a row-GEMM: dots(x_i, R_j) = (R x_i)_j */
+  const int64_t chunk = 16384;
+  for (int64_t r0 = 0; r0 < n; r0 += chunk) {
+    int64_t rn = std::min(chunk, n - r0);
+    if (gk::dots_mfma(s, x_dev + (size_t)r0 * d_, (int)rn,
+                      opq_R_.as<float>(), d_, d_,
+                      y_dev + (size_t)r0 * d_) != hipSuccess)
+      return -1;
+  }
+  return 0;
+}
+
+namespace {
+/* One-sided Jacobi SVD of a dense dxd matrix (double, in place):
+ * orthogonalizes A's columns with Givens rotations accumulated in V,
+ * so A_in = U diag(sigma) V^T with U = normalized columns of A_out.
+ * d <= ~1k, used once per OPQ iteration at train time. */
+void jacobi_svd(std::vector<double> &A, int d, std::vector<double> &V) {
+  V.assign((size_t)d * d, 0.0);
+  for (int i = 0; i < d; i++) V[(size_t)i * d + i] = 1.0;
+  const double eps = 1e-12;
+  for (int sweep = 0; sweep < 30; sweep++) {
+    double off = 0.0;
+    for (int p = 0; p < d - 1; p++) {
+      for (int q = p + 1; q < d; q++) {
+        double app = 0, aqq = 0, apq = 0;
+        for (int i = 0; i < d; i++) {
+          double x = A[(size_t)i * d + p], y = A[(size_t)i * d + q];
+          app += x * x;
+          aqq += y * y;
+          apq += x * y;
+        }
+        off += apq * apq;
+        if (fabs(apq) < eps * sqrt(app * aqq) || apq == 0.0) continue;
+        double tau = (aqq - app) / (2.0 * apq);
+        double t = (tau >= 0 ? 1.0 : -1.0) /
+                   (fabs(tau) + sqrt(1.0 + tau * tau));
+        double c = 1.0 / sqrt(1.0 + t * t), sn = c * t;
+        for (int i = 0; i < d; i++) {
+          double x = A[(size_t)i * d + p], y = A[(size_t)i * d + q];
+          A[(size_t)i * d + p] = c * x - sn * y;
+          A[(size_t)i * d + q] = sn * x + c * y;
+          double vx = V[(size_t)i * d + p], vy = V[(size_t)i * d + q];
+          V[(size_t)i * d + p] = c * vx - sn * vy;
+          V[(size_t)i * d + q] = sn * vx + c * vy;
+        }
+      }
+    }
+    if (off < 1e-20) break;
+  }
+}
+}  // namespace
+
+int IVFIndex::train_opq_(const float *xt, int64_t n, hipStream_t s,
+                         std::string *err) {
+  const int64_t n_o = std::min<int64_t>(n, 65536);
+  const int niter_opq = 8;
+  const int d = d_;
+  /* R init: seeded Gaussian rows, Gram-Schmidt -> orthonormal */
+  std::vector<double> R((size_t)d * d);
+  {
+    std::mt19937_64 rng(4321);
+    std::normal_distribution<double> g(0.0, 1.0);
+    for (auto &x : R) x = g(rng);
+    for (int i = 0; i < d; i++) {
+      double *ri = R.data() + (size_t)i * d;
+      for (int j = 0; j < i; j++) {
+        const double *rj = R.data() + (size_t)j * d;
+        double dot = 0;
+        for (int t = 0; t < d; t++) dot += ri[t] * rj[t];
+        for (int t = 0; t < d; t++) ri[t] -= dot * rj[t];
+      }
+      double nrm = 0;
+      for (int t = 0; t < d; t++) nrm += ri[t] * ri[t];
+      nrm = sqrt(std::max(nrm, 1e-30));
+      for (int t = 0; t < d; t++) ri[t] /= nrm;
+    }
+  }
+  DeviceBuf xd, xr, codes_d;
+  if (xd.reserve((size_t)n_o * d * 4)) return -1;
+  GAMMA_CHECK(hipMemcpy(xd.get(), xt, (size_t)n_o * d * 4,
+                        hipMemcpyHostToDevice));
+  if (xr.reserve((size_t)n_o * d * 4)) return -1;
+  if (codes_d.reserve((size_t)n_o * M_)) return -1;
+  if (opq_R_.reserve((size_t)d * d * 4)) return -1;
+  std::vector<float> Rf((size_t)d * d);
+  std::vector<float> xr_h((size_t)n_o * d);
+  std::vector<uint8_t> codes_h((size_t)n_o * M_);
+  std::vector<float> books((size_t)M_ * ksub_ * dsub_);
+  std::vector<float> sub((size_t)n_o * dsub_);
+  std::vector<float> Xt_h, Yt_h;
+  DeviceBuf Xt_d, Yt_d, B_d;
+
+  for (int it = 0; it < niter_opq; it++) {
+    for (size_t i = 0; i < Rf.size(); i++) Rf[i] = (float)R[i];
+    GAMMA_CHECK(hipMemcpy(opq_R_.get(), Rf.data(), Rf.size() * 4,
+                          hipMemcpyHostToDevice));
+    if (rotate_dev(xd.as<float>(), n_o, xr.as<float>(), s)) return -1;
+    GAMMA_CHECK(hipMemcpy(xr_h.data(), xr.get(), xr_h.size() * 4,
+                          hipMemcpyDeviceToHost));
+    /* fit a plain (non-residual) PQ to the rotated sample — the same
+     * independent-PQ objective OPQMatrix optimizes */
+    for (int m = 0; m < M_; m++) {
+      for (int64_t i = 0; i < n_o; i++)
+        memcpy(sub.data() + (size_t)i * dsub_,
+               xr_h.data() + (size_t)i * d + (size_t)m * dsub_,
+               (size_t)dsub_ * 4);
+      std::vector<float> cb;
+      if (pq_subspace_kmeans_(sub.data(), n_o, cb, s, m + 100 * it)) {
+        if (err) *err = "opq pq train failed";
+        return -1;
+      }
+      memcpy(books.data() + (size_t)m * ksub_ * dsub_, cb.data(),
+             (size_t)ksub_ * dsub_ * 4);
+    }
+    if (codebooks_.reserve(books.size() * 4)) return -1;
+    GAMMA_CHECK(hipMemcpy(codebooks_.get(), books.data(),
+                          books.size() * 4, hipMemcpyHostToDevice));
+    GAMMA_CHECK(gk::pq_encode(s, n_o, d, M_, ksub_, xr.as<float>(),
+                              codebooks_.as<float>(),
+                              codes_d.as<uint8_t>()));
+    GAMMA_CHECK(hipMemcpy(codes_h.data(), codes_d.get(), codes_h.size(),
+                          hipMemcpyDeviceToHost));
+    /* Y = decode(codes); B = X^T Y via a GEMM on transposed layouts */
+    Xt_h.assign((size_t)d * n_o, 0.f);
+    Yt_h.assign((size_t)d * n_o, 0.f);
+    for (int64_t i = 0; i < n_o; i++) {
+      for (int t = 0; t < d; t++)
+        Xt_h[(size_t)t * n_o + i] = xt[(size_t)i * d + t];
+      for (int m = 0; m < M_; m++) {
+        const float *cw =
+            books.data() +
+            ((size_t)m * ksub_ + codes_h[(size_t)i * M_ + m]) * dsub_;
+        for (int t = 0; t < dsub_; t++)
+          Yt_h[(size_t)(m * dsub_ + t) * n_o + i] = cw[t];
+      }
+    }
+    if (Xt_d.reserve(Xt_h.size() * 4) || Yt_d.reserve(Yt_h.size() * 4) ||
+        B_d.reserve((size_t)d * d * 4))
+      return -1;
+    GAMMA_CHECK(hipMemcpy(Xt_d.get(), Xt_h.data(), Xt_h.size() * 4,
+                          hipMemcpyHostToDevice));
+    GAMMA_CHECK(hipMemcpy(Yt_d.get(), Yt_h.data(), Yt_h.size() * 4,
+                          hipMemcpyHostToDevice));
+    GAMMA_CHECK(gk::dots_mfma(s, Xt_d.as<float>(), d, Yt_d.as<float>(), d,
+                              (int)n_o, B_d.as<float>()));
+    std::vector<float> Bf((size_t)d * d);
+    GAMMA_CHECK(hipMemcpy(Bf.data(), B_d.get(), Bf.size() * 4,
+                          hipMemcpyDeviceToHost));
+    /* orthogonal Procrustes: max tr(R B), B = U S V^T -> R = V U^T */
+    std::vector<double> A(Bf.begin(), Bf.end()), V;
+    jacobi_svd(A, d, V); /* A now holds U * diag(sigma) in columns */
+    for (int k = 0; k < d; k++) { /* normalize columns -> U */
+      double nrm = 0;
+      for (int i = 0; i < d; i++) {
+        double x = A[(size_t)i * d + k];
+        nrm += x * x;
+      }
+      nrm = sqrt(std::max(nrm, 1e-30));
+      for (int i = 0; i < d; i++) A[(size_t)i * d + k] /= nrm;
+    }
+    for (int i = 0; i < d; i++)
+      for (int j = 0; j < d; j++) {
+        double acc = 0;
+        for (int k = 0; k < d; k++)
+          acc += V[(size_t)i * d + k] * A[(size_t)j * d + k];
+        R[(size_t)i * d + j] = acc;
+      }
+  }
+  for (size_t i = 0; i < Rf.size(); i++) Rf[i] = (float)R[i];
+  GAMMA_CHECK(hipMemcpy(opq_R_.get(), Rf.data(), Rf.size() * 4,
+                        hipMemcpyHostToDevice));
+  opq_R_host_ = Rf;
+  return 0;
+}
+
 int IVFIndex::train(const float *xt, int64_t n, hipStream_t s,
                     std::string *err) {
+  std::vector<float> xrot_h;
+  if (has_opq()) {
+    /* train R, then train the IVFPQ model in rotated space
+     * (ivfpq.cc:362-364: opq_->train + xt = opq_->apply(train)) */
+    if (train_opq_(xt, n, s, err)) return -1;
+    DeviceBuf xd, xr;
+    if (xd.reserve((size_t)n * d_ * 4) || xr.reserve((size_t)n * d_ * 4))
+      return -1;
+    GAMMA_CHECK(hipMemcpy(xd.get(), xt, (size_t)n * d_ * 4,
+                          hipMemcpyHostToDevice));
+    if (rotate_dev(xd.as<float>(), n, xr.as<float>(), s)) return -1;
+    xrot_h.resize((size_t)n * d_);
+    GAMMA_CHECK(hipMemcpy(xrot_h.data(), xr.get(), xrot_h.size() * 4,
+                          hipMemcpyDeviceToHost));
+    xt = xrot_h.data();
+  }
   std::vector<float> cent;
   if (kmeans_gpu(xt, n, nlist_, 10, params_.metric_ip, cent, s)) {
     if (err) *err = "coarse k-means failed";
@@ -591,19 +801,26 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
   std::vector<int32_t> asg_h(std::min(n, chunk));
   std::vector<uint8_t> codes_h;
 
+  DeviceBuf xrot;
   for (int64_t c0 = 0; c0 < n; c0 += chunk) {
     int64_t cn = std::min(chunk, n - c0);
     if (xd.reserve((size_t)cn * d_ * 4)) return -1;
     (void)hipMemcpy(xd.get(), x_host + (size_t)c0 * d_, (size_t)cn * d_ * 4,
               hipMemcpyHostToDevice);
+    const float *xin = xd.as<float>();
+    if (has_opq()) { /* encode in rotated space (ivfpq.cc:470-471) */
+      if (xrot.reserve((size_t)cn * d_ * 4)) return -1;
+      if (rotate_dev(xin, cn, xrot.as<float>(), s)) return -1;
+      xin = xrot.as<float>();
+    }
     if (xnorm.reserve((size_t)cn * 4)) return -1;
-    (void)gk::row_norms(s, xd.as<float>(), cn, d_, xnorm.as<float>());
+    (void)gk::row_norms(s, xin, cn, d_, xnorm.as<float>());
     const int64_t sub = 16384;
     if (dots.reserve((size_t)std::min(cn, sub) * nlist_ * 4)) return -1;
     if (asg.reserve((size_t)cn * 4)) return -1;
     for (int64_t r0 = 0; r0 < cn; r0 += sub) {
       int64_t rn = std::min(sub, cn - r0);
-      if (gk::dots_mfma(s, xd.as<float>() + (size_t)r0 * d_, (int)rn,
+      if (gk::dots_mfma(s, xin + (size_t)r0 * d_, (int)rn,
                         centroids_.as<float>(), nlist_, d_,
                         dots.as<float>()) != hipSuccess)
         return -1;
@@ -619,7 +836,7 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
     const uint8_t *payload_h = nullptr;
     if (params_.kind == IndexKind::IVFPQ) {
       if (resid.reserve((size_t)cn * d_ * 4)) return -1;
-      if (gk::residuals(s, cn, d_, xd.as<float>(), centroids_.as<float>(),
+      if (gk::residuals(s, cn, d_, xin, centroids_.as<float>(),
                         asg.as<int32_t>(), resid.as<float>()) != hipSuccess)
         return -1;
       if (codes.reserve((size_t)cn * code_size_)) return -1;
@@ -767,6 +984,17 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
   if (sc.probes.reserve((size_t)nq * nprobe * 8)) return -1;
   if (sc.pdists.reserve((size_t)nq * nprobe * 4)) return -1;
   if (update_dev_buckets(s)) return -1;
+  if (has_opq()) {
+    /* coarse assign + ADC run in rotated space; the exact rerank leg
+     * (Engine level) stays in raw space (ivfpq.cc:585-588, :735) */
+    if (sc.rot_q.reserve((size_t)nq * d_ * 4)) return -1;
+    if (sc.rot_norms.reserve((size_t)nq * 4)) return -1;
+    if (rotate_dev(q_dev, nq, sc.rot_q.as<float>(), s)) return -1;
+    GAMMA_CHECK(gk::row_norms(s, sc.rot_q.as<float>(), nq, d_,
+                              sc.rot_norms.as<float>()));
+    q_dev = sc.rot_q.as<float>();
+    q_norms_dev = sc.rot_norms.as<float>();
+  }
 
   struct Ev3 { /* RAII so error paths cannot leak events */
     hipEvent_t e[3];
@@ -867,6 +1095,8 @@ int IVFIndex::dump(FILE *f, hipStream_t s) const {
     (void)hipMemcpy(books.data(), codebooks_.get(), books.size() * 4,
               hipMemcpyDeviceToHost);
     fwrite(books.data(), 4, books.size(), f);
+    if (has_opq()) /* write_opq analog (ivfpq.cc:1040) */
+      fwrite(opq_R_host_.data(), 4, opq_R_host_.size(), f);
   }
   const size_t entry =
       params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;
@@ -910,6 +1140,15 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
                         codebooks_.as<float>(),
                         btable_.as<float>()) != hipSuccess)
       return -1;
+    if (has_opq()) { /* read_opq analog (ivfpq.cc:1079-1080) */
+      opq_R_host_.resize((size_t)d_ * d_);
+      if (fread(opq_R_host_.data(), 4, opq_R_host_.size(), f) !=
+          opq_R_host_.size())
+        return -1;
+      if (opq_R_.reserve(opq_R_host_.size() * 4)) return -1;
+      (void)hipMemcpy(opq_R_.get(), opq_R_host_.data(),
+                opq_R_host_.size() * 4, hipMemcpyHostToDevice);
+    }
   }
   const size_t entry =
       params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;

@@ -145,6 +145,10 @@ struct IndexParams {
   int bucket_max_size = 1280000;
   int training_threshold = 0;
   bool metric_ip = true;       /* reference default INNER_PRODUCT */
+  /* OPQ pre-rotation (ivfpq.h:1043-1044 has_opq/opq_nsubvector;
+   * OPQMatrix(d, opq_nsubvector, d) at ivfpq.cc:177) */
+  bool has_opq = false;
+  int opq_nsubvector = 0;
   /* parse the index-params JSON (ivfpq.h:1065 Parse); empty ok */
   int parse(const std::string &json, std::string *err);
 };
@@ -168,6 +172,7 @@ struct SearchScratch {
   std::vector<uint32_t> filt_host;
   /* index-side (coarse assign + ADC scan) */
   DeviceBuf dots, sel_keys, probes, pdists, atab;
+  DeviceBuf rot_q, rot_norms; /* OPQ-rotated queries + their norms */
   bool in_use = false;
   ~SearchScratch() {
     if (stream) (void)hipStreamDestroy(stream);
@@ -202,6 +207,11 @@ class IVFIndex {
   int code_size() const { return code_size_; }
   int copy_model_to_host(float *centroids, float *codebooks,
                          hipStream_t s) const;
+  bool has_opq() const { return params_.has_opq && params_.kind == IndexKind::IVFPQ; }
+  const std::vector<float> &opq_R_host() const { return opq_R_host_; }
+  /* y = R x for n row vectors, device to device (chunked MFMA GEMM) */
+  int rotate_dev(const float *x_dev, int64_t n, float *y_dev,
+                 hipStream_t s) const;
   int64_t list_size(int64_t ln) const;
   int copy_list_to_host(int64_t ln, int64_t *ids, uint8_t *codes,
                         hipStream_t s) const;
@@ -216,12 +226,19 @@ class IVFIndex {
                           std::vector<float> &cb, hipStream_t s,
                           int seed_off);
   int update_dev_buckets(hipStream_t s);
+  /* OPQ-NP training: alternate PQ fit and orthogonal Procrustes
+   * (SVD), the algorithm behind faiss OPQMatrix::train (the reference
+   * trains/applies it at ivfpq.cc:362-364, 470-471, 585-588) */
+  int train_opq_(const float *xt, int64_t n, hipStream_t s,
+                 std::string *err);
   IndexParams params_;
   bool trained_ = false;
   int d_ = 0, M_ = 0, ksub_ = 256, dsub_ = 0, code_size_ = 0, nlist_ = 0;
   int64_t ntotal_ = 0;
   DeviceBuf centroids_, cent_norms_, codebooks_;
   DeviceBuf btable_; /* pct1 B table: nlist x M x ksub f32 */
+  DeviceBuf opq_R_;  /* d x d rotation, row-major: y_j = R[j] . x */
+  std::vector<float> opq_R_host_;
   struct Bucket {
     std::unique_ptr<DeviceBuf> ids, data;
     long long size = 0, cap = 0;

@@ -65,6 +65,8 @@ def lib():
         L.GammaDebugCoarseAssign.argtypes = [c.c_void_p, c.c_int, f32p,
                                              c.c_int, i64p, f32p]
         L.GammaDebugGetModel.argtypes = [c.c_void_p, f32p, f32p]
+        L.GammaDebugGetOPQ.argtypes = [c.c_void_p, f32p]
+        L.GammaDebugApplyOPQ.argtypes = [c.c_void_p, f32p, c.c_int, f32p]
         L.GammaDebugGetList.restype = c.c_int64
         L.GammaDebugGetList.argtypes = [c.c_void_p, c.c_int64, i64p, u8p]
         L.GammaDebugNumDocs.restype = c.c_int64
@@ -262,6 +264,22 @@ class GammaEngine:
         if rc != 0:
             raise RuntimeError("debug_model failed")
         return cent, books
+
+    def debug_opq(self, d):
+        """d x d OPQ rotation R (row-major, y = R x); raises if none."""
+        R = np.zeros((d, d), dtype=np.float32)
+        if lib().GammaDebugGetOPQ(self.h, _fp(R)) != 0:
+            raise RuntimeError("no OPQ")
+        return R
+
+    def debug_apply_opq(self, xq):
+        """engine-GPU-rotated queries (bit-exact input for the oracle)."""
+        xq = np.ascontiguousarray(xq, dtype=np.float32)
+        out = np.zeros_like(xq)
+        if lib().GammaDebugApplyOPQ(self.h, _fp(xq), xq.shape[0],
+                                    _fp(out)) != 0:
+            raise RuntimeError("no OPQ")
+        return out
 
     def debug_list(self, list_no, code_size):
         n = lib().GammaDebugGetList(self.h, list_no, None, None)

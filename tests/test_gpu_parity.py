@@ -514,7 +514,7 @@ def test_filter_index_incremental_and_stringarray(data):
     # incremental rows after the first (index-building) filter call
     for vid in range(1000, 2000):
         addd(vid, b"b", vid)
-    addd("7", b"zzz", 7)  # update: docid 7 re-keyed, old row dead
+    addd(7, b"zzz", 7)  # update: pkey "7" re-keyed, old row dead
     eng.delete_doc("1002")
     res = eng.search_pb(q[:4], topn=300, term_filters=[("tags", b"b")])
     got = set()

@@ -63,6 +63,16 @@ WORKLOADS = {
 
 HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
 
+# Measured HBM bytes per scan-kernel launch (rocprofv3 --pmc FETCH_SIZE,
+# counters-only pass per the MI355X guide; raw KB x2 gfx950 wide-read
+# correction). Collected on the exact workload shape keyed here; see
+# profiles/r01_fetch_size_c1.csv and profiles/README.md. Any other
+# shape reports traffic=null rather than guessing.
+MEASURED_TRAFFIC_BYTES = {
+    # 56.09 GB raw/dispatch x2 @ nq=10000/step (north-star workload)
+    "ivfpq_d128_n10m_nprobe32": 112.2e9,
+}
+
 
 def log(msg):
     if int(os.environ.get("RANK", "0")) == 0:
@@ -359,7 +369,9 @@ def main():
                 "peak": HBM_PEAK_GBS,
                 "unit": "GB/s",
                 "frac": round(achieved / HBM_PEAK_GBS, 4),
-                "traffic": None,
+                "traffic": MEASURED_TRAFFIC_BYTES.get(args.workload)
+                if cfg["nq"] == WORKLOADS.get(args.workload, {}).get("nq")
+                and not args.db_size else None,
             }
         cpu = None
         if not args.skip_cpu_baseline and world == 1:

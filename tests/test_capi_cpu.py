@@ -105,6 +105,50 @@ def test_doc_codec_roundtrip(L):
     assert rc == 0, f"doc roundtrip failed rc={rc}"
 
 
+def test_search_request_codec_randomized(L):
+    """Property-style: 60 random valid SearchRequests round-trip every
+    header field through the C parser (arbitrary names/sizes/filters —
+    the Go router composes requests with all of these)."""
+    import random
+    rng = random.Random(99)
+    L.GammaTestParseSearchRequest.argtypes = [
+        ctypes.c_char_p, ctypes.c_int, ctypes.POINTER(ctypes.c_char_p),
+        ctypes.POINTER(ctypes.c_int)]
+    for trial in range(60):
+        d = rng.choice([1, 4, 63, 128, 770])
+        nvec = 1
+        name = "".join(rng.choice("abz_09") for _ in range(
+            rng.randint(1, 24)))
+        rid = "".join(rng.choice("r0-9") for _ in range(rng.randint(0, 12)))
+        topn = rng.randint(1, 1000)
+        req_num = rng.randint(1, 64)
+        pid = rng.randint(0, 1 << 20)
+        nterm = rng.randint(0, 3)
+        nrange = rng.randint(0, 3)
+        terms = [("f%d" % i,
+                  bytes(rng.randrange(256) for _ in range(
+                      rng.randint(0, 9))))
+                 for i in range(nterm)]
+        ranges = [("g%d" % i,
+                   bytes(rng.randrange(256) for _ in range(4)),
+                   bytes(rng.randrange(256) for _ in range(4)),
+                   bool(rng.getrandbits(1)), bool(rng.getrandbits(1)))
+                  for i in range(nrange)]
+        q = np.arange(d * req_num, dtype=np.float32) + trial
+        buf = proto.encode_search_request(
+            name, q.tobytes(), topn=topn, req_num=req_num,
+            request_id=rid, partition_id=pid,
+            index_params='{"nprobe": %d}' % rng.randint(1, 512),
+            term_filters=terms, range_filters=ranges)
+        j = _parse(L, L.GammaTestParseSearchRequest, buf)
+        assert j["request_id"] == rid
+        assert j["partition_id"] == pid
+        assert j["req_num"] == req_num and j["topn"] == topn
+        assert j["vec_name"] == name
+        assert j["vec_bytes"] == d * req_num * 4
+        assert j["n_filters"] == nterm + nrange
+
+
 def test_init_without_gpu_fails_loudly(L):
     """On a GPU-less box the engine must refuse to start (no silent CPU
     fallback)."""

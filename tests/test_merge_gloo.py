@@ -22,6 +22,13 @@ def _worker(rank, world, port, out_q):
     dist.init_process_group("gloo", rank=rank, world_size=world)
     base = orc.gen_clustered(4000, 32, seed=42, ncl=50)
     q = orc.gen_queries(base, 16, seed=3)
+    try:
+        _worker_body(rank, world, base, q, out_q)
+    finally:
+        dist.destroy_process_group()
+
+
+def _worker_body(rank, world, base, q, out_q):
     # round-robin sharding: rank owns rows where i % world == rank
     my_rows = np.arange(rank, 4000, world)
     my_base = base[my_rows]
@@ -39,15 +46,16 @@ def _worker(rank, world, port, out_q):
         md, mi = merge_topk(dl, il, K)
         od, oi = orc.flat_search(base, q, K, "L2")
         out_q.put((np.array_equal(mi, oi), np.allclose(md, od)))
-    dist.destroy_process_group()
 
 
-def test_two_partition_merge_equals_global():
+@pytest.mark.parametrize("world,port", [(2, 29611), (4, 29617)])
+def test_partition_merge_equals_global(world, port):
+    """world partitions (2 and 4 — the 8-GPU merge is the same code
+    with a bigger gather) must merge to the global FLAT result."""
     ctx = mp.get_context("spawn")
     out_q = ctx.Queue()
-    port = 29611
-    procs = [ctx.Process(target=_worker, args=(r, 2, port, out_q))
-             for r in range(2)]
+    procs = [ctx.Process(target=_worker, args=(r, world, port, out_q))
+             for r in range(world)]
     for p in procs:
         p.start()
     ok_ids, ok_dists = out_q.get(timeout=300)

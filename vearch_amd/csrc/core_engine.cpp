@@ -488,12 +488,12 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
     hipEvent_t ev[6];
     hipStream_t s;
     explicit Timer(hipStream_t s_) : s(s_) {
-      for (auto &e : ev) hipEventCreate(&e);
+      for (auto &e : ev) (void)hipEventCreate(&e);
     }
     ~Timer() {
-      for (auto &e : ev) hipEventDestroy(e);
+      for (auto &e : ev) (void)hipEventDestroy(e);
     }
-    void rec(int i) { hipEventRecord(ev[i], s); }
+    void rec(int i) { (void)hipEventRecord(ev[i], s); }
     double ms(int a, int b) {
       float m = 0;
       (void)hipEventElapsedTime(&m, ev[a], ev[b]);

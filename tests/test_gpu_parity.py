@@ -662,6 +662,29 @@ def test_concurrent_searches_deterministic(data):
     eng.close()
 
 
+def test_flat_chunked_gemm_seeded_select_bitexact():
+    """FLAT at nq>=512 and n>=200k takes the chunked-GEMM route: one
+    seeded select per 65536-vector chunk accumulates the running top-k
+    across chunks. With small k (no rerank leg) that select runs on the
+    wave-per-query path with seeding — cover it bit-exactly vs the
+    oracle (ids AND canonicalized distances)."""
+    base = orc.gen_clustered(260000, 32, seed=11, ncl=300)
+    q = orc.gen_queries(base, 512, seed=12)
+    eng = make_engine("/tmp/gamma_flat_big")
+    eng.create_table(32, "FLAT", '{"metric_type": "L2"}')
+    eng.add(base)
+    gd, gi = eng.raw_search(q, 10, rerank=0)
+    od, oi = orc.flat_search(base, q, 10, "L2")
+    assert np.array_equal(gi, oi)
+    assert np.array_equal(gd, od)
+    # and the IP metric through the same route
+    gdi, gii = eng.raw_search(q, 10, rerank=0, metric=2)
+    odi, oii = orc.flat_search(base, q, 10, "IP")
+    assert np.array_equal(gii, oii)
+    assert np.array_equal(gdi, odi)
+    eng.close()
+
+
 def test_opq_train_search_parity(data):
     """OPQ pre-rotation (ivfpq.cc:168-177 params, :362-364 train,
     :585-588 search, :735 raw-space rerank): R is orthonormal, the

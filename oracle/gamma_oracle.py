@@ -334,7 +334,12 @@ def flat_search(base, queries, k, metric="L2", del_bitmap=None):
     nq = queries.shape[0]
     dists = np.empty((nq, k), dtype=np.float32)
     ids = np.empty((nq, k), dtype=np.int64)
-    mip = 1 if metric == "InnerProduct" else 0
+    if metric in ("InnerProduct", "IP"):
+        mip = 1
+    elif metric == "L2":
+        mip = 0
+    else:
+        raise ValueError(f"unknown metric {metric!r}")
     RefLib.lib().oracle_flat_search(
         base.shape[0], base.shape[1], _fp(base), nq, _fp(queries), k,
         _up8(del_bitmap), mip, _fp(dists), _ip64(ids))

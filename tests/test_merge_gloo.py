@@ -114,3 +114,43 @@ def test_torch_pack_matches_numpy_pack():
                                      torch.from_numpy(i), world,
                                      rank).numpy()
         assert np.array_equal(want, got)
+
+
+def test_merge_topk_property_random():
+    """Property check over random shapes: merge_topk == brute-force
+    concatenate+stable-sort by (dist, id), empties last, -1 padding;
+    both metric orders."""
+    rng = np.random.default_rng(42)
+    for trial in range(200):
+        nq = int(rng.integers(1, 8))
+        nparts = int(rng.integers(1, 5))
+        k = int(rng.integers(1, 12))
+        descending = bool(rng.integers(0, 2))
+        dl, il = [], []
+        for _ in range(nparts):
+            kp = int(rng.integers(0, 9))
+            d = rng.standard_normal((nq, kp)).astype(np.float32)
+            # inject exact ties across partitions
+            d[rng.random((nq, kp)) < 0.3] = np.float32(0.25)
+            i = rng.integers(0, 5000, (nq, kp)).astype(np.int64)
+            i[rng.random((nq, kp)) < 0.2] = -1  # empty slots
+            # partition results arrive sorted best-first like the PS
+            key = np.where(i < 0, np.inf,
+                           -d if descending else d)
+            order = np.argsort(key, axis=1, kind="stable")
+            dl.append(np.take_along_axis(d, order, 1))
+            il.append(np.take_along_axis(i, order, 1))
+        md, mi = merge_topk(dl, il, k, descending=descending)
+        assert md.shape == (nq, k) and mi.shape == (nq, k)
+        for t in range(nq):
+            cand = []
+            for d, i in zip(dl, il):
+                for j in range(d.shape[1]):
+                    if i[t, j] >= 0:
+                        cand.append((float(d[t, j]), int(i[t, j])))
+            cand.sort(key=lambda p: (-p[0] if descending else p[0], p[1]))
+            want = cand[:k]
+            got = [(float(md[t, j]), int(mi[t, j]))
+                   for j in range(k) if mi[t, j] >= 0]
+            assert got == want, (trial, t, got, want)
+            assert all(mi[t, j] == -1 for j in range(len(want), k))

@@ -55,7 +55,7 @@ static bool memory_exceeded() {
 
 extern "C" {
 
-void *Init(const char *config_str, int len) {
+static void *Init_unguarded(const char *config_str, int len) {
   auto *e = new Engine();
   std::string err;
   if (e->init(std::string(config_str ? config_str : "", len > 0 ? len : 0),
@@ -67,13 +67,22 @@ void *Init(const char *config_str, int len) {
   return e;
 }
 
+void *Init(const char *config_str, int len) {
+  try {
+    return Init_unguarded(config_str, len);
+  } catch (...) {
+    return nullptr;
+  }
+}
+
+
 int Close(void *engine) {
   if (!engine) return 1;
   delete static_cast<Engine *>(engine);
   return 0;
 }
 
-struct CStatus CreateTable(void *engine, const char *table_str, int len) {
+static struct CStatus CreateTable_unguarded(void *engine, const char *table_str, int len) {
   if (!engine) return err_status(1, "null engine");
   gfb::TableSchema ts;
   if (!ts.parse(table_str, (size_t)len))
@@ -105,7 +114,19 @@ struct CStatus CreateTable(void *engine, const char *table_str, int len) {
   return ok_status();
 }
 
-int AddOrUpdateDoc(void *engine, const char *doc_str, int len) {
+struct CStatus CreateTable(void *engine, const char *table_str, int len) {
+  try {
+    return CreateTable_unguarded(engine, table_str, len);
+  } catch (const std::exception &ex) {
+    return err_status(-1,
+                      std::string("internal error: ") + ex.what());
+  } catch (...) {
+    return err_status(-1, "internal error");
+  }
+}
+
+
+static int AddOrUpdateDoc_unguarded(void *engine, const char *doc_str, int len) {
   if (!engine) return -1;
   if (memory_exceeded()) return -1;
   auto *e = static_cast<Engine *>(engine);
@@ -129,11 +150,29 @@ int AddOrUpdateDoc(void *engine, const char *doc_str, int len) {
   return e->add_doc(p_key, fields, vec, vec_len);
 }
 
-int DeleteDoc(void *engine, const char *docid, int docid_len) {
+int AddOrUpdateDoc(void *engine, const char *doc_str, int len) {
+  try {
+    return AddOrUpdateDoc_unguarded(engine, doc_str, len);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int DeleteDoc_unguarded(void *engine, const char *docid, int docid_len) {
   if (!engine) return -1;
   return static_cast<Engine *>(engine)->delete_doc(
       std::string(docid, docid_len));
 }
+
+int DeleteDoc(void *engine, const char *docid, int docid_len) {
+  try {
+    return DeleteDoc_unguarded(engine, docid, docid_len);
+  } catch (...) {
+    return -1;
+  }
+}
+
 
 void GetEngineStatus(void *engine, char **status, int *len) {
   std::string s = engine ? static_cast<Engine *>(engine)->status_json()
@@ -175,7 +214,7 @@ static int serialize_doc(Engine *e, int64_t docid, char **doc_str,
   return 0;
 }
 
-int GetDocByID(void *engine, const char *docid, int docid_len,
+static int GetDocByID_unguarded(void *engine, const char *docid, int docid_len,
                char **doc_str, int *len) {
   if (!engine) return -1;
   auto *e = static_cast<Engine *>(engine);
@@ -184,7 +223,17 @@ int GetDocByID(void *engine, const char *docid, int docid_len,
   return serialize_doc(e, id, doc_str, len);
 }
 
-int GetDocByDocID(void *engine, int docid, char next, char **doc_str,
+int GetDocByID(void *engine, const char *docid, int docid_len,
+               char **doc_str, int *len) {
+  try {
+    return GetDocByID_unguarded(engine, docid, docid_len, doc_str, len);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int GetDocByDocID_unguarded(void *engine, int docid, char next, char **doc_str,
                   int *len) {
   if (!engine) return -1;
   auto *e = static_cast<Engine *>(engine);
@@ -197,7 +246,17 @@ int GetDocByDocID(void *engine, int docid, char next, char **doc_str,
   return serialize_doc(e, id, doc_str, len);
 }
 
-int BuildIndex(void *engine) {
+int GetDocByDocID(void *engine, int docid, char next, char **doc_str,
+                  int *len) {
+  try {
+    return GetDocByDocID_unguarded(engine, docid, next, doc_str, len);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int BuildIndex_unguarded(void *engine) {
   if (!engine) return -1;
   std::string err;
   int rc = static_cast<Engine *>(engine)->build_index(&err);
@@ -205,7 +264,16 @@ int BuildIndex(void *engine) {
   return rc;
 }
 
-int RebuildIndex(void *engine, int drop_before_rebuild, int limit_cpu,
+int BuildIndex(void *engine) {
+  try {
+    return BuildIndex_unguarded(engine);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int RebuildIndex_unguarded(void *engine, int drop_before_rebuild, int limit_cpu,
                  int describe) {
   (void)limit_cpu; /* GPU engine has no CPU throttle */
   (void)describe;
@@ -217,7 +285,17 @@ int RebuildIndex(void *engine, int drop_before_rebuild, int limit_cpu,
   return rc;
 }
 
-int Dump(void *engine) {
+int RebuildIndex(void *engine, int drop_before_rebuild, int limit_cpu,
+                 int describe) {
+  try {
+    return RebuildIndex_unguarded(engine, drop_before_rebuild, limit_cpu, describe);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int Dump_unguarded(void *engine) {
   if (!engine) return -1;
   std::string err;
   int rc = static_cast<Engine *>(engine)->dump(&err);
@@ -225,7 +303,16 @@ int Dump(void *engine) {
   return rc;
 }
 
-int Load(void *engine) {
+int Dump(void *engine) {
+  try {
+    return Dump_unguarded(engine);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int Load_unguarded(void *engine) {
   if (!engine) return -1;
   std::string err;
   int rc = static_cast<Engine *>(engine)->load(&err);
@@ -233,7 +320,16 @@ int Load(void *engine) {
   return rc;
 }
 
-struct CStatus Search(void *engine, const char *request_str, int req_len,
+int Load(void *engine) {
+  try {
+    return Load_unguarded(engine);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static struct CStatus Search_unguarded(void *engine, const char *request_str, int req_len,
                       char **response_str, int *res_len) {
   if (!engine) return err_status(1, "null engine");
   if (memory_exceeded())
@@ -336,7 +432,20 @@ struct CStatus Search(void *engine, const char *request_str, int req_len,
   return ok_status();
 }
 
-struct CStatus Query(void *engine, const char *request_str, int req_len,
+struct CStatus Search(void *engine, const char *request_str, int req_len,
+                      char **response_str, int *res_len) {
+  try {
+    return Search_unguarded(engine, request_str, req_len, response_str, res_len);
+  } catch (const std::exception &ex) {
+    return err_status(-1,
+                      std::string("internal error: ") + ex.what());
+  } catch (...) {
+    return err_status(-1, "internal error");
+  }
+}
+
+
+static struct CStatus Query_unguarded(void *engine, const char *request_str, int req_len,
                      char **response_str, int *res_len) {
   if (!engine) return err_status(1, "null engine");
   auto *e = static_cast<Engine *>(engine);
@@ -390,27 +499,70 @@ struct CStatus Query(void *engine, const char *request_str, int req_len,
   return ok_status();
 }
 
-int SetConfig(void *engine, const char *config_str, int len) {
+struct CStatus Query(void *engine, const char *request_str, int req_len,
+                     char **response_str, int *res_len) {
+  try {
+    return Query_unguarded(engine, request_str, req_len, response_str, res_len);
+  } catch (const std::exception &ex) {
+    return err_status(-1,
+                      std::string("internal error: ") + ex.what());
+  } catch (...) {
+    return err_status(-1, "internal error");
+  }
+}
+
+
+static int SetConfig_unguarded(void *engine, const char *config_str, int len) {
   (void)engine;
   (void)config_str;
   (void)len;
   return 0; /* cache sizes are N/A for the GPU engine */
 }
 
-int GetConfig(void *engine, char **config_str, int *len) {
+int SetConfig(void *engine, const char *config_str, int len) {
+  try {
+    return SetConfig_unguarded(engine, config_str, len);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int GetConfig_unguarded(void *engine, char **config_str, int *len) {
   std::string s = "{\"path\": \"\", \"cache_sizes\": []}";
   *config_str = dup_malloc(s);
   *len = (int)s.size();
   return 0;
 }
 
-struct CStatus Backup(void *engine, int command) {
+int GetConfig(void *engine, char **config_str, int *len) {
+  try {
+    return GetConfig_unguarded(engine, config_str, len);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static struct CStatus Backup_unguarded(void *engine, int command) {
   (void)engine;
   (void)command;
   return err_status(1, "backup not supported this round");
 }
 
-struct CStatus AddFieldIndexWithParams(
+struct CStatus Backup(void *engine, int command) {
+  try {
+    return Backup_unguarded(engine, command);
+  } catch (const std::exception &ex) {
+    return err_status(-1,
+                      std::string("internal error: ") + ex.what());
+  } catch (...) {
+    return err_status(-1, "internal error");
+  }
+}
+
+
+static struct CStatus AddFieldIndexWithParams_unguarded(
     void *engine, const char *index_name, int index_name_len,
     const char *const *field_names, const int *field_name_lens,
     int field_name_count, const char *index_type, int index_type_len,
@@ -430,13 +582,42 @@ struct CStatus AddFieldIndexWithParams(
                     "scope this round");
 }
 
-struct CStatus RemoveFieldIndex(void *engine, const char *index_name,
+struct CStatus AddFieldIndexWithParams(
+    void *engine, const char *index_name, int index_name_len,
+    const char *const *field_names, const int *field_name_lens,
+    int field_name_count, const char *index_type, int index_type_len,
+    const char *index_params, int index_params_len) {
+  try {
+    return AddFieldIndexWithParams_unguarded(engine, index_name, index_name_len, field_names, field_name_lens, field_name_count, index_type, index_type_len, index_params, index_params_len);
+  } catch (const std::exception &ex) {
+    return err_status(-1,
+                      std::string("internal error: ") + ex.what());
+  } catch (...) {
+    return err_status(-1, "internal error");
+  }
+}
+
+
+static struct CStatus RemoveFieldIndex_unguarded(void *engine, const char *index_name,
                                 int index_name_len) {
   (void)engine;
   (void)index_name;
   (void)index_name_len;
   return err_status(1, "dynamic field indexes are out of scope this round");
 }
+
+struct CStatus RemoveFieldIndex(void *engine, const char *index_name,
+                                int index_name_len) {
+  try {
+    return RemoveFieldIndex_unguarded(engine, index_name, index_name_len);
+  } catch (const std::exception &ex) {
+    return err_status(-1,
+                      std::string("internal error: ") + ex.what());
+  } catch (...) {
+    return err_status(-1, "internal error");
+  }
+}
+
 
 void SetMemoryLimitConfig(int memory_limit) {
   g_memory_limit_mb = memory_limit;
@@ -453,7 +634,7 @@ void DeleteKillStatus(const char *request_id, int partition_id) {
 
 /* ----------------------------------------------------- bench extensions */
 
-int GammaBulkAdd(void *engine, const char *field, int field_len, int n,
+static int GammaBulkAdd_unguarded(void *engine, const char *field, int field_len, int n,
                  const float *vecs) {
   (void)field;
   (void)field_len;
@@ -461,7 +642,17 @@ int GammaBulkAdd(void *engine, const char *field, int field_len, int n,
   return static_cast<Engine *>(engine)->bulk_add(n, vecs);
 }
 
-int GammaRawSearch(void *engine, int nq, const float *xq, int k, int nprobe,
+int GammaBulkAdd(void *engine, const char *field, int field_len, int n,
+                 const float *vecs) {
+  try {
+    return GammaBulkAdd_unguarded(engine, field, field_len, n, vecs);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int GammaRawSearch_unguarded(void *engine, int nq, const float *xq, int k, int nprobe,
                    int rerank, int metric, float *out_dists,
                    int64_t *out_ids) {
   if (!engine) return -1;
@@ -469,12 +660,32 @@ int GammaRawSearch(void *engine, int nq, const float *xq, int k, int nprobe,
       nq, xq, k, nprobe, rerank, metric, false, "", 0, out_dists, out_ids);
 }
 
-int GammaCacheQueries(void *engine, int nq, const float *xq) {
+int GammaRawSearch(void *engine, int nq, const float *xq, int k, int nprobe,
+                   int rerank, int metric, float *out_dists,
+                   int64_t *out_ids) {
+  try {
+    return GammaRawSearch_unguarded(engine, nq, xq, k, nprobe, rerank, metric, out_dists, out_ids);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int GammaCacheQueries_unguarded(void *engine, int nq, const float *xq) {
   if (!engine) return -1;
   return static_cast<Engine *>(engine)->cache_queries(nq, xq);
 }
 
-int GammaRawSearchCached(void *engine, int nq, int k, int nprobe,
+int GammaCacheQueries(void *engine, int nq, const float *xq) {
+  try {
+    return GammaCacheQueries_unguarded(engine, nq, xq);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int GammaRawSearchCached_unguarded(void *engine, int nq, int k, int nprobe,
                          int rerank, int metric, float *out_dists,
                          int64_t *out_ids) {
   if (!engine) return -1;
@@ -482,6 +693,17 @@ int GammaRawSearchCached(void *engine, int nq, int k, int nprobe,
       nq, nullptr, k, nprobe, rerank, metric, false, "", 0, out_dists,
       out_ids);
 }
+
+int GammaRawSearchCached(void *engine, int nq, int k, int nprobe,
+                         int rerank, int metric, float *out_dists,
+                         int64_t *out_ids) {
+  try {
+    return GammaRawSearchCached_unguarded(engine, nq, k, nprobe, rerank, metric, out_dists, out_ids);
+  } catch (...) {
+    return -1;
+  }
+}
+
 
 int GammaDebugCoarseAssign(void *engine, int nq, const float *xq,
                            int nprobe, int64_t *out_lists,

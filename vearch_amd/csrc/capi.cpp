@@ -801,7 +801,7 @@ int GammaLastSearchTiming(void *engine, double *us6) {
  * Used by tests/test_capi_cpu.py to exercise the hand-written protobuf /
  * FlatBuffers codecs without a GPU. */
 
-int GammaTestParseSearchRequest(const char *buf, int len, char **json_out,
+static int GammaTestParseSearchRequest_unguarded(const char *buf, int len, char **json_out,
                                 int *json_len) {
   gpb::SearchRequest req;
   if (!req.parse(buf, len)) return -1;
@@ -830,7 +830,17 @@ int GammaTestParseSearchRequest(const char *buf, int len, char **json_out,
   return 0;
 }
 
-int GammaTestParseTable(const char *buf, int len, char **json_out,
+int GammaTestParseSearchRequest(const char *buf, int len, char **json_out,
+                                int *json_len) {
+  try {
+    return GammaTestParseSearchRequest_unguarded(buf, len, json_out, json_len);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int GammaTestParseTable_unguarded(const char *buf, int len, char **json_out,
                         int *json_len) {
   gfb::TableSchema ts;
   if (!ts.parse(buf, (size_t)len)) return -1;
@@ -852,7 +862,17 @@ int GammaTestParseTable(const char *buf, int len, char **json_out,
   return 0;
 }
 
-int GammaTestDocRoundtrip(const char *buf, int len, char **out,
+int GammaTestParseTable(const char *buf, int len, char **json_out,
+                        int *json_len) {
+  try {
+    return GammaTestParseTable_unguarded(buf, len, json_out, json_len);
+  } catch (...) {
+    return -1;
+  }
+}
+
+
+static int GammaTestDocRoundtrip_unguarded(const char *buf, int len, char **out,
                           int *out_len) {
   gfb::Doc doc;
   if (!doc.parse(buf, (size_t)len)) return -1;
@@ -870,5 +890,15 @@ int GammaTestDocRoundtrip(const char *buf, int len, char **out,
   *out_len = (int)s.size();
   return 0;
 }
+
+int GammaTestDocRoundtrip(const char *buf, int len, char **out,
+                          int *out_len) {
+  try {
+    return GammaTestDocRoundtrip_unguarded(buf, len, out, out_len);
+  } catch (...) {
+    return -1;
+  }
+}
+
 
 } /* extern "C" */

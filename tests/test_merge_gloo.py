@@ -154,3 +154,30 @@ def test_merge_topk_property_random():
                    for j in range(k) if mi[t, j] >= 0]
             assert got == want, (trial, t, got, want)
             assert all(mi[t, j] == -1 for j in range(len(want), k))
+
+
+def test_f32_key_total_order_extremes():
+    """The u64 (dist-key | id) packing (select.hpp sign-flip trick,
+    mirrored in merge._f32_key) must be a strict monotone image of fp32
+    over the full finite+inf range, including denormals and max-mag
+    values. (+-0.0 note: the key image distinguishes -0.0 < +0.0;
+    computed L2^2/IP distances are never -0.0 — sums of products
+    rounding to zero give +0.0 under IEEE round-to-nearest — so this
+    never diverges from the oracle's float compare.)"""
+    from vearch_amd.merge import _f32_key
+    vals = np.array([-np.inf, -3.4e38, -np.pi, -1.0, -1e-45,
+                     0.0, 1e-45, 1.0, np.pi, 3.4e38, np.inf],
+                    dtype=np.float32)
+    keys = _f32_key(vals)
+    assert (np.diff(keys.astype(np.uint64)) > 0).all()
+    assert _f32_key(np.array([-0.0], np.float32))[0] < \
+        _f32_key(np.array([0.0], np.float32))[0]
+    # random finite floats: key order == value order
+    rng = np.random.default_rng(8)
+    v = (rng.standard_normal(5000) *
+         10.0 ** rng.integers(-30, 30, 5000)).astype(np.float32)
+    v = v[np.isfinite(v)]
+    order_v = np.argsort(v, kind="stable")
+    order_k = np.argsort(_f32_key(v).astype(np.uint64), kind="stable")
+    # compare by value sequence (equal values may permute, none here)
+    assert np.array_equal(v[order_v], v[order_k])

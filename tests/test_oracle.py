@@ -255,3 +255,40 @@ def test_pct1_search_recall_equivalent():
     assert overlap >= 0.9, overlap
     _, gti = flat_topk_f64(base, q, 10)
     assert abs(recall_at(gti, i0, 10) - recall_at(gti, i1, 10)) < 0.05
+
+
+def test_pq_encode_tie_lowest_index():
+    """Equal-distance codewords: the encoder must pick the LOWEST index
+    (faiss/our argmin scans first-minimum). Pinned with duplicated
+    codewords so ties are guaranteed."""
+    d, M, ksub = 8, 2, 4
+    rng = np.random.default_rng(5)
+    books = rng.standard_normal((M, ksub, d // M)).astype(np.float32)
+    books[0][2] = books[0][0]  # duplicate codeword: index 0 must win
+    books[1][3] = books[1][1]  # index 1 must win over 3
+    x = (books[0][0].tolist() + books[1][1].tolist())
+    x = np.array([x], dtype=np.float32)
+    codes = np.empty((1, M), dtype=np.uint8)
+    RefLib.lib().oracle_pq_encode(
+        1, d, M, ksub, _fp(_c(x, np.float32)),
+        _fp(_c(books, np.float32)), _up8(codes))
+    assert codes[0, 0] == 0
+    assert codes[0, 1] == 1
+
+
+def test_kmeans_deterministic():
+    """Same data + seed -> identical centroids (training is part of the
+    dump/load parity surface)."""
+    x = gen_clustered(3000, 16, seed=3, ncl=20)
+    c1 = kmeans(x, 8, niter=5, seed=9)
+    c2 = kmeans(x, 8, niter=5, seed=9)
+    assert np.array_equal(c1, c2)
+    c3 = kmeans(x, 8, niter=5, seed=10)
+    assert not np.array_equal(c1, c3)
+
+
+def test_flat_search_rejects_unknown_metric():
+    base = gen_clustered(100, 8, seed=1, ncl=4)
+    q = gen_queries(base, 2, seed=2)
+    with pytest.raises(ValueError):
+        flat_search(base, q, 5, "cosine")

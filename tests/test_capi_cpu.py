@@ -69,6 +69,28 @@ def test_search_request_codec(L):
     assert j["index_params"] == '{"nprobe": 32}'
     assert j["l2_sqrt"] == 1
     assert j["n_fields"] == 2 and j["n_filters"] == 0
+    assert j["op"] == 0 and j["term_unions"] == [] and j["range_unions"] == []
+
+
+def test_search_request_codec_filter_operators(L):
+    """is_union on term/range filters (FilterOperator And=0 Or=1 Not=2,
+    engine.cc:475) and the request-level `operator` (field 17) survive
+    the wire round trip."""
+    q = np.arange(128, dtype=np.float32)
+    buf = proto.encode_search_request(
+        "emb", q.tobytes(), topn=5, req_num=1,
+        term_filters=[("tag", b"a", 2), ("tag2", b"b")],
+        range_filters=[("num", b"\x00" * 4, b"\x00" * 4, True, True, 2),
+                       ("num2", b"\x00" * 4, b"\x01" * 4, False, True)],
+        operator=1)
+    L.GammaTestParseSearchRequest.argtypes = [
+        ctypes.c_char_p, ctypes.c_int, ctypes.POINTER(ctypes.c_char_p),
+        ctypes.POINTER(ctypes.c_int)]
+    j = _parse(L, L.GammaTestParseSearchRequest, buf)
+    assert j["op"] == 1
+    assert j["term_unions"] == [2, 0]
+    assert j["range_unions"] == [2, 0]
+    assert j["n_filters"] == 4
 
 
 def test_table_codec(L):

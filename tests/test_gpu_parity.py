@@ -410,6 +410,16 @@ def test_edge_cases(data):
     eng2.build_index()
     gd, gi = eng2.raw_search(q[:4], 5, nprobe=999)
     assert (gi[:, :5] >= 0).all()
+    # unknown metric_type is a parse error, not a silent InnerProduct
+    with pytest.raises(RuntimeError, match="metric_type"):
+        eng2.search_pb(q[:2], topn=5,
+                       index_params='{"metric_type": "cosine"}')
+    # k2 over the selector cap names the limit instead of "search failed"
+    with pytest.raises(RuntimeError, match="1024"):
+        eng2.search_pb(q[:2], topn=1100)
+    with pytest.raises(RuntimeError, match="1024"):
+        eng2.search_pb(q[:2], topn=5,
+                       index_params='{"recall_num": 2000}')
     eng2.close()
 
 
@@ -484,6 +494,71 @@ def test_filtered_search_term_and_range(data):
         assert all(i % 4 == 1 and 10 <= i % 100 < 40 for i in ids)
     with pytest.raises(RuntimeError):
         eng.search_pb(q[:2], topn=5, term_filters=[("nope", b"x")])
+    eng.close()
+
+
+def test_filter_not_and_or(data):
+    """FilterOperator semantics (scalar_index_types.h:44 And=0 Or=1
+    Not=2): per-filter Not on terms (BitmapIndex::NotIn — docs missing
+    the value, including docs with no match at all, pass), numeric
+    NotEqual (Not + equal inclusive bounds, bitmap_index.cc:196), and
+    the request-level `operator` OR union
+    (scalar_index_manager.cc:1188-1190)."""
+    import struct
+    from vearch_amd import fbsenc
+    base, q = data
+    eng = make_engine("/tmp/gamma_filter_notor")
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}',
+                     scalar_fields=[("tag", fbsenc.DATA_STRING),
+                                    ("num", fbsenc.DATA_INT)])
+    n = 2000
+    for vid in range(n):
+        eng.add_doc(str(vid), base[vid],
+                    fields=[("tag", b"grp%d" % (vid % 4),
+                             fbsenc.DATA_STRING),
+                            ("num", struct.pack("<i", vid % 50),
+                             fbsenc.DATA_INT)])
+
+    def check(res, nq, pred):
+        for t in range(nq):
+            keep = [vid for vid in range(n) if pred(vid)]
+            bm = np.zeros((n + 7) // 8, dtype=np.uint8)
+            for vid in range(n):
+                if not pred(vid):
+                    bm[vid >> 3] |= 1 << (vid & 7)
+            od, oi = orc.flat_search(base[:n], q[t:t + 1], 10, "L2",
+                                     del_bitmap=bm)
+            ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+            assert ids == [i for i in oi[0].tolist() if i >= 0]
+            assert all(pred(i) for i in ids)
+
+    # NOT-IN term filter (is_union=2): everything except grp1/grp2
+    res = eng.search_pb(q[:4], topn=10,
+                        term_filters=[("tag", b"grp1\x01grp2", 2)])
+    check(res, 4, lambda v: v % 4 not in (1, 2))
+    # numeric NotEqual: num != 7
+    seven = struct.pack("<i", 7)
+    res = eng.search_pb(q[:4], topn=10,
+                        range_filters=[("num", seven, seven, True, True,
+                                        2)])
+    check(res, 4, lambda v: v % 50 != 7)
+    # request-level OR: tag==grp1 OR 10<=num<=12
+    res = eng.search_pb(
+        q[:4], topn=10, operator=1,
+        term_filters=[("tag", b"grp1")],
+        range_filters=[("num", struct.pack("<i", 10),
+                        struct.pack("<i", 12), True, True)])
+    check(res, 4, lambda v: v % 4 == 1 or 10 <= v % 50 <= 12)
+    # AND of a Not with a plain term: tag==grp1 AND num != 7
+    res = eng.search_pb(
+        q[:4], topn=10, term_filters=[("tag", b"grp1")],
+        range_filters=[("num", seven, seven, True, True, 2)])
+    check(res, 4, lambda v: v % 4 == 1 and v % 50 != 7)
+    # Not filter through the Query browse path too
+    docs = eng.query_pb(term_filters=[("tag", b"grp0\x01grp1\x01grp2", 2)],
+                        limit=50)
+    ids = [int(it["fields"]["_id"]) for it in docs[0]["items"]]
+    assert ids == [v for v in range(n) if v % 4 == 3][:50]
     eng.close()
 
 

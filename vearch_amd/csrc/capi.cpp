@@ -218,6 +218,7 @@ static int GetDocByID_unguarded(void *engine, const char *docid, int docid_len,
                char **doc_str, int *len) {
   if (!engine) return -1;
   auto *e = static_cast<Engine *>(engine);
+  auto rlock = e->read_lock(); /* stable doc state during serialization */
   int64_t id = e->docid_of(std::string(docid, docid_len));
   if (id < 0 || e->bitmap().test(id)) return -1;
   return serialize_doc(e, id, doc_str, len);
@@ -237,6 +238,7 @@ static int GetDocByDocID_unguarded(void *engine, int docid, char next, char **do
                   int *len) {
   if (!engine) return -1;
   auto *e = static_cast<Engine *>(engine);
+  auto rlock = e->read_lock(); /* stable doc state during serialization */
   int64_t id = docid;
   if (next) { /* next undeleted doc after docid (gamma_api.h:86) */
     id++;
@@ -358,8 +360,17 @@ static struct CStatus Search_unguarded(void *engine, const char *request_str, in
       v.get_int("nprobe", nprobe);
       v.get_int("recall_num", recall_num);
       std::string mt;
-      if (v.get_str("metric_type", mt))
-        metric = strcasecmp(mt.c_str(), "L2") == 0 ? 1 : 2;
+      if (v.get_str("metric_type", mt)) {
+        /* only the metric names the reference knows (ivfpq.cc:253-257,
+         * DistanceMetricType): anything else is a parse error, not a
+         * silent InnerProduct */
+        if (strcasecmp(mt.c_str(), "L2") == 0) metric = 1;
+        else if (strcasecmp(mt.c_str(), "InnerProduct") == 0 ||
+                 strcasecmp(mt.c_str(), "IP") == 0)
+          metric = 2;
+        else
+          return err_status(1, ("unknown metric_type: " + mt).c_str());
+      }
     }
   }
   /* scalar filters -> exclusion bitmap consumed by the scan kernels
@@ -371,18 +382,22 @@ static struct CStatus Search_unguarded(void *engine, const char *request_str, in
   std::vector<vgamma::RangeFilterSpec> ranges;
   for (auto &t : req.range_filters)
     ranges.push_back({t.field, t.lower, t.upper, t.include_lower,
-                      t.include_upper});
-  if (req.op == 1 && req.n_filters > 1)
-    return err_status(1, "filter operator OR not supported this round");
+                      t.include_upper, t.is_union});
 
   int k = req.topn + req.offset;
   std::vector<float> dists((size_t)nq * k);
   std::vector<int64_t> ids((size_t)nq * k);
   std::string ferr;
+  /* hold the shared lock across search AND response assembly so the
+   * doc/table state read below (pkey_of, field_value, raw().host_row,
+   * bitmap) cannot be reallocated by a concurrent AddOrUpdateDoc /
+   * BuildIndex (gamma_api.h threading contract) */
+  auto rlock = e->read_lock();
   int rc = e->search(nq, (const float *)vq.value.data(), k, nprobe,
                      recall_num, metric, req.brute == 1, req.request_id,
                      req.partition_id, dists.data(), ids.data(),
-                     req.l2_sqrt, &terms, &ranges, &ferr);
+                     req.l2_sqrt, &terms, &ranges, &ferr, req.op,
+                     /*prelocked=*/true);
   if (rc == -2) return err_status(-2, "request killed");
   if (rc == -3) return err_status(1, ferr);
   if (rc != 0) return err_status(1, "search failed");
@@ -452,6 +467,8 @@ static struct CStatus Query_unguarded(void *engine, const char *request_str, int
   gpb::QueryRequest req;
   if (!req.parse(request_str, req_len))
     return err_status(1, "parse query request failed");
+  /* stable doc/table state across filtering AND serialization */
+  auto rlock = e->read_lock();
   std::vector<int64_t> docids;
   if (!req.document_ids.empty()) {
     for (auto &pk : req.document_ids) {
@@ -466,10 +483,11 @@ static struct CStatus Query_unguarded(void *engine, const char *request_str, int
     std::vector<vgamma::RangeFilterSpec> ranges;
     for (auto &t : req.range_filters)
       ranges.push_back({t.field, t.lower, t.upper, t.include_lower,
-                        t.include_upper});
+                        t.include_upper, t.is_union});
     std::string ferr;
     if (e->filter_docids(terms, ranges, 0,
-                         req.limit > 0 ? req.limit : 50, &docids, &ferr))
+                         req.limit > 0 ? req.limit : 50, &docids, &ferr,
+                         req.op, /*prelocked=*/true))
       return err_status(1, ferr);
   }
   std::vector<gpb::SearchResult> results(1);
@@ -824,6 +842,15 @@ static int GammaTestParseSearchRequest_unguarded(const char *buf, int len, char 
   s += "\"index_params\": \"" + gjson::escape(req.index_params) + "\",";
   s += "\"l2_sqrt\": " + std::to_string(req.l2_sqrt ? 1 : 0) + ",";
   s += "\"n_fields\": " + std::to_string(req.fields.size()) + ",";
+  s += "\"op\": " + std::to_string(req.op) + ",";
+  s += "\"term_unions\": [";
+  for (size_t i = 0; i < req.term_filters.size(); i++)
+    s += (i ? "," : "") + std::to_string(req.term_filters[i].is_union);
+  s += "],";
+  s += "\"range_unions\": [";
+  for (size_t i = 0; i < req.range_filters.size(); i++)
+    s += (i ? "," : "") + std::to_string(req.range_filters[i].is_union);
+  s += "],";
   s += "\"n_filters\": " + std::to_string(req.n_filters) + "}";
   *json_out = dup_malloc(s);
   *json_len = (int)s.size();

@@ -286,6 +286,10 @@ struct TermFilterSpec {
 struct RangeFilterSpec {
   std::string field, lower, upper; /* raw binary of the field type */
   bool inc_l = false, inc_u = false;
+  /* FilterOperator per filter (scalar_index_types.h:44 And=0,Or=1,Not=2;
+   * engine.cc:475 casts is_union straight to it). Not + lower==upper
+   * inclusive = NotEqual (bitmap_index.cc:196). */
+  int is_union = 0;
 };
 
 /* The engine: one per Init() (one Vearch partition). Single vector field
@@ -321,7 +325,18 @@ class Engine {
              bool l2_sqrt = false,
              const std::vector<TermFilterSpec> *term_filters = nullptr,
              const std::vector<RangeFilterSpec> *range_filters = nullptr,
-             std::string *filter_err = nullptr);
+             std::string *filter_err = nullptr, int filter_op = 0,
+             bool prelocked = false);
+
+  /* Shared (read) lock on the engine. The C ABI takes this across
+   * Search/Query/GetDoc* response assembly so the doc/table state read
+   * while serializing (pkey_of, field_value, raw().host_row, bitmap)
+   * cannot be reallocated by a concurrent AddOrUpdateDoc/BuildIndex
+   * (which take the unique lock); pass prelocked=true to search() while
+   * holding it. */
+  std::shared_lock<std::shared_mutex> read_lock() const {
+    return std::shared_lock<std::shared_mutex>(rw_);
+  }
   /* upload queries once; later search(nq, nullptr, ...) reuses them */
   int cache_queries(int nq, const float *xq);
   int cached_nq() const { return cached_nq_; }
@@ -333,7 +348,8 @@ class Engine {
    * engine.cc:420+): docids passing all filters, offset/limit applied */
   int filter_docids(const std::vector<TermFilterSpec> &terms,
                     const std::vector<RangeFilterSpec> &ranges, int offset,
-                    int limit, std::vector<int64_t> *out, std::string *err);
+                    int limit, std::vector<int64_t> *out, std::string *err,
+                    int filter_op = 0, bool prelocked = false);
 
   int64_t num_docs() const { return max_docid_; }
   IVFIndex *index() { return index_.get(); }
@@ -363,7 +379,7 @@ class Engine {
   int build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
                            const std::vector<RangeFilterSpec> &ranges,
                            SearchScratch &sc, const uint32_t **dev_out,
-                           std::string *err);
+                           std::string *err, int filter_op = 0);
   std::string path_, log_dir_, space_name_;
   std::string index_type_ = "IVFPQ";
   std::string vec_name_;

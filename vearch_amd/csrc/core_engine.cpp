@@ -284,7 +284,7 @@ int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
                                  const std::vector<RangeFilterSpec> &ranges,
                                  SearchScratch &sc,
                                  const uint32_t **dev_out,
-                                 std::string *err) {
+                                 std::string *err, int filter_op) {
   const int64_t n = max_docid_;
   const int64_t words = (n + 31) / 32;
   /* resolve fields + lazily extend their scalar indexes. Semantics are
@@ -366,18 +366,40 @@ int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
     }
   }
 
-  /* included = AND over filters of each filter's match set */
-  std::vector<uint32_t> inc((size_t)words, 0xffffffffu);
+  /* per-filter match sets, combined by the request-level FilterOperator
+   * (scalar_index_manager.cc:1188-1190: And -> Intersection,
+   * Or -> Union). Per-filter is_union follows engine.cc:475 /
+   * scalar_index_types.h:44 (And=0, Or=1, Not=2): a term filter with
+   * Not takes the complement of the matched set over [0, maxdoc)
+   * (BitmapIndex::NotIn, bitmap_index.cc:133 — docs missing the field
+   * DO match Not); a numeric range filter with Not and equal inclusive
+   * bounds is NotEqual (bitmap_index.cc:196). */
+  const bool op_or = filter_op == 1;
+  std::vector<uint32_t> inc((size_t)words, op_or ? 0u : 0xffffffffu);
   std::vector<uint32_t> mt;
   for (auto &q : qs) {
     mt.assign((size_t)words, 0);
+    bool negate = false;
     if (q.tf) {
+      negate = q.tf->is_union == 2;
       for (auto &t : split_x01(q.tf->value)) {
         auto pit = q.ix->postings.find(t);
         if (pit == q.ix->postings.end()) continue;
         for (int64_t id : pit->second)
           if (id < n) mt[id >> 5] |= 1u << (id & 31);
       }
+    } else if (q.rf->is_union == 2 && q.dt != 4 && q.dt != 8 &&
+               q.rf->lower == q.rf->upper && q.rf->inc_l && q.rf->inc_u) {
+      /* numeric NotEqual: complement of Equal(lower) */
+      negate = true;
+      const auto &nv = q.ix->nvals;
+      auto b = std::lower_bound(
+          nv.begin(), nv.end(), q.lo,
+          [](const std::pair<double, int64_t> &a, double v) {
+            return a.first < v;
+          });
+      for (; b < nv.end() && b->first == q.lo; ++b)
+        if (b->second < n) mt[b->second >> 5] |= 1u << (b->second & 31);
     } else if (q.dt == 4) { /* lexicographic string range */
       const auto &sv = q.ix->svals;
       auto cmp = [](const std::pair<std::string, int64_t> &a,
@@ -415,7 +437,15 @@ int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
       for (; b < e; ++b)
         if (b->second < n) mt[b->second >> 5] |= 1u << (b->second & 31);
     }
-    for (int64_t w = 0; w < words; w++) inc[w] &= mt[w];
+    if (negate) {
+      /* complement over [0, n): flip, then clear the tail bits past n */
+      for (int64_t w = 0; w < words; w++) mt[w] = ~mt[w];
+      if (n & 31) mt[words - 1] &= (1u << (n & 31)) - 1;
+    }
+    if (op_or)
+      for (int64_t w = 0; w < words; w++) inc[w] |= mt[w];
+    else
+      for (int64_t w = 0; w < words; w++) inc[w] &= mt[w];
   }
 
   sc.filt_host.assign((size_t)words, 0);
@@ -467,15 +497,18 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
                    float *out_dists, int64_t *out_ids, bool l2_sqrt,
                    const std::vector<TermFilterSpec> *term_filters,
                    const std::vector<RangeFilterSpec> *range_filters,
-                   std::string *filter_err) {
+                   std::string *filter_err, int filter_op, bool prelocked) {
   if (!table_created_ || nq <= 0 || k <= 0) return -1;
   /* Concurrent searches: each takes the read lock (Add/Build/Load are
    * write-locked) plus one SearchScratch context from the pool — its
    * own HIP stream and device buffers — so arbitrary cgo threads can
    * search while a background thread indexes (engine.cc:1108-1127,
    * SURVEY 8b). The ScratchGuard synchronizes the context's stream
-   * before release, so no async work survives the read lock. */
-  std::shared_lock<std::shared_mutex> g(rw_);
+   * before release, so no async work survives the read lock. The C ABI
+   * passes prelocked=true when it already holds read_lock() across
+   * search + response assembly. */
+  std::shared_lock<std::shared_mutex> g(rw_, std::defer_lock);
+  if (!prelocked) g.lock();
   SearchScratch *scp = acquire_scratch_();
   if (!scp) return -1;
   ScratchGuard sg{this, scp};
@@ -507,7 +540,12 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
    * (search_preassigned, ivfpq.cc:765-776) */
   int k2 = std::max(k, recall_num);
   bool rerank = recall_num > 0;
-  if (k2 > 1024) return -1;
+  if (k2 > 1024) {
+    if (filter_err)
+      *filter_err = "topN+offset (or recall_num) = " + std::to_string(k2) +
+                    " exceeds this engine's supported maximum of 1024";
+    return -3;
+  }
 
   tm.rec(0);
   const float *qptr = nullptr;
@@ -536,7 +574,7 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
     static const std::vector<RangeFilterSpec> kNoR;
     if (build_filter_bitmap_(term_filters ? *term_filters : kNoT,
                              range_filters ? *range_filters : kNoR, sc,
-                             &bm, filter_err))
+                             &bm, filter_err, filter_op))
       return -3;
   }
 
@@ -619,12 +657,14 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
 int Engine::filter_docids(const std::vector<TermFilterSpec> &terms,
                           const std::vector<RangeFilterSpec> &ranges,
                           int offset, int limit, std::vector<int64_t> *out,
-                          std::string *err) {
-  std::shared_lock<std::shared_mutex> g(rw_);
+                          std::string *err, int filter_op, bool prelocked) {
+  std::shared_lock<std::shared_mutex> g(rw_, std::defer_lock);
+  if (!prelocked) g.lock();
   SearchScratch *scp = acquire_scratch_();
   if (!scp) return -1;
   ScratchGuard sg{this, scp};
-  if (build_filter_bitmap_(terms, ranges, *scp, nullptr, err)) return -1;
+  if (build_filter_bitmap_(terms, ranges, *scp, nullptr, err, filter_op))
+    return -1;
   int64_t skipped = 0;
   for (int64_t id = 0; id < max_docid_; id++) {
     if ((scp->filt_host[id >> 5] >> (id & 31)) & 1u) continue;
@@ -751,9 +791,17 @@ int Engine::load(std::string *err) {
   }
   int dim = 0, tt = 0;
   int64_t maxdoc = 0;
-  fread(&dim, 4, 1, f);
-  fread(&tt, 4, 1, f);
-  fread(&maxdoc, 8, 1, f);
+  auto bad = [&](const char *what) {
+    fclose(f);
+    if (err) *err = std::string("bad dump file: ") + what;
+    return -1;
+  };
+  if (fread(&dim, 4, 1, f) != 1 || fread(&tt, 4, 1, f) != 1 ||
+      fread(&maxdoc, 8, 1, f) != 1)
+    return bad("truncated header");
+  if (dim <= 0 || dim > (1 << 20) || maxdoc < 0 ||
+      maxdoc > ((int64_t)1 << 40))
+    return bad("implausible dim/maxdoc");
   if (table_created_ && dim != dim_) {
     fclose(f);
     if (err)
@@ -768,11 +816,12 @@ int Engine::load(std::string *err) {
     return -1;
   }
   int nfields = 0;
-  fread(&nfields, 4, 1, f);
+  if (fread(&nfields, 4, 1, f) != 1 || nfields < 0 || nfields > 65536)
+    return bad("field count");
   std::vector<FieldMeta> fms(nfields);
   for (auto &fm : fms) {
-    if (rstr(fm.name)) { fclose(f); return -1; }
-    fread(&fm.data_type, 4, 1, f);
+    if (rstr(fm.name)) return bad("field name");
+    if (fread(&fm.data_type, 4, 1, f) != 1) return bad("field type");
   }
   if (!table_created_) {
     /* table should have been created before Load (reference flow);
@@ -800,7 +849,7 @@ int Engine::load(std::string *err) {
   if (bitmap_.load(f, stream_)) { fclose(f); return -1; }
   bitmap_.ensure(max_docid_, stream_);
   int has_index = 0;
-  fread(&has_index, 4, 1, f);
+  if (fread(&has_index, 4, 1, f) != 1) return bad("index flag");
   if (has_index && index_) {
     if (index_->load(f, stream_)) { fclose(f); return -1; }
     indexed_count_ = index_->ntotal();

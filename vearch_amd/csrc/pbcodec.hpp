@@ -143,6 +143,7 @@ struct RangeFilter { /* router_grpc.proto:116-123; values = raw binary of
                         table/inverted_index.cc:263) */
   std::string field, lower, upper;
   bool include_lower = false, include_upper = false;
+  int is_union = 0; /* FilterOperator: And=0 Or=1 Not=2 (engine.cc:475) */
 };
 
 struct SearchRequest { /* router_grpc.proto:168-192 */
@@ -236,6 +237,8 @@ struct SearchRequest { /* router_grpc.proto:168-192 */
                         rf.include_lower = v != 0; break; }
               case 5: { uint64_t v; if (!fr.varint(v)) return false;
                         rf.include_upper = v != 0; break; }
+              case 6: { uint64_t v; if (!fr.varint(v)) return false;
+                        rf.is_union = (int)v; break; }
               default: if (!fr.skip(fwt)) return false;
             }
           }
@@ -293,6 +296,7 @@ struct QueryRequest { /* router_grpc.proto:146-166 */
   bool is_vector_value = false;
   int limit = 0;
   int n_filters = 0;
+  int op = 0; /* request-level FilterOperator (`operator` field 15) */
 
   bool parse(const char *buf, int len) {
     Reader r(buf, (size_t)len);
@@ -340,6 +344,8 @@ struct QueryRequest { /* router_grpc.proto:146-166 */
                         rf.include_lower = v != 0; break; }
               case 5: { uint64_t v; if (!fr.varint(v)) return false;
                         rf.include_upper = v != 0; break; }
+              case 6: { uint64_t v; if (!fr.varint(v)) return false;
+                        rf.is_union = (int)v; break; }
               default: if (!fr.skip(fwt)) return false;
             }
           }
@@ -372,6 +378,8 @@ struct QueryRequest { /* router_grpc.proto:146-166 */
                   is_vector_value = v != 0; break; }
         case 9: { uint64_t v; if (!r.varint(v)) return false;
                   limit = (int)v; break; }
+        case 15: { uint64_t v; if (!r.varint(v)) return false;
+                   op = (int)v; break; }
         default: if (!r.skip(wt)) return false;
       }
     }

@@ -200,7 +200,7 @@ class GammaEngine:
     def search_pb(self, queries, topn, index_params="", fields=("_id",),
                   request_id="req1", partition_id=1, brute=0,
                   min_score=None, max_score=None, l2_sqrt=False,
-                  term_filters=(), range_filters=()):
+                  term_filters=(), range_filters=(), operator=0):
         """The real C-ABI Search with protobuf marshalling (reader.go
         path)."""
         from . import proto
@@ -210,7 +210,8 @@ class GammaEngine:
             request_id=request_id, partition_id=partition_id,
             index_params=index_params, brute=brute, fields=fields,
             min_score=min_score, max_score=max_score, l2_sqrt=l2_sqrt,
-            term_filters=term_filters, range_filters=range_filters)
+            term_filters=term_filters, range_filters=range_filters,
+            operator=operator)
         out = ctypes.c_char_p()
         n = ctypes.c_int()
         st = lib().Search(self.h, req, len(req), ctypes.byref(out),
@@ -222,12 +223,12 @@ class GammaEngine:
         return proto.decode_search_response(buf)
 
     def query_pb(self, document_ids=(), fields=("_id",), term_filters=(),
-                 range_filters=(), limit=0):
+                 range_filters=(), limit=0, operator=0):
         """The C-ABI Query (doc fetch / filtered browse)."""
         from . import proto
         req = proto.encode_query_request(
             list(document_ids), fields=fields, term_filters=term_filters,
-            range_filters=range_filters, limit=limit)
+            range_filters=range_filters, limit=limit, operator=operator)
         out = ctypes.c_char_p()
         n = ctypes.c_int()
         st = lib().Query(self.h, req, len(req), ctypes.byref(out),

@@ -44,11 +44,15 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
                           index_params="", min_score=None, max_score=None,
                           brute=0, fields=("_id",), l2_sqrt=False,
                           is_vector_value=False, term_filters=(),
-                          range_filters=()):
+                          range_filters=(), operator=0):
     """vearchpb.SearchRequest (router_grpc.proto:168-192).
     term_filters: (field, value_bytes) or (field, value_bytes, is_union).
     range_filters: (field, lower_bytes, upper_bytes, incl_lower,
-    incl_upper) — values are the raw binary of the field type."""
+    incl_upper[, is_union]) — values are the raw binary of the field
+    type. is_union is the per-filter FilterOperator (And=0 Or=1 Not=2,
+    engine.cc:475); `operator` is the request-level combiner (And=0
+    intersects the filters' match sets, Or=1 unions them,
+    scalar_index_manager.cc:1188)."""
     out = b""
     out += _ld(1, encode_head(request_id, partition_id))
     out += _vint(2, req_num)
@@ -63,9 +67,11 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
     for f in fields:
         out += _ld(6, f.encode())
     for rf in range_filters:
-        field, lower, upper, il, iu = rf
+        field, lower, upper, il, iu = rf[:5]
         body = _ld(1, field.encode()) + _ld(2, lower) + _ld(3, upper)
         body += _vint(4, 1 if il else 0) + _vint(5, 1 if iu else 0)
+        if len(rf) > 5:
+            body += _vint(6, rf[5])
         out += _ld(7, body)
     for tf in term_filters:
         field, value = tf[0], tf[1]
@@ -79,23 +85,30 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
         out += _vint(11, 1)
     if is_vector_value:
         out += _vint(12, 1)
+    if operator:
+        out += _vint(17, operator)
     return out
 
 
 def encode_query_request(document_ids, fields=("_id",), request_id="req1",
                          partition_id=1, is_vector_value=False,
-                         term_filters=(), range_filters=(), limit=0):
+                         term_filters=(), range_filters=(), limit=0,
+                         operator=0):
     out = b""
     out += _ld(1, encode_head(request_id, partition_id))
     for d in document_ids:
         out += _ld(2, d.encode())
     for rf in range_filters:
-        field, lower, upper, il, iu = rf
+        field, lower, upper, il, iu = rf[:5]
         body = _ld(1, field.encode()) + _ld(2, lower) + _ld(3, upper)
         body += _vint(4, 1 if il else 0) + _vint(5, 1 if iu else 0)
+        if len(rf) > 5:
+            body += _vint(6, rf[5])
         out += _ld(5, body)
     for tf in term_filters:
         body = _ld(1, tf[0].encode()) + _ld(2, tf[1])
+        if len(tf) > 2:
+            body += _vint(3, tf[2])
         out += _ld(6, body)
     for f in fields:
         out += _ld(7, f.encode())
@@ -103,6 +116,8 @@ def encode_query_request(document_ids, fields=("_id",), request_id="req1",
         out += _vint(8, 1)
     if limit:
         out += _vint(9, limit)
+    if operator:
+        out += _vint(15, operator)
     return out
 
 

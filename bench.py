@@ -188,6 +188,9 @@ def main():
     ap.add_argument("--queries", type=int, default=0,
                     help="override nq per step")
     ap.add_argument("--nprobe", type=int, default=0)
+    ap.add_argument("--nlist", type=int, default=0,
+                    help="override ncentroids (training_threshold scales "
+                         "with it, reference ivfpq.cc:142 nlist*39 rule)")
     ap.add_argument("--rerank", type=int, default=-1,
                     help="recall_num (-1 = workload default)")
     ap.add_argument("--skip-recall", action="store_true")
@@ -207,6 +210,10 @@ def main():
         cfg["nq"] = args.queries
     if args.nprobe:
         cfg["nprobe"] = args.nprobe
+    if args.nlist and cfg["nlist"]:
+        cfg["nlist"] = args.nlist
+        cfg["train_n"] = min(cfg["n"], max(cfg["train_n"],
+                                           39 * args.nlist))
     if args.rerank >= 0:
         cfg["rerank"] = args.rerank
 
@@ -371,7 +378,8 @@ def main():
                 "frac": round(achieved / HBM_PEAK_GBS, 4),
                 "traffic": MEASURED_TRAFFIC_BYTES.get(args.workload)
                 if cfg["nq"] == WORKLOADS.get(args.workload, {}).get("nq")
-                and not args.db_size else None,
+                and not args.db_size and not args.nlist
+                and not args.nprobe else None,
             }
         cpu = None
         if not args.skip_cpu_baseline and world == 1:

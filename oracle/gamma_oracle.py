@@ -227,7 +227,12 @@ class OracleIVFPQ:
 
     @property
     def metric_ip(self):
-        return 1 if self.metric == "InnerProduct" else 0
+        m = self.metric.lower()
+        if m in ("innerproduct", "ip"):
+            return 1
+        if m == "l2":
+            return 0
+        raise ValueError(f"unknown metric: {self.metric}")
 
     def train(self, xt, seed=42):
         spherical = self.metric_ip == 1

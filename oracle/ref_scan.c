@@ -340,7 +340,6 @@ EXPORT void oracle_ivfpq_search_pct1(
   for (int qi = 0; qi < nq; qi++) {
     const float *q = queries + (size_t)qi * d;
     float *atab = (float *)malloc((size_t)M * ksub * 4);
-    float *table = (float *)malloc((size_t)M * ksub * 4);
     uint64_t *heap = (uint64_t *)malloc((size_t)k * 8);
     int size = 0;
     oracle_pct1_a_table(d, M, ksub, q, codebooks, atab);
@@ -351,16 +350,32 @@ EXPORT void oracle_ivfpq_search_pct1(
       if (end <= beg) continue;
       float dis0 = probe_dists[(size_t)qi * nprobe + p];
       const float *btab = btabs + (size_t)ln * M * ksub;
-      for (size_t e = 0; e < (size_t)M * ksub; e++)
-        table[e] = atab[e] + btab[e];
-      oracle_adc_scan_list(end - beg, M, ksub, codes + (size_t)beg * M,
-                           ids + beg, table, dis0, del_bitmap, 0, k, heap,
-                           &size);
+      /* S-term grouping (the engine's formulation, kernels.h
+       * pq_sterm): the list half of T = A + B is summed per code
+       * first — S = sum_m B[m][code_m], plain adds in m order,
+       * exactly the engine's encode-time k_pq_sterm — then
+       * dis = dis0 + S + sum_m A[m][code_m] (plain adds, m order,
+       * the h:945-948 loop over A instead of T). Algebraically the
+       * same T = A+B sum, grouped per vector; fp32-rounding-only
+       * difference vs the per-element A+B add. */
+      for (int64_t j = beg; j < end; j++) {
+        uint64_t id = (uint64_t)ids[j];
+        if (id & DEL_MASK) continue;               /* h:930 */
+        if (bitmap_test(del_bitmap, id)) continue; /* h:935 IsValid */
+        const uint8_t *code = codes + (size_t)j * M;
+        float sv = 0.0f;
+        for (int m = 0; m < M; m++)
+          sv += btab[(size_t)m * ksub + code[m]];
+        float dis = dis0 + sv;
+        for (int m = 0; m < M; m++) /* h:945-948 over A */
+          dis += atab[(size_t)m * ksub + code[m]];
+        heap_push_or_replace(heap, &size, k,
+                             make_key(dis, (uint32_t)id, 0));
+      }
     }
     emit_sorted(heap, size, k, 0, out_dists + (size_t)qi * k,
                 out_ids + (size_t)qi * k);
     free(atab);
-    free(table);
     free(heap);
   }
   free(btabs);

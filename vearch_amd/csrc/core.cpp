@@ -779,6 +779,8 @@ int IVFIndex::update_dev_buckets(hipStream_t s) {
   for (int i = 0; i < nlist_; i++) {
     h[i].ids = buckets_[i].ids ? buckets_[i].ids->as<uint32_t>() : nullptr;
     h[i].data = buckets_[i].data ? buckets_[i].data->get() : nullptr;
+    h[i].svals =
+        buckets_[i].svals ? buckets_[i].svals->as<float>() : nullptr;
     h[i].size = buckets_[i].size;
   }
   if (dev_buckets_.reserve(nlist_ * sizeof(GammaBucketDev))) return -1;
@@ -797,9 +799,10 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
   const size_t entry =
       params_.kind == IndexKind::IVFPQ ? (size_t)code_size_ : (size_t)d_ * 4;
 
-  DeviceBuf xd, xnorm, dots, asg, resid, codes;
+  DeviceBuf xd, xnorm, dots, asg, resid, codes, sterm;
   std::vector<int32_t> asg_h(std::min(n, chunk));
   std::vector<uint8_t> codes_h;
+  std::vector<float> sterm_h;
 
   DeviceBuf xrot;
   for (int64_t c0 = 0; c0 < n; c0 += chunk) {
@@ -847,26 +850,41 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
       codes_h.resize((size_t)cn * code_size_);
       (void)hipMemcpy(codes_h.data(), codes.get(), (size_t)cn * code_size_,
                 hipMemcpyDeviceToHost);
+      /* per-vector S term against the vector's own list's B table */
+      if (sterm.reserve((size_t)cn * 4)) return -1;
+      if (gk::pq_sterm(s, cn, M_, nlist_, codes.as<uint8_t>(),
+                       asg.as<int32_t>(), 0, btable_.as<float>(),
+                       sterm.as<float>()) != hipSuccess)
+        return -1;
+      sterm_h.resize(cn);
+      (void)hipMemcpy(sterm_h.data(), sterm.get(), (size_t)cn * 4,
+                hipMemcpyDeviceToHost);
       payload_h = codes_h.data();
     } else {
       payload_h = (const uint8_t *)(x_host + (size_t)c0 * d_);
     }
 
     /* group by bucket (AddKeys analog, realtime_mem_data.cc) */
-    std::map<int32_t, std::pair<std::vector<uint32_t>, std::vector<uint8_t>>>
-        groups;
+    const bool pq = params_.kind == IndexKind::IVFPQ;
+    struct Group {
+      std::vector<uint32_t> ids;
+      std::vector<uint8_t> data;
+      std::vector<float> svals;
+    };
+    std::map<int32_t, Group> groups;
     for (int64_t i = 0; i < cn; i++) {
       int32_t b = asg_h[i];
       if (b < 0 || b >= nlist_) b = (int32_t)(vids[c0 + i] % nlist_);
       auto &g = groups[b];
-      g.first.push_back((uint32_t)vids[c0 + i]); /* vid < 2^31 enforced */
-      size_t off = g.second.size();
-      g.second.resize(off + entry);
-      memcpy(g.second.data() + off, payload_h + (size_t)i * entry, entry);
+      g.ids.push_back((uint32_t)vids[c0 + i]); /* vid < 2^31 enforced */
+      size_t off = g.data.size();
+      g.data.resize(off + entry);
+      memcpy(g.data.data() + off, payload_h + (size_t)i * entry, entry);
+      if (pq) g.svals.push_back(sterm_h[i]);
     }
     for (auto &kv : groups) {
       Bucket &bk = buckets_[kv.first];
-      int64_t add_n = (int64_t)kv.second.first.size();
+      int64_t add_n = (int64_t)kv.second.ids.size();
       if (bk.size + add_n > bk.cap) {
         long long ncap =
             std::max<long long>({(long long)params_.bucket_init_size,
@@ -877,26 +895,36 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
                                 bk.size + add_n));
         auto nids = std::make_unique<DeviceBuf>();
         auto ndata = std::make_unique<DeviceBuf>();
+        auto nsv = std::make_unique<DeviceBuf>();
         if (nids->reserve((size_t)ncap * 4)) return -1;
         if (ndata->reserve((size_t)ncap * entry)) return -1;
+        if (pq && nsv->reserve((size_t)ncap * 4)) return -1;
         if (bk.size > 0) {
           (void)hipMemcpy(nids->get(), bk.ids->get(), (size_t)bk.size * 4,
                     hipMemcpyDeviceToDevice);
           (void)hipMemcpy(ndata->get(), bk.data->get(), (size_t)bk.size * entry,
                     hipMemcpyDeviceToDevice);
+          if (pq)
+            (void)hipMemcpy(nsv->get(), bk.svals->get(), (size_t)bk.size * 4,
+                      hipMemcpyDeviceToDevice);
         }
         bk.ids = std::move(nids);
         bk.data = std::move(ndata);
+        if (pq) bk.svals = std::move(nsv);
         bk.cap = ncap;
         dev_buckets_dirty_ = true;
       }
-      (void)hipMemcpy(bk.ids->as<uint32_t>() + bk.size, kv.second.first.data(),
+      (void)hipMemcpy(bk.ids->as<uint32_t>() + bk.size, kv.second.ids.data(),
                 (size_t)add_n * 4, hipMemcpyHostToDevice);
       (void)hipMemcpy((uint8_t *)bk.data->get() + (size_t)bk.size * entry,
-                kv.second.second.data(), (size_t)add_n * entry,
+                kv.second.data.data(), (size_t)add_n * entry,
                 hipMemcpyHostToDevice);
+      if (pq)
+        (void)hipMemcpy(bk.svals->as<float>() + bk.size,
+                  kv.second.svals.data(), (size_t)add_n * 4,
+                  hipMemcpyHostToDevice);
       for (int64_t i = 0; i < add_n; i++) {
-        int64_t vid = (int64_t)kv.second.first[i];
+        int64_t vid = (int64_t)kv.second.ids[i];
         if (vid >= (int64_t)vid_loc_.size())
           vid_loc_.resize((size_t)vid + 1024, -1);
         vid_loc_[vid] = ((int64_t)kv.first << 40) | (bk.size + i);
@@ -1020,7 +1048,6 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
     GAMMA_CHECK(gk::ivfpq_scan(s, nq, S, d_, M_, nprobe, k2, q_dev,
                                centroids_.as<float>(),
                                codebooks_.as<float>(), atab,
-                               btable_.as<float>(),
                                sc.pdists.as<float>(),
                                dev_buckets_.as<GammaBucketDev>(), nlist_,
                                sc.probes.as<int64_t>(), bitmap_dev,
@@ -1178,6 +1205,17 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
                 hipMemcpyHostToDevice);
       (void)hipMemcpy(bk.data->get(), data.data(), data.size(),
                 hipMemcpyHostToDevice);
+      if (params_.kind == IndexKind::IVFPQ) {
+        /* S terms are recomputable from codes + the B table, so the
+         * dump format is unchanged: rebuild them here (asg == the
+         * bucket number by construction) */
+        bk.svals = std::make_unique<DeviceBuf>();
+        if (bk.svals->reserve((size_t)sz * 4)) return -1;
+        if (gk::pq_sterm(s, sz, M_, nlist_, bk.data->as<uint8_t>(),
+                         nullptr, i, btable_.as<float>(),
+                         bk.svals->as<float>()) != hipSuccess)
+          return -1;
+      }
       bk.size = bk.cap = sz;
       for (long long j = 0; j < sz; j++) {
         int64_t vid = ids[j];

@@ -241,6 +241,9 @@ class IVFIndex {
   std::vector<float> opq_R_host_;
   struct Bucket {
     std::unique_ptr<DeviceBuf> ids, data;
+    /* IVFPQ: per-entry S term (see kernels.h pq_sterm); recomputable
+     * from codes + the B table, so Dump stays format-compatible */
+    std::unique_ptr<DeviceBuf> svals;
     long long size = 0, cap = 0;
   };
   std::vector<Bucket> buckets_;

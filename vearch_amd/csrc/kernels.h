@@ -14,6 +14,11 @@
 struct GammaBucketDev {
   const uint32_t *ids;
   const void *data;
+  /* IVFPQ only: per-entry S term, S_v = sum_m btab[ln][m][code_m]
+   * (the list-dependent half of the decomposed ADC tables, folded into
+   * one float per vector at encode time so the scan never touches the
+   * B table — see kernels.h pq_sterm). Null for IVFFLAT. */
+  const float *svals;
   long long size;
 };
 
@@ -64,17 +69,30 @@ hipError_t pq_tables_a(hipStream_t s, int nq, int d, int M,
                        const float *queries, const float *codebooks,
                        float *atab);
 
-/* IVFPQ fused search: one workgroup per query; per probed list stages
- * T = A_q + B_list in LDS (L2; dis0 = the coarse probe distance) or the
- * query-level IP table (h:164-167), then scans the list (h:923-953).
- * out_keys: nq x k2. */
+/* Per-vector S term (encode/add/load time):
+ *   out[i] = sum_m btab[asg_i][m][code_i[m]]   (plain adds, m order)
+ * where asg_i = asg ? asg[i] : asg_const. Folding the list half of the
+ * pct1 decomposition into one float per vector removes the per-(query,
+ * list) B-table staging (nprobe x M x ksub floats per query) from the
+ * scan: dis = coarse_dis + S_v + sum_m atab[q][m][code_m], which is
+ * the same T = A + B sum grouped per vector — fp32-rounding-only
+ * difference, same class as the documented pct1 deviation
+ * (DESIGN.md "The L2 ADC table mode"). */
+hipError_t pq_sterm(hipStream_t s, int64_t n, int M, int nlist,
+                    const uint8_t *codes, const int32_t *asg,
+                    int asg_const, const float *btab, float *out);
+
+/* IVFPQ fused search: one workgroup per query; stages the query-level
+ * table ONCE (L2: A_q from atab, dis = coarse_dis + S_v + sum A[c_m];
+ * IP: the h:164-167 query table, dis = dis0 + sum T[c_m]), then scans
+ * the probed lists (h:923-953 semantics). out_keys: nq x k2. */
 /* S = probe-split factor: S sub-workgroups per query (out_keys is
  * nq x S x k2; merge with sort_rows). S>1 serves small batches. */
 hipError_t ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
                       int nprobe,
                       int k2, const float *queries, const float *centroids,
                       const float *codebooks, const float *atab,
-                      const float *btab, const float *probe_dists,
+                      const float *probe_dists,
                       const GammaBucketDev *buckets,
                       int nlist, const int64_t *probes,
                       const uint32_t *bitmap, bool ip, uint64_t *out_keys,

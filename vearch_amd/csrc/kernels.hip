@@ -409,7 +409,6 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
              const float *__restrict__ centroids,
              const float *__restrict__ codebooks,
              const float *__restrict__ atab,
-             const float *__restrict__ btab,
              const float *__restrict__ probe_dists,
              const GammaBucketDev *__restrict__ buckets, int nlist,
              const int64_t *__restrict__ probes,
@@ -449,6 +448,16 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
       lut[e] = acc;
     }
     __syncthreads();
+  } else {
+    /* L2: the query-level A table ONCE per workgroup. The list half of
+     * the decomposition (B) is pre-folded into one float per vector
+     * (bucket svals, gk::pq_sterm), so no per-(query,list) table is
+     * ever staged: dis = coarse_dis + S_v + sum_m A[m][code_m]. */
+    const float4 *Aq = (const float4 *)(atab + (size_t)q * M * ksub);
+    float4 *lut4 = (float4 *)lut;
+    for (int e = threadIdx.x; e < (M * ksub) >> 2; e += blockDim.x)
+      lut4[e] = Aq[e];
+    __syncthreads();
   }
 
   /* in-flight kill (is_killed_every<1024> analog, ivfpq.h:927): poll a
@@ -476,22 +485,14 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
       __syncthreads();
       dis0 = dis0s[0];
     } else {
-      /* T = A_q + B_list, float4-vectorized (M*ksub % 4 == 0);
-       * dis0 = the coarse probe distance (ivfpq.h:255 dis0=coarse_dis) */
-      const float4 *Aq =
-          (const float4 *)(atab + (size_t)q * M * ksub);
-      const float4 *Bl = (const float4 *)(btab + (size_t)ln * M * ksub);
-      float4 *lut4 = (float4 *)lut;
-      for (int e = threadIdx.x; e < (M * ksub) >> 2; e += blockDim.x) {
-        float4 a = Aq[e], b = Bl[e];
-        lut4[e] = make_float4(a.x + b.x, a.y + b.y, a.z + b.z, a.w + b.w);
-      }
-      __syncthreads();
+      /* dis0 = the coarse probe distance (ivfpq.h:255 dis0=coarse_dis);
+       * the per-list table half arrives as the per-vector S term */
       dis0 = probe_dists[(int64_t)q * nprobe + p];
     }
 
     const uint32_t *ids = bk.ids;
     const uint8_t *codes = (const uint8_t *)bk.data;
+    const float *svals = (const float *)bk.svals;
     if (MW > 0) {
       const int C = GAMMA_ADC_C;
       for (long long j0 = 0; j0 < bk.size;
@@ -499,16 +500,19 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
         long long jb = j0 + (long long)threadIdx.x * C;
         uint32_t w[C][MW ? MW : 1]; /* compile-time bounds -> registers */
         int64_t idv[C];
+        float sv[C];
 #pragma unroll
         for (int c = 0; c < C; c++) {
           long long j = jb + c;
           if (j < bk.size) {
             idv[c] = (int64_t)(int32_t)ids[j]; /* bit31 -> negative */
+            if (!IP) sv[c] = svals[j];
             const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
 #pragma unroll
             for (int mw = 0; mw < MW; mw++) w[c][mw] = cw[mw];
           } else {
             idv[c] = -1; /* bit 63 set -> skipped below */
+            sv[c] = 0.0f;
           }
         }
 #pragma unroll
@@ -516,7 +520,7 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
           int64_t id = idv[c];
           if (!((uint64_t)id >> 63) &&
               !gamma_bitmap_test(bitmap, (uint64_t)id)) {
-            float dis = dis0;
+            float dis = IP ? dis0 : dis0 + sv[c];
             const float *tab = lut;
 #pragma unroll
             for (int mw = 0; mw < MW; mw++) {
@@ -540,7 +544,7 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
           if (!((uint64_t)id >> 63) &&
               !gamma_bitmap_test(bitmap, (uint64_t)id)) {
             const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
-            float dis = dis0;
+            float dis = IP ? dis0 : dis0 + svals[j];
             const float *tab = lut;
             for (int mw = 0; mw < mwords; mw++) {
               uint32_t wv = cw[mw];
@@ -555,7 +559,7 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
         sel.maybe_flush(blockDim.x);
       }
     }
-    __syncthreads();  /* lut rebuilt next list: scan readers done */
+    if (IP) __syncthreads(); /* dis0s rewritten next list */
   }
   sel.finish();
   for (int i = threadIdx.x; i < k2; i += blockDim.x)
@@ -566,7 +570,7 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
                           int nprobe,
                           int k2, const float *queries,
                           const float *centroids, const float *codebooks,
-                          const float *atab, const float *btab,
+                          const float *atab,
                           const float *probe_dists,
                           const GammaBucketDev *buckets, int nlist,
                           const int64_t *probes, const uint32_t *bitmap,
@@ -585,7 +589,7 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
 #define GAMMA_LAUNCH_SCAN(IPV, MWV, BSV)                                  \
   k_ivfpq_scan<IPV, MWV, BSV><<<g, dim3(BSV), smem, s>>>(                 \
       nq, S, d, M, nprobe, k2, queries, centroids, codebooks, atab,       \
-      btab, probe_dists, buckets, nlist, probes, bitmap, out_keys,        \
+      probe_dists, buckets, nlist, probes, bitmap, out_keys,              \
       kill_flag)
   if (ip) {
     if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG);
@@ -840,24 +844,106 @@ __global__ void k_rerank(int nq, int ncand, int d,
   keys_out[idx] = gamma_make_key<IP>(acc, id);
 }
 
+/* LDS-staged re-rank (tools/rerank_bench.hip "v2": 0.354 ms vs 0.698 ms
+ * for the thread-per-candidate version at nq=10k ncand=200 d=128 —
+ * 2.95 TB/s algorithmic). One block per (query, candidate chunk); the
+ * block stages BS candidate rows through LDS in 128 B slices with
+ * 8 threads cooperating per row (coalesced float4s), then each thread
+ * accumulates ITS candidate's distance from the staged slice in the
+ * same canonical element order as the scalar chain — bit-identical
+ * output, ~4x less time stalled on scattered row gathers. */
+template <bool IP, int DV, int BS>
+__global__ void __launch_bounds__(BS)
+k_rerank_lds(int nq, int ncand, const float *__restrict__ queries,
+             const float *const *__restrict__ segs, int seg_shift,
+             const uint64_t *__restrict__ keys_in,
+             uint64_t *__restrict__ keys_out) {
+  const int CH = 32; /* floats per slice = 128 B */
+  __shared__ float rows[BS][CH + 1]; /* +1: bank-shift */
+  __shared__ float qs[DV];
+  const int nchunk = (ncand + BS - 1) / BS;
+  const int q = blockIdx.x / nchunk;
+  const int c0 = (blockIdx.x - q * nchunk) * BS;
+  if (q >= nq) return;
+  for (int i = threadIdx.x; i < DV; i += BS)
+    qs[i] = queries[(size_t)q * DV + i];
+  const int64_t seg_mask = ((int64_t)1 << seg_shift) - 1;
+  const int my = c0 + threadIdx.x;
+  uint64_t mykey = GAMMA_KEY_EMPTY;
+  if (my < ncand) mykey = keys_in[(size_t)q * ncand + my];
+  const bool live = my < ncand && mykey != GAMMA_KEY_EMPTY;
+  const uint32_t myid = (uint32_t)(mykey & 0xffffffffu);
+  float acc = 0.0f;
+  const int f4 = threadIdx.x & 7, rg = threadIdx.x >> 3;
+  for (int sl = 0; sl < DV / CH; sl++) {
+    __syncthreads();
+    for (int r = rg; r < BS; r += BS / 8) {
+      int cand = c0 + r;
+      if (cand < ncand) {
+        uint64_t key = keys_in[(size_t)q * ncand + cand];
+        if (key != GAMMA_KEY_EMPTY) {
+          uint32_t id = (uint32_t)(key & 0xffffffffu);
+          const float *v = segs[id >> seg_shift] +
+                           (size_t)(id & seg_mask) * DV;
+          float4 x = *(const float4 *)(v + sl * CH + f4 * 4);
+          rows[r][f4 * 4 + 0] = x.x;
+          rows[r][f4 * 4 + 1] = x.y;
+          rows[r][f4 * 4 + 2] = x.z;
+          rows[r][f4 * 4 + 3] = x.w;
+        }
+      }
+    }
+    __syncthreads();
+    if (live) {
+      const float *qm = qs + sl * CH;
+      const float *vm = rows[threadIdx.x];
+#pragma unroll
+      for (int t = 0; t < CH; t++) {
+        if (IP) {
+          acc = fmaf(qm[t], vm[t], acc);
+        } else {
+          float dx = qm[t] - vm[t];
+          acc = fmaf(dx, dx, acc);
+        }
+      }
+    }
+  }
+  if (my < ncand)
+    keys_out[(size_t)q * ncand + my] =
+        live ? gamma_make_key<IP>(acc, myid) : GAMMA_KEY_EMPTY;
+}
+
 hipError_t gk::rerank(hipStream_t s, int nq, int ncand, int d,
                       const float *queries, const float *const *segs,
                       int seg_shift, bool ip, const uint64_t *keys_in,
                       uint64_t *keys_out) {
   int64_t total = (int64_t)nq * ncand;
   if (total == 0) return hipSuccess;
+  if (d == 128 || d == 768) {
+    const int BS = 256;
+    int nchunk = (ncand + BS - 1) / BS;
+    dim3 g((uint32_t)((int64_t)nq * nchunk));
+#define GAMMA_LAUNCH_RERANK_LDS(IPV, DVV)                                 \
+  k_rerank_lds<IPV, DVV, BS><<<g, dim3(BS), 0, s>>>(                      \
+      nq, ncand, queries, segs, seg_shift, keys_in, keys_out)
+    if (ip) {
+      if (d == 128) GAMMA_LAUNCH_RERANK_LDS(true, 128);
+      else GAMMA_LAUNCH_RERANK_LDS(true, 768);
+    } else {
+      if (d == 128) GAMMA_LAUNCH_RERANK_LDS(false, 128);
+      else GAMMA_LAUNCH_RERANK_LDS(false, 768);
+    }
+#undef GAMMA_LAUNCH_RERANK_LDS
+    return hipGetLastError();
+  }
   int64_t blocks = (total + WG - 1) / WG;
 #define GAMMA_LAUNCH_RERANK(IPV, DVV)                                     \
   k_rerank<IPV, DVV><<<dim3((uint32_t)blocks), dim3(WG), 0, s>>>(         \
       nq, ncand, d, queries, segs, seg_shift, keys_in, keys_out)
   if (ip) {
-    if (d == 128) GAMMA_LAUNCH_RERANK(true, 128);
-    else if (d == 768) GAMMA_LAUNCH_RERANK(true, 768);
-    else GAMMA_LAUNCH_RERANK(true, 0);
+    GAMMA_LAUNCH_RERANK(true, 0);
   } else {
-    if (d == 128) GAMMA_LAUNCH_RERANK(false, 128);
-    else if (d == 768) GAMMA_LAUNCH_RERANK(false, 768);
-    else GAMMA_LAUNCH_RERANK(false, 0);
+    GAMMA_LAUNCH_RERANK(false, 0);
   }
 #undef GAMMA_LAUNCH_RERANK
   return hipGetLastError();
@@ -1009,6 +1095,39 @@ hipError_t gk::pq_tables_a(hipStream_t s, int nq, int d, int M,
   int64_t total = (int64_t)nq * M * 256;
   k_pq_atable<<<dim3((uint32_t)((total + WG - 1) / WG)), dim3(WG), 0, s>>>(
       nq, d, M, queries, codebooks, atab);
+  return hipGetLastError();
+}
+
+/* per-vector S term: out[i] = sum_m btab[asg_i][m][code_i[m]], plain
+ * adds in m order (the oracle mirrors this exactly —
+ * oracle_ivfpq_search_pct1's per-code B sum) */
+__global__ void k_pq_sterm(int64_t n, int M, int nlist,
+                           const uint8_t *__restrict__ codes,
+                           const int32_t *__restrict__ asg, int asg_const,
+                           const float *__restrict__ btab,
+                           float *__restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  int ln = asg ? asg[i] : asg_const;
+  if (ln < 0 || ln >= nlist) { /* defensive: matches the host-side
+                                  bucket fixup for degenerate
+                                  assignments (core.cpp add) */
+    out[i] = 0.0f;
+    return;
+  }
+  const float *B = btab + (size_t)ln * M * 256;
+  const uint8_t *c = codes + (size_t)i * M;
+  float acc = 0.0f;
+  for (int m = 0; m < M; m++) acc += B[(size_t)m * 256 + c[m]];
+  out[i] = acc;
+}
+
+hipError_t gk::pq_sterm(hipStream_t s, int64_t n, int M, int nlist,
+                        const uint8_t *codes, const int32_t *asg,
+                        int asg_const, const float *btab, float *out) {
+  if (n <= 0) return hipSuccess;
+  k_pq_sterm<<<dim3((uint32_t)((n + WG - 1) / WG)), dim3(WG), 0, s>>>(
+      n, M, nlist, codes, asg, asg_const, btab, out);
   return hipGetLastError();
 }
 

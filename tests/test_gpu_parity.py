@@ -497,6 +497,54 @@ def test_filtered_search_term_and_range(data):
     eng.close()
 
 
+def test_field_index_lifecycle_and_backup(data):
+    """AddFieldIndexWithParams / RemoveFieldIndex route SCALAR names
+    onto the ScalarFieldIndex machinery (engine.cc:1561,1648 semantics:
+    duplicate add and unknown remove are OK no-ops, unknown fields are
+    errors); Backup(0) creates a full dump under <path>/backup
+    (engine.cc:1529)."""
+    import struct
+    from vearch_amd import fbsenc
+    base, q = data
+    path = "/tmp/gamma_fieldindex"
+    eng = make_engine(path)
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}',
+                     scalar_fields=[("tag", fbsenc.DATA_STRING),
+                                    ("num", fbsenc.DATA_INT)])
+    for vid in range(500):
+        eng.add_doc(str(vid), base[vid],
+                    fields=[("tag", b"grp%d" % (vid % 3),
+                             fbsenc.DATA_STRING),
+                            ("num", struct.pack("<i", vid),
+                             fbsenc.DATA_INT)])
+    # scalar index on both fields; duplicate add ignored; vector ok
+    eng.add_field_index("ix_tag", ["tag"], "SCALAR")
+    eng.add_field_index("ix_tag", ["tag"], "SCALAR")  # no-op
+    eng.add_field_index("ix_num", ["num"], "SCALAR")
+    eng.add_field_index("ix_vec", [eng.vec_name], "IVFPQ")
+    with pytest.raises(RuntimeError, match="not found"):
+        eng.add_field_index("ix_bad", ["nope"], "SCALAR")
+    # filters work through the (now eagerly built) indexes
+    res = eng.search_pb(q[:4], topn=10, term_filters=[("tag", b"grp1")])
+    for t in range(4):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert ids and all(i % 3 == 1 for i in ids)
+    # remove + idempotent remove; filters still correct (lazy rebuild)
+    eng.remove_field_index("ix_tag")
+    eng.remove_field_index("ix_tag")
+    eng.remove_field_index("never_existed")
+    res = eng.search_pb(q[:2], topn=10, term_filters=[("tag", b"grp2")])
+    for t in range(2):
+        ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert ids and all(i % 3 == 2 for i in ids)
+    # backup: command 0 writes a loadable dump under <path>/backup
+    import os
+    eng.backup(0)
+    assert os.path.exists(os.path.join(path, "backup", "gamma.dump"))
+    eng.backup(1)  # non-create commands: accepted no-ops
+    eng.close()
+
+
 def test_filter_not_and_or(data):
     """FilterOperator semantics (scalar_index_types.h:44 And=0 Or=1
     Not=2): per-filter Not on terms (BitmapIndex::NotIn — docs missing

@@ -563,9 +563,11 @@ int GetConfig(void *engine, char **config_str, int *len) {
 
 
 static struct CStatus Backup_unguarded(void *engine, int command) {
-  (void)engine;
-  (void)command;
-  return err_status(1, "backup not supported this round");
+  if (!engine) return err_status(1, "null engine");
+  auto *e = static_cast<Engine *>(engine);
+  std::string err;
+  if (e->backup(command, &err)) return err_status(1, err.c_str());
+  return ok_status();
 }
 
 struct CStatus Backup(void *engine, int command) {
@@ -585,19 +587,18 @@ static struct CStatus AddFieldIndexWithParams_unguarded(
     const char *const *field_names, const int *field_name_lens,
     int field_name_count, const char *index_type, int index_type_len,
     const char *index_params, int index_params_len) {
-  (void)engine;
-  (void)index_name;
-  (void)index_name_len;
-  (void)field_names;
-  (void)field_name_lens;
-  (void)field_name_count;
-  (void)index_type;
-  (void)index_type_len;
-  (void)index_params;
-  (void)index_params_len;
-  return err_status(1,
-                    "dynamic field indexes (scalar/composite) are out of "
-                    "scope this round");
+  if (!engine) return err_status(1, "null engine");
+  auto *e = static_cast<Engine *>(engine);
+  std::vector<std::string> fields;
+  for (int i = 0; i < field_name_count; i++)
+    fields.emplace_back(field_names[i], field_name_lens[i]);
+  std::string err;
+  if (e->add_field_index(std::string(index_name, index_name_len), fields,
+                         std::string(index_type, index_type_len),
+                         std::string(index_params, index_params_len),
+                         &err))
+    return err_status(1, err.c_str());
+  return ok_status();
 }
 
 struct CStatus AddFieldIndexWithParams(
@@ -618,10 +619,12 @@ struct CStatus AddFieldIndexWithParams(
 
 static struct CStatus RemoveFieldIndex_unguarded(void *engine, const char *index_name,
                                 int index_name_len) {
-  (void)engine;
-  (void)index_name;
-  (void)index_name_len;
-  return err_status(1, "dynamic field indexes are out of scope this round");
+  if (!engine) return err_status(1, "null engine");
+  auto *e = static_cast<Engine *>(engine);
+  std::string err;
+  if (e->remove_field_index(std::string(index_name, index_name_len), &err))
+    return err_status(1, err.c_str());
+  return ok_status();
 }
 
 struct CStatus RemoveFieldIndex(void *engine, const char *index_name,

@@ -347,6 +347,22 @@ class Engine {
   int dump(std::string *err);
   int load(std::string *err);
 
+  /* Named field indexes (gamma_api.h:107-116, Engine::AddFieldIndex /
+   * RemoveFieldIndex — engine.cc:1561,1648). Scalar names route onto
+   * the existing ScalarFieldIndex machinery (built eagerly here, kept
+   * current lazily by filtered search); the single vector field's name
+   * registers against the vector index. Duplicate add and unknown
+   * remove are OK no-ops, as in the reference. */
+  int add_field_index(const std::string &name,
+                      const std::vector<std::string> &fields,
+                      const std::string &index_type,
+                      const std::string &params, std::string *err);
+  int remove_field_index(const std::string &name, std::string *err);
+  /* Backup (gamma_api.h:104, engine.cc:1515): command 0 = create (a
+   * full dump into <path>/backup); other commands are accepted no-ops
+   * as in the reference's BackupThread. */
+  int backup(int command, std::string *err);
+
   /* doc browse by scalar predicate (Engine::Query filter path,
    * engine.cc:420+): docids passing all filters, offset/limit applied */
   int filter_docids(const std::vector<TermFilterSpec> &terms,
@@ -392,6 +408,13 @@ class Engine {
   std::unordered_map<std::string, std::vector<std::string>> field_vals_;
   std::unordered_map<std::string, ScalarFieldIndex> scalar_idx_;
   std::mutex scalar_mu_; /* lazy index appends under the shared lock */
+  /* user-named indexes: name -> covered field names (vector or scalar) */
+  std::map<std::string, std::vector<std::string>> named_indexes_;
+  const ScalarFieldIndex *extend_scalar_index_(const std::string &fname,
+                                               int dt, bool want_terms,
+                                               bool want_range,
+                                               int64_t upto);
+  int dump_to_(const std::string &dir, std::string *err);
   std::unordered_map<std::string, int64_t> pkey2docid_;
   std::vector<std::string> docid2pkey_;
   RawStore raw_;

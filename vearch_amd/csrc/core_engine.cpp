@@ -280,6 +280,49 @@ static std::vector<std::string> split_x01(const std::string &s) {
 }
 }  // namespace
 
+/* Append rows [ix.*_upto, upto) of field `fname` into its scalar index
+ * (term postings and/or value-sorted arrays). Caller holds scalar_mu_.
+ * The lazy path (filtered search) and the explicit path
+ * (AddFieldIndexWithParams) share this. */
+const ScalarFieldIndex *Engine::extend_scalar_index_(
+    const std::string &fname, int dt, bool want_terms, bool want_range,
+    int64_t upto) {
+  auto cit = field_vals_.find(fname);
+  if (cit == field_vals_.end()) return nullptr;
+  const std::vector<std::string> &col = cit->second;
+  ScalarFieldIndex &ix = scalar_idx_[fname];
+  static const std::string kNone;
+  const int64_t n = upto;
+  if (want_terms && ix.terms_upto < n) {
+    for (int64_t id = ix.terms_upto; id < n; id++) {
+      const std::string &v = id < (int64_t)col.size() ? col[id] : kNone;
+      if (dt == 8) {
+        for (auto &el : split_x01(v))
+          if (!el.empty()) ix.postings[el].push_back(id);
+      } else {
+        ix.postings[v].push_back(id);
+      }
+    }
+    ix.terms_upto = n;
+  }
+  if (want_range && ix.range_upto < n) {
+    for (int64_t id = ix.range_upto; id < n; id++) {
+      const std::string &v = id < (int64_t)col.size() ? col[id] : kNone;
+      if (dt == 4) {
+        ix.svals.emplace_back(v, id);
+      } else {
+        double x;
+        if (decode_num(dt, v, &x)) ix.nvals.emplace_back(x, id);
+        /* decode failure: absent from nvals -> fails the range */
+      }
+    }
+    ix.range_upto = n;
+    std::sort(ix.svals.begin(), ix.svals.end());
+    std::sort(ix.nvals.begin(), ix.nvals.end());
+  }
+  return &ix;
+}
+
 int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
                                  const std::vector<RangeFilterSpec> &ranges,
                                  SearchScratch &sc,
@@ -304,39 +347,7 @@ int Engine::build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
     };
     auto extend = [&](const std::string &fname, int dt, bool want_terms,
                       bool want_range) -> const ScalarFieldIndex * {
-      auto cit = field_vals_.find(fname);
-      if (cit == field_vals_.end()) return nullptr;
-      const std::vector<std::string> &col = cit->second;
-      ScalarFieldIndex &ix = scalar_idx_[fname];
-      static const std::string kNone;
-      if (want_terms && ix.terms_upto < n) {
-        for (int64_t id = ix.terms_upto; id < n; id++) {
-          const std::string &v = id < (int64_t)col.size() ? col[id] : kNone;
-          if (dt == 8) {
-            for (auto &el : split_x01(v))
-              if (!el.empty()) ix.postings[el].push_back(id);
-          } else {
-            ix.postings[v].push_back(id);
-          }
-        }
-        ix.terms_upto = n;
-      }
-      if (want_range && ix.range_upto < n) {
-        for (int64_t id = ix.range_upto; id < n; id++) {
-          const std::string &v = id < (int64_t)col.size() ? col[id] : kNone;
-          if (dt == 4) {
-            ix.svals.emplace_back(v, id);
-          } else {
-            double x;
-            if (decode_num(dt, v, &x)) ix.nvals.emplace_back(x, id);
-            /* decode failure: absent from nvals -> fails the range */
-          }
-        }
-        ix.range_upto = n;
-        std::sort(ix.svals.begin(), ix.svals.end());
-        std::sort(ix.nvals.begin(), ix.nvals.end());
-      }
-      return &ix;
+      return extend_scalar_index_(fname, dt, want_terms, want_range, n);
     };
     for (auto &t : terms) {
       int dt = dtype_of(t.field);
@@ -716,10 +727,12 @@ std::string Engine::status_json() const {
 
 static const uint32_t kDumpMagic = 0x47414D41; /* "GAMA" */
 
-int Engine::dump(std::string *err) {
+int Engine::dump(std::string *err) { return dump_to_(path_, err); }
+
+int Engine::dump_to_(const std::string &dir, std::string *err) {
   std::unique_lock<std::shared_mutex> g(rw_);
-  mkdir(path_.c_str(), 0755);
-  std::string fn = path_ + "/gamma.dump";
+  mkdir(dir.c_str(), 0755);
+  std::string fn = dir + "/gamma.dump";
   FILE *f = fopen((fn + ".tmp").c_str(), "wb");
   if (!f) {
     if (err) *err = "cannot open dump file " + fn;
@@ -855,6 +868,87 @@ int Engine::load(std::string *err) {
     indexed_count_ = index_->ntotal();
   }
   fclose(f);
+  return 0;
+}
+
+int Engine::add_field_index(const std::string &name,
+                            const std::vector<std::string> &fields,
+                            const std::string &index_type,
+                            const std::string &params, std::string *err) {
+  (void)index_type;
+  (void)params;
+  if (!table_created_) {
+    if (err) *err = "table not initialized";
+    return -1;
+  }
+  if (name.empty()) {
+    if (err) *err = "index_name is empty";
+    return -1;
+  }
+  if (fields.empty()) {
+    if (err) *err = "field_names is empty";
+    return -1;
+  }
+  std::unique_lock<std::shared_mutex> g(rw_);
+  if (named_indexes_.count(name)) return 0; /* exists: ignore
+                                               (engine.cc:1578) */
+  if (fields[0] == vec_name_) {
+    /* vector index: exactly one field; the vector index itself is
+     * built by BuildIndex — this registers the name (engine.cc:1593) */
+    if (fields.size() != 1) {
+      if (err) *err = "vector index must reference exactly one field";
+      return -1;
+    }
+  } else {
+    /* scalar path: validate every field synchronously (engine.cc:1611),
+     * then build its ScalarFieldIndex eagerly so the first filtered
+     * search pays nothing */
+    for (auto &fname : fields) {
+      int dt = -1;
+      for (auto &fm : fields_)
+        if (fm.name == fname) dt = fm.data_type;
+      if (dt < 0) {
+        if (err) *err = "field [" + fname + "] not found in table";
+        return -1;
+      }
+      std::lock_guard<std::mutex> lk(scalar_mu_);
+      extend_scalar_index_(fname, dt, true, true, max_docid_);
+    }
+  }
+  named_indexes_[name] = fields;
+  return 0;
+}
+
+int Engine::remove_field_index(const std::string &name, std::string *err) {
+  if (!table_created_) {
+    if (err) *err = "table not initialized";
+    return -1;
+  }
+  if (name.empty()) {
+    if (err) *err = "index_name is empty";
+    return -1;
+  }
+  std::unique_lock<std::shared_mutex> g(rw_);
+  auto it = named_indexes_.find(name);
+  if (it == named_indexes_.end()) return 0; /* idempotent
+                                               (scalar_index_manager.h
+                                               RemoveIndex) */
+  for (auto &fname : it->second) {
+    if (fname == vec_name_) continue; /* the vector index itself stays;
+                                         only the name is dropped */
+    std::lock_guard<std::mutex> lk(scalar_mu_);
+    scalar_idx_.erase(fname); /* filtered search rebuilds lazily */
+  }
+  named_indexes_.erase(it);
+  return 0;
+}
+
+int Engine::backup(int command, std::string *err) {
+  if (command == 0) { /* create (BackupThread command 0) */
+    return dump_to_(path_ + "/backup", err);
+  }
+  /* other commands: the reference's BackupThread does nothing and
+   * reports success (engine.cc:1529-1553) */
   return 0;
 }
 

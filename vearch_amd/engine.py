@@ -54,6 +54,15 @@ def lib():
                                  c.POINTER(c.c_char_p), c.POINTER(c.c_int)]
         L.SetKillStatus.argtypes = [c.c_char_p, c.c_int, c.c_int]
         L.DeleteKillStatus.argtypes = [c.c_char_p, c.c_int]
+        L.Backup.restype = CStatus
+        L.Backup.argtypes = [c.c_void_p, c.c_int]
+        L.AddFieldIndexWithParams.restype = CStatus
+        L.AddFieldIndexWithParams.argtypes = [
+            c.c_void_p, c.c_char_p, c.c_int,
+            c.POINTER(c.c_char_p), c.POINTER(c.c_int), c.c_int,
+            c.c_char_p, c.c_int, c.c_char_p, c.c_int]
+        L.RemoveFieldIndex.restype = CStatus
+        L.RemoveFieldIndex.argtypes = [c.c_void_p, c.c_char_p, c.c_int]
         L.GammaBulkAdd.argtypes = [c.c_void_p, c.c_char_p, c.c_int, c.c_int,
                                    f32p]
         L.GammaRawSearch.argtypes = [c.c_void_p, c.c_int, f32p, c.c_int,
@@ -154,6 +163,29 @@ class GammaEngine:
         rc = lib().Load(self.h)
         if rc != 0:
             raise RuntimeError("Load failed")
+
+    def backup(self, command=0):
+        """Backup (gamma_api.h:104): command 0 = create."""
+        _check(lib().Backup(self.h, command), "Backup")
+
+    def add_field_index(self, name, fields, index_type="SCALAR",
+                        params=""):
+        """AddFieldIndexWithParams (gamma_api.h:107)."""
+        arr = (ctypes.c_char_p * len(fields))(
+            *[f.encode() for f in fields])
+        lens = (ctypes.c_int * len(fields))(
+            *[len(f.encode()) for f in fields])
+        st = lib().AddFieldIndexWithParams(
+            self.h, name.encode(), len(name.encode()), arr, lens,
+            len(fields), index_type.encode(), len(index_type.encode()),
+            params.encode(), len(params.encode()))
+        _check(st, "AddFieldIndexWithParams")
+
+    def remove_field_index(self, name):
+        """RemoveFieldIndex (gamma_api.h:114)."""
+        _check(lib().RemoveFieldIndex(self.h, name.encode(),
+                                      len(name.encode())),
+               "RemoveFieldIndex")
 
     def num_docs(self):
         return lib().GammaDebugNumDocs(self.h)

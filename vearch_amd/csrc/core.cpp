@@ -121,11 +121,12 @@ int RawStore::ensure_capacity(int64_t n_new, hipStream_t s) {
 
 int RawStore::add(const float *x, int64_t cnt, hipStream_t s) {
   if (cnt <= 0) return 0;
-  if (ensure_capacity(n_ + cnt, s)) return -1;
+  const int64_t n0 = n_.load(std::memory_order_relaxed);
+  if (ensure_capacity(n0 + cnt, s)) return -1;
   const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
   int64_t done = 0;
   while (done < cnt) {
-    int64_t vid = n_ + done;
+    int64_t vid = n0 + done;
     int64_t seg = vid >> SEG_SHIFT;
     int64_t off = vid & (seg_n - 1);
     int64_t room = seg_n - off;
@@ -142,7 +143,9 @@ int RawStore::add(const float *x, int64_t cnt, hipStream_t s) {
     done += take;
   }
   if (hipStreamSynchronize(s) != hipSuccess) return -1;
-  n_ += cnt;
+  /* publish AFTER the rows + norms are device-resident (the
+   * retrieve_idx_pos_ order, realtime_mem_data.cc:57-68) */
+  n_.store(n0 + cnt, std::memory_order_release);
   return 0;
 }
 
@@ -165,17 +168,19 @@ void RawStore::host_copy(int64_t start, int64_t cnt, float *out) const {
 const float *RawStore::dev_run(int64_t vid, int64_t *run_len) const {
   const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
   int64_t seg = vid >> SEG_SHIFT, off = vid & (seg_n - 1);
-  int64_t end = std::min(n_, (seg + 1) << SEG_SHIFT);
+  int64_t end = std::min(n_.load(std::memory_order_acquire),
+                         (seg + 1) << SEG_SHIFT);
   *run_len = end - vid;
   return (const float *)dev_segs_[seg] + (size_t)off * d_;
 }
 
 int RawStore::dump(FILE *f) const {
   fwrite(&d_, 4, 1, f);
-  fwrite(&n_, 8, 1, f);
+  int64_t n = n_.load(std::memory_order_acquire);
+  fwrite(&n, 8, 1, f);
   const int64_t seg_n = (int64_t)1 << SEG_SHIFT;
-  for (int64_t s = 0; s * seg_n < n_; s++) {
-    int64_t take = std::min(seg_n, n_ - s * seg_n);
+  for (int64_t s = 0; s * seg_n < n; s++) {
+    int64_t take = std::min(seg_n, n - s * seg_n);
     fwrite(host_segs_[s].data(), 4, (size_t)take * d_, f);
   }
   return 0;
@@ -935,6 +940,105 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
   }
   ntotal_ += n;
   return update_dev_buckets(s);
+}
+
+int IVFIndex::prepare_fast_one(const float *vec_h, hipStream_t s,
+                               int32_t *out_bucket, uint8_t *code_out,
+                               float *sval_out) {
+  if (!trained_) return 1;
+  if (dev_buckets_dirty_) return 1; /* device table stale: slow path
+                                       rebuilds it under the write lock */
+  DeviceBuf xd, xrot, xnorm, dots, asg, resid, codes, sterm;
+  if (xd.reserve((size_t)d_ * 4)) return -1;
+  GAMMA_CHECK(hipMemcpy(xd.get(), vec_h, (size_t)d_ * 4,
+                        hipMemcpyHostToDevice));
+  const float *xin = xd.as<float>();
+  if (has_opq()) { /* encode in rotated space (ivfpq.cc:470-471) */
+    if (xrot.reserve((size_t)d_ * 4)) return -1;
+    if (rotate_dev(xin, 1, xrot.as<float>(), s)) return -1;
+    xin = xrot.as<float>();
+  }
+  if (xnorm.reserve(4)) return -1;
+  if (gk::row_norms(s, xin, 1, d_, xnorm.as<float>()) != hipSuccess)
+    return -1;
+  if (dots.reserve((size_t)nlist_ * 4)) return -1;
+  if (gk::dots_mfma(s, xin, 1, centroids_.as<float>(), nlist_, d_,
+                    dots.as<float>()) != hipSuccess)
+    return -1;
+  if (asg.reserve(4)) return -1;
+  if (gk::argmin_rows(s, 1, nlist_, dots.as<float>(), xnorm.as<float>(),
+                      cent_norms_.as<float>(), !params_.metric_ip,
+                      asg.as<int32_t>()) != hipSuccess)
+    return -1;
+  GAMMA_CHECK(hipStreamSynchronize(s));
+  int32_t b = -1;
+  GAMMA_CHECK(hipMemcpy(&b, asg.get(), 4, hipMemcpyDeviceToHost));
+  if (b < 0 || b >= nlist_) return 1; /* degenerate assign: slow path */
+  if (buckets_[b].size + 1 > buckets_[b].cap) return 1; /* extension */
+  if (params_.kind == IndexKind::IVFPQ) {
+    if (resid.reserve((size_t)d_ * 4)) return -1;
+    if (gk::residuals(s, 1, d_, xin, centroids_.as<float>(),
+                      asg.as<int32_t>(), resid.as<float>()) != hipSuccess)
+      return -1;
+    if (codes.reserve((size_t)code_size_)) return -1;
+    if (gk::pq_encode(s, 1, d_, M_, ksub_, resid.as<float>(),
+                      codebooks_.as<float>(),
+                      codes.as<uint8_t>()) != hipSuccess)
+      return -1;
+    if (sterm.reserve(4)) return -1;
+    if (gk::pq_sterm(s, 1, M_, nlist_, codes.as<uint8_t>(),
+                     asg.as<int32_t>(), 0, btable_.as<float>(),
+                     sterm.as<float>()) != hipSuccess)
+      return -1;
+    GAMMA_CHECK(hipStreamSynchronize(s));
+    GAMMA_CHECK(hipMemcpy(code_out, codes.get(), (size_t)code_size_,
+                          hipMemcpyDeviceToHost));
+    GAMMA_CHECK(hipMemcpy(sval_out, sterm.get(), 4,
+                          hipMemcpyDeviceToHost));
+  }
+  *out_bucket = b;
+  return 0;
+}
+
+int IVFIndex::commit_fast_one(int32_t b, int64_t vid, const float *vec_h,
+                              const uint8_t *code, float sval,
+                              hipStream_t s) {
+  (void)s; /* synchronous small copies carry the ordering */
+  Bucket &bk = buckets_[b];
+  if (bk.size + 1 > bk.cap) return -1; /* prepare checked; appenders
+                                          are serialized */
+  const size_t entry = params_.kind == IndexKind::IVFPQ
+                           ? (size_t)code_size_
+                           : (size_t)d_ * 4;
+  /* data first ... */
+  if (params_.kind == IndexKind::IVFPQ) {
+    GAMMA_CHECK(hipMemcpy((uint8_t *)bk.data->get() +
+                              (size_t)bk.size * entry,
+                          code, entry, hipMemcpyHostToDevice));
+    GAMMA_CHECK(hipMemcpy(bk.svals->as<float>() + bk.size, &sval, 4,
+                          hipMemcpyHostToDevice));
+  } else {
+    GAMMA_CHECK(hipMemcpy((uint8_t *)bk.data->get() +
+                              (size_t)bk.size * entry,
+                          vec_h, entry, hipMemcpyHostToDevice));
+  }
+  uint32_t id32 = (uint32_t)vid;
+  GAMMA_CHECK(hipMemcpy(bk.ids->as<uint32_t>() + bk.size, &id32, 4,
+                        hipMemcpyHostToDevice));
+  /* ... size last: one aligned 8-byte write into this bucket's device
+   * descriptor (concurrent scans read the old or the new size — the
+   * retrieve_idx_pos_ publication, realtime_mem_data.cc:57-68) */
+  long long nsz = bk.size + 1;
+  char *size_dev = (char *)dev_buckets_.get() +
+                   (size_t)b * sizeof(GammaBucketDev) +
+                   offsetof(GammaBucketDev, size);
+  GAMMA_CHECK(hipMemcpy(size_dev, &nsz, 8, hipMemcpyHostToDevice));
+  bk.size = nsz;
+  if (vid >= (int64_t)vid_loc_.size())
+    vid_loc_.resize((size_t)vid + 1024, -1);
+  vid_loc_[vid] = ((int64_t)b << 40) | (nsz - 1);
+  ntotal_ += 1;
+  return 0;
 }
 
 int IVFIndex::del(int64_t vid, hipStream_t s) {

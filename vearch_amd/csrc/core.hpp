@@ -13,6 +13,7 @@
 #include <hip/hip_runtime.h>
 #include <stdint.h>
 
+#include <atomic>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -77,6 +78,51 @@ class DeviceBuf {
   size_t bytes_ = 0;
 };
 
+/* Append-only string column with stable element storage: readers under
+ * the engine's shared lock may index any element below the published
+ * doc count while an appender (also under the shared lock — the
+ * lock-free add path, realtime_mem_data.cc:57-68 publication order)
+ * pushes new rows. Blocks never move once allocated and the block
+ * table's storage is reserved up front, so existing elements are never
+ * relocated by growth. */
+class StableStrCol {
+  static constexpr int SHIFT = 13; /* 8192 strings per block */
+  static constexpr size_t BMASK = ((size_t)1 << SHIFT) - 1;
+  using Block = std::vector<std::string>;
+
+ public:
+  StableStrCol() { blocks_.reserve((size_t)1 << (31 - SHIFT)); }
+  size_t size() const { return size_.load(std::memory_order_acquire); }
+  void push_back(std::string v) {
+    size_t i = size_.load(std::memory_order_relaxed);
+    size_t b = i >> SHIFT;
+    if (b == blocks_.size()) {
+      blocks_.push_back(std::make_unique<Block>());
+      blocks_.back()->resize((size_t)1 << SHIFT);
+    }
+    (*blocks_[b])[i & BMASK] = std::move(v);
+    size_.store(i + 1, std::memory_order_release);
+  }
+  void resize(size_t n) { /* grow-only */
+    while (size_.load(std::memory_order_relaxed) < n)
+      push_back(std::string());
+  }
+  /* idempotent write-at-index (an aborted append may retry the same
+   * docid; a blind push_back would shift every later row) */
+  void set(size_t i, std::string v) {
+    resize(i + 1);
+    (*blocks_[i >> SHIFT])[i & BMASK] = std::move(v);
+  }
+  std::string &operator[](size_t i) { return (*blocks_[i >> SHIFT])[i & BMASK]; }
+  const std::string &operator[](size_t i) const {
+    return (*blocks_[i >> SHIFT])[i & BMASK];
+  }
+
+ private:
+  std::vector<std::unique_ptr<Block>> blocks_;
+  std::atomic<size_t> size_{0};
+};
+
 /* Raw vectors: fp32 row-major, append-only segments of 2^SEG_SHIFT
  * vectors (memory_raw_vector.h:58 segment array). Device segments are
  * the product store; host shadow serves training + Dump. */
@@ -85,7 +131,15 @@ class RawStore {
   static constexpr int SEG_SHIFT = 19; /* 512k vectors per segment */
   int init(int d);
   int add(const float *x, int64_t cnt, hipStream_t s); /* host pointer */
-  int64_t size() const { return n_; }
+  /* true if add(cnt) would allocate (new segment / norms growth /
+   * seg-table upload) — the lock-free append path requires false */
+  bool would_grow(int64_t cnt) const {
+    int64_t n_new = n_.load(std::memory_order_relaxed) + cnt;
+    return (size_t)((n_new + ((int64_t)1 << SEG_SHIFT) - 1) >> SEG_SHIFT) >
+               dev_segs_.size() ||
+           n_new > norms_cap_;
+  }
+  int64_t size() const { return n_.load(std::memory_order_acquire); }
   int dim() const { return d_; }
   const float *host_row(int64_t vid) const;
   /* copy [start,start+cnt) rows into a contiguous host buffer */
@@ -102,7 +156,9 @@ class RawStore {
  private:
   int ensure_capacity(int64_t n_new, hipStream_t s);
   int d_ = 0;
-  int64_t n_ = 0;
+  /* published AFTER data + norms are resident (release), so a reader
+   * holding only the shared lock sees a consistent prefix */
+  std::atomic<int64_t> n_{0};
   std::vector<void *> dev_segs_;
   std::vector<std::vector<float>> host_segs_;
   DeviceBuf seg_table_;
@@ -117,6 +173,7 @@ class Bitmap {
   bool test(int64_t vid) const;
   bool any() const { return set_count_ > 0; }
   int ensure(int64_t nbits, hipStream_t s);
+  bool has_capacity(int64_t nbits) const { return nbits <= bits_; }
   const uint32_t *dev() const { return dev_.as<uint32_t>(); }
   /* host shadow word (0 when out of range) for filter-bitmap merges */
   uint32_t host_word(int64_t w) const {
@@ -186,6 +243,21 @@ class IVFIndex {
   int train(const float *xt, int64_t n, hipStream_t s, std::string *err);
   int add(const float *x_host, const int64_t *vids, int64_t n,
           hipStream_t s);
+  /* Lock-free-append support (realtime_mem_data.cc:57-68 publication
+   * order): prepare_fast_one assigns/encodes ONE vector
+   * without touching any shared structure and bails (returns 1) when
+   * the target bucket would need extension (or anything else needs the
+   * write-locked slow path); commit_fast_one then appends into the
+   * bucket's existing capacity and publishes the new size LAST — host
+   * AND the device GammaBucketDev entry — so concurrent read-locked
+   * searches always see a consistent prefix. Appenders are serialized
+   * by the engine's append mutex. Returns 0 ok, 1 = use slow path,
+   * -1 error. */
+  int prepare_fast_one(const float *vec_h, hipStream_t s,
+                       int32_t *out_bucket, uint8_t *code_out,
+                       float *sval_out);
+  int commit_fast_one(int32_t b, int64_t vid, const float *vec_h,
+                      const uint8_t *code, float sval, hipStream_t s);
   int del(int64_t vid, hipStream_t s); /* set bit 63 in the bucket slot */
   /* search: writes keys (nq x k2) into out_keys (device) */
   /* S = probe-split (out_keys must hold nq*S*k2 keys); see
@@ -370,7 +442,9 @@ class Engine {
                     int limit, std::vector<int64_t> *out, std::string *err,
                     int filter_op = 0, bool prelocked = false);
 
-  int64_t num_docs() const { return max_docid_; }
+  int64_t num_docs() const {
+    return max_docid_.load(std::memory_order_acquire);
+  }
   IVFIndex *index() { return index_.get(); }
   RawStore &raw() { return raw_; }
   Bitmap &bitmap() { return bitmap_; }
@@ -405,7 +479,7 @@ class Engine {
   int dim_ = 0;
   int training_threshold_ = 0;
   std::vector<FieldMeta> fields_;
-  std::unordered_map<std::string, std::vector<std::string>> field_vals_;
+  std::unordered_map<std::string, StableStrCol> field_vals_;
   std::unordered_map<std::string, ScalarFieldIndex> scalar_idx_;
   std::mutex scalar_mu_; /* lazy index appends under the shared lock */
   /* user-named indexes: name -> covered field names (vector or scalar) */
@@ -415,14 +489,25 @@ class Engine {
                                                bool want_range,
                                                int64_t upto);
   int dump_to_(const std::string &dir, std::string *err);
+  /* lock-free append (caller holds the SHARED lock + append_mu_):
+   * 0 = done, 1 = caller must retry under the write lock, -1 error */
+  int add_doc_fast_(
+      const std::string &p_key,
+      const std::vector<std::pair<std::string, std::string>> &fields,
+      const float *vec);
   std::unordered_map<std::string, int64_t> pkey2docid_;
-  std::vector<std::string> docid2pkey_;
+  StableStrCol docid2pkey_;
+  mutable std::mutex pkey_mu_; /* pkey2docid_: read-locked appenders
+                                  insert while Query/GetDocByID read */
+  std::mutex append_mu_;       /* serializes lock-free appenders */
   RawStore raw_;
   Bitmap bitmap_;
   std::unique_ptr<IVFIndex> index_;
   IndexParams params_;
-  int64_t max_docid_ = 0;
-  int64_t indexed_count_ = 0;
+  /* published with release AFTER all row state (columns, pkey, raw
+   * vectors, bucket entry) is visible — the lock-free append contract */
+  std::atomic<int64_t> max_docid_{0};
+  std::atomic<int64_t> indexed_count_{0};
   bool table_created_ = false;
   hipStream_t stream_ = nullptr; /* mutation stream (add/build/load) */
   int cached_nq_ = 0;

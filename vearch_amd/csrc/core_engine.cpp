@@ -52,7 +52,7 @@ int Engine::create_table(const std::string &name,
   }
   space_name_ = space_name_.empty() ? name : space_name_;
   fields_ = scalar_fields;
-  for (auto &f : fields_) field_vals_[f.name] = {};
+  for (auto &f : fields_) field_vals_[f.name]; /* default-construct */
   vec_name_ = vec_name;
   dim_ = dimension;
   index_type_ = index_type.empty() ? "IVFPQ" : index_type;
@@ -92,18 +92,35 @@ int Engine::add_doc(
     const std::vector<std::pair<std::string, std::string>> &fields,
     const float *vec, int vec_len) {
   if (!table_created_ || vec_len != dim_) return -1;
-  std::unique_lock<std::shared_mutex> g(rw_);
-  auto it = pkey2docid_.find(p_key);
-  if (it != pkey2docid_.end()) {
-    /* update = delete old + add new (engine.cc AddOrUpdate semantics) */
-    int64_t old = it->second;
-    bitmap_.set(old, stream_);
-    if (index_) index_->del(old, stream_);
+  /* Lock-free common append (realtime_mem_data.cc:57-68
+   * retrieve_idx_pos_ pattern): a pure append that fits every existing
+   * capacity runs under the SHARED lock — concurrent searches are
+   * never blocked — with all row state written before the new docid /
+   * bucket size is published. Anything structural (update of an
+   * existing pkey, raw-segment or bucket growth, bitmap growth)
+   * retries under the write lock. */
+  {
+    std::shared_lock<std::shared_mutex> g(rw_);
+    std::lock_guard<std::mutex> ap(append_mu_);
+    int rc = add_doc_fast_(p_key, fields, vec);
+    if (rc != 1) return rc;
   }
-  int64_t docid = max_docid_++;
-  pkey2docid_[p_key] = docid;
-  docid2pkey_.push_back(p_key);
-  bitmap_.ensure(max_docid_, stream_);
+  std::unique_lock<std::shared_mutex> g(rw_);
+  std::lock_guard<std::mutex> ap(append_mu_);
+  {
+    std::lock_guard<std::mutex> pk(pkey_mu_);
+    auto it = pkey2docid_.find(p_key);
+    if (it != pkey2docid_.end()) {
+      /* update = delete old + add new (engine.cc AddOrUpdate
+       * semantics) */
+      int64_t old = it->second;
+      bitmap_.set(old, stream_);
+      if (index_) index_->del(old, stream_);
+    }
+  }
+  int64_t docid = max_docid_.load(std::memory_order_relaxed);
+  docid2pkey_.set(docid, p_key);
+  bitmap_.ensure(docid + 1, stream_);
   for (auto &f : fields_) {
     auto &col = field_vals_[f.name];
     col.resize(docid + 1);
@@ -115,24 +132,75 @@ int Engine::add_doc(
     if (index_->add(vec, &docid, 1, stream_)) return -1;
     indexed_count_++;
   }
+  {
+    std::lock_guard<std::mutex> pk(pkey_mu_);
+    pkey2docid_[p_key] = docid;
+  }
+  max_docid_.store(docid + 1, std::memory_order_release);
+  return 0;
+}
+
+int Engine::add_doc_fast_(
+    const std::string &p_key,
+    const std::vector<std::pair<std::string, std::string>> &fields,
+    const float *vec) {
+  {
+    std::lock_guard<std::mutex> pk(pkey_mu_);
+    if (pkey2docid_.count(p_key)) return 1; /* update -> slow path */
+  }
+  int64_t docid = max_docid_.load(std::memory_order_relaxed);
+  if (docid + 1 > 0x7fffffff) return -1; /* 31-bit device vid */
+  if (raw_.would_grow(1)) return 1;
+  if (!bitmap_.has_capacity(docid + 1)) return 1;
+  int32_t bucket = -1;
+  uint8_t code[256];
+  float sval = 0.0f;
+  bool commit_index = false;
+  if (index_ && index_->trained()) {
+    if (index_->code_size() > (int)sizeof(code)) return 1;
+    int rc = index_->prepare_fast_one(vec, stream_, &bucket, code, &sval);
+    if (rc != 0) return rc; /* 1 = bucket needs extension -> slow */
+    commit_index = true;
+  }
+  /* all row state BEFORE publication */
+  docid2pkey_.set(docid, p_key);
+  for (auto &f : fields_) {
+    auto &col = field_vals_[f.name];
+    col.resize(docid + 1);
+    for (auto &kv : fields)
+      if (kv.first == f.name) col[docid] = kv.second;
+  }
+  if (raw_.add(vec, 1, stream_)) return -1; /* publishes raw n_ last */
+  if (commit_index) {
+    if (index_->commit_fast_one(bucket, docid, vec, code, sval, stream_))
+      return -1;
+    indexed_count_++;
+  }
+  {
+    std::lock_guard<std::mutex> pk(pkey_mu_);
+    pkey2docid_[p_key] = docid;
+  }
+  max_docid_.store(docid + 1, std::memory_order_release);
   return 0;
 }
 
 int Engine::bulk_add(int64_t n, const float *vecs) {
   if (!table_created_ || n <= 0) return -1;
   std::unique_lock<std::shared_mutex> g(rw_);
-  int64_t base = max_docid_;
+  int64_t base = max_docid_.load(std::memory_order_relaxed);
   if ((uint64_t)(base + n) > 0x7fffffffull) return -1; /* 31-bit vids */
-  docid2pkey_.reserve(base + n);
   char buf[24];
-  for (int64_t i = 0; i < n; i++) {
-    int64_t docid = base + i;
-    snprintf(buf, sizeof buf, "%lld", (long long)docid);
-    pkey2docid_.emplace(buf, docid);
-    docid2pkey_.push_back(buf);
+  {
+    std::lock_guard<std::mutex> pk(pkey_mu_);
+    for (int64_t i = 0; i < n; i++) {
+      int64_t docid = base + i;
+      snprintf(buf, sizeof buf, "%lld", (long long)docid);
+      pkey2docid_.emplace(buf, docid);
+      docid2pkey_.set(docid, buf);
+    }
   }
-  max_docid_ += n;
-  bitmap_.ensure(max_docid_, stream_);
+  max_docid_.store(base + n, std::memory_order_release);
+  bitmap_.ensure(base + n, stream_);
   if (raw_.add(vecs, n, stream_)) return -1;
   if (index_ && index_->trained()) {
     std::vector<int64_t> vids(n);
@@ -145,6 +213,7 @@ int Engine::bulk_add(int64_t n, const float *vecs) {
 
 int Engine::delete_doc(const std::string &p_key) {
   std::unique_lock<std::shared_mutex> g(rw_);
+  std::lock_guard<std::mutex> pk(pkey_mu_);
   auto it = pkey2docid_.find(p_key);
   if (it == pkey2docid_.end()) return -1;
   bitmap_.set(it->second, stream_);
@@ -289,7 +358,7 @@ const ScalarFieldIndex *Engine::extend_scalar_index_(
     int64_t upto) {
   auto cit = field_vals_.find(fname);
   if (cit == field_vals_.end()) return nullptr;
-  const std::vector<std::string> &col = cit->second;
+  const StableStrCol &col = cit->second;
   ScalarFieldIndex &ix = scalar_idx_[fname];
   static const std::string kNone;
   const int64_t n = upto;
@@ -696,6 +765,7 @@ int Engine::cache_queries(int nq, const float *xq) {
 }
 
 int64_t Engine::docid_of(const std::string &p_key) const {
+  std::lock_guard<std::mutex> pk(pkey_mu_);
   auto it = pkey2docid_.find(p_key);
   return it == pkey2docid_.end() ? -1 : it->second;
 }
@@ -718,9 +788,10 @@ std::string Engine::status_json() const {
   snprintf(buf, sizeof buf,
            "{\"doc_count\": %lld, \"index_status\": %d, \"min_indexed_num\": "
            "%lld, \"max_docid\": %lld, \"table_name\": \"%s\"}",
-           (long long)(max_docid_ - bitmap_.popcount()),
+           (long long)(max_docid_.load() - bitmap_.popcount()),
            (index_ && index_->trained()) ? 2 : 0,
-           (long long)indexed_count_, (long long)(max_docid_ - 1),
+           (long long)indexed_count_.load(),
+           (long long)(max_docid_.load() - 1),
            gjson::escape(space_name_).c_str());
   return buf;
 }
@@ -751,17 +822,18 @@ int Engine::dump_to_(const std::string &dir, std::string *err) {
   wstr(vec_name_);
   fwrite(&dim_, 4, 1, f);
   fwrite(&training_threshold_, 4, 1, f);
-  fwrite(&max_docid_, 8, 1, f);
+  int64_t maxdoc_now = max_docid_.load(std::memory_order_acquire);
+  fwrite(&maxdoc_now, 8, 1, f);
   int nfields = (int)fields_.size();
   fwrite(&nfields, 4, 1, f);
   for (auto &fm : fields_) {
     wstr(fm.name);
     fwrite(&fm.data_type, 4, 1, f);
   }
-  for (int64_t i = 0; i < max_docid_; i++) wstr(docid2pkey_[i]);
+  for (int64_t i = 0; i < maxdoc_now; i++) wstr(docid2pkey_[i]);
   for (auto &fm : fields_) {
     auto &col = field_vals_.at(fm.name);
-    for (int64_t i = 0; i < max_docid_; i++)
+    for (int64_t i = 0; i < maxdoc_now; i++)
       wstr(i < (int64_t)col.size() ? col[i] : kEmpty);
   }
   raw_.dump(f);
@@ -846,11 +918,14 @@ int Engine::load(std::string *err) {
       return -1;
     }
   }
-  max_docid_ = maxdoc;
+  max_docid_.store(maxdoc, std::memory_order_release);
   docid2pkey_.resize(maxdoc);
-  for (int64_t i = 0; i < maxdoc; i++) {
-    if (rstr(docid2pkey_[i])) { fclose(f); return -1; }
-    pkey2docid_[docid2pkey_[i]] = i;
+  {
+    std::lock_guard<std::mutex> pk(pkey_mu_);
+    for (int64_t i = 0; i < maxdoc; i++) {
+      if (rstr(docid2pkey_[i])) { fclose(f); return -1; }
+      pkey2docid_[docid2pkey_[i]] = i;
+    }
   }
   for (auto &fm : fms) {
     auto &col = field_vals_[fm.name];
@@ -860,7 +935,7 @@ int Engine::load(std::string *err) {
   }
   if (raw_.load(f, stream_)) { fclose(f); return -1; }
   if (bitmap_.load(f, stream_)) { fclose(f); return -1; }
-  bitmap_.ensure(max_docid_, stream_);
+  bitmap_.ensure(maxdoc, stream_);
   int has_index = 0;
   if (fread(&has_index, 4, 1, f) != 1) return bad("index flag");
   if (has_index && index_) {

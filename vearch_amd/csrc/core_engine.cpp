@@ -1020,6 +1020,7 @@ int Engine::remove_field_index(const std::string &name, std::string *err) {
 
 int Engine::backup(int command, std::string *err) {
   if (command == 0) { /* create (BackupThread command 0) */
+    mkdir(path_.c_str(), 0755); /* parent first (utils::make_dir) */
     return dump_to_(path_ + "/backup", err);
   }
   /* other commands: the reference's BackupThread does nothing and

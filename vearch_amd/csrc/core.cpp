@@ -6,6 +6,7 @@
 
 #include <math.h>
 #include <stdio.h>
+#include <stdlib.h>
 #include <string.h>
 
 #include <algorithm>
@@ -1098,6 +1099,12 @@ int IVFIndex::probe_split(int nq, int k2, int nprobe) const {
    * probes across S sub-workgroups (merged by sort_rows; S*k2 <= 2048
    * keeps the merge a single row sort) */
   if (params_.kind != IndexKind::IVFPQ) return 1;
+  const char *force = getenv("GAMMA_SCAN_S"); /* perf experiments */
+  if (force && atoi(force) > 0) {
+    int S = atoi(force);
+    while ((int64_t)S * k2 > 2048 && S > 1) S /= 2;
+    return S;
+  }
   int S = 1;
   while (nq * S < 1024 && S < nprobe && (int64_t)(2 * S) * k2 <= 2048)
     S *= 2;

@@ -10,6 +10,7 @@
  */
 #include <hip/hip_runtime.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 #include "kernels.h"
 #include "select.hpp"
@@ -582,7 +583,15 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
   /* batched path needs flush margin blockDim*C inside the selector cap.
    * 512-thread blocks put 24 waves on a CU at the same LDS/WG (the ADC
    * gather stream wants ~4 waves/SIMD — microarch §LDS). */
-  const int BS = 512;
+  /* BS=256 wins when lists are short (nlist=16384: ~N/nlist codes per
+   * list barely fills 512 threads; 256 halves the tail waste and puts
+   * more independent workgroups on each CU). GAMMA_SCAN_BS overrides
+   * for experiments. */
+  int BS = 512;
+  {
+    const char *e = getenv("GAMMA_SCAN_BS");
+    if (e && (atoi(e) == 256 || atoi(e) == 512)) BS = atoi(e);
+  }
   bool fast = (k2 + BS * GAMMA_ADC_C) <= GAMMA_SORT_CAP &&
               (M == 16 || M == 32 || M == 64 || M == 96);
   dim3 g((uint32_t)nq * (uint32_t)S);
@@ -591,19 +600,25 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
       nq, S, d, M, nprobe, k2, queries, centroids, codebooks, atab,       \
       probe_dists, buckets, nlist, probes, bitmap, out_keys,              \
       kill_flag)
+#define GAMMA_LAUNCH_SCAN_BS(IPV, MWV)                                    \
+  do {                                                                    \
+    if (BS == 256) GAMMA_LAUNCH_SCAN(IPV, MWV, 256);                      \
+    else GAMMA_LAUNCH_SCAN(IPV, MWV, 512);                                \
+  } while (0)
   if (ip) {
     if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG);
-    else if (M == 16) GAMMA_LAUNCH_SCAN(true, 4, 512);
-    else if (M == 32) GAMMA_LAUNCH_SCAN(true, 8, 512);
-    else if (M == 64) GAMMA_LAUNCH_SCAN(true, 16, 512);
-    else GAMMA_LAUNCH_SCAN(true, 24, 512);
+    else if (M == 16) GAMMA_LAUNCH_SCAN_BS(true, 4);
+    else if (M == 32) GAMMA_LAUNCH_SCAN_BS(true, 8);
+    else if (M == 64) GAMMA_LAUNCH_SCAN_BS(true, 16);
+    else GAMMA_LAUNCH_SCAN_BS(true, 24);
   } else {
     if (!fast) GAMMA_LAUNCH_SCAN(false, 0, WG);
-    else if (M == 16) GAMMA_LAUNCH_SCAN(false, 4, 512);
-    else if (M == 32) GAMMA_LAUNCH_SCAN(false, 8, 512);
-    else if (M == 64) GAMMA_LAUNCH_SCAN(false, 16, 512);
-    else GAMMA_LAUNCH_SCAN(false, 24, 512);
+    else if (M == 16) GAMMA_LAUNCH_SCAN_BS(false, 4);
+    else if (M == 32) GAMMA_LAUNCH_SCAN_BS(false, 8);
+    else if (M == 64) GAMMA_LAUNCH_SCAN_BS(false, 16);
+    else GAMMA_LAUNCH_SCAN_BS(false, 24);
   }
+#undef GAMMA_LAUNCH_SCAN_BS
 #undef GAMMA_LAUNCH_SCAN
   return hipGetLastError();
 }

@@ -822,10 +822,10 @@ def test_lockfree_add_under_search(data):
     docs = eng.query_pb(document_ids=[str(n0 + n_new - 1)])
     assert docs[0]["items"]
     # a query equal to an appended vector finds it (or its duplicate
-    # source row — same vector, ties by id): check the top hit's
-    # distance is 0 at nprobe=nlist
+    # source row — same vector, ties by id): with the exact rerank leg
+    # the top hit's canonical distance is exactly 0 at nprobe=nlist
     probe = base[(n0 + 17) % len(base)]
-    gd, gi = eng.raw_search(probe[None, :], 3, nprobe=32)
+    gd, gi = eng.raw_search(probe[None, :], 3, nprobe=32, rerank=32)
     assert gd[0, 0] == 0.0
     eng.close()
 

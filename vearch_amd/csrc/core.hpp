@@ -147,6 +147,7 @@ class RawStore {
   const float *const *dev_seg_table() const {
     return (const float *const *)seg_table_.get();
   }
+  int num_segs() const { return (int)dev_segs_.size(); }
   const float *dev_norms() const { return norms_.as<float>(); }
   /* contiguous device run beginning at vid (within one segment) */
   const float *dev_run(int64_t vid, int64_t *run_len) const;

@@ -703,7 +703,7 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
 
   if (need_canonical_rerank) {
     GAMMA_CHECK(gk::rerank(s, nq, k2, dim_, qptr, raw_.dev_seg_table(),
-                           RawStore::SEG_SHIFT, ip,
+                           raw_.num_segs(), RawStore::SEG_SHIFT, ip,
                            sc.keys.as<uint64_t>(),
                            sc.keys.as<uint64_t>()));
   }

@@ -110,8 +110,8 @@ hipError_t ivfflat_scan(hipStream_t s, int nq, int d, int nprobe, int k2,
  * keys_out may alias keys_in. */
 hipError_t rerank(hipStream_t s, int nq, int ncand, int d,
                   const float *queries, const float *const *segs,
-                  int seg_shift, bool ip, const uint64_t *keys_in,
-                  uint64_t *keys_out);
+                  int n_segs, int seg_shift, bool ip,
+                  const uint64_t *keys_in, uint64_t *keys_out);
 
 /* Per-row sort of ncand keys (<= 2048), emit top-k (dists, ids). */
 hipError_t sort_rows(hipStream_t s, int nq, int ncand, int k,

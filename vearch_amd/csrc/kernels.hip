@@ -251,17 +251,44 @@ k_select_from_dots_wave(int nq, int64_t ncols, int64_t col_base,
     }
     sel.finish();
   }
-  for (int64_t c0 = 0; c0 < ncols; c0 += 64) {
-    int64_t c = c0 + lane;
-    if (c < ncols) {
-      int64_t id = col_base + c;
-      if (!gamma_bitmap_test(bitmap, (uint64_t)id)) {
-        float dot = row[c];
-        float dist = l2 ? fmaf(-2.0f, dot, qn + bnorms[id]) : dot;
-        sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
+  if ((ncols & 3) == 0 && (col_base & 3) == 0 && ((uintptr_t)row & 15) == 0) {
+    /* vectorized: 4 dists (and 4 bnorms) per lane per pass — quarters
+     * the serial trip count over a 16k-wide row (nlist=16384 coarse
+     * assign reads 64 KB/query; the scalar loop was latency-bound) */
+    const float4 *row4 = (const float4 *)row;
+    for (int64_t c0 = 0; c0 < (ncols >> 2); c0 += 64) {
+      int64_t c4 = c0 + lane;
+      if (c4 < (ncols >> 2)) {
+        int64_t id0 = col_base + c4 * 4;
+        float4 d4 = row4[c4];
+        float4 b4 = l2 ? *(const float4 *)(bnorms + id0)
+                       : make_float4(0, 0, 0, 0);
+        float dv[4] = {d4.x, d4.y, d4.z, d4.w};
+        float bv[4] = {b4.x, b4.y, b4.z, b4.w};
+#pragma unroll
+        for (int t = 0; t < 4; t++) {
+          int64_t id = id0 + t;
+          if (!gamma_bitmap_test(bitmap, (uint64_t)id)) {
+            float dist = l2 ? fmaf(-2.0f, dv[t], qn + bv[t]) : dv[t];
+            sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
+          }
+        }
       }
+      sel.maybe_flush(64 * 4);
     }
-    sel.maybe_flush(64);
+  } else {
+    for (int64_t c0 = 0; c0 < ncols; c0 += 64) {
+      int64_t c = c0 + lane;
+      if (c < ncols) {
+        int64_t id = col_base + c;
+        if (!gamma_bitmap_test(bitmap, (uint64_t)id)) {
+          float dot = row[c];
+          float dist = l2 ? fmaf(-2.0f, dot, qn + bnorms[id]) : dot;
+          sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
+        }
+      }
+      sel.maybe_flush(64);
+    }
   }
   sel.finish();
   for (int i = lane; i < k2; i += 64)
@@ -870,18 +897,25 @@ __global__ void k_rerank(int nq, int ncand, int d,
 template <bool IP, int DV, int BS>
 __global__ void __launch_bounds__(BS)
 k_rerank_lds(int nq, int ncand, const float *__restrict__ queries,
-             const float *const *__restrict__ segs, int seg_shift,
-             const uint64_t *__restrict__ keys_in,
+             const float *const *__restrict__ segs, int n_segs,
+             int seg_shift, const uint64_t *__restrict__ keys_in,
              uint64_t *__restrict__ keys_out) {
   const int CH = 32; /* floats per slice = 128 B */
+  const int SEG_LDS = 64; /* seg-table cache (64 segs = 33M vectors) */
   __shared__ float rows[BS][CH + 1]; /* +1: bank-shift */
   __shared__ float qs[DV];
+  __shared__ const float *seg_lds[SEG_LDS];
   const int nchunk = (ncand + BS - 1) / BS;
   const int q = blockIdx.x / nchunk;
   const int c0 = (blockIdx.x - q * nchunk) * BS;
   if (q >= nq) return;
   for (int i = threadIdx.x; i < DV; i += BS)
     qs[i] = queries[(size_t)q * DV + i];
+  /* the row gathers chase segs[id>>shift] before every load; serve the
+   * tiny pointer table from LDS instead of L2 */
+  const bool seg_cached = n_segs <= SEG_LDS;
+  if (seg_cached)
+    for (int i = threadIdx.x; i < n_segs; i += BS) seg_lds[i] = segs[i];
   const int64_t seg_mask = ((int64_t)1 << seg_shift) - 1;
   const int my = c0 + threadIdx.x;
   uint64_t mykey = GAMMA_KEY_EMPTY;
@@ -898,7 +932,8 @@ k_rerank_lds(int nq, int ncand, const float *__restrict__ queries,
         uint64_t key = keys_in[(size_t)q * ncand + cand];
         if (key != GAMMA_KEY_EMPTY) {
           uint32_t id = (uint32_t)(key & 0xffffffffu);
-          const float *v = segs[id >> seg_shift] +
+          const float *v = (seg_cached ? seg_lds[id >> seg_shift]
+                                       : segs[id >> seg_shift]) +
                            (size_t)(id & seg_mask) * DV;
           float4 x = *(const float4 *)(v + sl * CH + f4 * 4);
           rows[r][f4 * 4 + 0] = x.x;
@@ -930,8 +965,8 @@ k_rerank_lds(int nq, int ncand, const float *__restrict__ queries,
 
 hipError_t gk::rerank(hipStream_t s, int nq, int ncand, int d,
                       const float *queries, const float *const *segs,
-                      int seg_shift, bool ip, const uint64_t *keys_in,
-                      uint64_t *keys_out) {
+                      int n_segs, int seg_shift, bool ip,
+                      const uint64_t *keys_in, uint64_t *keys_out) {
   int64_t total = (int64_t)nq * ncand;
   if (total == 0) return hipSuccess;
   if (d == 128 || d == 768) {
@@ -940,7 +975,7 @@ hipError_t gk::rerank(hipStream_t s, int nq, int ncand, int d,
     dim3 g((uint32_t)((int64_t)nq * nchunk));
 #define GAMMA_LAUNCH_RERANK_LDS(IPV, DVV)                                 \
   k_rerank_lds<IPV, DVV, BS><<<g, dim3(BS), 0, s>>>(                      \
-      nq, ncand, queries, segs, seg_shift, keys_in, keys_out)
+      nq, ncand, queries, segs, n_segs, seg_shift, keys_in, keys_out)
     if (ip) {
       if (d == 128) GAMMA_LAUNCH_RERANK_LDS(true, 128);
       else GAMMA_LAUNCH_RERANK_LDS(true, 768);

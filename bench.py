@@ -80,9 +80,14 @@ HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
 # profiles/r01_fetch_size_c1.csv and profiles/README.md. Any other
 # shape reports traffic=null rather than guessing.
 MEASURED_TRAFFIC_BYTES = {
-    # r01 at nlist=4096 with per-list B staging measured 112.2 GB
-    # (profiles/r01_fetch_size_c1.csv); the r02 S-term scan at
-    # nlist=16384 awaits its own PMC pass — null until measured.
+    # r02 S-term scan at nlist=16384: rocprofv3 --pmc FETCH_SIZE
+    # 4.5613 GB raw/dispatch x2 gfx950 wide-read correction = 9.12 GB
+    # true per 10k-query step (profiles/r02_fetch_size.csv) — equal to
+    # the 9.11 GB of algorithmic bytes actually scanned (22.8k
+    # codes/query x 40 B), i.e. zero table/re-read waste. r01 at
+    # nlist=4096 with per-list B staging was 112.2 GB
+    # (profiles/r01_fetch_size_c1.csv).
+    "ivfpq_d128_n10m_nprobe32": 9.12e9,
 }
 
 

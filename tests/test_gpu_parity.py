@@ -749,6 +749,132 @@ def test_concurrent_search_and_mutation(data):
     eng.close()
 
 
+def test_multi_vector_search(data):
+    """Multi-vector-field search (vector_manager.cc:851-1090): per-field
+    top-n, docid-intersection merge, WeightedRanker combined scores,
+    docid-order vs multi_vector_rank score-order — checked against a
+    python replica of the reference merge over the oracle's exact
+    per-field FLAT results."""
+    base, q = data  # d=64
+    n, nq, topn, d2 = 3000, 5, 20, 32
+    rng = np.random.default_rng(97)
+    base2 = rng.standard_normal((n, d2)).astype(np.float32)
+    q2 = base2[:nq] + 0.05 * rng.standard_normal((nq, d2)).astype(
+        np.float32)
+    eng = make_engine("/tmp/gamma_multivec")
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}',
+                     extra_vecs=[("emb2", d2)])
+    for vid in range(n):
+        eng.add_doc(str(vid), base[vid],
+                    extra_vecs=[("emb2", base2[vid])])
+
+    # expected: per-field exact results from the oracle, merged exactly
+    # as the reference does (intersection, weights, order)
+    d1o, i1o = orc.flat_search(base[:n], q[:nq], topn, "L2")
+    d2o, i2o = orc.flat_search(base2, q2, topn, "L2")
+
+    def merge(t, weights, by_score):
+        l1 = sorted((int(i), float(d)) for i, d in zip(i1o[t], d1o[t])
+                    if i >= 0)
+        m2 = {int(i): float(d) for i, d in zip(i2o[t], d2o[t]) if i >= 0}
+        out = []
+        for i, dd in l1:
+            if i in m2:
+                score = dd * weights[0] + m2[i] * weights[1]
+                out.append((score, i))
+        if by_score:
+            out.sort(key=lambda x: x[0])
+        return out
+
+    # score-ordered (multi_vector_rank), default weights 1/2
+    res = eng.search_pb(q[:nq], topn=topn,
+                        extra_vec_queries=[("emb2", q2)],
+                        multi_vector_rank=1)
+    for t in range(nq):
+        exp = merge(t, (0.5, 0.5), True)
+        got = [(it["score"], int(it["fields"]["_id"]))
+               for it in res[t]["items"]]
+        assert [g[1] for g in got] == [e[1] for e in exp]
+        for g, e in zip(got, exp):
+            assert abs(g[0] - e[0]) < 1e-6 * max(1.0, abs(e[0]))
+
+    # docid-ordered (no multi_vector_rank)
+    res = eng.search_pb(q[:nq], topn=topn,
+                        extra_vec_queries=[("emb2", q2)])
+    for t in range(nq):
+        exp = merge(t, (0.5, 0.5), False)
+        got_ids = [int(it["fields"]["_id"]) for it in res[t]["items"]]
+        assert got_ids == [e[1] for e in exp]
+        assert got_ids == sorted(got_ids)
+
+    # WeightedRanker weights
+    res = eng.search_pb(
+        q[:nq], topn=topn, extra_vec_queries=[("emb2", q2)],
+        multi_vector_rank=1,
+        ranker='{"type": "WeightedRanker", "params": [0.9, 0.1]}')
+    for t in range(nq):
+        exp = merge(t, (0.9, 0.1), True)
+        got = [(it["score"], int(it["fields"]["_id"]))
+               for it in res[t]["items"]]
+        assert [g[1] for g in got] == [e[1] for e in exp]
+    # bad ranker length is an error (common_query_data.h:291)
+    with pytest.raises(RuntimeError, match="length"):
+        eng.search_pb(q[:2], topn=5, extra_vec_queries=[("emb2", q2[:2])],
+                      ranker='{"type": "WeightedRanker", "params": [1.0]}')
+    # unknown field name is an error
+    with pytest.raises(RuntimeError, match="unknown vector field"):
+        eng.search_pb(q[:2], topn=5,
+                      extra_vec_queries=[("nope", q2[:2])])
+
+    # dump/load keeps the extra field (v2 dump format)
+    eng.dump()
+    eng.close()
+    eng2 = make_engine("/tmp/gamma_multivec")
+    eng2.create_table(64, "FLAT", '{"metric_type": "L2"}',
+                      extra_vecs=[("emb2", d2)])
+    eng2.load()
+    res = eng2.search_pb(q[:nq], topn=topn,
+                         extra_vec_queries=[("emb2", q2)],
+                         multi_vector_rank=1)
+    for t in range(nq):
+        exp = merge(t, (0.5, 0.5), True)
+        assert [int(it["fields"]["_id"]) for it in res[t]["items"]] == \
+            [e[1] for e in exp]
+    eng2.close()
+
+
+def test_multi_vector_ivfpq(data):
+    """Multi-vector over trained IVFPQ indexes: both fields carry the
+    same vectors, so the merged ids must echo the single-field search's
+    ids (intersection of identical top-n sets) with score = the
+    weighted sum of the two identical distances."""
+    base, q = data
+    n, nq, topn = 6000, 4, 10
+    eng = make_engine("/tmp/gamma_multivec_pq")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 32, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 6000}',
+        extra_vecs=[("emb2", 64)])
+    for vid in range(n):
+        eng.add_doc(str(vid), base[vid], extra_vecs=[("emb2", base[vid])])
+    eng.build_index()
+    single = eng.search_pb(q[:nq], topn=topn,
+                           index_params='{"recall_num": 50}')
+    multi = eng.search_pb(q[:nq], topn=topn,
+                          index_params='{"recall_num": 50}',
+                          extra_vec_queries=[("emb2", q[:nq])],
+                          multi_vector_rank=1)
+    for t in range(nq):
+        sids = [int(it["fields"]["_id"]) for it in single[t]["items"]]
+        mids = [int(it["fields"]["_id"]) for it in multi[t]["items"]]
+        assert mids == sids
+        for si, mi in zip(single[t]["items"], multi[t]["items"]):
+            assert abs(mi["score"] - si["score"]) < 1e-5 * max(
+                1.0, abs(si["score"]))
+    eng.close()
+
+
 def test_lockfree_add_under_search(data):
     """§8f-3 lock-free realtime add: a writer thread appends fresh docs
     (the common pure-append path runs under the SHARED lock — the

@@ -89,10 +89,13 @@ static struct CStatus CreateTable_unguarded(void *engine, const char *table_str,
     return err_status(1, "bad table flatbuffer");
   if (ts.vectors.empty())
     return err_status(1, "table has no vector field");
-  if (ts.vectors.size() > 1)
-    return err_status(1, "multi-vector tables not supported this round");
   std::vector<FieldMeta> fields;
   for (auto &f : ts.fields) fields.push_back({f.name, f.data_type});
+  /* extra vector fields of a multi-vector table (vector_manager.cc
+   * keeps one index per field) */
+  std::vector<std::pair<std::string, int>> extra_vecs;
+  for (size_t i = 1; i < ts.vectors.size(); i++)
+    extra_vecs.push_back({ts.vectors[i].name, ts.vectors[i].dimension});
   std::string index_type = ts.index_type;
   std::string index_params = ts.index_params;
   if (index_type.empty() && !ts.indexes.empty()) {
@@ -109,7 +112,7 @@ static struct CStatus CreateTable_unguarded(void *engine, const char *table_str,
   std::string err;
   if (static_cast<Engine *>(engine)->create_table(
           ts.name, fields, ts.vectors[0].name, ts.vectors[0].dimension,
-          index_type, index_params, tt, &err))
+          index_type, index_params, tt, &err, extra_vecs))
     return err_status(1, err);
   return ok_status();
 }
@@ -136,18 +139,25 @@ static int AddOrUpdateDoc_unguarded(void *engine, const char *doc_str, int len) 
   const float *vec = nullptr;
   int vec_len = 0;
   std::vector<std::pair<std::string, std::string>> fields;
+  std::vector<vgamma::MultiVecQuery> extra;
   for (auto &f : doc.fields) {
     if (f.name == "_id") {
       p_key = f.value;
     } else if (f.data_type == gfb::VECTOR) {
-      vec = (const float *)f.value.data();
-      vec_len = (int)(f.value.size() / 4);
+      if (f.name == e->vec_field_name() ||
+          (vec == nullptr && e->vec_dim_of(f.name) < 0)) {
+        vec = (const float *)f.value.data();
+        vec_len = (int)(f.value.size() / 4);
+      } else {
+        extra.push_back({f.name, (const float *)f.value.data()});
+      }
     } else {
       fields.emplace_back(f.name, f.value);
     }
   }
   if (p_key.empty() || !vec) return -1;
-  return e->add_doc(p_key, fields, vec, vec_len);
+  return e->add_doc(p_key, fields, vec, vec_len,
+                    extra.empty() ? nullptr : &extra);
 }
 
 int AddOrUpdateDoc(void *engine, const char *doc_str, int len) {
@@ -331,6 +341,121 @@ int Load(void *engine) {
 }
 
 
+/* Multi-vector search (vector_manager.cc:851-1090): one query batch
+ * over several vector fields, docid-intersection merge with
+ * WeightedRanker scores. Request surface: vec_fields[] (one per
+ * field), `ranker` = {"type":"WeightedRanker","params":[w...]}
+ * (request.cc:86-92, default weights 1/vec_num),
+ * `multi_vector_rank` != 0 -> results ordered by combined score
+ * (:1073-1086), else docid-ascending. */
+static struct CStatus SearchMulti_unguarded(Engine *e,
+                                            const gpb::SearchRequest &req,
+                                            char **response_str,
+                                            int *res_len) {
+  const size_t vn = req.vec_fields.size();
+  int nprobe = 0, recall_num = 0, metric = 0;
+  if (!req.index_params.empty()) {
+    gjson::Value v;
+    if (gjson::parse(req.index_params, v)) {
+      v.get_int("nprobe", nprobe);
+      v.get_int("recall_num", recall_num);
+      std::string mt;
+      if (v.get_str("metric_type", mt)) {
+        if (strcasecmp(mt.c_str(), "L2") == 0) metric = 1;
+        else if (strcasecmp(mt.c_str(), "InnerProduct") == 0 ||
+                 strcasecmp(mt.c_str(), "IP") == 0)
+          metric = 2;
+        else
+          return err_status(1, ("unknown metric_type: " + mt).c_str());
+      }
+    }
+  }
+  /* per-field queries; nq must agree across fields */
+  std::vector<vgamma::MultiVecQuery> queries;
+  int nq = -1;
+  for (auto &vq : req.vec_fields) {
+    int fd = e->vec_dim_of(vq.name);
+    if (fd <= 0)
+      return err_status(1, "unknown vector field " + vq.name);
+    int fn = (int)(vq.value.size() / ((size_t)fd * 4));
+    if (fn <= 0) return err_status(1, "Search n shouldn't less than 0!");
+    if (nq < 0) nq = fn;
+    else if (nq != fn)
+      return err_status(1, "vector queries disagree on batch size");
+    queries.push_back({vq.name, (const float *)vq.value.data()});
+  }
+  /* WeightedRanker (common_query_data.h:251-302) */
+  std::vector<double> weights;
+  if (!req.ranker.empty()) {
+    gjson::Value rv;
+    std::string msg = "weighted ranker params err: " + req.ranker;
+    if (!gjson::parse(req.ranker, rv)) return err_status(1, msg);
+    std::string rtype;
+    if (!rv.get_str("type", rtype)) return err_status(1, msg);
+    const gjson::Value *pv = rv.get("params");
+    if (!pv || pv->type != gjson::Value::ARR)
+      return err_status(1, msg);
+    if (pv->arr.size() != vn)
+      return err_status(1, "weighted ranker params: " + req.ranker +
+                               ", length don't equal to " +
+                               std::to_string(vn));
+    for (auto &w : pv->arr) {
+      if (w.type != gjson::Value::NUM) return err_status(1, msg);
+      weights.push_back(w.num);
+    }
+  }
+  std::vector<vgamma::TermFilterSpec> terms;
+  for (auto &t : req.term_filters)
+    terms.push_back({t.field, t.value, t.is_union});
+  std::vector<vgamma::RangeFilterSpec> ranges;
+  for (auto &t : req.range_filters)
+    ranges.push_back({t.field, t.lower, t.upper, t.include_lower,
+                      t.include_upper, t.is_union});
+
+  int k = req.topn; /* vector_manager.cc:955: depth = topn when merging */
+  std::vector<double> scores((size_t)nq * k);
+  std::vector<int64_t> ids((size_t)nq * k);
+  std::string err;
+  auto rlock = e->read_lock();
+  int rc = e->search_multi(nq, queries, k, nprobe, recall_num, metric,
+                           req.brute == 1, req.request_id,
+                           req.partition_id, weights,
+                           req.multi_vector_rank != 0, scores.data(),
+                           ids.data(), &err, &terms, &ranges, req.op,
+                           /*prelocked=*/true);
+  if (rc == -2) return err_status(-2, "request killed");
+  if (rc != 0)
+    return err_status(1, err.empty() ? "search failed" : err);
+
+  int nres = req.req_num > 0 ? std::min(req.req_num, nq) : nq;
+  std::vector<gpb::SearchResult> results(nres);
+  int64_t total = e->num_docs() - e->bitmap().popcount();
+  for (int i = 0; i < nres; i++) {
+    gpb::SearchResult &res = results[i];
+    res.total = (int)total;
+    for (int j = req.offset; j < k; j++) {
+      int64_t id = ids[(size_t)i * k + j];
+      if (id < 0) continue;
+      gpb::ResultItem item;
+      item.score = scores[(size_t)i * k + j];
+      item.fields.push_back({"_id", e->pkey_of(id)});
+      for (auto &fname : req.fields) {
+        if (fname == "_id") continue;
+        if (e->vec_dim_of(fname) > 0) continue;
+        const std::string *v = e->field_value(id, fname);
+        if (v) item.fields.push_back({fname, *v});
+      }
+      res.max_score = std::max(res.max_score, item.score);
+      res.items.push_back(std::move(item));
+    }
+  }
+  std::string out = gpb::encode_search_response(results);
+  *response_str = (char *)malloc(out.size());
+  memcpy(*response_str, out.data(), out.size());
+  *res_len = (int)out.size();
+  return ok_status();
+}
+
 static struct CStatus Search_unguarded(void *engine, const char *request_str, int req_len,
                       char **response_str, int *res_len) {
   if (!engine) return err_status(1, "null engine");
@@ -342,9 +467,9 @@ static struct CStatus Search_unguarded(void *engine, const char *request_str, in
     return err_status(1, "parse search request failed");
   if (req.vec_fields.empty())
     return err_status(1, "no vector query (scalar-only search via Query)");
-  if (req.vec_fields.size() > 1)
-    return err_status(1, "multi-vector ranking not supported this round");
   if (req.topn <= 0) return err_status(1, "limit[topN] is zero");
+  if (req.vec_fields.size() > 1)
+    return SearchMulti_unguarded(e, req, response_str, res_len);
 
   const gpb::VectorQuery &vq = req.vec_fields[0];
   int d = e->dimension();

@@ -370,6 +370,24 @@ struct RangeFilterSpec {
 
 /* The engine: one per Init() (one Vearch partition). Single vector field
  * round 1 (vector_manager multi-field merging is a later row). */
+/* one additional vector field of a multi-vector table (the primary
+ * field stays in the Engine's legacy members; extra fields each carry
+ * their own raw store + index, sharing the table's index params —
+ * vector_manager.cc:898 keeps one IndexModel per field) */
+struct ExtraVecField {
+  std::string name;
+  int dim = 0;
+  RawStore raw;
+  std::unique_ptr<IVFIndex> index;
+};
+
+/* one per-field query of a multi-vector search (vector_manager.cc:851
+ * dispatch); value points at nq*dim floats */
+struct MultiVecQuery {
+  std::string name;
+  const float *vecs = nullptr;
+};
+
 class Engine {
  public:
   int init(const std::string &config_json, std::string *err);
@@ -378,12 +396,17 @@ class Engine {
                    const std::string &vec_name, int dimension,
                    const std::string &index_type,
                    const std::string &index_params_json,
-                   int training_threshold, std::string *err);
+                   int training_threshold, std::string *err,
+                   const std::vector<std::pair<std::string, int>>
+                       &extra_vec_fields = {});
 
-  /* add one doc (p_key + scalar field bytes + vector) */
+  /* add one doc (p_key + scalar field bytes + vector). extra_vecs:
+   * one entry per extra vector field (name -> nq=1 row), required for
+   * every extra field of a multi-vector table. */
   int add_doc(const std::string &p_key,
               const std::vector<std::pair<std::string, std::string>> &fields,
-              const float *vec, int vec_len);
+              const float *vec, int vec_len,
+              const std::vector<MultiVecQuery> *extra_vecs = nullptr);
   int bulk_add(int64_t n, const float *vecs);
   int delete_doc(const std::string &p_key);
   int build_index(std::string *err);
@@ -413,6 +436,32 @@ class Engine {
   std::shared_lock<std::shared_mutex> read_lock() const {
     return std::shared_lock<std::shared_mutex>(rw_);
   }
+  /* Multi-vector search (vector_manager.cc:851-1090): one query batch
+   * across several vector fields. Per field: the field's own index
+   * searched to depth topn; merge: docid intersection across ALL
+   * fields (seek loop :1025-1070), score = sum_j weight_j * dist_j
+   * (WeightedRanker, default 1/vec_num). Output per query: up to topn
+   * (docid, score) pairs — docid-ascending, or score-ordered when
+   * multi_vector_rank is set (:1073-1086). Unfilled slots id = -1. */
+  int search_multi(int nq, const std::vector<MultiVecQuery> &queries,
+                   int topn, int nprobe, int recall_num, int metric,
+                   bool brute_force, const std::string &request_id,
+                   int partition_id, const std::vector<double> &weights,
+                   bool rank_by_score, double *out_scores,
+                   int64_t *out_ids, std::string *err,
+                   const std::vector<TermFilterSpec> *term_filters = nullptr,
+                   const std::vector<RangeFilterSpec> *range_filters = nullptr,
+                   int filter_op = 0, bool prelocked = false);
+
+  /* dimension of a named vector field (primary or extra); -1 unknown */
+  int vec_dim_of(const std::string &name) const {
+    if (name == vec_name_) return dim_;
+    for (auto &e : extra_vecs_)
+      if (e->name == name) return e->dim;
+    return -1;
+  }
+  size_t num_vec_fields() const { return 1 + extra_vecs_.size(); }
+
   /* upload queries once; later search(nq, nullptr, ...) reuses them */
   int cache_queries(int nq, const float *xq);
   int cached_nq() const { return cached_nq_; }
@@ -464,10 +513,17 @@ class Engine {
   std::string status_json() const;
 
  private:
-  int flat_search_keys(const float *q_dev, int nq, int k2,
-                       const float *q_norms_dev, bool ip, hipStream_t s,
-                       uint64_t *out_keys_dev, const uint32_t *bm,
-                       SearchScratch &sc);
+  int flat_search_keys(RawStore &raw, int dim, const float *q_dev,
+                       int nq, int k2, const float *q_norms_dev, bool ip,
+                       hipStream_t s, uint64_t *out_keys_dev,
+                       const uint32_t *bm, SearchScratch &sc);
+  /* one field of a multi-vector search: that field's index (or FLAT
+   * fallback) to depth topn, canonical scores, host output */
+  int search_field_(RawStore &raw, IVFIndex *idx, int dim,
+                    const float *xq, int nq, int topn, int nprobe,
+                    int recall_num, bool ip, bool brute_force,
+                    SearchScratch &sc, const uint32_t *bm,
+                    float *host_dists, int64_t *host_ids);
   /* 1 = excluded (deleted or fails a filter); dev_out == nullptr skips
    * the device upload (filter_docids browse path) */
   int build_filter_bitmap_(const std::vector<TermFilterSpec> &terms,
@@ -504,6 +560,12 @@ class Engine {
   RawStore raw_;
   Bitmap bitmap_;
   std::unique_ptr<IVFIndex> index_;
+  std::vector<std::unique_ptr<ExtraVecField>> extra_vecs_;
+  ExtraVecField *extra_vec_(const std::string &name) {
+    for (auto &e : extra_vecs_)
+      if (e->name == name) return e.get();
+    return nullptr;
+  }
   IndexParams params_;
   /* published with release AFTER all row state (columns, pkey, raw
    * vectors, bucket entry) is visible — the lock-free append contract */

@@ -41,7 +41,9 @@ int Engine::create_table(const std::string &name,
                          const std::string &vec_name, int dimension,
                          const std::string &index_type,
                          const std::string &index_params_json,
-                         int training_threshold, std::string *err) {
+                         int training_threshold, std::string *err,
+                         const std::vector<std::pair<std::string, int>>
+                             &extra_vec_fields) {
   if (table_created_) {
     if (err) *err = "table already created";
     return -1;
@@ -76,6 +78,30 @@ int Engine::create_table(const std::string &name,
       return -1;
     }
   }
+  /* extra vector fields (multi-vector table, vector_manager.cc:898):
+   * each gets its own raw store + index of the table's index type;
+   * nsubvector falls back to the per-field default when the table's
+   * value does not divide the field's dimension */
+  for (auto &ev : extra_vec_fields) {
+    if (ev.second <= 0 || ev.second % 4 != 0) {
+      if (err) *err = "extra vector field " + ev.first + ": bad dim";
+      return -1;
+    }
+    auto e = std::make_unique<ExtraVecField>();
+    e->name = ev.first;
+    e->dim = ev.second;
+    if (e->raw.init(e->dim)) return -1;
+    if (params_.kind != IndexKind::FLAT) {
+      IndexParams pe = params_;
+      if (pe.nsubvector && e->dim % pe.nsubvector) pe.nsubvector = 0;
+      e->index = std::make_unique<IVFIndex>();
+      if (e->index->init(e->dim, pe)) {
+        if (err) *err = "extra vector index init failed for " + ev.first;
+        return -1;
+      }
+    }
+    extra_vecs_.push_back(std::move(e));
+  }
   /* training_threshold default: max(nlist*39, 256*39) vectors
    * (ivfpq.cc:137-144 with default_points_per_centroid=39) */
   if (training_threshold > 0) training_threshold_ = training_threshold;
@@ -90,8 +116,16 @@ int Engine::create_table(const std::string &name,
 int Engine::add_doc(
     const std::string &p_key,
     const std::vector<std::pair<std::string, std::string>> &fields,
-    const float *vec, int vec_len) {
+    const float *vec, int vec_len,
+    const std::vector<MultiVecQuery> *extra_vecs) {
   if (!table_created_ || vec_len != dim_) return -1;
+  /* every extra vector field of the table must be supplied */
+  if (!extra_vecs_.empty()) {
+    size_t have = extra_vecs ? extra_vecs->size() : 0;
+    if (have != extra_vecs_.size()) return -1;
+    for (auto &q : *extra_vecs)
+      if (!extra_vec_(q.name) || !q.vecs) return -1;
+  }
   /* Lock-free common append (realtime_mem_data.cc:57-68
    * retrieve_idx_pos_ pattern): a pure append that fits every existing
    * capacity runs under the SHARED lock — concurrent searches are
@@ -99,7 +133,7 @@ int Engine::add_doc(
    * bucket size is published. Anything structural (update of an
    * existing pkey, raw-segment or bucket growth, bitmap growth)
    * retries under the write lock. */
-  {
+  if (extra_vecs_.empty()) { /* multi-vector rows take the slow path */
     std::shared_lock<std::shared_mutex> g(rw_);
     std::lock_guard<std::mutex> ap(append_mu_);
     int rc = add_doc_fast_(p_key, fields, vec);
@@ -116,6 +150,8 @@ int Engine::add_doc(
       int64_t old = it->second;
       bitmap_.set(old, stream_);
       if (index_) index_->del(old, stream_);
+      for (auto &e : extra_vecs_)
+        if (e->index) e->index->del(old, stream_);
     }
   }
   int64_t docid = max_docid_.load(std::memory_order_relaxed);
@@ -131,6 +167,15 @@ int Engine::add_doc(
   if (index_ && index_->trained()) {
     if (index_->add(vec, &docid, 1, stream_)) return -1;
     indexed_count_++;
+  }
+  for (auto &e : extra_vecs_) { /* parallel rows, same docid */
+    const float *v = nullptr;
+    for (auto &q : *extra_vecs)
+      if (q.name == e->name) v = q.vecs;
+    if (e->raw.add(v, 1, stream_)) return -1;
+    if (e->index && e->index->trained() &&
+        e->index->add(v, &docid, 1, stream_))
+      return -1;
   }
   {
     std::lock_guard<std::mutex> pk(pkey_mu_);
@@ -218,6 +263,8 @@ int Engine::delete_doc(const std::string &p_key) {
   if (it == pkey2docid_.end()) return -1;
   bitmap_.set(it->second, stream_);
   if (index_) index_->del(it->second, stream_);
+  for (auto &e : extra_vecs_)
+    if (e->index) e->index->del(it->second, stream_);
   pkey2docid_.erase(it);
   return 0;
 }
@@ -260,6 +307,22 @@ int Engine::build_index(std::string *err) {
     if (index_->add(buf.data(), vids.data(), cn, stream_)) return -1;
   }
   indexed_count_ = raw_.size();
+  /* extra vector fields: train + index each on its own raw store */
+  for (auto &e : extra_vecs_) {
+    if (!e->index || e->index->trained()) continue;
+    int64_t enum_ = std::min<int64_t>(num, e->raw.size());
+    std::vector<float> ext((size_t)enum_ * e->dim);
+    e->raw.host_copy(0, enum_, ext.data());
+    if (e->index->train(ext.data(), enum_, stream_, err)) return -1;
+    for (int64_t c0 = 0; c0 < e->raw.size(); c0 += chunk) {
+      int64_t cn = std::min(chunk, e->raw.size() - c0);
+      buf.resize((size_t)cn * e->dim);
+      e->raw.host_copy(c0, cn, buf.data());
+      vids.resize(cn);
+      for (int64_t i = 0; i < cn; i++) vids[i] = c0 + i;
+      if (e->index->add(buf.data(), vids.data(), cn, stream_)) return -1;
+    }
+  }
   return 0;
 }
 
@@ -283,14 +346,15 @@ int Engine::rebuild_index(bool drop_before_rebuild, std::string *err) {
   return build_index(err);
 }
 
-int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
-                             const float *q_norms_dev, bool ip,
-                             hipStream_t s, uint64_t *out_keys_dev,
-                             const uint32_t *bm, SearchScratch &sc) {
-  const int64_t n = raw_.size();
+int Engine::flat_search_keys(RawStore &raw, int dim, const float *q_dev,
+                             int nq, int k2, const float *q_norms_dev,
+                             bool ip, hipStream_t s,
+                             uint64_t *out_keys_dev, const uint32_t *bm,
+                             SearchScratch &sc) {
+  const int64_t n = raw.size();
   if (nq < 512 || n < 200000) {
     GAMMA_CHECK(gk::flat_stream_scan(
-        s, nq, n, dim_, k2, q_dev, raw_.dev_seg_table(),
+        s, nq, n, dim, k2, q_dev, raw.dev_seg_table(),
         RawStore::SEG_SHIFT, bm, ip, out_keys_dev));
     return 0;
   }
@@ -300,13 +364,13 @@ int Engine::flat_search_keys(const float *q_dev, int nq, int k2,
   bool seeded = false;
   for (int64_t v0 = 0; v0 < n;) {
     int64_t run = 0;
-    const float *seg = raw_.dev_run(v0, &run);
+    const float *seg = raw.dev_run(v0, &run);
     int64_t take = std::min(run, chunk);
-    GAMMA_CHECK(gk::dots_mfma(s, q_dev, nq, seg, take, dim_,
+    GAMMA_CHECK(gk::dots_mfma(s, q_dev, nq, seg, take, dim,
                               sc.flat_dots.as<float>()));
     GAMMA_CHECK(gk::select_from_dots(
         s, nq, take, v0, take, sc.flat_dots.as<float>(), q_norms_dev,
-        raw_.dev_norms(), !ip, ip, bm, k2, out_keys_dev, seeded));
+        raw.dev_norms(), !ip, ip, bm, k2, out_keys_dev, seeded));
     seeded = true;
     v0 += take;
   }
@@ -668,7 +732,8 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
     int kf = std::min<int64_t>((int64_t)k2 + 64, 1088);
     kf = (int)std::min<int64_t>(kf, std::max<int64_t>(raw_.size(), 1));
     if (sc.keys.reserve((size_t)nq * kf * 8)) return -1;
-    if (flat_search_keys(qptr, nq, kf, sc.q_norms.as<float>(), ip, s,
+    if (flat_search_keys(raw_, dim_, qptr, nq, kf,
+                         sc.q_norms.as<float>(), ip, s,
                          sc.keys.as<uint64_t>(), bm, sc))
       return -1;
     k2 = kf;
@@ -730,6 +795,181 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
     last_timing[3] = tm.ms(2, 3) * 1000.0;
     last_timing[4] = tm.ms(3, 4) * 1000.0;
     last_timing[5] = tm.ms(0, 4) * 1000.0;
+  }
+  return 0;
+}
+
+int Engine::search_field_(RawStore &raw, IVFIndex *idx, int dim,
+                          const float *xq, int nq, int topn, int nprobe,
+                          int recall_num, bool ip, bool brute_force,
+                          SearchScratch &sc, const uint32_t *bm,
+                          float *host_dists, int64_t *host_ids) {
+  hipStream_t s = sc.stream;
+  int k2 = std::max(topn, recall_num);
+  bool rr = recall_num > 0;
+  if (k2 > 1024) return -1;
+  if (sc.q_dev.reserve((size_t)nq * dim * 4)) return -1;
+  GAMMA_CHECK(hipMemcpyAsync(sc.q_dev.get(), xq, (size_t)nq * dim * 4,
+                             hipMemcpyHostToDevice, s));
+  const float *qptr = sc.q_dev.as<float>();
+  if (sc.q_norms.reserve((size_t)nq * 4)) return -1;
+  GAMMA_CHECK(gk::row_norms(s, qptr, nq, dim, sc.q_norms.as<float>()));
+  if (sc.out_d.reserve((size_t)nq * topn * 4)) return -1;
+  if (sc.out_i.reserve((size_t)nq * topn * 8)) return -1;
+  bool use_flat = brute_force || !idx || !idx->trained();
+  bool canon = true;
+  if (use_flat) {
+    int kf = std::min<int64_t>((int64_t)k2 + 64, 1088);
+    kf = (int)std::min<int64_t>(kf, std::max<int64_t>(raw.size(), 1));
+    if (sc.keys.reserve((size_t)nq * kf * 8)) return -1;
+    if (flat_search_keys(raw, dim, qptr, nq, kf, sc.q_norms.as<float>(),
+                         ip, s, sc.keys.as<uint64_t>(), bm, sc))
+      return -1;
+    k2 = kf;
+  } else {
+    int S = idx->probe_split(nq, k2, nprobe > 0 ? nprobe : 1024);
+    if (sc.keys.reserve((size_t)nq * S * k2 * 8)) return -1;
+    double ta = 0, ts = 0;
+    if (idx->search(qptr, nq, k2, nprobe, bm, ip, s,
+                    sc.keys.as<uint64_t>(), sc.q_norms.as<float>(), &ta,
+                    &ts, sc, S, nullptr))
+      return -1;
+    k2 *= S;
+    canon = rr || params_.kind == IndexKind::IVFFLAT;
+  }
+  if (canon)
+    GAMMA_CHECK(gk::rerank(s, nq, k2, dim, qptr, raw.dev_seg_table(),
+                           raw.num_segs(), RawStore::SEG_SHIFT, ip,
+                           sc.keys.as<uint64_t>(),
+                           sc.keys.as<uint64_t>()));
+  GAMMA_CHECK(gk::sort_rows(s, nq, k2, topn, sc.keys.as<uint64_t>(), ip,
+                            sc.out_d.as<float>(), sc.out_i.as<int64_t>()));
+  GAMMA_CHECK(hipMemcpyAsync(host_dists, sc.out_d.get(),
+                             (size_t)nq * topn * 4, hipMemcpyDeviceToHost,
+                             s));
+  GAMMA_CHECK(hipMemcpyAsync(host_ids, sc.out_i.get(),
+                             (size_t)nq * topn * 8, hipMemcpyDeviceToHost,
+                             s));
+  GAMMA_CHECK(hipStreamSynchronize(s));
+  return 0;
+}
+
+int Engine::search_multi(int nq, const std::vector<MultiVecQuery> &queries,
+                         int topn, int nprobe, int recall_num, int metric,
+                         bool brute_force, const std::string &request_id,
+                         int partition_id,
+                         const std::vector<double> &weights,
+                         bool rank_by_score, double *out_scores,
+                         int64_t *out_ids, std::string *err,
+                         const std::vector<TermFilterSpec> *term_filters,
+                         const std::vector<RangeFilterSpec> *range_filters,
+                         int filter_op, bool prelocked) {
+  if (!table_created_ || nq <= 0 || topn <= 0) return -1;
+  const size_t vn = queries.size();
+  if (vn < 2) {
+    if (err) *err = "search_multi needs >= 2 vector queries";
+    return -1;
+  }
+  bool ip = metric == 0 ? params_.metric_ip : (metric == 2);
+  struct FieldRef {
+    RawStore *raw;
+    IVFIndex *idx;
+    int dim;
+  };
+  std::vector<FieldRef> fs;
+  for (auto &q : queries) {
+    if (q.name == vec_name_) {
+      fs.push_back({&raw_, index_.get(), dim_});
+    } else if (ExtraVecField *e = extra_vec_(q.name)) {
+      fs.push_back({&e->raw, e->index.get(), e->dim});
+    } else {
+      if (err) *err = "unknown vector field " + q.name;
+      return -1;
+    }
+  }
+  std::shared_lock<std::shared_mutex> g(rw_, std::defer_lock);
+  if (!prelocked) g.lock();
+  if (KillRegistry::inst().killed(request_id, partition_id)) return -2;
+  SearchScratch *scp = acquire_scratch_();
+  if (!scp) return -1;
+  ScratchGuard sg{this, scp};
+  /* docid-level filters apply to every field's scan identically */
+  const uint32_t *bm = bitmap_.any() ? bitmap_.dev() : nullptr;
+  bool have_filters = (term_filters && !term_filters->empty()) ||
+                      (range_filters && !range_filters->empty());
+  if (have_filters) {
+    static const std::vector<TermFilterSpec> kNoT;
+    static const std::vector<RangeFilterSpec> kNoR;
+    if (build_filter_bitmap_(term_filters ? *term_filters : kNoT,
+                             range_filters ? *range_filters : kNoR, *scp,
+                             &bm, err, filter_op))
+      return -1;
+  }
+  /* per field: the field's own index to depth topn
+   * (vector_manager.cc:955: topN = condition->topn when merging) */
+  std::vector<std::vector<float>> fd(vn);
+  std::vector<std::vector<int64_t>> fi(vn);
+  for (size_t j = 0; j < vn; j++) {
+    fd[j].resize((size_t)nq * topn);
+    fi[j].resize((size_t)nq * topn);
+    if (search_field_(*fs[j].raw, fs[j].idx, fs[j].dim, queries[j].vecs,
+                      nq, topn, nprobe, recall_num, ip, brute_force,
+                      *scp, bm, fd[j].data(), fi[j].data()))
+      return -1;
+    if (KillRegistry::inst().killed(request_id, partition_id)) return -2;
+  }
+  /* merge (vector_manager.cc:1025-1086): a doc survives iff it is in
+   * EVERY field's top-n; score = sum_j weight_j * dist_j
+   * (WeightedRanker, default 1/vec_num); docid-ascending order, or
+   * combined-score order when multi_vector_rank is set */
+  for (int i = 0; i < nq; i++) {
+    std::vector<std::vector<std::pair<int64_t, float>>> lists(vn);
+    for (size_t j = 0; j < vn; j++) {
+      for (int t = 0; t < topn; t++) {
+        int64_t id = fi[j][(size_t)i * topn + t];
+        if (id >= 0)
+          lists[j].push_back({id, fd[j][(size_t)i * topn + t]});
+      }
+      std::sort(lists[j].begin(), lists[j].end());
+    }
+    std::vector<std::pair<double, int64_t>> merged; /* (score, docid) */
+    for (auto &pr : lists[0]) {
+      double score = 0;
+      bool common = true;
+      for (size_t j = 0; j < vn && common; j++) {
+        auto it = std::lower_bound(
+            lists[j].begin(), lists[j].end(), pr.first,
+            [](const std::pair<int64_t, float> &a, int64_t b) {
+              return a.first < b;
+            });
+        if (it == lists[j].end() || it->first != pr.first) {
+          common = false;
+        } else {
+          double w = weights.size() == vn ? weights[j] : 1.0 / vn;
+          score += it->second * w;
+        }
+      }
+      if (common) merged.push_back({score, pr.first});
+    }
+    if (rank_by_score) {
+      if (ip) /* InnerProductCmp: best (largest) first */
+        std::stable_sort(merged.begin(), merged.end(),
+                         [](const std::pair<double, int64_t> &a,
+                            const std::pair<double, int64_t> &b) {
+                           return a.first > b.first;
+                         });
+      else /* L2Cmp: smallest first */
+        std::stable_sort(merged.begin(), merged.end());
+    } /* else: docid-ascending (lists[0] order is already sorted) */
+    for (int t = 0; t < topn; t++) {
+      if (t < (int)merged.size()) {
+        out_scores[(size_t)i * topn + t] = merged[t].first;
+        out_ids[(size_t)i * topn + t] = merged[t].second;
+      } else {
+        out_scores[(size_t)i * topn + t] = 0;
+        out_ids[(size_t)i * topn + t] = -1;
+      }
+    }
   }
   return 0;
 }
@@ -810,7 +1050,7 @@ int Engine::dump_to_(const std::string &dir, std::string *err) {
     return -1;
   }
   fwrite(&kDumpMagic, 4, 1, f);
-  int ver = 1;
+  int ver = 2; /* v2 appends the extra vector fields after the index */
   fwrite(&ver, 4, 1, f);
   auto wstr = [&](const std::string &s_) {
     int64_t n = (int64_t)s_.size();
@@ -841,6 +1081,17 @@ int Engine::dump_to_(const std::string &dir, std::string *err) {
   int has_index = index_ ? 1 : 0;
   fwrite(&has_index, 4, 1, f);
   if (index_) index_->dump(f, stream_);
+  /* v2: extra vector fields */
+  int n_extra = (int)extra_vecs_.size();
+  fwrite(&n_extra, 4, 1, f);
+  for (auto &e : extra_vecs_) {
+    wstr(e->name);
+    fwrite(&e->dim, 4, 1, f);
+    e->raw.dump(f);
+    int hi = (e->index && e->index->trained()) ? 1 : 0;
+    fwrite(&hi, 4, 1, f);
+    if (hi) e->index->dump(f, stream_);
+  }
   fclose(f);
   rename((fn + ".tmp").c_str(), fn.c_str());
   return 0;
@@ -941,6 +1192,29 @@ int Engine::load(std::string *err) {
   if (has_index && index_) {
     if (index_->load(f, stream_)) { fclose(f); return -1; }
     indexed_count_ = index_->ntotal();
+  }
+  if (ver >= 2) {
+    int n_extra = 0;
+    if (fread(&n_extra, 4, 1, f) != 1 || n_extra < 0 || n_extra > 64)
+      return bad("extra vec count");
+    if (n_extra != (int)extra_vecs_.size())
+      return bad("extra vector field count mismatch with table");
+    for (int i = 0; i < n_extra; i++) {
+      std::string ename;
+      int edim = 0;
+      if (rstr(ename)) return bad("extra vec name");
+      if (fread(&edim, 4, 1, f) != 1) return bad("extra vec dim");
+      ExtraVecField *e = extra_vec_(ename);
+      if (!e || e->dim != edim)
+        return bad("extra vector field mismatch with table");
+      if (e->raw.load(f, stream_)) return bad("extra raw vectors");
+      int hi = 0;
+      if (fread(&hi, 4, 1, f) != 1) return bad("extra index flag");
+      if (hi && e->index && e->index->load(f, stream_))
+        return bad("extra index");
+    }
+  } else if (!extra_vecs_.empty()) {
+    return bad("v1 dump has no extra vector fields");
   }
   fclose(f);
   return 0;

@@ -118,12 +118,16 @@ class GammaEngine:
             self.h = None
 
     def create_table(self, d, index_type="IVFPQ", index_params="",
-                     name="bench_space", scalar_fields=(), vec_name="emb"):
+                     name="bench_space", scalar_fields=(), vec_name="emb",
+                     extra_vecs=()):
+        """extra_vecs: additional (name, dim) vector fields of a
+        multi-vector table (vector_manager.cc multi-field dispatch)."""
         from . import fbsenc
         self.d = d
         self.vec_name = vec_name
         buf = fbsenc.build_table(name, list(scalar_fields), vec_name, d,
-                                 index_type, index_params)
+                                 index_type, index_params,
+                                 extra_vecs=extra_vecs)
         _check(lib().CreateTable(self.h, buf, len(buf)), "CreateTable")
 
     def add(self, vecs):
@@ -134,13 +138,18 @@ class GammaEngine:
         if rc != 0:
             raise RuntimeError(f"GammaBulkAdd failed rc={rc}")
 
-    def add_doc(self, p_key, vec, fields=()):
+    def add_doc(self, p_key, vec, fields=(), extra_vecs=()):
+        """extra_vecs: [(name, array)] — one entry per extra vector
+        field of a multi-vector table."""
         from . import fbsenc
         vec = np.ascontiguousarray(vec, dtype=np.float32)
         fl = [("_id", p_key.encode(), fbsenc.DATA_STRING)]
         for name, value, dt in fields:
             fl.append((name, value, dt))
         fl.append((self.vec_name, vec.tobytes(), fbsenc.DATA_VECTOR))
+        for name, v in extra_vecs:
+            v = np.ascontiguousarray(v, dtype=np.float32)
+            fl.append((name, v.tobytes(), fbsenc.DATA_VECTOR))
         buf = fbsenc.build_doc(fl)
         rc = lib().AddOrUpdateDoc(self.h, buf, len(buf))
         if rc != 0:
@@ -232,18 +241,24 @@ class GammaEngine:
     def search_pb(self, queries, topn, index_params="", fields=("_id",),
                   request_id="req1", partition_id=1, brute=0,
                   min_score=None, max_score=None, l2_sqrt=False,
-                  term_filters=(), range_filters=(), operator=0):
+                  term_filters=(), range_filters=(), operator=0,
+                  extra_vec_queries=(), multi_vector_rank=0, ranker=""):
         """The real C-ABI Search with protobuf marshalling (reader.go
-        path)."""
+        path). extra_vec_queries: [(field_name, query_array)] for
+        multi-vector search; multi_vector_rank orders by combined
+        score; ranker = WeightedRanker JSON."""
         from . import proto
         q = np.ascontiguousarray(queries, dtype=np.float32)
+        extra = [(n, np.ascontiguousarray(v, np.float32).tobytes())
+                 for n, v in extra_vec_queries]
         req = proto.encode_search_request(
             self.vec_name, q.tobytes(), topn, q.shape[0],
             request_id=request_id, partition_id=partition_id,
             index_params=index_params, brute=brute, fields=fields,
             min_score=min_score, max_score=max_score, l2_sqrt=l2_sqrt,
             term_filters=term_filters, range_filters=range_filters,
-            operator=operator)
+            operator=operator, extra_vec_fields=extra,
+            multi_vector_rank=multi_vector_rank, ranker=ranker)
         out = ctypes.c_char_p()
         n = ctypes.c_int()
         st = lib().Search(self.h, req, len(req), ctypes.byref(out),

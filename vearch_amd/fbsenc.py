@@ -104,9 +104,11 @@ def build_doc(fields):
 
 
 def build_table(name, scalar_fields, vec_name, dimension, index_type,
-                index_params, store_type="MemoryOnly"):
+                index_params, store_type="MemoryOnly", extra_vecs=()):
     """table.fbs Table: name(0), fields(1), vectors_info(2), index_type(3),
-    index_params(4). scalar_fields: list of (name, data_type)."""
+    index_params(4). scalar_fields: list of (name, data_type).
+    extra_vecs: additional (name, dimension) vector fields of a
+    multi-vector table."""
     B = _Buf()
     B.w32(0)
     B.pad4()
@@ -140,26 +142,31 @@ def build_table(name, scalar_fields, vec_name, dimension, index_type,
         B.pad4()
         B.w32at(fv_elems + 4 * i, ft - (fv_elems + 4 * i))
 
-    # vectors_info vector with one VectorInfo
+    # vectors_info vector of VectorInfo (primary first, then extras)
+    all_vecs = [(vec_name, dimension)] + list(extra_vecs)
     B.pad4()
     vv = len(B.b)
-    B.w32(1)
-    vv_elem = len(B.b)
-    B.w32(0)
+    B.w32(len(all_vecs))
+    vv_elems = len(B.b)
+    for _ in all_vecs:
+        B.w32(0)
     B.w32at(slots["vectors"], vv - slots["vectors"])
 
-    B.pad4()
-    vt = len(B.b)
-    B.w32(0)                    # soffset
-    vname_slot = len(B.b)
-    B.w32(0)                    # name
-    B.b += struct.pack("<i", dimension)  # dimension inline @8
-    B.b.append(DATA_VECTOR)     # data_type @12
-    B.b.append(1)               # is_index @13
-    B.pad4()
-    vstore_slot = len(B.b)
-    B.w32(0)                    # store_type @16
-    B.w32at(vv_elem, vt - vv_elem)
+    v_tabs, v_name_slots, v_store_slots = [], [], []
+    for i, (vn, vd) in enumerate(all_vecs):
+        B.pad4()
+        vt = len(B.b)
+        v_tabs.append(vt)
+        B.w32(0)                    # soffset
+        v_name_slots.append(len(B.b))
+        B.w32(0)                    # name
+        B.b += struct.pack("<i", vd)  # dimension inline @8
+        B.b.append(DATA_VECTOR)     # data_type @12
+        B.b.append(1)               # is_index @13
+        B.pad4()
+        v_store_slots.append(len(B.b))
+        B.w32(0)                    # store_type @16
+        B.w32at(vv_elems + 4 * i, vt - (vv_elems + 4 * i))
 
     # strings
     s = B.string(name)
@@ -167,10 +174,11 @@ def build_table(name, scalar_fields, vec_name, dimension, index_type,
     for i, (fname, dt) in enumerate(scalar_fields):
         s = B.string(fname)
         B.w32at(fi_name_slots[i], s - fi_name_slots[i])
-    s = B.string(vec_name)
-    B.w32at(vname_slot, s - vname_slot)
-    s = B.string(store_type)
-    B.w32at(vstore_slot, s - vstore_slot)
+    for i, (vn, vd) in enumerate(all_vecs):
+        s = B.string(vn)
+        B.w32at(v_name_slots[i], s - v_name_slots[i])
+        s = B.string(store_type)
+        B.w32at(v_store_slots[i], s - v_store_slots[i])
     s = B.string(index_type)
     B.w32at(slots["index_type"], s - slots["index_type"])
     s = B.string(index_params)
@@ -188,5 +196,6 @@ def build_table(name, scalar_fields, vec_name, dimension, index_type,
     # VectorInfo: name=0@4, data_type=1@12, is_index=2@13, dimension=3@8,
     #             store_type=4@16
     vi_vt = B.vtable([14, 20, 4, 12, 13, 8, 16])
-    B.ws32at(vt, vt - vi_vt)
+    for vt in v_tabs:
+        B.ws32at(vt, vt - vi_vt)
     return bytes(B.b)

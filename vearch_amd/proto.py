@@ -44,7 +44,9 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
                           index_params="", min_score=None, max_score=None,
                           brute=0, fields=("_id",), l2_sqrt=False,
                           is_vector_value=False, term_filters=(),
-                          range_filters=(), operator=0):
+                          range_filters=(), operator=0,
+                          extra_vec_fields=(), multi_vector_rank=0,
+                          ranker=""):
     """vearchpb.SearchRequest (router_grpc.proto:168-192).
     term_filters: (field, value_bytes) or (field, value_bytes, is_union).
     range_filters: (field, lower_bytes, upper_bytes, incl_lower,
@@ -64,6 +66,8 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
     if max_score is not None:
         vq += _vdouble(4, max_score)
     out += _ld(5, vq)
+    for en, ev in extra_vec_fields:
+        out += _ld(5, _ld(1, en.encode()) + _ld(2, ev))
     for f in fields:
         out += _ld(6, f.encode())
     for rf in range_filters:
@@ -81,10 +85,14 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
         out += _ld(8, body)
     if index_params:
         out += _ld(9, index_params.encode())
+    if multi_vector_rank:
+        out += _vint(10, multi_vector_rank)
     if l2_sqrt:
         out += _vint(11, 1)
     if is_vector_value:
         out += _vint(12, 1)
+    if ranker:
+        out += _ld(15, ranker.encode())
     if operator:
         out += _vint(17, operator)
     return out

@@ -450,8 +450,9 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
   uint64_t *sortbuf = (uint64_t *)(smem + ((size_t)M * ksub * 4 + 7) / 8 * 8);
   uint64_t *res = sortbuf + GAMMA_SORT_CAP;
   float *qs = (float *)(res + k2);                  /* d (IP table build) */
-  float *dis0s = qs + d;                            /* 1 */
-  int *state = (int *)(dis0s + 1) + 1;              /* int[2] */
+  float *dis0s = qs + d;                            /* 2 (one per half) */
+  long long *szsh = (long long *)(dis0s + 2);       /* 2 (list sizes) */
+  int *state = (int *)(szsh + 2) + 1;               /* int[2] */
 
   /* probe-split: S sub-workgroups per query, sub-block handles probes
    * p ≡ sub (mod S); partials merged by sort_rows afterwards. S>1 is
@@ -492,47 +493,69 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
    * device flag between lists with an agent-scope load (L2-served, so a
    * host write during the kernel is visible — plain loads can stay
    * L1-stale, microarch §Workgroup dispatch). */
-  for (int p = sub; p < nprobe; p += S) {
-    if (kill_flag &&
-        __hip_atomic_load(kill_flag, __ATOMIC_RELAXED,
-                          __HIP_MEMORY_SCOPE_AGENT))
-      break;
-    int64_t ln = probes[(int64_t)q * nprobe + p];
-    if (ln < 0 || ln >= nlist) continue;
-    GammaBucketDev bk = buckets[ln];
-    if (bk.size <= 0) continue;
-    const float *cent = centroids + (size_t)ln * d;
-
-    float dis0;
-    if (IP) {
-      if (threadIdx.x == 0) {  /* dis0 = dot(q, c), canonical order */
-        float acc = 0.0f;
-        for (int t = 0; t < d; t++) acc = fmaf(qs[t], cent[t], acc);
-        dis0s[0] = acc;
+  if (MW > 0) {
+    /* TWO probed lists in flight per workgroup: halves of the block
+     * scan consecutive probes of this sub-workgroup's sequence. At
+     * large nlist a list holds ~N/nlist codes (~600 at the headline
+     * config) — against a 512-thread block that wastes ~30% of the
+     * last pass; 2x(BS/2) halves the tail waste and doubles the
+     * independent HBM streams. Push order changes; the selector's
+     * top-k2 is exact under the (dist,id) total order, so results are
+     * unchanged. */
+    const int C = GAMMA_ADC_C;
+    const int HB = BS / 2;
+    const int half = threadIdx.x >= HB ? 1 : 0;
+    const int tid = threadIdx.x - half * HB;
+    for (int t = 0;; t += 2) {
+      int p0 = sub + t * S;
+      if (p0 >= nprobe) break;
+      if (kill_flag &&
+          __hip_atomic_load(kill_flag, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT))
+        break;
+      int p = sub + (t + half) * S;
+      int64_t ln = -1;
+      if (p < nprobe) ln = probes[(int64_t)q * nprobe + p];
+      if (ln >= nlist) ln = -1;
+      GammaBucketDev bk;
+      bk.size = 0;
+      if (ln >= 0) bk = buckets[ln];
+      float dis0 = 0.0f;
+      if (IP) {
+        if (tid == 0) { /* dis0 = dot(q, c), canonical order, per half */
+          float acc = 0.0f;
+          if (ln >= 0) {
+            const float *cent = centroids + (size_t)ln * d;
+            for (int e = 0; e < d; e++) acc = fmaf(qs[e], cent[e], acc);
+          }
+          dis0s[half] = acc;
+          szsh[half] = bk.size;
+        }
+      } else {
+        /* dis0 = the coarse probe distance (ivfpq.h:255
+         * dis0=coarse_dis); the per-list table half arrives as the
+         * per-vector S term */
+        if (ln >= 0 && p < nprobe)
+          dis0 = probe_dists[(int64_t)q * nprobe + p];
+        if (tid == 0) szsh[half] = bk.size;
       }
       __syncthreads();
-      dis0 = dis0s[0];
-    } else {
-      /* dis0 = the coarse probe distance (ivfpq.h:255 dis0=coarse_dis);
-       * the per-list table half arrives as the per-vector S term */
-      dis0 = probe_dists[(int64_t)q * nprobe + p];
-    }
-
-    const uint32_t *ids = bk.ids;
-    const uint8_t *codes = (const uint8_t *)bk.data;
-    const float *svals = (const float *)bk.svals;
-    if (MW > 0) {
-      const int C = GAMMA_ADC_C;
-      for (long long j0 = 0; j0 < bk.size;
-           j0 += (long long)blockDim.x * C) {
-        long long jb = j0 + (long long)threadIdx.x * C;
+      if (IP) dis0 = dis0s[half];
+      /* both halves iterate in lockstep to the longer list so the
+       * selector's block-wide flush barriers stay aligned */
+      long long maxsz = szsh[0] > szsh[1] ? szsh[0] : szsh[1];
+      const uint32_t *ids = bk.ids;
+      const uint8_t *codes = (const uint8_t *)bk.data;
+      const float *svals = (const float *)bk.svals;
+      for (long long j0 = 0; j0 < maxsz; j0 += (long long)HB * C) {
+        long long jb = j0 + (long long)tid * C;
         uint32_t w[C][MW ? MW : 1]; /* compile-time bounds -> registers */
         int64_t idv[C];
         float sv[C];
 #pragma unroll
         for (int c = 0; c < C; c++) {
           long long j = jb + c;
-          if (j < bk.size) {
+          if (ln >= 0 && j < bk.size) {
             idv[c] = (int64_t)(int32_t)ids[j]; /* bit31 -> negative */
             if (!IP) sv[c] = svals[j];
             const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
@@ -563,7 +586,36 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
         }
         sel.maybe_flush(blockDim.x * C);
       }
-    } else {
+      __syncthreads(); /* dis0s/szsh rewritten next pair */
+    }
+  } else {
+    for (int p = sub; p < nprobe; p += S) {
+      if (kill_flag &&
+          __hip_atomic_load(kill_flag, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT))
+        break;
+      int64_t ln = probes[(int64_t)q * nprobe + p];
+      if (ln < 0 || ln >= nlist) continue;
+      GammaBucketDev bk = buckets[ln];
+      if (bk.size <= 0) continue;
+      const float *cent = centroids + (size_t)ln * d;
+
+      float dis0;
+      if (IP) {
+        if (threadIdx.x == 0) {  /* dis0 = dot(q, c), canonical order */
+          float acc = 0.0f;
+          for (int e = 0; e < d; e++) acc = fmaf(qs[e], cent[e], acc);
+          dis0s[0] = acc;
+        }
+        __syncthreads();
+        dis0 = dis0s[0];
+      } else {
+        dis0 = probe_dists[(int64_t)q * nprobe + p];
+      }
+
+      const uint32_t *ids = bk.ids;
+      const uint8_t *codes = (const uint8_t *)bk.data;
+      const float *svals = (const float *)bk.svals;
       const int mwords = M >> 2;
       for (long long j0 = 0; j0 < bk.size; j0 += blockDim.x) {
         long long j = j0 + threadIdx.x;
@@ -586,8 +638,8 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
         }
         sel.maybe_flush(blockDim.x);
       }
+      if (IP) __syncthreads(); /* dis0s rewritten next list */
     }
-    if (IP) __syncthreads(); /* dis0s rewritten next list */
   }
   sel.finish();
   for (int i = threadIdx.x; i < k2; i += blockDim.x)
@@ -605,7 +657,8 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
                           bool ip, uint64_t *out_keys,
                           const int *kill_flag) {
   size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
-                (GAMMA_SORT_CAP + k2) * 8 + (d + 1) * 4 + 4 * sizeof(int);
+                (GAMMA_SORT_CAP + k2) * 8 + (d + 2) * 4 +
+                2 * sizeof(long long) + 4 * sizeof(int);
   if (smem > 160 * 1024) return hipErrorInvalidValue;
   /* batched path needs flush margin blockDim*C inside the selector cap.
    * 512-thread blocks put 24 waves on a CU at the same LDS/WG (the ADC

@@ -888,6 +888,17 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
       memcpy(g.data.data() + off, payload_h + (size_t)i * entry, entry);
       if (pq) g.svals.push_back(sterm_h[i]);
     }
+    /* capacity pass, then ONE staged upload + scatter kernel for the
+     * whole chunk (3 tiny hipMemcpys per bucket per chunk were the
+     * N=50M build-time bottleneck at large nlist) */
+    std::vector<GammaScatterSeg> segs;
+    std::vector<uint32_t> ids_cat;
+    std::vector<uint8_t> data_cat;
+    std::vector<float> svals_cat;
+    segs.reserve(groups.size());
+    ids_cat.reserve(cn);
+    data_cat.reserve((size_t)cn * entry);
+    if (pq) svals_cat.reserve(cn);
     for (auto &kv : groups) {
       Bucket &bk = buckets_[kv.first];
       int64_t add_n = (int64_t)kv.second.ids.size();
@@ -918,17 +929,21 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
         bk.data = std::move(ndata);
         if (pq) bk.svals = std::move(nsv);
         bk.cap = ncap;
-        dev_buckets_dirty_ = true;
       }
-      (void)hipMemcpy(bk.ids->as<uint32_t>() + bk.size, kv.second.ids.data(),
-                (size_t)add_n * 4, hipMemcpyHostToDevice);
-      (void)hipMemcpy((uint8_t *)bk.data->get() + (size_t)bk.size * entry,
-                kv.second.data.data(), (size_t)add_n * entry,
-                hipMemcpyHostToDevice);
+      GammaScatterSeg sg;
+      sg.ids_dst = bk.ids->as<uint32_t>() + bk.size;
+      sg.data_dst = (uint8_t *)bk.data->get() + (size_t)bk.size * entry;
+      sg.sval_dst = pq ? bk.svals->as<float>() + bk.size : nullptr;
+      sg.src_start = (long long)ids_cat.size();
+      sg.count = add_n;
+      segs.push_back(sg);
+      ids_cat.insert(ids_cat.end(), kv.second.ids.begin(),
+                     kv.second.ids.end());
+      data_cat.insert(data_cat.end(), kv.second.data.begin(),
+                      kv.second.data.end());
       if (pq)
-        (void)hipMemcpy(bk.svals->as<float>() + bk.size,
-                  kv.second.svals.data(), (size_t)add_n * 4,
-                  hipMemcpyHostToDevice);
+        svals_cat.insert(svals_cat.end(), kv.second.svals.begin(),
+                         kv.second.svals.end());
       for (int64_t i = 0; i < add_n; i++) {
         int64_t vid = (int64_t)kv.second.ids[i];
         if (vid >= (int64_t)vid_loc_.size())
@@ -937,6 +952,31 @@ int IVFIndex::add(const float *x_host, const int64_t *vids, int64_t n,
       }
       bk.size += add_n;
       dev_buckets_dirty_ = true;
+    }
+    if (!segs.empty()) {
+      if (scat_segs_.reserve(segs.size() * sizeof(GammaScatterSeg)))
+        return -1;
+      if (scat_ids_.reserve(ids_cat.size() * 4)) return -1;
+      if (scat_data_.reserve(data_cat.size())) return -1;
+      if (pq && scat_svals_.reserve(svals_cat.size() * 4)) return -1;
+      GAMMA_CHECK(hipMemcpyAsync(scat_segs_.get(), segs.data(),
+                                 segs.size() * sizeof(GammaScatterSeg),
+                                 hipMemcpyHostToDevice, s));
+      GAMMA_CHECK(hipMemcpyAsync(scat_ids_.get(), ids_cat.data(),
+                                 ids_cat.size() * 4,
+                                 hipMemcpyHostToDevice, s));
+      GAMMA_CHECK(hipMemcpyAsync(scat_data_.get(), data_cat.data(),
+                                 data_cat.size(), hipMemcpyHostToDevice,
+                                 s));
+      if (pq)
+        GAMMA_CHECK(hipMemcpyAsync(scat_svals_.get(), svals_cat.data(),
+                                   svals_cat.size() * 4,
+                                   hipMemcpyHostToDevice, s));
+      GAMMA_CHECK(gk::bucket_scatter(
+          s, (int)segs.size(), scat_segs_.as<GammaScatterSeg>(),
+          scat_ids_.as<uint32_t>(), scat_data_.as<uint8_t>(),
+          pq ? scat_svals_.as<float>() : nullptr, (int)entry));
+      GAMMA_CHECK(hipStreamSynchronize(s));
     }
   }
   ntotal_ += n;

@@ -327,6 +327,8 @@ class IVFIndex {
   /* scratch for add/train (grow-only; search scratch lives in
    * SearchScratch so searches can run concurrently) */
   DeviceBuf scratch_i32_, scratch_f32_;
+  /* bulk-ingest staging (gk::bucket_scatter) */
+  DeviceBuf scat_segs_, scat_ids_, scat_data_, scat_svals_;
   std::mutex bk_mu_; /* guards the one lazy write under shared lock:
                         first update_dev_buckets with no prior add */
 };

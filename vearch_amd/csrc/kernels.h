@@ -22,7 +22,28 @@ struct GammaBucketDev {
   long long size;
 };
 
+/* one bucket's slice of a bulk-ingest batch (gk::bucket_scatter):
+ * copy `count` entries starting at src_start of the staged arrays into
+ * this bucket's device arrays at the recorded append position */
+struct GammaScatterSeg {
+  uint32_t *ids_dst;
+  uint8_t *data_dst;
+  float *sval_dst; /* null for IVFFLAT */
+  long long src_start;
+  long long count;
+};
+
 namespace gk {
+
+/* Bulk bucket append: instead of 3 small hipMemcpys per bucket per
+ * chunk (~100k calls per 64k-vector chunk at nlist=32k — the N=50M
+ * build-time killer), the host stages the grouped chunk once and one
+ * kernel scatters it into every bucket. */
+hipError_t bucket_scatter(hipStream_t s, int nseg,
+                          const GammaScatterSeg *segs_dev,
+                          const uint32_t *ids_src,
+                          const uint8_t *data_src,
+                          const float *svals_src, int entry_bytes);
 
 /* out[i*n+j] = dot(Q_i, B_j); f32 MFMA 16x16x4. */
 hipError_t dots_mfma(hipStream_t s, const float *Q, int nq, const float *B,

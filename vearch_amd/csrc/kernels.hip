@@ -1234,6 +1234,44 @@ hipError_t gk::pq_sterm(hipStream_t s, int64_t n, int M, int nlist,
   return hipGetLastError();
 }
 
+/* ------------------------------------------------------ bucket scatter */
+__global__ void k_bucket_scatter(int nseg,
+                                 const GammaScatterSeg *__restrict__ segs,
+                                 const uint32_t *__restrict__ ids_src,
+                                 const uint8_t *__restrict__ data_src,
+                                 const float *__restrict__ svals_src,
+                                 int entry_bytes) {
+  const GammaScatterSeg sg = segs[blockIdx.x];
+  for (long long i = threadIdx.x; i < sg.count; i += blockDim.x) {
+    sg.ids_dst[i] = ids_src[sg.src_start + i];
+    if (sg.sval_dst && svals_src)
+      sg.sval_dst[i] = svals_src[sg.src_start + i];
+  }
+  /* entry payloads as u32 words when aligned, bytes otherwise */
+  long long total = sg.count * entry_bytes;
+  const uint8_t *src = data_src + sg.src_start * entry_bytes;
+  if ((entry_bytes & 3) == 0) {
+    const uint32_t *s4 = (const uint32_t *)src;
+    uint32_t *d4 = (uint32_t *)sg.data_dst;
+    for (long long i = threadIdx.x; i < (total >> 2); i += blockDim.x)
+      d4[i] = s4[i];
+  } else {
+    for (long long i = threadIdx.x; i < total; i += blockDim.x)
+      sg.data_dst[i] = src[i];
+  }
+}
+
+hipError_t gk::bucket_scatter(hipStream_t s, int nseg,
+                              const GammaScatterSeg *segs_dev,
+                              const uint32_t *ids_src,
+                              const uint8_t *data_src,
+                              const float *svals_src, int entry_bytes) {
+  if (nseg <= 0) return hipSuccess;
+  k_bucket_scatter<<<dim3((uint32_t)nseg), dim3(256), 0, s>>>(
+      nseg, segs_dev, ids_src, data_src, svals_src, entry_bytes);
+  return hipGetLastError();
+}
+
 /* ----------------------------------------------------------- residuals */
 __global__ void k_residuals(int64_t n, int d, const float *__restrict__ x,
                             const float *__restrict__ centroids,

@@ -989,7 +989,12 @@ int IVFIndex::prepare_fast_one(const float *vec_h, hipStream_t s,
   if (!trained_) return 1;
   if (dev_buckets_dirty_) return 1; /* device table stale: slow path
                                        rebuilds it under the write lock */
-  DeviceBuf xd, xrot, xnorm, dots, asg, resid, codes, sterm;
+  /* persistent grow-only scratch: a hipFree here would synchronize the
+   * whole device and stall the concurrent read-locked searches this
+   * path exists for; appenders are serialized so sharing is safe */
+  DeviceBuf &xd = fast_xd_, &xrot = fast_xrot_, &xnorm = fast_xnorm_,
+            &dots = fast_dots_, &asg = fast_asg_, &resid = fast_resid_,
+            &codes = fast_codes_, &sterm = fast_sterm_;
   if (xd.reserve((size_t)d_ * 4)) return -1;
   GAMMA_CHECK(hipMemcpy(xd.get(), vec_h, (size_t)d_ * 4,
                         hipMemcpyHostToDevice));

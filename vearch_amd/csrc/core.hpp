@@ -329,6 +329,9 @@ class IVFIndex {
   DeviceBuf scratch_i32_, scratch_f32_;
   /* bulk-ingest staging (gk::bucket_scatter) */
   DeviceBuf scat_segs_, scat_ids_, scat_data_, scat_svals_;
+  /* lock-free-append scratch (prepare_fast_one; appenders serialize) */
+  DeviceBuf fast_xd_, fast_xrot_, fast_xnorm_, fast_dots_, fast_asg_,
+      fast_resid_, fast_codes_, fast_sterm_;
   std::mutex bk_mu_; /* guards the one lazy write under shared lock:
                         first update_dev_buckets with no prior add */
 };

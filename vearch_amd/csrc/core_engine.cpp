@@ -341,6 +341,16 @@ int Engine::rebuild_index(bool drop_before_rebuild, std::string *err) {
         return -1;
       }
       indexed_count_ = 0;
+      for (auto &e : extra_vecs_) { /* extras retrain with the table */
+        if (!e->index) continue;
+        IndexParams pe = params_;
+        if (pe.nsubvector && e->dim % pe.nsubvector) pe.nsubvector = 0;
+        e->index = std::make_unique<IVFIndex>();
+        if (e->index->init(e->dim, pe)) {
+          if (err) *err = "extra index re-init failed";
+          return -1;
+        }
+      }
     }
   }
   return build_index(err);

@@ -1201,13 +1201,50 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                                   sc.atab.as<float>()));
       atab = sc.atab.as<float>();
     }
+    /* cache-clustered schedule: sort the batch by first probed list
+     * so workgroups touching the same lists run together while those
+     * lists are L2/LLC-resident (counting sort on the host, two 40 KB
+     * copies; only worth the stream sync at large batches) */
+    const int32_t *qmap_dev = nullptr;
+    if (nq >= 2048 && getenv("GAMMA_NO_QSORT") == nullptr) {
+      if (sc.qcol.reserve((size_t)nq * 4)) return -1;
+      if (sc.qmap.reserve((size_t)nq * 4)) return -1;
+      GAMMA_CHECK(gk::extract_probe0(s, nq, nprobe,
+                                     sc.probes.as<int64_t>(),
+                                     sc.qcol.as<int32_t>()));
+      sc.qcol_h.resize(nq);
+      GAMMA_CHECK(hipMemcpyAsync(sc.qcol_h.data(), sc.qcol.get(),
+                                 (size_t)nq * 4, hipMemcpyDeviceToHost,
+                                 s));
+      GAMMA_CHECK(hipStreamSynchronize(s));
+      sc.qmap_h.resize(nq);
+      {
+        std::vector<int32_t> cnt((size_t)nlist_ + 2, 0);
+        for (int i = 0; i < nq; i++) {
+          int32_t ln = sc.qcol_h[i];
+          if (ln < 0 || ln >= nlist_) ln = nlist_;
+          cnt[ln + 1]++;
+        }
+        for (int i = 1; i <= nlist_ + 1; i++) cnt[i] += cnt[i - 1];
+        for (int i = 0; i < nq; i++) {
+          int32_t ln = sc.qcol_h[i];
+          if (ln < 0 || ln >= nlist_) ln = nlist_;
+          sc.qmap_h[cnt[ln]++] = i;
+        }
+      }
+      GAMMA_CHECK(hipMemcpyAsync(sc.qmap.get(), sc.qmap_h.data(),
+                                 (size_t)nq * 4, hipMemcpyHostToDevice,
+                                 s));
+      qmap_dev = sc.qmap.as<int32_t>();
+    }
     GAMMA_CHECK(gk::ivfpq_scan(s, nq, S, d_, M_, nprobe, k2, q_dev,
                                centroids_.as<float>(),
                                codebooks_.as<float>(), atab,
                                sc.pdists.as<float>(),
                                dev_buckets_.as<GammaBucketDev>(), nlist_,
                                sc.probes.as<int64_t>(), bitmap_dev,
-                               metric_ip, out_keys_dev, kill_flag_dev));
+                               metric_ip, out_keys_dev, kill_flag_dev,
+                               qmap_dev));
   } else {
     GAMMA_CHECK(gk::ivfflat_scan(s, nq, d_, nprobe, k2, q_dev,
                                  dev_buckets_.as<GammaBucketDev>(), nlist_,

@@ -230,6 +230,9 @@ struct SearchScratch {
   std::vector<uint32_t> filt_host;
   /* index-side (coarse assign + ADC scan) */
   DeviceBuf dots, sel_keys, probes, pdists, atab;
+  /* cache-clustered query schedule (see kernels.h ivfpq_scan qmap) */
+  DeviceBuf qcol, qmap;
+  std::vector<int32_t> qcol_h, qmap_h;
   DeviceBuf rot_q, rot_norms; /* OPQ-rotated queries + their norms */
   bool in_use = false;
   ~SearchScratch() {

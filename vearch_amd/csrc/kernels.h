@@ -109,6 +109,14 @@ hipError_t pq_sterm(hipStream_t s, int64_t n, int M, int nlist,
  * the probed lists (h:923-953 semantics). out_keys: nq x k2. */
 /* S = probe-split factor: S sub-workgroups per query (out_keys is
  * nq x S x k2; merge with sort_rows). S>1 serves small batches. */
+/* qmap: optional query schedule — workgroup b serves query
+ * qmap[b/S]. The engine sorts large batches by their first probed
+ * list so neighbouring workgroups scan the same lists while they are
+ * L2/LLC-resident (the 10M-doc code store is ~400 MB against the
+ * 256 MiB Infinity Cache — unordered queries re-fetch every list
+ * ~nq*nprobe/nlist times from DRAM). Results are identical: the
+ * selector's top-k2 is order-independent and outputs are written at
+ * the ORIGINAL query row. */
 hipError_t ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
                       int nprobe,
                       int k2, const float *queries, const float *centroids,
@@ -117,7 +125,11 @@ hipError_t ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
                       const GammaBucketDev *buckets,
                       int nlist, const int64_t *probes,
                       const uint32_t *bitmap, bool ip, uint64_t *out_keys,
-                      const int *kill_flag);
+                      const int *kill_flag, const int32_t *qmap);
+
+/* out[q] = (int32) probes[q*nprobe] (the schedule sort key) */
+hipError_t extract_probe0(hipStream_t s, int nq, int nprobe,
+                          const int64_t *probes, int32_t *out);
 
 /* IVFFLAT fused search (gamma_index_ivfflat.h:36-91). */
 hipError_t ivfflat_scan(hipStream_t s, int nq, int d, int nprobe, int k2,

@@ -442,7 +442,8 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
              const int64_t *__restrict__ probes,
              const uint32_t *__restrict__ bitmap,
              uint64_t *__restrict__ out_keys,
-             const int *__restrict__ kill_flag) {
+             const int *__restrict__ kill_flag,
+             const int32_t *__restrict__ qmap) {
   extern __shared__ char smem[];
   const int ksub = 256;
   const int dsub = d / M;
@@ -457,9 +458,12 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
   /* probe-split: S sub-workgroups per query, sub-block handles probes
    * p ≡ sub (mod S); partials merged by sort_rows afterwards. S>1 is
    * the small-batch/serving path (a lone query still fills S CUs). */
-  const int q = blockIdx.x / S;
-  const int sub = blockIdx.x - q * S;
-  if (q >= nq) return;
+  const int bq = blockIdx.x / S;
+  const int sub = blockIdx.x - bq * S;
+  if (bq >= nq) return;
+  const int q = qmap ? qmap[bq] : bq; /* scheduled query (cache
+                                         clustering); outputs go to the
+                                         ORIGINAL row below */
   const float *qg = queries + (int64_t)q * d;
   for (int i = threadIdx.x; i < d; i += blockDim.x) qs[i] = qg[i];
 
@@ -643,7 +647,21 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
   }
   sel.finish();
   for (int i = threadIdx.x; i < k2; i += blockDim.x)
-    out_keys[(int64_t)blockIdx.x * k2 + i] = res[i];
+    out_keys[((int64_t)q * S + sub) * k2 + i] = res[i];
+}
+
+__global__ void k_extract_probe0(int nq, int nprobe,
+                                 const int64_t *__restrict__ probes,
+                                 int32_t *__restrict__ out) {
+  int q = blockIdx.x * blockDim.x + threadIdx.x;
+  if (q < nq) out[q] = (int32_t)probes[(int64_t)q * nprobe];
+}
+
+hipError_t gk::extract_probe0(hipStream_t s, int nq, int nprobe,
+                              const int64_t *probes, int32_t *out) {
+  k_extract_probe0<<<dim3((uint32_t)((nq + WG - 1) / WG)), dim3(WG), 0,
+                     s>>>(nq, nprobe, probes, out);
+  return hipGetLastError();
 }
 
 hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
@@ -655,7 +673,7 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
                           const GammaBucketDev *buckets, int nlist,
                           const int64_t *probes, const uint32_t *bitmap,
                           bool ip, uint64_t *out_keys,
-                          const int *kill_flag) {
+                          const int *kill_flag, const int32_t *qmap) {
   size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
                 (GAMMA_SORT_CAP + k2) * 8 + (d + 2) * 4 +
                 2 * sizeof(long long) + 4 * sizeof(int);
@@ -679,7 +697,7 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
   k_ivfpq_scan<IPV, MWV, BSV><<<g, dim3(BSV), smem, s>>>(                 \
       nq, S, d, M, nprobe, k2, queries, centroids, codebooks, atab,       \
       probe_dists, buckets, nlist, probes, bitmap, out_keys,              \
-      kill_flag)
+      kill_flag, qmap)
 #define GAMMA_LAUNCH_SCAN_BS(IPV, MWV)                                    \
   do {                                                                    \
     if (BS == 256) GAMMA_LAUNCH_SCAN(IPV, MWV, 256);                      \

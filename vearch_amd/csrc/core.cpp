@@ -1201,12 +1201,16 @@ int IVFIndex::search(const float *q_dev, int nq, int k2, int nprobe,
                                   sc.atab.as<float>()));
       atab = sc.atab.as<float>();
     }
-    /* cache-clustered schedule: sort the batch by first probed list
-     * so workgroups touching the same lists run together while those
-     * lists are L2/LLC-resident (counting sort on the host, two 40 KB
-     * copies; only worth the stream sync at large batches) */
+    /* cache-clustered schedule (opt-in): sort the batch by first
+     * probed list so workgroups touching the same lists run together
+     * while those lists are L2/LLC-resident. MEASURED NEUTRAL on the
+     * headline workload (scan 2.78 vs 2.72 ms with/without,
+     * profiles/README round-2 notes): the scan sits at its
+     * access-pattern ceiling, not on DRAM re-reads, so the schedule's
+     * stream sync only adds cost. Kept behind GAMMA_QSORT=1 for
+     * workloads with heavier list skew. */
     const int32_t *qmap_dev = nullptr;
-    if (nq >= 2048 && getenv("GAMMA_NO_QSORT") == nullptr) {
+    if (nq >= 2048 && getenv("GAMMA_QSORT") != nullptr) {
       if (sc.qcol.reserve((size_t)nq * 4)) return -1;
       if (sc.qmap.reserve((size_t)nq * 4)) return -1;
       GAMMA_CHECK(gk::extract_probe0(s, nq, nprobe,

@@ -429,8 +429,8 @@ hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
  * 9.84 ms for C=2 on the uniform 10M/nprobe=32 microbench; 24 waves/CU
  * already cover the load latency, and C=1 halves the register-staging
  * pressure) — see tools/adc_bench.hip. */
-#define GAMMA_ADC_C 1
-template <bool IP, int MW, int BS>
+#define GAMMA_ADC_C_DEFAULT 1
+template <bool IP, int MW, int BS, int CP>
 __global__ void __launch_bounds__(BS)
 k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
              const float *__restrict__ queries,
@@ -506,7 +506,7 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
      * independent HBM streams. Push order changes; the selector's
      * top-k2 is exact under the (dist,id) total order, so results are
      * unchanged. */
-    const int C = GAMMA_ADC_C;
+    const int C = CP;
     const int HB = BS / 2;
     const int half = threadIdx.x >= HB ? 1 : 0;
     const int tid = threadIdx.x - half * HB;
@@ -690,27 +690,39 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
     const char *e = getenv("GAMMA_SCAN_BS");
     if (e && (atoi(e) == 256 || atoi(e) == 512)) BS = atoi(e);
   }
-  bool fast = (k2 + BS * GAMMA_ADC_C) <= GAMMA_SORT_CAP &&
+  /* C = codes register-staged per thread per flush interval: deeper
+   * staging doubles the outstanding VMEM requests per wave — the lever
+   * when short lists (large nlist) cap the per-wave pipelining.
+   * GAMMA_ADC_C overrides; default policy below. */
+  int CSEL = GAMMA_ADC_C_DEFAULT;
+  {
+    const char *e = getenv("GAMMA_ADC_C");
+    if (e && (atoi(e) == 1 || atoi(e) == 2)) CSEL = atoi(e);
+  }
+  bool fast = (k2 + BS * CSEL) <= GAMMA_SORT_CAP &&
               (M == 16 || M == 32 || M == 64 || M == 96);
+  if (!((k2 + BS * CSEL) <= GAMMA_SORT_CAP)) CSEL = 1;
   dim3 g((uint32_t)nq * (uint32_t)S);
-#define GAMMA_LAUNCH_SCAN(IPV, MWV, BSV)                                  \
-  k_ivfpq_scan<IPV, MWV, BSV><<<g, dim3(BSV), smem, s>>>(                 \
+#define GAMMA_LAUNCH_SCAN(IPV, MWV, BSV, CPV)                             \
+  k_ivfpq_scan<IPV, MWV, BSV, CPV><<<g, dim3(BSV), smem, s>>>(            \
       nq, S, d, M, nprobe, k2, queries, centroids, codebooks, atab,       \
       probe_dists, buckets, nlist, probes, bitmap, out_keys,              \
       kill_flag, qmap)
 #define GAMMA_LAUNCH_SCAN_BS(IPV, MWV)                                    \
   do {                                                                    \
-    if (BS == 256) GAMMA_LAUNCH_SCAN(IPV, MWV, 256);                      \
-    else GAMMA_LAUNCH_SCAN(IPV, MWV, 512);                                \
+    if (BS == 256 && CSEL == 2) GAMMA_LAUNCH_SCAN(IPV, MWV, 256, 2);      \
+    else if (BS == 256) GAMMA_LAUNCH_SCAN(IPV, MWV, 256, 1);              \
+    else if (CSEL == 2) GAMMA_LAUNCH_SCAN(IPV, MWV, 512, 2);              \
+    else GAMMA_LAUNCH_SCAN(IPV, MWV, 512, 1);                             \
   } while (0)
   if (ip) {
-    if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG);
+    if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG, 1);
     else if (M == 16) GAMMA_LAUNCH_SCAN_BS(true, 4);
     else if (M == 32) GAMMA_LAUNCH_SCAN_BS(true, 8);
     else if (M == 64) GAMMA_LAUNCH_SCAN_BS(true, 16);
     else GAMMA_LAUNCH_SCAN_BS(true, 24);
   } else {
-    if (!fast) GAMMA_LAUNCH_SCAN(false, 0, WG);
+    if (!fast) GAMMA_LAUNCH_SCAN(false, 0, WG, 1);
     else if (M == 16) GAMMA_LAUNCH_SCAN_BS(false, 4);
     else if (M == 32) GAMMA_LAUNCH_SCAN_BS(false, 8);
     else if (M == 64) GAMMA_LAUNCH_SCAN_BS(false, 16);

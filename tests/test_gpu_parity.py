@@ -749,6 +749,37 @@ def test_concurrent_search_and_mutation(data):
     eng.close()
 
 
+def test_scan_launch_knobs_bitexact(data, ivfpq_engine):
+    """The perf-experiment knobs (GAMMA_ADC_C staging depth,
+    GAMMA_SCAN_BS block size, GAMMA_SCAN_S probe split, GAMMA_QSORT
+    query schedule) must not change results: the selector's top-k is
+    exact under the (dist,id) total order and outputs are written at
+    the original query row."""
+    import os
+    base, q = data
+    eng = ivfpq_engine
+    # >= 2048 queries so GAMMA_QSORT actually activates its schedule
+    qq = np.repeat(q, 32, axis=0)
+    ref_d, ref_i = eng.raw_search(qq, 10, nprobe=16, rerank=64)
+    knobs = [
+        {"GAMMA_ADC_C": "2"},
+        {"GAMMA_SCAN_BS": "256"},
+        {"GAMMA_SCAN_S": "2"},
+        {"GAMMA_QSORT": "1"},
+        {"GAMMA_ADC_C": "2", "GAMMA_SCAN_BS": "256", "GAMMA_SCAN_S": "2"},
+    ]
+    for kv in knobs:
+        for k, v in kv.items():
+            os.environ[k] = v
+        try:
+            gd, gi = eng.raw_search(qq, 10, nprobe=16, rerank=64)
+        finally:
+            for k in kv:
+                del os.environ[k]
+        assert np.array_equal(gi, ref_i), kv
+        assert np.array_equal(gd, ref_d), kv
+
+
 def test_multi_vector_search(data):
     """Multi-vector-field search (vector_manager.cc:851-1090): per-field
     top-n, docid-intersection merge, WeightedRanker combined scores,

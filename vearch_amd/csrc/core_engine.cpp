@@ -155,6 +155,7 @@ int Engine::add_doc(
     }
   }
   int64_t docid = max_docid_.load(std::memory_order_relaxed);
+  if (raw_.size() != docid) return -1; /* see add_doc_fast_ */
   docid2pkey_.set(docid, p_key);
   bitmap_.ensure(docid + 1, stream_);
   for (auto &f : fields_) {
@@ -195,6 +196,9 @@ int Engine::add_doc_fast_(
   }
   int64_t docid = max_docid_.load(std::memory_order_relaxed);
   if (docid + 1 > 0x7fffffff) return -1; /* 31-bit device vid */
+  if (raw_.size() != docid) return -1; /* row/docid drift (a prior
+                                          failed append): fail loudly
+                                          rather than mis-map rows */
   if (raw_.would_grow(1)) return 1;
   if (!bitmap_.has_capacity(docid + 1)) return 1;
   int32_t bucket = -1;

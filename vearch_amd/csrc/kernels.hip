@@ -295,98 +295,6 @@ k_select_from_dots_wave(int nq, int64_t ncols, int64_t col_base,
     state_keys[(int64_t)q * k2 + i] = sel.buf[i];
 }
 
-/* Split-row variant: TWO waves per query, each scanning half the row
- * (the coarse assign reads a 16k-wide row per query at nlist=16384 —
- * one wave's 64-lane float4 walk is a 64-trip serial chain; halving it
- * doubles the per-query memory-level parallelism). After both halves
- * finish, the even wave merges the pair's sorted top-k2 regions with
- * one wave-level bitonic pass — exact top-k2 under the same
- * (dist,id) total order. Dispatched for wide aligned rows only. */
-template <bool IP>
-__global__ void __launch_bounds__(512)
-k_select_from_dots_wave2(int nq, int64_t ncols, int64_t col_base,
-                         int64_t ld, const float *__restrict__ dots,
-                         const float *__restrict__ qnorms,
-                         const float *__restrict__ bnorms, int l2,
-                         const uint32_t *__restrict__ bitmap, int k2,
-                         uint64_t *__restrict__ state_keys, int seeded) {
-  extern __shared__ char smem[];
-  const int nw = 512 / 64; /* 8 waves = 4 queries x 2 halves */
-  uint64_t *wb = (uint64_t *)smem;
-  int *cnts = (int *)(wb + (size_t)nw * GAMMA_WSEL_CAP);
-  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int half = wave & 1;
-  const int q = blockIdx.x * 4 + (wave >> 1);
-  uint64_t *reg = wb + (size_t)wave * GAMMA_WSEL_CAP;
-  GammaWaveSelector sel;
-  sel.init(reg, cnts + wave, k2);
-  if (q < nq) {
-    const float *row = dots + (int64_t)q * ld;
-    const float qn = l2 ? qnorms[q] : 0.0f;
-    if (seeded && half == 0) { /* seeds fold into half 0 only */
-      for (int i = lane; i < k2; i += 64) {
-        uint64_t kk = state_keys[(int64_t)q * k2 + i];
-        if (kk != GAMMA_KEY_EMPTY) {
-          int idx = atomicAdd(sel.cnt, 1);
-          sel.buf[sel.k + idx] = kk;
-        }
-      }
-      sel.finish();
-    }
-    const int64_t nc4 = ncols >> 2; /* dispatch guards ncols%4==0 */
-    const int64_t h4 = (nc4 + 1) >> 1;
-    const int64_t beg4 = (int64_t)half * h4;
-    const int64_t end4 = nc4 < beg4 + h4 ? nc4 : beg4 + h4;
-    const float4 *row4 = (const float4 *)row;
-    for (int64_t c0 = beg4; c0 < end4; c0 += 64) {
-      int64_t c4 = c0 + lane;
-      if (c4 < end4) {
-        int64_t id0 = col_base + c4 * 4;
-        float4 d4 = row4[c4];
-        float4 b4 = l2 ? *(const float4 *)(bnorms + id0)
-                       : make_float4(0, 0, 0, 0);
-        float dv[4] = {d4.x, d4.y, d4.z, d4.w};
-        float bv[4] = {b4.x, b4.y, b4.z, b4.w};
-#pragma unroll
-        for (int t = 0; t < 4; t++) {
-          int64_t id = id0 + t;
-          if (!gamma_bitmap_test(bitmap, (uint64_t)id)) {
-            float dist = l2 ? fmaf(-2.0f, dv[t], qn + bv[t]) : dv[t];
-            sel.push(gamma_make_key<IP>(dist, (uint32_t)id));
-          }
-        }
-      }
-      sel.maybe_flush(64 * 4);
-    }
-    sel.finish();
-  }
-  __syncthreads(); /* both halves' regions final */
-  if (half == 0 && q < nq) {
-    uint64_t *other = wb + (size_t)(wave + 1) * GAMMA_WSEL_CAP;
-    for (int i = lane; i < k2; i += 64) reg[k2 + i] = other[i];
-    __builtin_amdgcn_wave_barrier();
-    int n = 1;
-    while (n < 2 * k2) n <<= 1; /* <= GAMMA_WSEL_CAP for k2 <= 256 */
-    for (int i = 2 * k2 + lane; i < n; i += 64) reg[i] = GAMMA_KEY_EMPTY;
-    __builtin_amdgcn_wave_barrier();
-    for (int len = 2; len <= n; len <<= 1) {
-      for (int inc = len >> 1; inc > 0; inc >>= 1) {
-        for (int i = lane; i < n; i += 64) {
-          int j = i ^ inc;
-          if (j > i) {
-            bool up = (i & len) == 0;
-            uint64_t a = reg[i], b = reg[j];
-            if ((a > b) == up) { reg[i] = b; reg[j] = a; }
-          }
-        }
-        __builtin_amdgcn_wave_barrier();
-      }
-    }
-    for (int i = lane; i < k2; i += 64)
-      state_keys[(int64_t)q * k2 + i] = reg[i];
-  }
-}
-
 hipError_t gk::select_from_dots(hipStream_t s, int nq, int64_t ncols,
                                 int64_t col_base, int64_t ld,
                                 const float *dots, const float *qnorms,
@@ -399,19 +307,6 @@ hipError_t gk::select_from_dots(hipStream_t s, int nq, int64_t ncols,
   if (k2 <= 192) {
     const int nw = 512 / 64;
     size_t smem = (size_t)nw * GAMMA_WSEL_CAP * 8 + nw * sizeof(int);
-    if (ncols >= 4096 && (ncols & 3) == 0 && (col_base & 3) == 0) {
-      /* wide aligned rows: split each row over two waves */
-      dim3 g2((uint32_t)((nq + 3) / 4));
-      if (ip_order)
-        k_select_from_dots_wave2<true><<<g2, dim3(512), smem, s>>>(
-            nq, ncols, col_base, ld, dots, qnorms, bnorms, l2 ? 1 : 0,
-            bitmap, k2, state_keys, seeded ? 1 : 0);
-      else
-        k_select_from_dots_wave2<false><<<g2, dim3(512), smem, s>>>(
-            nq, ncols, col_base, ld, dots, qnorms, bnorms, l2 ? 1 : 0,
-            bitmap, k2, state_keys, seeded ? 1 : 0);
-      return hipGetLastError();
-    }
     dim3 g((uint32_t)((nq + nw - 1) / nw));
     if (ip_order)
       k_select_from_dots_wave<true><<<g, dim3(512), smem, s>>>(

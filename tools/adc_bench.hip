@@ -1,3 +1,7 @@
+// NOTE: r1-era harness — replicates the ROUND-1 scan formulation
+// (per-list T = A + B staging, pre S-term). Kept so the r1 ablation
+// record (profiles/r01_adc_ablation_selectors.txt) stays reproducible;
+// the production kernel has since moved to the per-vector S term.
 /*
  * adc_bench.hip — standalone ablation bench for the IVFPQ ADC scan kernel
  * (north-star shape: d=128, M=32, nlist=4096, N=10M, nprobe=32, nq

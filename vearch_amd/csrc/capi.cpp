@@ -8,6 +8,7 @@
 #include <unistd.h>
 #include <string.h>
 
+#include <algorithm>
 #include <string>
 #include <vector>
 
@@ -413,6 +414,11 @@ static struct CStatus SearchMulti_unguarded(Engine *e,
                       t.include_upper, t.is_union});
 
   int k = req.topn; /* vector_manager.cc:955: depth = topn when merging */
+  if (std::max(k, recall_num) > 1024)
+    return err_status(1, "topN (or recall_num) = " +
+                             std::to_string(std::max(k, recall_num)) +
+                             " exceeds this engine's supported maximum "
+                             "of 1024");
   std::vector<double> scores((size_t)nq * k);
   std::vector<int64_t> ids((size_t)nq * k);
   std::string err;

@@ -857,6 +857,14 @@ def test_multi_vector_search(data):
         eng.search_pb(q[:2], topn=5,
                       extra_vec_queries=[("nope", q2[:2])])
 
+    # doc fetch returns EVERY vector field of a multi-vector table
+    docs = eng.query_pb(document_ids=["7"], is_vector_value=True)
+    fields = docs[0]["items"][0]["fields"]
+    got1 = np.frombuffer(fields[eng.vec_name], dtype=np.float32)
+    got2 = np.frombuffer(fields["emb2"], dtype=np.float32)
+    assert np.array_equal(got1, base[7])
+    assert np.array_equal(got2, base2[7])
+
     # dump/load keeps the extra field (v2 dump format)
     eng.dump()
     eng.close()

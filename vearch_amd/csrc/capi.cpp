@@ -219,6 +219,12 @@ static int serialize_doc(Engine *e, int64_t docid, char **doc_str,
       {e->vec_field_name(),
        std::string((const char *)vec, (size_t)e->dimension() * 4),
        gfb::VECTOR});
+  for (auto &ev : e->extra_vec_fields()) {
+    const float *v = ev->raw.host_row(docid);
+    doc.fields.push_back(
+        {ev->name, std::string((const char *)v, (size_t)ev->dim * 4),
+         gfb::VECTOR});
+  }
   std::string out = doc.serialize();
   *doc_str = dup_malloc(out);
   *len = (int)out.size();
@@ -566,6 +572,12 @@ static struct CStatus Search_unguarded(void *engine, const char *request_str, in
         item.fields.push_back(
             {e->vec_field_name(),
              std::string((const char *)vec, (size_t)d * 4)});
+        for (auto &ev : e->extra_vec_fields()) {
+          const float *v2 = ev->raw.host_row(id);
+          item.fields.push_back(
+              {ev->name,
+               std::string((const char *)v2, (size_t)ev->dim * 4)});
+        }
       }
       res.max_score = std::max(res.max_score, item.score);
       res.items.push_back(std::move(item));
@@ -638,6 +650,12 @@ static struct CStatus Query_unguarded(void *engine, const char *request_str, int
       item.fields.push_back(
           {e->vec_field_name(),
            std::string((const char *)vec, (size_t)e->dimension() * 4)});
+      for (auto &ev : e->extra_vec_fields()) {
+        const float *v2 = ev->raw.host_row(id);
+        item.fields.push_back(
+            {ev->name,
+             std::string((const char *)v2, (size_t)ev->dim * 4)});
+      }
     }
     res.items.push_back(std::move(item));
   }

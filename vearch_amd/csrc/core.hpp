@@ -461,6 +461,10 @@ class Engine {
                    const std::vector<RangeFilterSpec> *range_filters = nullptr,
                    int filter_op = 0, bool prelocked = false);
 
+  const std::vector<std::unique_ptr<ExtraVecField>> &extra_vec_fields()
+      const {
+    return extra_vecs_;
+  }
   /* dimension of a named vector field (primary or extra); -1 unknown */
   int vec_dim_of(const std::string &name) const {
     if (name == vec_name_) return dim_;

@@ -270,12 +270,14 @@ class GammaEngine:
         return proto.decode_search_response(buf)
 
     def query_pb(self, document_ids=(), fields=("_id",), term_filters=(),
-                 range_filters=(), limit=0, operator=0):
+                 range_filters=(), limit=0, operator=0,
+                 is_vector_value=False):
         """The C-ABI Query (doc fetch / filtered browse)."""
         from . import proto
         req = proto.encode_query_request(
             list(document_ids), fields=fields, term_filters=term_filters,
-            range_filters=range_filters, limit=limit, operator=operator)
+            range_filters=range_filters, limit=limit, operator=operator,
+            is_vector_value=is_vector_value)
         out = ctypes.c_char_p()
         n = ctypes.c_int()
         st = lib().Query(self.h, req, len(req), ctypes.byref(out),

@@ -430,12 +430,7 @@ hipError_t gk::argmin_rows(hipStream_t s, int64_t nrows, int ncols,
  * already cover the load latency, and C=1 halves the register-staging
  * pressure) — see tools/adc_bench.hip. */
 #define GAMMA_ADC_C_DEFAULT 1
-/* GLUT: serve the query-level L2 A-table from GLOBAL memory instead of
- * LDS. At M=96 (d=768) the 96 KB LDS table caps occupancy at one
- * workgroup (8 waves) per CU; the global table is a hot 96 KB region
- * (L1/L2-resident) and dropping it from LDS lifts occupancy ~6x. Only
- * instantiated for the large-M L2 path. */
-template <bool IP, int MW, int BS, int CP, bool GLUT = false>
+template <bool IP, int MW, int BS, int CP>
 __global__ void __launch_bounds__(BS)
 k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
              const float *__restrict__ queries,
@@ -452,9 +447,8 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
   extern __shared__ char smem[];
   const int ksub = 256;
   const int dsub = d / M;
-  float *lut = (float *)smem;                       /* M*ksub (unless GLUT) */
-  uint64_t *sortbuf = (uint64_t *)(
-      smem + (GLUT ? 0 : ((size_t)M * ksub * 4 + 7) / 8 * 8));
+  float *lut = (float *)smem;                       /* M*ksub */
+  uint64_t *sortbuf = (uint64_t *)(smem + ((size_t)M * ksub * 4 + 7) / 8 * 8);
   uint64_t *res = sortbuf + GAMMA_SORT_CAP;
   float *qs = (float *)(res + k2);                  /* d (IP table build) */
   float *dis0s = qs + d;                            /* 2 (one per half) */
@@ -487,7 +481,7 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
       lut[e] = acc;
     }
     __syncthreads();
-  } else if (!GLUT) {
+  } else {
     /* L2: the query-level A table ONCE per workgroup. The list half of
      * the decomposition (B) is pre-folded into one float per vector
      * (bucket svals, gk::pq_sterm), so no per-(query,list) table is
@@ -498,9 +492,6 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
       lut4[e] = Aq[e];
     __syncthreads();
   }
-  const float *lutp = (!IP && GLUT)
-                          ? atab + (size_t)q * (size_t)M * ksub
-                          : lut;
 
   /* in-flight kill (is_killed_every<1024> analog, ivfpq.h:927): poll a
    * device flag between lists with an agent-scope load (L2-served, so a
@@ -585,7 +576,7 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
           if (!((uint64_t)id >> 63) &&
               !gamma_bitmap_test(bitmap, (uint64_t)id)) {
             float dis = IP ? dis0 : dis0 + sv[c];
-            const float *tab = lutp;
+            const float *tab = lut;
 #pragma unroll
             for (int mw = 0; mw < MW; mw++) {
               uint32_t wv = w[c][mw];
@@ -638,7 +629,7 @@ k_ivfpq_scan(int nq, int S, int d, int M, int nprobe, int k2,
               !gamma_bitmap_test(bitmap, (uint64_t)id)) {
             const uint32_t *cw = (const uint32_t *)(codes + (size_t)j * M);
             float dis = IP ? dis0 : dis0 + svals[j];
-            const float *tab = lutp;
+            const float *tab = lut;
             for (int mw = 0; mw < mwords; mw++) {
               uint32_t wv = cw[mw];
               dis += tab[wv & 255u];         tab += ksub;
@@ -686,8 +677,6 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
   size_t smem = ((size_t)M * 256 * 4 + 7) / 8 * 8 +
                 (GAMMA_SORT_CAP + k2) * 8 + (d + 2) * 4 +
                 2 * sizeof(long long) + 4 * sizeof(int);
-  size_t smem_glut = (GAMMA_SORT_CAP + k2) * 8 + (d + 2) * 4 +
-                     2 * sizeof(long long) + 4 * sizeof(int);
   if (smem > 160 * 1024) return hipErrorInvalidValue;
   /* batched path needs flush margin blockDim*C inside the selector cap.
    * 512-thread blocks put 24 waves on a CU at the same LDS/WG (the ADC
@@ -713,12 +702,6 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
   bool fast = (k2 + BS * CSEL) <= GAMMA_SORT_CAP &&
               (M == 16 || M == 32 || M == 64 || M == 96);
   if (!((k2 + BS * CSEL) <= GAMMA_SORT_CAP)) CSEL = 1;
-  /* global-LUT experiment (GAMMA_GLOBAL_LUT=1): large-M L2 only */
-  bool glut = false;
-  {
-    const char *e = getenv("GAMMA_GLOBAL_LUT");
-    if (e && atoi(e) == 1 && !ip && M == 96 && fast) glut = true;
-  }
   dim3 g((uint32_t)nq * (uint32_t)S);
 #define GAMMA_LAUNCH_SCAN(IPV, MWV, BSV, CPV)                             \
   k_ivfpq_scan<IPV, MWV, BSV, CPV><<<g, dim3(BSV), smem, s>>>(            \
@@ -732,13 +715,6 @@ hipError_t gk::ivfpq_scan(hipStream_t s, int nq, int S, int d, int M,
     else if (CSEL == 2) GAMMA_LAUNCH_SCAN(IPV, MWV, 512, 2);              \
     else GAMMA_LAUNCH_SCAN(IPV, MWV, 512, 1);                             \
   } while (0)
-  if (glut) {
-    k_ivfpq_scan<false, 24, 512, 1, true><<<g, dim3(512), smem_glut, s>>>(
-        nq, S, d, M, nprobe, k2, queries, centroids, codebooks, atab,
-        probe_dists, buckets, nlist, probes, bitmap, out_keys, kill_flag,
-        qmap);
-    return hipGetLastError();
-  }
   if (ip) {
     if (!fast) GAMMA_LAUNCH_SCAN(true, 0, WG, 1);
     else if (M == 16) GAMMA_LAUNCH_SCAN_BS(true, 4);

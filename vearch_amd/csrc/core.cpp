@@ -1404,14 +1404,10 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
                 hipMemcpyHostToDevice);
       if (params_.kind == IndexKind::IVFPQ) {
         /* S terms are recomputable from codes + the B table, so the
-         * dump format is unchanged: rebuild them here (asg == the
-         * bucket number by construction) */
+         * dump format is unchanged; allocated here, filled by ONE
+         * pq_sterm_buckets launch after update_dev_buckets below */
         bk.svals = std::make_unique<DeviceBuf>();
         if (bk.svals->reserve((size_t)sz * 4)) return -1;
-        if (gk::pq_sterm(s, sz, M_, nlist_, bk.data->as<uint8_t>(),
-                         nullptr, i, btable_.as<float>(),
-                         bk.svals->as<float>()) != hipSuccess)
-          return -1;
       }
       bk.size = bk.cap = sz;
       for (long long j = 0; j < sz; j++) {
@@ -1426,7 +1422,15 @@ int IVFIndex::load(FILE *f, hipStream_t s) {
   }
   dev_buckets_dirty_ = true;
   trained_ = true;
-  return update_dev_buckets(s);
+  if (update_dev_buckets(s)) return -1;
+  if (params_.kind == IndexKind::IVFPQ) {
+    if (gk::pq_sterm_buckets(s, nlist_,
+                             M_, dev_buckets_.as<GammaBucketDev>(),
+                             btable_.as<float>()) != hipSuccess)
+      return -1;
+    GAMMA_CHECK(hipStreamSynchronize(s));
+  }
+  return 0;
 }
 
 }  // namespace vgamma

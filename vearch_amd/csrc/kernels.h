@@ -102,6 +102,9 @@ hipError_t pq_tables_a(hipStream_t s, int nq, int d, int M,
 hipError_t pq_sterm(hipStream_t s, int64_t n, int M, int nlist,
                     const uint8_t *codes, const int32_t *asg,
                     int asg_const, const float *btab, float *out);
+/* Load-path batch: one launch rebuilds every bucket's S terms */
+hipError_t pq_sterm_buckets(hipStream_t s, int nlist, int M,
+                            const GammaBucketDev *bks, const float *btab);
 
 /* IVFPQ fused search: one workgroup per query; stages the query-level
  * table ONCE (L2: A_q from atab, dis = coarse_dis + S_v + sum A[c_m];

@@ -914,6 +914,58 @@ def test_multi_vector_ivfpq(data):
     eng.close()
 
 
+def test_multi_vector_update_delete_rebuild(data):
+    """Multi-vector tables under mutation: updating a doc replaces its
+    row in EVERY field's index (old rows deleted in each extra index),
+    deletes hide the doc from merged results, and RebuildIndex retrains
+    the extra fields' indexes together with the primary."""
+    import ctypes as c
+    from vearch_amd.engine import lib
+    base, q = data
+    n, topn = 5000, 10
+    eng = make_engine("/tmp/gamma_mv_mut")
+    eng.create_table(
+        64, "IVFPQ",
+        '{"ncentroids": 32, "nsubvector": 16, "metric_type": "L2", '
+        '"training_threshold": 4000}',
+        extra_vecs=[("emb2", 64)])
+    for vid in range(n):
+        eng.add_doc(str(vid), base[vid], extra_vecs=[("emb2", base[vid])])
+    eng.build_index()
+
+    def top_ids(t=0):
+        res = eng.search_pb(q[t:t + 1], topn=topn,
+                            index_params='{"recall_num": 50}',
+                            extra_vec_queries=[("emb2", q[t:t + 1])],
+                            multi_vector_rank=1)
+        return [int(it["fields"]["_id"]) for it in res[0]["items"]]
+
+    ids0 = top_ids()
+    assert ids0
+    victim = ids0[0]
+    # update: move the top doc's vectors far away in BOTH fields
+    far = base[victim] + 100.0
+    eng.add_doc(str(victim), far, extra_vecs=[("emb2", far)])
+    ids1 = top_ids()
+    assert victim not in ids1
+    # delete the next top doc entirely
+    victim2 = ids1[0]
+    eng.delete_doc(str(victim2))
+    ids2 = top_ids()
+    assert victim2 not in ids2
+    # rebuild from scratch: extras retrain too; merged results stay sane
+    lib().RebuildIndex.argtypes = [c.c_void_p, c.c_int, c.c_int, c.c_int]
+    assert lib().RebuildIndex(eng.h, 1, 0, 0) == 0
+    ids3 = top_ids()
+    assert ids3 and victim not in ids3 and victim2 not in ids3
+    # identical fields => merged order echoes a single-field search
+    single = eng.search_pb(q[:1], topn=topn,
+                           index_params='{"recall_num": 50}')
+    sids = [int(it["fields"]["_id"]) for it in single[0]["items"]]
+    assert ids3 == sids
+    eng.close()
+
+
 def test_lockfree_add_under_search(data):
     """§8f-3 lock-free realtime add: a writer thread appends fresh docs
     (the common pure-append path runs under the SHARED lock — the

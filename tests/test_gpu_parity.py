@@ -907,6 +907,7 @@ def test_multi_vector_ivfpq(data):
     for t in range(nq):
         sids = [int(it["fields"]["_id"]) for it in single[t]["items"]]
         mids = [int(it["fields"]["_id"]) for it in multi[t]["items"]]
+        assert sids, "single-field search returned nothing (vacuous)"
         assert mids == sids
         for si, mi in zip(single[t]["items"], multi[t]["items"]):
             assert abs(mi["score"] - si["score"]) < 1e-5 * max(

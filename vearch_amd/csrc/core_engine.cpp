@@ -752,7 +752,11 @@ int Engine::search(int nq, const float *xq, int k, int nprobe,
       return -1;
     k2 = kf;
   } else {
-    int S = index_->probe_split(nq, k2, nprobe > 0 ? nprobe : 1024);
+    /* nprobe unset -> the index's configured nprobe (the reference's
+     * retrieval-params default, ivfpq.cc:253-294 / ivfpq.h:1049) */
+    if (nprobe <= 0) nprobe = index_->params().nprobe;
+    if (nprobe <= 0) nprobe = 80;
+    int S = index_->probe_split(nq, k2, nprobe);
     if (S > 1 && sc.keys.reserve((size_t)nq * S * k2 * 8)) return -1;
     /* arm the in-flight kill flag so SetKillStatus stops the scan
      * between lists (request_context.h:83 semantics) */
@@ -841,7 +845,9 @@ int Engine::search_field_(RawStore &raw, IVFIndex *idx, int dim,
       return -1;
     k2 = kf;
   } else {
-    int S = idx->probe_split(nq, k2, nprobe > 0 ? nprobe : 1024);
+    if (nprobe <= 0) nprobe = idx->params().nprobe;
+    if (nprobe <= 0) nprobe = 80;
+    int S = idx->probe_split(nq, k2, nprobe);
     if (sc.keys.reserve((size_t)nq * S * k2 * 8)) return -1;
     double ta = 0, ts = 0;
     if (idx->search(qptr, nq, k2, nprobe, bm, ip, s,

@@ -915,6 +915,85 @@ def test_multi_vector_ivfpq(data):
     eng.close()
 
 
+def test_score_range_docwalk_status(data, ivfpq_engine):
+    """Request-surface corners: min/max score ranges (single AND
+    per-field multi-vector), GetDocByDocID(next) walking over deleted
+    docs, and the EngineStatus / MemoryInfo JSON."""
+    import json as _json
+    import ctypes as c
+    from vearch_amd.engine import lib
+    base, q = data
+    eng = ivfpq_engine
+    # single-path score range: min/max bracket the true score window
+    full = eng.search_pb(q[:4], topn=10,
+                         index_params='{"recall_num": 50}')
+    for t in range(4):
+        scores = [it["score"] for it in full[t]["items"]]
+        assert scores == sorted(scores)
+        lo, hi = scores[2], scores[6]
+        ranged = eng.search_pb(q[t:t + 1], topn=10,
+                               index_params='{"recall_num": 50}',
+                               min_score=lo, max_score=hi)
+        rs = [it["score"] for it in ranged[0]["items"]]
+        assert rs and all(lo <= s <= hi for s in rs)
+    # status / memory JSON are well-formed and consistent
+    st = _json.loads(eng.status())
+    assert st["doc_count"] > 0 and st["index_status"] == 2
+    out = c.c_char_p()
+    ln = c.c_int()
+    lib().GetMemoryInfo.argtypes = [c.c_void_p, c.POINTER(c.c_char_p),
+                                    c.POINTER(c.c_int)]
+    lib().GetMemoryInfo(eng.h, c.byref(out), c.byref(ln))
+    mi = _json.loads(c.string_at(out, ln.value).decode())
+    assert mi["vector_mem_bytes"] > 0
+    # GetDocByDocID(next=1) walks past deleted docs (gamma_api.h:86)
+    eng2 = make_engine("/tmp/gamma_docwalk")
+    eng2.create_table(64, "FLAT", '{"metric_type": "L2"}')
+    for vid in range(10):
+        eng2.add_doc(str(vid), base[vid])
+    eng2.delete_doc("3")
+    eng2.delete_doc("4")
+    lib().GetDocByDocID.argtypes = [c.c_void_p, c.c_int, c.c_char,
+                                    c.POINTER(c.c_char_p),
+                                    c.POINTER(c.c_int)]
+    rc = lib().GetDocByDocID(eng2.h, 2, b"\x01", c.byref(out),
+                             c.byref(ln))
+    assert rc == 0  # next live doc after 2 is 5; payload is a Doc fbs
+    rc = lib().GetDocByDocID(eng2.h, 3, b"\x00", c.byref(out),
+                             c.byref(ln))
+    assert rc != 0  # deleted doc, no next-walk
+    eng2.close()
+
+
+def test_multi_vector_score_range(data):
+    """Per-field min/max in a multi-vector request: a doc must pass
+    EVERY field's score window to survive the intersection
+    (per-field IsSimilarScoreValid, gamma_common_data.h:94)."""
+    base, q = data
+    n, topn = 3000, 20
+    eng = make_engine("/tmp/gamma_mv_range")
+    eng.create_table(64, "FLAT", '{"metric_type": "L2"}',
+                     extra_vecs=[("emb2", 64)])
+    for vid in range(n):
+        eng.add_doc(str(vid), base[vid], extra_vecs=[("emb2", base[vid])])
+    # identical fields: per-field dists equal; cap field 2's max so the
+    # worse half of the top-20 drops out of the intersection
+    full = eng.search_pb(q[:2], topn=topn,
+                         extra_vec_queries=[("emb2", q[:2])],
+                         multi_vector_rank=1)
+    for t in range(2):
+        scores = [it["score"] for it in full[t]["items"]]
+        cut = scores[9]  # combined == per-field dist here
+        ranged = eng.search_pb(
+            q[t:t + 1], topn=topn,
+            extra_vec_queries=[("emb2", q[t:t + 1], None, cut)],
+            multi_vector_rank=1)
+        rids = [int(it["fields"]["_id"]) for it in ranged[0]["items"]]
+        fids = [int(it["fields"]["_id"]) for it in full[t]["items"]]
+        assert rids == fids[:10]
+    eng.close()
+
+
 def test_multi_vector_update_delete_rebuild(data):
     """Multi-vector tables under mutation: updating a doc replaces its
     row in EVERY field's index (old rows deleted in each extra index),

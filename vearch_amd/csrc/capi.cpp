@@ -389,7 +389,14 @@ static struct CStatus SearchMulti_unguarded(Engine *e,
     if (nq < 0) nq = fn;
     else if (nq != fn)
       return err_status(1, "vector queries disagree on batch size");
-    queries.push_back({vq.name, (const float *)vq.value.data()});
+    vgamma::MultiVecQuery mq;
+    mq.name = vq.name;
+    mq.vecs = (const float *)vq.value.data();
+    mq.has_min = vq.has_min;
+    mq.has_max = vq.has_max;
+    mq.min_score = vq.min_score;
+    mq.max_score = vq.max_score;
+    queries.push_back(std::move(mq));
   }
   /* WeightedRanker (common_query_data.h:251-302) */
   std::vector<double> weights;

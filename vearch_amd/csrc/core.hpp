@@ -390,10 +390,15 @@ struct ExtraVecField {
 };
 
 /* one per-field query of a multi-vector search (vector_manager.cc:851
- * dispatch); value points at nq*dim floats */
+ * dispatch); value points at nq*dim floats. min/max mirror the
+ * per-field SearchCondition::IsSimilarScoreValid range
+ * (gamma_common_data.h:94-96), applied to the field's candidates
+ * before the docid-intersection merge. */
 struct MultiVecQuery {
   std::string name;
   const float *vecs = nullptr;
+  bool has_min = false, has_max = false;
+  double min_score = 0, max_score = 0;
 };
 
 class Engine {

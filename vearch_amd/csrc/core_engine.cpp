@@ -945,10 +945,15 @@ int Engine::search_multi(int nq, const std::vector<MultiVecQuery> &queries,
   for (int i = 0; i < nq; i++) {
     std::vector<std::vector<std::pair<int64_t, float>>> lists(vn);
     for (size_t j = 0; j < vn; j++) {
+      const MultiVecQuery &mq = queries[j];
       for (int t = 0; t < topn; t++) {
         int64_t id = fi[j][(size_t)i * topn + t];
-        if (id >= 0)
-          lists[j].push_back({id, fd[j][(size_t)i * topn + t]});
+        if (id < 0) continue;
+        float dj = fd[j][(size_t)i * topn + t];
+        /* per-field score range (IsSimilarScoreValid) */
+        if (mq.has_min && dj < mq.min_score) continue;
+        if (mq.has_max && dj > mq.max_score) continue;
+        lists[j].push_back({id, dj});
       }
       std::sort(lists[j].begin(), lists[j].end());
     }

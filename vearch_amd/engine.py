@@ -249,8 +249,12 @@ class GammaEngine:
         score; ranker = WeightedRanker JSON."""
         from . import proto
         q = np.ascontiguousarray(queries, dtype=np.float32)
-        extra = [(n, np.ascontiguousarray(v, np.float32).tobytes())
-                 for n, v in extra_vec_queries]
+        extra = []
+        for ef in extra_vec_queries:
+            row = [ef[0],
+                   np.ascontiguousarray(ef[1], np.float32).tobytes()]
+            row += list(ef[2:])  # optional (min_score, max_score)
+            extra.append(tuple(row))
         req = proto.encode_search_request(
             self.vec_name, q.tobytes(), topn, q.shape[0],
             request_id=request_id, partition_id=partition_id,

@@ -66,8 +66,13 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
     if max_score is not None:
         vq += _vdouble(4, max_score)
     out += _ld(5, vq)
-    for en, ev in extra_vec_fields:
-        out += _ld(5, _ld(1, en.encode()) + _ld(2, ev))
+    for ef in extra_vec_fields:
+        evq = _ld(1, ef[0].encode()) + _ld(2, ef[1])
+        if len(ef) > 2 and ef[2] is not None:
+            evq += _vdouble(3, ef[2])
+        if len(ef) > 3 and ef[3] is not None:
+            evq += _vdouble(4, ef[3])
+        out += _ld(5, evq)
     for f in fields:
         out += _ld(6, f.encode())
     for rf in range_filters:

@@ -946,6 +946,16 @@ def test_score_range_docwalk_status(data, ivfpq_engine):
     lib().GetMemoryInfo(eng.h, c.byref(out), c.byref(ln))
     mi = _json.loads(c.string_at(out, ln.value).decode())
     assert mi["vector_mem_bytes"] > 0
+    # SetConfig/GetConfig round-trip (engine.cc:2071-2114 fields)
+    lib().SetConfig.argtypes = [c.c_void_p, c.c_char_p, c.c_int]
+    lib().GetConfig.argtypes = [c.c_void_p, c.POINTER(c.c_char_p),
+                                c.POINTER(c.c_int)]
+    cfg = b'{"slow_search_time": 77, "refresh_interval": 300}'
+    assert lib().SetConfig(eng.h, cfg, len(cfg)) == 0
+    lib().GetConfig(eng.h, c.byref(out), c.byref(ln))
+    got = _json.loads(c.string_at(out, ln.value).decode())
+    assert got["slow_search_time"] == 77
+    assert got["refresh_interval"] == 300
     # GetDocByDocID(next=1) walks past deleted docs (gamma_api.h:86)
     eng2 = make_engine("/tmp/gamma_docwalk")
     eng2.create_table(64, "FLAT", '{"metric_type": "L2"}')

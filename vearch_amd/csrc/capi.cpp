@@ -687,10 +687,9 @@ struct CStatus Query(void *engine, const char *request_str, int req_len,
 
 
 static int SetConfig_unguarded(void *engine, const char *config_str, int len) {
-  (void)engine;
-  (void)config_str;
-  (void)len;
-  return 0; /* cache sizes are N/A for the GPU engine */
+  if (!engine) return -1;
+  return static_cast<Engine *>(engine)->set_config(
+      std::string(config_str ? config_str : "", len > 0 ? len : 0));
 }
 
 int SetConfig(void *engine, const char *config_str, int len) {
@@ -703,7 +702,8 @@ int SetConfig(void *engine, const char *config_str, int len) {
 
 
 static int GetConfig_unguarded(void *engine, char **config_str, int *len) {
-  std::string s = "{\"path\": \"\", \"cache_sizes\": []}";
+  std::string s = engine ? static_cast<Engine *>(engine)->get_config()
+                         : std::string("{}");
   *config_str = dup_malloc(s);
   *len = (int)s.size();
   return 0;

@@ -519,6 +519,11 @@ class Engine {
   const std::string &space() const { return space_name_; }
   bool metric_ip_default() const;
   const std::string &index_type() const { return index_type_; }
+  /* SetConfig/GetConfig (gamma_api.h, engine.cc:2071-2114): the
+   * RocksDB cache knob is N/A on the GPU engine (reported as 0); the
+   * other fields round-trip. */
+  int set_config(const std::string &json);
+  std::string get_config() const;
   int64_t docid_of(const std::string &p_key) const;
   const std::string &pkey_of(int64_t docid) const;
   const std::string *field_value(int64_t docid, const std::string &f) const;
@@ -587,6 +592,8 @@ class Engine {
   /* published with release AFTER all row state (columns, pkey, raw
    * vectors, bucket entry) is visible — the lock-free append contract */
   std::atomic<int64_t> max_docid_{0};
+  int slow_search_time_ = 1000;  /* ms, engine.cc SetConfig field */
+  int refresh_interval_ = 1000;
   std::atomic<int64_t> indexed_count_{0};
   bool table_created_ = false;
   hipStream_t stream_ = nullptr; /* mutation stream (add/build/load) */

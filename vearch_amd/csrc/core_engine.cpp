@@ -1317,6 +1317,30 @@ int Engine::remove_field_index(const std::string &name, std::string *err) {
   return 0;
 }
 
+int Engine::set_config(const std::string &json) {
+  gjson::Value v;
+  if (!gjson::parse(json, v)) return -1;
+  /* engine_cache_size / enable_id_cache: RocksDB-side knobs, accepted
+   * and ignored on the GPU engine (engine.cc:2087-2112 fields) */
+  std::string p;
+  if (v.get_str("path", p) && !p.empty()) path_ = p;
+  int x;
+  if (v.get_int("slow_search_time", x)) slow_search_time_ = x;
+  if (v.get_int("refresh_interval", x)) refresh_interval_ = x;
+  return 0;
+}
+
+std::string Engine::get_config() const {
+  char buf[512];
+  snprintf(buf, sizeof buf,
+           "{\"engine_cache_size\": 0, \"path\": \"%s\", "
+           "\"slow_search_time\": %d, \"refresh_interval\": %d, "
+           "\"enable_id_cache\": true}",
+           gjson::escape(path_).c_str(), slow_search_time_,
+           refresh_interval_);
+  return buf;
+}
+
 int Engine::backup(int command, std::string *err) {
   if (command == 0) { /* create (BackupThread command 0) */
     mkdir(path_.c_str(), 0755); /* parent first (utils::make_dir) */

@@ -936,6 +936,15 @@ def test_score_range_docwalk_status(data, ivfpq_engine):
                                min_score=lo, max_score=hi)
         rs = [it["score"] for it in ranged[0]["items"]]
         assert rs and all(lo <= s <= hi for s in rs)
+    # pagination: offset=o returns ranks [o, o+topn) of the full order
+    full10 = eng.search_pb(q[:2], topn=10,
+                           index_params='{"recall_num": 50}')
+    for t in range(2):
+        paged = eng.search_pb(q[t:t + 1], topn=5, offset=3,
+                              index_params='{"recall_num": 50}')
+        pids = [int(it["fields"]["_id"]) for it in paged[0]["items"]]
+        fids = [int(it["fields"]["_id"]) for it in full10[t]["items"]]
+        assert pids == fids[3:8]
     # status / memory JSON are well-formed and consistent
     st = _json.loads(eng.status())
     assert st["doc_count"] > 0 and st["index_status"] == 2

@@ -242,7 +242,8 @@ class GammaEngine:
                   request_id="req1", partition_id=1, brute=0,
                   min_score=None, max_score=None, l2_sqrt=False,
                   term_filters=(), range_filters=(), operator=0,
-                  extra_vec_queries=(), multi_vector_rank=0, ranker=""):
+                  extra_vec_queries=(), multi_vector_rank=0, ranker="",
+                  offset=0):
         """The real C-ABI Search with protobuf marshalling (reader.go
         path). extra_vec_queries: [(field_name, query_array)] for
         multi-vector search; multi_vector_rank orders by combined
@@ -262,7 +263,8 @@ class GammaEngine:
             min_score=min_score, max_score=max_score, l2_sqrt=l2_sqrt,
             term_filters=term_filters, range_filters=range_filters,
             operator=operator, extra_vec_fields=extra,
-            multi_vector_rank=multi_vector_rank, ranker=ranker)
+            multi_vector_rank=multi_vector_rank, ranker=ranker,
+            offset=offset)
         out = ctypes.c_char_p()
         n = ctypes.c_int()
         st = lib().Search(self.h, req, len(req), ctypes.byref(out),

@@ -46,7 +46,7 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
                           is_vector_value=False, term_filters=(),
                           range_filters=(), operator=0,
                           extra_vec_fields=(), multi_vector_rank=0,
-                          ranker=""):
+                          ranker="", offset=0):
     """vearchpb.SearchRequest (router_grpc.proto:168-192).
     term_filters: (field, value_bytes) or (field, value_bytes, is_union).
     range_filters: (field, lower_bytes, upper_bytes, incl_lower,
@@ -100,6 +100,8 @@ def encode_search_request(vec_name, queries_f32_bytes, topn, req_num,
         out += _ld(15, ranker.encode())
     if operator:
         out += _vint(17, operator)
+    if offset:
+        out += _vint(20, offset)
     return out
 
 
